@@ -1,0 +1,69 @@
+"""Checkpoint/resume in SafeTensors format.
+
+The reference's de-facto durable state is SafeTensors files
+(`0_global_weights.pt`, per-round `{n}_local_gradients.pt`, PS `momentum` /
+`avg-final` — training.py:61-63, parameter_server.rs:393-398) with no actual
+resume. Here: {theta_global, outer momentum, inner AdamW state, counters} as
+safetensors + a JSON manifest, and real resume (SURVEY.md §5 Checkpoint).
+"""
+
+from __future__ import annotations
+
+import json
+import os
+
+import torch
+from safetensors.torch import load_file, save_file
+
+MANIFEST = "manifest.json"
+
+
+def save_checkpoint(worker, out_dir: str) -> None:
+    """Save a DiLoCoWorker's full training state."""
+    os.makedirs(out_dir, exist_ok=True)
+    fp = worker.fp
+    save_file({"theta_global": fp.theta0.cpu()}, os.path.join(out_dir, "0_global_weights.safetensors"))
+    save_file(
+        {
+            "outer_momentum": fp.outer_momentum.cpu(),
+            "exp_avg": fp.exp_avg.cpu(),
+            "exp_avg_sq": fp.exp_avg_sq.cpu(),
+            "master": fp.master.cpu(),
+        },
+        os.path.join(out_dir, "optimizer_state.safetensors"),
+    )
+    manifest = {
+        "format": "hypha_amd.checkpoint.v1",
+        "numel": fp.numel,
+        "inner_step_count": worker.inner_step_count,
+        "round": worker.round,
+        "steps_in_round": worker.steps_in_round,
+        "h": worker.cfg.h,
+        "outer_lr": worker.cfg.outer.lr,
+        "outer_momentum": worker.cfg.outer.momentum,
+    }
+    with open(os.path.join(out_dir, MANIFEST), "w") as f:
+        json.dump(manifest, f, indent=2)
+
+
+def load_checkpoint(worker, ckpt_dir: str) -> dict:
+    """Restore a DiLoCoWorker's state in place; returns the manifest."""
+    with open(os.path.join(ckpt_dir, MANIFEST)) as f:
+        manifest = json.load(f)
+    if manifest["numel"] != worker.fp.numel:
+        raise ValueError(
+            f"checkpoint numel {manifest['numel']} != model numel {worker.fp.numel}"
+        )
+    fp = worker.fp
+    gw = load_file(os.path.join(ckpt_dir, "0_global_weights.safetensors"))
+    fp.theta0.copy_(gw["theta_global"].to(fp.theta0.device))
+    opt = load_file(os.path.join(ckpt_dir, "optimizer_state.safetensors"))
+    fp.outer_momentum.copy_(opt["outer_momentum"].to(fp.master.device))
+    fp.exp_avg.copy_(opt["exp_avg"].to(fp.master.device))
+    fp.exp_avg_sq.copy_(opt["exp_avg_sq"].to(fp.master.device))
+    fp.master.copy_(opt["master"].to(fp.master.device))
+    fp.flat.copy_(fp.master.to(fp.work_dtype))
+    worker.inner_step_count = manifest["inner_step_count"]
+    worker.round = manifest["round"]
+    worker.steps_in_round = manifest["steps_in_round"]
+    return manifest

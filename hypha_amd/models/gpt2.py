@@ -1,0 +1,98 @@
+"""GPT-2 architecture (LayerNorm + learned positions + GELU MLP), from scratch.
+
+This is the BASELINE.json config-1 plumbing model ("GPT-2-small DiLoCo H=10,
+2 CPU worker processes on gloo"): faithful GPT-2 wiring so the framework's
+model registry covers the reference's GPT2 ModelType
+(/root/reference/crates/messages/src/lib.rs:419-488 ModelType list) without
+pulling HF weights. Runs on the reference op path on CPU and on library GEMMs
++ our attention kernel on GPU.
+"""
+
+from __future__ import annotations
+
+import math
+from dataclasses import dataclass
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+from hypha_amd import ops
+
+
+@dataclass
+class GPT2Config:
+    vocab_size: int = 50304  # 50257 rounded up for GEMM-friendly vocab
+    n_positions: int = 1024
+    hidden_size: int = 768
+    n_layers: int = 12
+    n_heads: int = 12
+    norm_eps: float = 1e-5
+    init_std: float = 0.02
+
+
+PRESETS = {
+    "gpt2-small": GPT2Config(),
+    "gpt2-tiny": GPT2Config(vocab_size=512, n_positions=128, hidden_size=64, n_layers=2, n_heads=4),
+}
+
+
+class GPT2Block(nn.Module):
+    def __init__(self, cfg: GPT2Config):
+        super().__init__()
+        h = cfg.hidden_size
+        self.ln1 = nn.LayerNorm(h, eps=cfg.norm_eps)
+        self.attn_qkv = nn.Linear(h, 3 * h)
+        self.attn_out = nn.Linear(h, h)
+        self.ln2 = nn.LayerNorm(h, eps=cfg.norm_eps)
+        self.mlp_fc = nn.Linear(h, 4 * h)
+        self.mlp_proj = nn.Linear(4 * h, h)
+        self.n_heads = cfg.n_heads
+
+    def forward(self, x):
+        b, s, h = x.shape
+        hd = h // self.n_heads
+        q, k, v = self.attn_qkv(self.ln1(x)).split(h, dim=-1)
+        q = q.view(b, s, self.n_heads, hd).transpose(1, 2)
+        k = k.view(b, s, self.n_heads, hd).transpose(1, 2)
+        v = v.view(b, s, self.n_heads, hd).transpose(1, 2)
+        o = ops.flash_attention(q, k, v, causal=True)
+        x = x + self.attn_out(o.transpose(1, 2).reshape(b, s, h))
+        x = x + self.mlp_proj(F.gelu(self.mlp_fc(self.ln2(x)), approximate="tanh"))
+        return x
+
+
+class GPT2ForCausalLM(nn.Module):
+    def __init__(self, cfg: GPT2Config):
+        super().__init__()
+        self.cfg = cfg
+        self.wte = nn.Embedding(cfg.vocab_size, cfg.hidden_size)
+        self.wpe = nn.Embedding(cfg.n_positions, cfg.hidden_size)
+        self.blocks = nn.ModuleList(GPT2Block(cfg) for _ in range(cfg.n_layers))
+        self.ln_f = nn.LayerNorm(cfg.hidden_size, eps=cfg.norm_eps)
+        self.apply(self._init)
+        for blk in self.blocks:
+            for lin in (blk.attn_out, blk.mlp_proj):
+                nn.init.normal_(lin.weight, std=cfg.init_std / math.sqrt(2 * cfg.n_layers))
+
+    def _init(self, m):
+        if isinstance(m, (nn.Linear, nn.Embedding)):
+            nn.init.normal_(m.weight, std=self.cfg.init_std)
+            if isinstance(m, nn.Linear) and m.bias is not None:
+                nn.init.zeros_(m.bias)
+
+    def forward(self, input_ids: torch.Tensor, labels: torch.Tensor | None = None):
+        b, s = input_ids.shape
+        pos = torch.arange(s, device=input_ids.device)
+        x = self.wte(input_ids) + self.wpe(pos)
+        for blk in self.blocks:
+            x = blk(x)
+        x = self.ln_f(x)
+        logits = F.linear(x, self.wte.weight)  # tied head
+        if labels is None:
+            return logits
+        return ops.cross_entropy_loss(logits[:, :-1], labels[:, 1:])
+
+
+def build_model(name: str) -> GPT2ForCausalLM:
+    return GPT2ForCausalLM(PRESETS[name])

@@ -1,0 +1,203 @@
+"""Public op API: autograd wrappers that dispatch HIP kernels (GPU) or the
+PyTorch reference path (CPU).
+
+The HIP extension ABI (hypha_amd._C):
+  rmsnorm_fwd(x2d, w, eps) -> (y, rstd)
+  rmsnorm_bwd(dy, x2d, w, rstd) -> (dx, dw)
+  rope_fwd(q, k, cos, sin, inverse) -> (q_out, k_out)      # [B,H,S,D]
+  swiglu_fwd(gate, up) -> out
+  swiglu_bwd(dout, gate, up) -> (dgate, dup)
+  ce_fwd(logits2d, targets) -> (loss_sum, lse, n_valid)    # fp32 scalars/rows
+  ce_bwd_(logits2d, targets, lse, scale) -> dlogits        # overwrites logits
+  attn_fwd(q, k, v, causal) -> (o, lse)
+  attn_bwd(q, k, v, o, do, lse, causal) -> (dq, dk, dv)
+  adamw_step_(master, param, grad, m, v, lr, b1, b2, eps, wd, step)
+  nesterov_step_(master, delta, momentum, lr, mu)
+  extract_delta(master, theta0, out_bf16)
+"""
+
+from __future__ import annotations
+
+import torch
+
+from . import reference
+from . import use_native
+
+
+def _c():
+    from hypha_amd import _C
+
+    return _C
+
+
+# ---------------------------------------------------------------------------
+# RMSNorm
+# ---------------------------------------------------------------------------
+
+
+class _RMSNorm(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, weight, eps):
+        shape = x.shape
+        x2d = x.reshape(-1, shape[-1]).contiguous()
+        y, rstd = _c().rmsnorm_fwd(x2d, weight, eps)
+        ctx.save_for_backward(x2d, weight, rstd)
+        ctx.shape = shape
+        return y.view(shape)
+
+    @staticmethod
+    def backward(ctx, dy):
+        x2d, weight, rstd = ctx.saved_tensors
+        dx, dw = _c().rmsnorm_bwd(dy.reshape(x2d.shape).contiguous(), x2d, weight, rstd)
+        return dx.view(ctx.shape), dw.to(weight.dtype), None
+
+
+def rmsnorm(x: torch.Tensor, weight: torch.Tensor, eps: float = 1e-5) -> torch.Tensor:
+    if use_native(x):
+        return _RMSNorm.apply(x, weight, eps)
+    return reference.rmsnorm(x, weight, eps)
+
+
+# ---------------------------------------------------------------------------
+# RoPE on q and k together (one fused kernel launch)
+# ---------------------------------------------------------------------------
+
+
+class _RoPE(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, q, k, cos, sin):
+        qo, ko = _c().rope_fwd(q.contiguous(), k.contiguous(), cos, sin, False)
+        ctx.save_for_backward(cos, sin)
+        return qo, ko
+
+    @staticmethod
+    def backward(ctx, dq, dk):
+        cos, sin = ctx.saved_tensors
+        dqo, dko = _c().rope_fwd(dq.contiguous(), dk.contiguous(), cos, sin, True)
+        return dqo, dko, None, None
+
+
+def apply_rope_qk(q, k, cos, sin):
+    """q [B,Hq,S,D], k [B,Hkv,S,D]; cos/sin [S_max, D/2] fp32."""
+    if use_native(q):
+        return _RoPE.apply(q, k, cos, sin)
+    return reference.apply_rope(q, cos, sin), reference.apply_rope(k, cos, sin)
+
+
+# ---------------------------------------------------------------------------
+# SwiGLU
+# ---------------------------------------------------------------------------
+
+
+class _SwiGLU(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, gate, up):
+        gate = gate.contiguous()
+        up = up.contiguous()
+        ctx.save_for_backward(gate, up)
+        return _c().swiglu_fwd(gate, up)
+
+    @staticmethod
+    def backward(ctx, dout):
+        gate, up = ctx.saved_tensors
+        dgate, dup = _c().swiglu_bwd(dout.contiguous(), gate, up)
+        return dgate, dup
+
+
+def swiglu(gate: torch.Tensor, up: torch.Tensor) -> torch.Tensor:
+    if use_native(gate):
+        return _SwiGLU.apply(gate, up)
+    return reference.swiglu(gate, up)
+
+
+# ---------------------------------------------------------------------------
+# Flash attention (causal, GQA)
+# ---------------------------------------------------------------------------
+
+
+class _FlashAttention(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, q, k, v, causal):
+        q, k, v = q.contiguous(), k.contiguous(), v.contiguous()
+        o, lse = _c().attn_fwd(q, k, v, causal)
+        ctx.save_for_backward(q, k, v, o, lse)
+        ctx.causal = causal
+        return o
+
+    @staticmethod
+    def backward(ctx, do):
+        q, k, v, o, lse = ctx.saved_tensors
+        dq, dk, dv = _c().attn_bwd(q, k, v, o, do.contiguous(), lse, ctx.causal)
+        return dq, dk, dv, None
+
+
+def flash_attention(q, k, v, causal: bool = True) -> torch.Tensor:
+    """q [B,Hq,S,D] bf16, k/v [B,Hkv,S,D] bf16 -> [B,Hq,S,D]."""
+    if use_native(q):
+        return _FlashAttention.apply(q, k, v, causal)
+    return reference.attention(q, k, v, causal)
+
+
+# ---------------------------------------------------------------------------
+# Cross-entropy over vocab (memory-frugal: backward overwrites the logits)
+# ---------------------------------------------------------------------------
+
+
+class _CrossEntropy(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, logits2d, targets):
+        loss_sum, lse, n_valid = _c().ce_fwd(logits2d, targets)
+        ctx.save_for_backward(logits2d, targets, lse)
+        ctx.n_valid = max(int(n_valid), 1)
+        return loss_sum / ctx.n_valid
+
+    @staticmethod
+    def backward(ctx, dloss):
+        logits2d, targets, lse = ctx.saved_tensors
+        scale = float(dloss) / ctx.n_valid
+        dlogits = _c().ce_bwd_(logits2d, targets, lse, scale)
+        return dlogits, None
+
+
+def cross_entropy_loss(logits: torch.Tensor, targets: torch.Tensor) -> torch.Tensor:
+    """logits [..., V], targets [...] int64 with -100 = ignore. Mean loss."""
+    v = logits.shape[-1]
+    logits2d = logits.reshape(-1, v)
+    t = targets.reshape(-1)
+    if use_native(logits):
+        return _CrossEntropy.apply(logits2d.contiguous(), t.contiguous())
+    return reference.cross_entropy(logits2d, t)
+
+
+# ---------------------------------------------------------------------------
+# Fused optimizers (no autograd; flat-buffer in-place)
+# ---------------------------------------------------------------------------
+
+
+@torch.no_grad()
+def fused_adamw(master, param, grad, exp_avg, exp_avg_sq, *, lr, beta1, beta2, eps, weight_decay, step):
+    if use_native(master):
+        _c().adamw_step_(master, param, grad, exp_avg, exp_avg_sq, lr, beta1, beta2, eps, weight_decay, step)
+    else:
+        reference.adamw_step(
+            master, param, grad, exp_avg, exp_avg_sq,
+            lr=lr, beta1=beta1, beta2=beta2, eps=eps, weight_decay=weight_decay, step=step,
+        )
+
+
+@torch.no_grad()
+def fused_nesterov(master, delta, momentum, *, lr, mu):
+    if use_native(master):
+        _c().nesterov_step_(master, delta, momentum, lr, mu)
+    else:
+        reference.nesterov_outer_step(master, delta, momentum, lr=lr, mu=mu)
+
+
+@torch.no_grad()
+def extract_delta(master: torch.Tensor, theta0: torch.Tensor, out: torch.Tensor) -> torch.Tensor:
+    """out (bf16 or fp32) = master - theta0, fused sub+cast on GPU."""
+    if use_native(master):
+        _c().extract_delta(master, theta0, out)
+        return out
+    out.copy_(master - theta0)
+    return out

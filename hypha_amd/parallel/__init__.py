@@ -1,0 +1,11 @@
+from .comm import Comm
+from .diloco import DiLoCoConfig, DiLoCoWorker, InnerOptConfig, OuterOptConfig, lr_at
+
+__all__ = [
+    "Comm",
+    "DiLoCoConfig",
+    "DiLoCoWorker",
+    "InnerOptConfig",
+    "OuterOptConfig",
+    "lr_at",
+]

@@ -1,0 +1,100 @@
+"""Collective communication over RCCL (xGMI) / gloo.
+
+One process per GPU; backend "nccl" IS RCCL on ROCm. The outer-sync
+all-reduce is bucketed: xGMI is 7 point-to-point links per GPU, so ring
+collectives are per-link bound — medium-size buckets issued back-to-back let
+RCCL pipeline across channels instead of serialising one giant ring pass
+(SURVEY.md §2.12). Replaces the reference's libp2p tensor push/pull streams
+(C1/C2 in SURVEY.md §2.10; ~1 GB/s ceiling per
+/root/reference/rfc/2025-03-25-libp2p_network_stack.md).
+"""
+
+from __future__ import annotations
+
+import datetime
+import os
+
+import torch
+import torch.distributed as dist
+
+DEFAULT_BUCKET_BYTES = 128 * 1024 * 1024
+
+
+class Comm:
+    """Process-group wrapper with byte accounting and single-process fallback."""
+
+    def __init__(self, backend: str | None = None, timeout_s: float = 600.0):
+        self.rank = int(os.environ.get("RANK", "0"))
+        self.world_size = int(os.environ.get("WORLD_SIZE", "1"))
+        self.local_rank = int(os.environ.get("LOCAL_RANK", str(self.rank)))
+        self.bytes_sent_payload = 0  # payload bytes offered to collectives
+        self.syncs = 0
+        if self.world_size > 1 and not dist.is_initialized():
+            if backend is None:
+                backend = "nccl" if torch.cuda.is_available() else "gloo"
+            os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+            os.environ.setdefault("MASTER_PORT", "29531")
+            dist.init_process_group(
+                backend=backend,
+                rank=self.rank,
+                world_size=self.world_size,
+                timeout=datetime.timedelta(seconds=timeout_s),
+            )
+        self.backend = dist.get_backend() if dist.is_initialized() else "none"
+        if torch.cuda.is_available():
+            torch.cuda.set_device(self.local_rank)
+
+    @property
+    def is_distributed(self) -> bool:
+        return self.world_size > 1
+
+    def barrier(self) -> None:
+        if self.is_distributed:
+            if torch.cuda.is_available() and self.backend == "nccl":
+                dist.barrier(device_ids=[self.local_rank])
+            else:
+                dist.barrier()
+
+    def all_reduce_mean_flat(
+        self, flat: torch.Tensor, bucket_bytes: int = DEFAULT_BUCKET_BYTES
+    ) -> None:
+        """In-place mean all-reduce of a flat tensor, bucketed + async."""
+        self.syncs += 1
+        self.bytes_sent_payload += flat.numel() * flat.element_size()
+        if not self.is_distributed:
+            return
+        n_per_bucket = max(1, bucket_bytes // flat.element_size())
+        handles = []
+        for start in range(0, flat.numel(), n_per_bucket):
+            chunk = flat.narrow(0, start, min(n_per_bucket, flat.numel() - start))
+            handles.append(dist.all_reduce(chunk, op=dist.ReduceOp.SUM, async_op=True))
+        for h in handles:
+            h.wait()
+        flat.div_(self.world_size)
+
+    def broadcast_flat(self, flat: torch.Tensor, src: int = 0) -> None:
+        if self.is_distributed:
+            dist.broadcast(flat, src=src)
+
+    def all_reduce_scalar(self, value: float, op: str = "max") -> float:
+        if not self.is_distributed:
+            return value
+        device = (
+            torch.device("cuda", self.local_rank)
+            if (torch.cuda.is_available() and self.backend == "nccl")
+            else torch.device("cpu")
+        )
+        t = torch.tensor([value], dtype=torch.float64, device=device)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX if op == "max" else dist.ReduceOp.SUM)
+        return float(t.item())
+
+    def wire_bytes_per_rank(self, payload_bytes: int) -> int:
+        """Ring all-reduce on-wire bytes per rank for a given payload."""
+        w = self.world_size
+        if w <= 1:
+            return 0
+        return int(2 * (w - 1) / w * payload_bytes)
+
+    def shutdown(self) -> None:
+        if dist.is_initialized():
+            dist.destroy_process_group()

@@ -1,0 +1,46 @@
+import torch
+
+from hypha_amd import models
+
+
+def test_llama_tiny_forward_backward():
+    torch.manual_seed(0)
+    m = models.build("llama-tiny")
+    ids = torch.randint(0, 512, (2, 32))
+    loss = m(ids, labels=ids)
+    assert loss.dim() == 0 and torch.isfinite(loss)
+    loss.backward()
+    grads = [p.grad for p in m.parameters() if p.requires_grad]
+    assert all(g is not None for g in grads)
+    assert all(torch.isfinite(g).all() for g in grads)
+
+
+def test_gpt2_tiny_forward_backward():
+    torch.manual_seed(0)
+    m = models.build("gpt2-tiny")
+    ids = torch.randint(0, 512, (2, 32))
+    loss = m(ids, labels=ids)
+    assert torch.isfinite(loss)
+    loss.backward()
+
+
+def test_llama_param_count_8b():
+    cfg = models.llama.PRESETS["llama3-8b"]
+    n = cfg.num_params()
+    assert 7.5e9 < n < 8.5e9, n
+
+
+def test_loss_near_uniform_at_init():
+    import math
+
+    torch.manual_seed(0)
+    m = models.build("llama-tiny")
+    ids = torch.randint(0, 512, (2, 64))
+    loss = float(m(ids, labels=ids))
+    assert abs(loss - math.log(512)) < 1.0, loss
+
+
+def test_registry_lists_flagships():
+    av = models.available()
+    for name in ("llama3-8b", "llama3-70b", "gpt2-small", "llama-tiny"):
+        assert name in av
