@@ -1,0 +1,523 @@
+// Fused causal GQA flash attention, forward + backward, for CDNA4 (gfx950).
+// Covers K1's attention piece natively — the reference
+// delegates this to HF/PyTorch SDPA inside its Python executor.
+//
+// Forward structure (one workgroup = 4 waves = 128 q rows; KV tile = 64):
+//  * swapped QK^T: S^T = mfma(A=K, B=Q) so each lane owns ONE q column and
+//    the online-softmax state (m, l) is lane-local — no cross-lane reduce
+//    beyond the in-reg accumulator sweep + one __shfl_xor(32) half combine.
+//  * O is accumulated TRANSPOSED: O^T = mfma(A=V^T, B=P^T), keeping the
+//    per-q rescale factor lane-local as well; the epilogue transposes O
+//    back through LDS once per q block for coalesced stores.
+//  * P^T (f32 accum regs) is packed to bf16 in-register (pack pairs +
+//    __builtin_amdgcn_permlane32_swap half exchange) and feeds the PV MFMA
+//    B operand directly — no LDS round trip for P.
+//  * K tile lives in LDS row-major with an XOR-16 swizzle (bank-conflict
+//    free ds_read_b128); V is transposed into LDS at staging time (pitch
+//    padded to 72 elems, conflict-free b128 rows).
+//
+// Backward (one workgroup = 4 waves = one 32-row KV tile, q-tiles split
+// round-robin across waves): recomputes P^T from q/k/lse, stages P^T, dS
+// tiles and transposed operand images through LDS, accumulates dK/dV in
+// f32 MFMA accumulators (cross-wave combine through LDS at the end), and
+// accumulates dQ with global f32 atomics (FA2-style).
+
+#include <torch/extension.h>
+
+#include "hip_common.h"
+
+namespace {
+
+constexpr int KVB = 64;   // kv tile (fwd)
+constexpr int QB = 32;    // q rows per wave
+constexpr int NWAVE = 4;  // waves per workgroup
+constexpr int WGQ = QB * NWAVE;
+constexpr int PADV = 8;   // V^T pitch pad (72 elems -> conflict-free b128)
+
+__device__ __forceinline__ unsigned pack_bf16x2(float lo, float hi) {
+  return (unsigned)(unsigned short)f2bf(lo) | ((unsigned)(unsigned short)f2bf(hi) << 16);
+}
+
+// Swizzled byte offset inside the K tile: row-major [KVB][HD] bf16 with
+// byte ^= (row & 7) << 4 (guide G4 XOR swizzle, <=2-way on ds_read_b128).
+template <int HD>
+__device__ __forceinline__ int k_lds_off(int row, int elem) {
+  return (row * HD + elem) * 2 ^ ((row & 7) << 4);
+}
+
+template <int HD>
+__global__ __launch_bounds__(256, 1) void attn_fwd_kernel(
+    const short* __restrict__ qg, const short* __restrict__ kg,
+    const short* __restrict__ vg, short* __restrict__ og, float* __restrict__ lseg,
+    int B, int Hq, int Hkv, int S, float scale, bool causal) {
+  constexpr int KC = HD / 16;        // 16-wide k chunks over head dim
+  constexpr int DBLK = HD / 32;      // 32-row d blocks
+  constexpr int VT_PITCH = KVB + PADV;
+  __shared__ __attribute__((aligned(16))) short k_lds[KVB * HD];
+  __shared__ __attribute__((aligned(16))) short vt_lds[HD * VT_PITCH];
+  __shared__ __attribute__((aligned(16))) short ot_lds[NWAVE][QB * (HD + 8)];  // epilogue transpose
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int hi = lane >> 5;  // half-wave
+  const int ln = lane & 31;
+
+  const int bh = blockIdx.y;
+  const int b = bh / Hq;
+  const int hq = bh % Hq;
+  const int hkv = hq / (Hq / Hkv);
+  const long long qbase = ((long long)(b * Hq + hq) * S) * HD;
+  const long long kvbase = ((long long)(b * Hkv + hkv) * S) * HD;
+
+  const int q0wg = blockIdx.x * WGQ;
+  const int q0 = q0wg + wid * QB;   // this wave's q rows
+  const int qrow = q0 + ln;         // this lane's q row (its S^T column)
+
+  // ---- Q fragments: B operand of S^T = mfma(K, Q); lane ln = q column ----
+  s16x8 qfrag[KC];
+#pragma unroll
+  for (int kc = 0; kc < KC; ++kc)
+    qfrag[kc] = *reinterpret_cast<const s16x8*>(qg + qbase + (long long)qrow * HD +
+                                                16 * kc + 8 * hi);
+
+  f32x16 ot[DBLK] = {};
+  float m_run = -INFINITY, l_run = 0.f;
+
+  const int kv_end = causal ? (q0wg + WGQ) : S;  // exclusive upper bound
+  for (int kv0 = 0; kv0 < kv_end; kv0 += KVB) {
+    // ---- stage K [KVB][HD] (swizzled) and V^T [HD][KVB] ----
+    __syncthreads();
+    constexpr int CH = KVB * HD / 8;  // 16B chunks
+    for (int c = tid; c < CH; c += 256) {
+      int row = c / (HD / 8), e0 = (c % (HD / 8)) * 8;
+      s16x8 kv8 = *reinterpret_cast<const s16x8*>(kg + kvbase + (long long)(kv0 + row) * HD + e0);
+      *reinterpret_cast<s16x8*>((char*)k_lds + k_lds_off<HD>(row, e0)) = kv8;
+      s16x8 vv8 = *reinterpret_cast<const s16x8*>(vg + kvbase + (long long)(kv0 + row) * HD + e0);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) vt_lds[(e0 + j) * VT_PITCH + row] = vv8[j];
+    }
+    __syncthreads();
+
+    if (causal && kv0 > q0 + QB - 1) continue;  // wave fully above the diagonal
+
+    // ---- S^T = K Q^T : two 32-kv accumulators ----
+    f32x16 st[2];
+#pragma unroll
+    for (int mb = 0; mb < 2; ++mb) {
+      f32x16 acc = {};
+#pragma unroll
+      for (int kc = 0; kc < KC; ++kc) {
+        s16x8 kf = *reinterpret_cast<const s16x8*>(
+            (char*)k_lds + k_lds_off<HD>(32 * mb + ln, 16 * kc + 8 * hi));
+        acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qfrag[kc], acc, 0, 0, 0);
+      }
+      st[mb] = acc;
+    }
+
+    // ---- online softmax over the 32 regs (all kv of this tile, own q) ----
+    float tmax = -INFINITY;
+#pragma unroll
+    for (int mb = 0; mb < 2; ++mb)
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        int kv = kv0 + 32 * mb + (r & 3) + 8 * (r >> 2) + 4 * hi;
+        float s = st[mb][r] * scale;
+        if (causal && kv > qrow) s = -INFINITY;
+        st[mb][r] = s;
+        tmax = fmaxf(tmax, s);
+      }
+    tmax = fmaxf(tmax, __shfl_xor(tmax, 32, 64));
+    const float m_new = fmaxf(m_run, tmax);
+    const float alpha = __expf(m_run - m_new);
+    float psum = 0.f;
+#pragma unroll
+    for (int mb = 0; mb < 2; ++mb)
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        float p = __expf(st[mb][r] - m_new);
+        st[mb][r] = p;
+        psum += p;
+      }
+    psum += __shfl_xor(psum, 32, 64);
+    l_run = l_run * alpha + psum;
+    m_run = m_new;
+#pragma unroll
+    for (int db = 0; db < DBLK; ++db)
+#pragma unroll
+      for (int r = 0; r < 16; ++r) ot[db][r] *= alpha;
+
+    // ---- pack P^T to bf16 B-fragments (pair-pack + half swap) ----
+    s16x8 pfrag[4];
+#pragma unroll
+    for (int mb = 0; mb < 2; ++mb) {
+#pragma unroll
+      for (int half = 0; half < 2; ++half) {
+        unsigned c0 = pack_bf16x2(st[mb][8 * half + 0], st[mb][8 * half + 1]);
+        unsigned c1 = pack_bf16x2(st[mb][8 * half + 2], st[mb][8 * half + 3]);
+        unsigned c2 = pack_bf16x2(st[mb][8 * half + 4], st[mb][8 * half + 5]);
+        unsigned c3 = pack_bf16x2(st[mb][8 * half + 6], st[mb][8 * half + 7]);
+        auto r02 = __builtin_amdgcn_permlane32_swap(c0, c2, false, false);
+        auto r13 = __builtin_amdgcn_permlane32_swap(c1, c3, false, false);
+        unsigned frag[4] = {(unsigned)r02[0], (unsigned)r13[0], (unsigned)r02[1],
+                            (unsigned)r13[1]};
+        pfrag[2 * mb + half] = *reinterpret_cast<s16x8*>(frag);
+      }
+    }
+
+    // ---- O^T += V^T P^T ----
+#pragma unroll
+    for (int db = 0; db < DBLK; ++db) {
+#pragma unroll
+      for (int kc = 0; kc < 4; ++kc) {
+        s16x8 vf = *reinterpret_cast<const s16x8*>(
+            vt_lds + (32 * db + ln) * VT_PITCH + 16 * kc + 8 * hi);
+        ot[db] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vf, pfrag[kc], ot[db], 0, 0, 0);
+      }
+    }
+  }
+
+  // ---- epilogue: normalize, transpose through LDS, coalesced store ----
+  const float inv_l = l_run > 0.f ? 1.f / l_run : 0.f;
+  if (lane < 32) lseg[(long long)bh * S + qrow] = m_run + __logf(fmaxf(l_run, 1e-30f));
+
+  constexpr int OPITCH = HD + 8;
+#pragma unroll
+  for (int db = 0; db < DBLK; ++db)
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      int d = 32 * db + (r & 3) + 8 * (r >> 2) + 4 * hi;
+      ot_lds[wid][ln * OPITCH + d] = f2bf(ot[db][r] * inv_l);
+    }
+  __builtin_amdgcn_s_waitcnt(0);  // lgkmcnt(0): own-wave LDS writes visible
+#pragma unroll
+  for (int it = 0; it < QB * HD / 8 / 64; ++it) {
+    int c = it * 64 + lane;
+    int row = c / (HD / 8), e0 = (c % (HD / 8)) * 8;
+    s16x8 o8 = *reinterpret_cast<const s16x8*>(ot_lds[wid] + row * OPITCH + e0);
+    *reinterpret_cast<s16x8*>(og + qbase + (long long)(q0 + row) * HD + e0) = o8;
+  }
+}
+
+}  // namespace
+
+std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
+                                    bool causal) {
+  TORCH_CHECK(q.dim() == 4 && q.dtype() == torch::kBFloat16 && q.is_contiguous());
+  const int B = q.size(0), Hq = q.size(1), S = q.size(2), HD = q.size(3);
+  const int Hkv = k.size(1);
+  TORCH_CHECK(Hq % Hkv == 0 && (HD == 64 || HD == 128));
+  TORCH_CHECK(S % WGQ == 0, "seq len must be a multiple of 128");
+  auto o = torch::empty_like(q);
+  auto lse = torch::empty({B, Hq, S}, q.options().dtype(torch::kFloat32));
+  const float scale = 1.0f / sqrtf((float)HD);
+  dim3 grid(S / WGQ, B * Hq);
+  hipStream_t stream = hypha_stream();
+  if (HD == 128)
+    hipLaunchKernelGGL(attn_fwd_kernel<128>, grid, dim3(256), 0, stream,
+                       (const short*)q.data_ptr(), (const short*)k.data_ptr(),
+                       (const short*)v.data_ptr(), (short*)o.data_ptr(),
+                       lse.data_ptr<float>(), B, Hq, Hkv, S, scale, causal);
+  else
+    hipLaunchKernelGGL(attn_fwd_kernel<64>, grid, dim3(256), 0, stream,
+                       (const short*)q.data_ptr(), (const short*)k.data_ptr(),
+                       (const short*)v.data_ptr(), (short*)o.data_ptr(),
+                       lse.data_ptr<float>(), B, Hq, Hkv, S, scale, causal);
+  return {o, lse};
+}
+
+// ===========================================================================
+// Backward
+// ===========================================================================
+
+namespace {
+
+// Di = rowsum(dO * O) per (b,h,q) row — FA2 preprocess. One wave per row.
+template <int HD>
+__global__ void attn_bwd_di_kernel(const short* __restrict__ dog,
+                                   const short* __restrict__ og,
+                                   float* __restrict__ dig, long long nrows) {
+  const long long row = (long long)blockIdx.x * 4 + (threadIdx.x >> 6);
+  if (row >= nrows) return;
+  const int lane = threadIdx.x & 63;
+  float acc = 0.f;
+#pragma unroll
+  for (int i = 0; i < HD / 64; ++i) {
+    int d = lane + 64 * i;
+    acc += bf2f(dog[row * HD + d]) * bf2f(og[row * HD + d]);
+  }
+  acc = wave_reduce_sum(acc);
+  if (lane == 0) dig[row] = acc;
+}
+
+__global__ void cast_f32_to_bf16_kernel(const float* __restrict__ in,
+                                        short* __restrict__ out, long long n) {
+  const long long stride = (long long)gridDim.x * blockDim.x * 4;
+  for (long long base = ((long long)blockIdx.x * blockDim.x + threadIdx.x) * 4; base < n;
+       base += stride) {
+    if (base + 4 <= n) {
+      f32x4 v = *reinterpret_cast<const f32x4*>(in + base);
+      s16x4 o;
+#pragma unroll
+      for (int j = 0; j < 4; ++j) o[j] = f2bf(v[j]);
+      *reinterpret_cast<s16x4*>(out + base) = o;
+    } else {
+      for (long long i = base; i < n; ++i) out[i] = f2bf(in[i]);
+    }
+  }
+}
+
+// Main backward: one workgroup = 4 waves sharing ONE 32-row kv tile; q-tiles
+// (and the GQA q-head group) are walked by all waves with a round-robin
+// split; dK/dV accumulate in per-wave MFMA accumulators and are combined
+// through LDS at the end; dQ accumulates via global f32 atomics (FA2-style).
+template <int HD>
+__global__ __launch_bounds__(256, 1) void attn_bwd_kernel(
+    const short* __restrict__ qg, const short* __restrict__ kg,
+    const short* __restrict__ vg, const short* __restrict__ dog,
+    const float* __restrict__ lseg, const float* __restrict__ dig,
+    float* __restrict__ dqg, short* __restrict__ dkg, short* __restrict__ dvg,
+    int B, int Hq, int Hkv, int S, float scale, bool causal) {
+  constexpr int KVT = 32;            // kv rows per workgroup
+  constexpr int QT = 32;             // q rows per tile
+  constexpr int KC = HD / 16;        // chunks over head dim
+  constexpr int DBLK = HD / 32;      // d blocks
+  constexpr int TP = 40;             // transposed-image pitch (16B-aligned rows)
+  __shared__ __attribute__((aligned(16))) short k_img[KVT * HD];        // [kv][d], XOR-swizzled
+  __shared__ __attribute__((aligned(16))) short v_img[KVT * HD];        // [kv][d], XOR-swizzled
+  __shared__ __attribute__((aligned(16))) short kt_img[HD * TP];        // [d][kv]
+  __shared__ __attribute__((aligned(16))) short qt_img[NWAVE][HD * TP];   // per wave: [d][q]
+  __shared__ __attribute__((aligned(16))) short dot_img[NWAVE][HD * TP];  // per wave: [d][q]
+  __shared__ __attribute__((aligned(16))) short ptds_img[NWAVE][KVT * TP];  // per wave: ptT then ds reuse
+  __shared__ __attribute__((aligned(16))) short dst_img[NWAVE][KVT * TP];   // per wave: dsT
+  __shared__ __attribute__((aligned(16))) float comb[KVT * HD];         // cross-wave f32 combine
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int hi = lane >> 5;
+  const int ln = lane & 31;
+
+  const int bhkv = blockIdx.y;
+  const int b = bhkv / Hkv;
+  const int hkv = bhkv % Hkv;
+  const int G = Hq / Hkv;
+  const int kv0 = blockIdx.x * KVT;
+  const long long kvbase = ((long long)(b * Hkv + hkv) * S) * HD;
+
+  // ---- stage K, V (swizzled row-major) and K^T ----
+  {
+    constexpr int CH = KVT * HD / 8;
+    for (int c = tid; c < CH; c += 256) {
+      int row = c / (HD / 8), e0 = (c % (HD / 8)) * 8;
+      s16x8 k8 = *reinterpret_cast<const s16x8*>(kg + kvbase + (long long)(kv0 + row) * HD + e0);
+      *reinterpret_cast<s16x8*>((char*)k_img + k_lds_off<HD>(row, e0)) = k8;
+      s16x8 v8 = *reinterpret_cast<const s16x8*>(vg + kvbase + (long long)(kv0 + row) * HD + e0);
+      *reinterpret_cast<s16x8*>((char*)v_img + k_lds_off<HD>(row, e0)) = v8;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) kt_img[(e0 + j) * TP + row] = k8[j];
+    }
+  }
+  __syncthreads();
+
+  f32x16 dk_acc[DBLK] = {};  // D[m=kv][n=d]
+  f32x16 dv_acc[DBLK] = {};  // D[m=d][n=kv]
+
+  short* qt = qt_img[wid];
+  short* dot = dot_img[wid];
+  short* ptds = ptds_img[wid];
+  short* dst = dst_img[wid];
+
+  const int t0 = causal ? kv0 / QT : 0;
+  const int Tq = S / QT;
+
+  for (int hq = hkv * G; hq < (hkv + 1) * G; ++hq) {
+    const long long qbase = ((long long)(b * Hq + hq) * S) * HD;
+    const long long lsebase = (long long)(b * Hq + hq) * S;
+    for (int t = t0 + wid; t < Tq; t += NWAVE) {
+      const int q0 = t * QT;
+      const int qrow = q0 + ln;  // this lane's q (for B-operand frags)
+
+      // ---- per-wave staging: Q^T and dO^T images (transpose scatter) ----
+#pragma unroll
+      for (int it = 0; it < QT * HD / 8 / 64; ++it) {
+        int c = it * 64 + lane;
+        int row = c / (HD / 8), e0 = (c % (HD / 8)) * 8;
+        s16x8 q8 = *reinterpret_cast<const s16x8*>(qg + qbase + (long long)(q0 + row) * HD + e0);
+        s16x8 d8 = *reinterpret_cast<const s16x8*>(dog + qbase + (long long)(q0 + row) * HD + e0);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          qt[(e0 + j) * TP + row] = q8[j];
+          dot[(e0 + j) * TP + row] = d8[j];
+        }
+      }
+
+      // ---- S^T = K Q^T (Q frags straight from global) ----
+      f32x16 st = {};
+#pragma unroll
+      for (int kc = 0; kc < KC; ++kc) {
+        s16x8 qf = *reinterpret_cast<const s16x8*>(qg + qbase + (long long)qrow * HD +
+                                                   16 * kc + 8 * hi);
+        s16x8 kf = *reinterpret_cast<const s16x8*>(
+            (char*)k_img + k_lds_off<HD>(ln, 16 * kc + 8 * hi));
+        st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qf, st, 0, 0, 0);
+      }
+
+      // ---- P^T = exp(scale*S^T - lse[q]) with causal mask ----
+      const float lse = lseg[lsebase + qrow];
+      const float di = dig[lsebase + qrow];
+      f32x16 pt;
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        int kv = kv0 + (r & 3) + 8 * (r >> 2) + 4 * hi;
+        float p = __expf(st[r] * scale - lse);
+        if (causal && kv > qrow) p = 0.f;
+        pt[r] = p;
+      }
+      // write ptT image [kv][q]
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        int kv = (r & 3) + 8 * (r >> 2) + 4 * hi;
+        ptds[kv * TP + ln] = f2bf(pt[r]);
+      }
+
+      // ---- dV^T += dO^T P  (A = dO^T image, B = ptT image) ----
+#pragma unroll
+      for (int db = 0; db < DBLK; ++db) {
+#pragma unroll
+        for (int kcq = 0; kcq < QT / 16; ++kcq) {
+          s16x8 af = *reinterpret_cast<const s16x8*>(dot + (32 * db + ln) * TP +
+                                                     16 * kcq + 8 * hi);
+          s16x8 bf = *reinterpret_cast<const s16x8*>(ptds + ln * TP + 16 * kcq + 8 * hi);
+          dv_acc[db] =
+              __builtin_amdgcn_mfma_f32_32x32x16_bf16(af, bf, dv_acc[db], 0, 0, 0);
+        }
+      }
+
+      // ---- dP^T = V dO^T (A = V image, B = dO frags from global) ----
+      f32x16 dpt = {};
+#pragma unroll
+      for (int kc = 0; kc < KC; ++kc) {
+        s16x8 df = *reinterpret_cast<const s16x8*>(dog + qbase + (long long)qrow * HD +
+                                                   16 * kc + 8 * hi);
+        s16x8 vf = *reinterpret_cast<const s16x8*>(
+            (char*)v_img + k_lds_off<HD>(ln, 16 * kc + 8 * hi));
+        dpt = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vf, df, dpt, 0, 0, 0);
+      }
+
+      // ---- dS^T = P^T * (dP^T - Di) * scale; write dsT [kv][q] + ds [q][kv]
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        int kv = (r & 3) + 8 * (r >> 2) + 4 * hi;
+        float v = pt[r] * (dpt[r] - di) * scale;
+        short vb = f2bf(v);
+        dst[kv * TP + ln] = vb;
+        ptds[ln * TP + kv] = vb;  // ds image reuses the ptT buffer (pt consumed)
+      }
+
+      // ---- dK += dS^T Q (A = dsT image, B = Q^T image) ----
+#pragma unroll
+      for (int db = 0; db < DBLK; ++db) {
+#pragma unroll
+        for (int kcq = 0; kcq < QT / 16; ++kcq) {
+          s16x8 af = *reinterpret_cast<const s16x8*>(dst + ln * TP + 16 * kcq + 8 * hi);
+          s16x8 bf = *reinterpret_cast<const s16x8*>(qt + (32 * db + ln) * TP +
+                                                     16 * kcq + 8 * hi);
+          dk_acc[db] =
+              __builtin_amdgcn_mfma_f32_32x32x16_bf16(af, bf, dk_acc[db], 0, 0, 0);
+        }
+      }
+
+      // ---- dQ(tile) = dS K (A = ds image, B = K^T image); global atomics --
+#pragma unroll
+      for (int db = 0; db < DBLK; ++db) {
+        f32x16 dq = {};
+#pragma unroll
+        for (int kck = 0; kck < KVT / 16; ++kck) {
+          s16x8 af = *reinterpret_cast<const s16x8*>(ptds + ln * TP + 16 * kck + 8 * hi);
+          s16x8 bf = *reinterpret_cast<const s16x8*>(kt_img + (32 * db + ln) * TP +
+                                                     16 * kck + 8 * hi);
+          dq = __builtin_amdgcn_mfma_f32_32x32x16_bf16(af, bf, dq, 0, 0, 0);
+        }
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          int qi = q0 + (r & 3) + 8 * (r >> 2) + 4 * hi;
+          atomicAdd(dqg + (qbase + (long long)qi * HD + 32 * db + ln), dq[r]);
+        }
+      }
+    }
+  }
+
+  // ---- combine dK across waves (phased, no atomics), store bf16 ----
+  auto combine_store = [&](const f32x16* acc, short* outg, bool transposed) {
+    for (int c = tid; c < KVT * HD; c += 256) comb[c] = 0.f;
+    __syncthreads();
+    for (int w = 0; w < NWAVE; ++w) {
+      if (w == wid) {
+#pragma unroll
+        for (int db = 0; db < DBLK; ++db)
+#pragma unroll
+          for (int r = 0; r < 16; ++r) {
+            int m = (r & 3) + 8 * (r >> 2) + 4 * hi;
+            int kv = transposed ? ln : m;
+            int d = transposed ? (32 * db + m) : (32 * db + ln);
+            comb[kv * HD + d] += acc[db][r];
+          }
+      }
+      __syncthreads();
+    }
+    for (int c = tid; c < KVT * HD / 8; c += 256) {
+      int row = c / (HD / 8), e0 = (c % (HD / 8)) * 8;
+      s16x8 o8;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) o8[j] = f2bf(comb[row * HD + e0 + j]);
+      *reinterpret_cast<s16x8*>(outg + kvbase + (long long)(kv0 + row) * HD + e0) = o8;
+    }
+    __syncthreads();
+  };
+  combine_store(dk_acc, dkg, false);
+  combine_store(dv_acc, dvg, true);
+}
+
+}  // namespace
+
+std::vector<torch::Tensor> attn_bwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
+                                    torch::Tensor o, torch::Tensor dout,
+                                    torch::Tensor lse, bool causal) {
+  const int B = q.size(0), Hq = q.size(1), S = q.size(2), HD = q.size(3);
+  const int Hkv = k.size(1);
+  TORCH_CHECK(S % 32 == 0 && (HD == 64 || HD == 128));
+  auto dq32 = torch::zeros({B, Hq, S, HD}, q.options().dtype(torch::kFloat32));
+  auto dk = torch::empty_like(k);
+  auto dv = torch::empty_like(v);
+  auto di = torch::empty({B, Hq, S}, q.options().dtype(torch::kFloat32));
+  const float scale = 1.0f / sqrtf((float)HD);
+  hipStream_t stream = hypha_stream();
+  long long nrows = (long long)B * Hq * S;
+
+#define DISPATCH(HDV)                                                                   \
+  do {                                                                                  \
+    hipLaunchKernelGGL(attn_bwd_di_kernel<HDV>, dim3((unsigned)((nrows + 3) / 4)),      \
+                       dim3(256), 0, stream, (const short*)dout.data_ptr(),             \
+                       (const short*)o.data_ptr(), di.data_ptr<float>(), nrows);        \
+    hipLaunchKernelGGL(attn_bwd_kernel<HDV>, dim3(S / 32, B * Hkv), dim3(256), 0,       \
+                       stream, (const short*)q.data_ptr(), (const short*)k.data_ptr(),  \
+                       (const short*)v.data_ptr(), (const short*)dout.data_ptr(),       \
+                       lse.data_ptr<float>(), di.data_ptr<float>(),                     \
+                       dq32.data_ptr<float>(), (short*)dk.data_ptr(),                   \
+                       (short*)dv.data_ptr(), B, Hq, Hkv, S, scale, causal);            \
+  } while (0)
+
+  if (HD == 128)
+    DISPATCH(128);
+  else
+    DISPATCH(64);
+#undef DISPATCH
+
+  auto dq = torch::empty_like(q);
+  long long n = dq32.numel();
+  hipLaunchKernelGGL(cast_f32_to_bf16_kernel, dim3(elementwise_grid((n + 3) / 4)),
+                     dim3(256), 0, stream, dq32.data_ptr<float>(), (short*)dq.data_ptr(),
+                     n);
+  return {dq, dk, dv};
+}

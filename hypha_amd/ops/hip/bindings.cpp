@@ -1,0 +1,53 @@
+// pybind11 bindings for the hypha_amd CDNA4 HIP extension (_C).
+
+#include <torch/extension.h>
+
+#include <vector>
+
+// elementwise.hip
+void adamw_step_(torch::Tensor master, torch::Tensor param, torch::Tensor grad,
+                 torch::Tensor m, torch::Tensor v, double lr, double beta1, double beta2,
+                 double eps, double wd, long step);
+void nesterov_step_(torch::Tensor theta, torch::Tensor delta, torch::Tensor mom, double lr,
+                    double mu);
+void extract_delta(torch::Tensor master, torch::Tensor theta0, torch::Tensor out);
+torch::Tensor swiglu_fwd(torch::Tensor gate, torch::Tensor up);
+std::vector<torch::Tensor> swiglu_bwd(torch::Tensor dout, torch::Tensor gate,
+                                      torch::Tensor up);
+// rmsnorm.hip
+std::vector<torch::Tensor> rmsnorm_fwd(torch::Tensor x, torch::Tensor w, double eps);
+std::vector<torch::Tensor> rmsnorm_bwd(torch::Tensor dy, torch::Tensor x, torch::Tensor w,
+                                       torch::Tensor rstd);
+// rope.hip
+std::vector<torch::Tensor> rope_fwd(torch::Tensor q, torch::Tensor k, torch::Tensor cos_t,
+                                    torch::Tensor sin_t, bool inverse);
+// cross_entropy.hip
+std::vector<torch::Tensor> ce_fwd(torch::Tensor logits, torch::Tensor targets);
+torch::Tensor ce_bwd_(torch::Tensor logits, torch::Tensor targets, torch::Tensor lse,
+                      double scale);
+// attention.hip
+std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
+                                    bool causal);
+std::vector<torch::Tensor> attn_bwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
+                                    torch::Tensor o, torch::Tensor dout, torch::Tensor lse,
+                                    bool causal);
+// probe.hip (MFMA fragment-layout verification)
+torch::Tensor mfma_probe_32x32x16(torch::Tensor a, torch::Tensor b);
+torch::Tensor mfma_probe_16x16x32(torch::Tensor a, torch::Tensor b);
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("adamw_step_", &adamw_step_);
+  m.def("nesterov_step_", &nesterov_step_);
+  m.def("extract_delta", &extract_delta);
+  m.def("swiglu_fwd", &swiglu_fwd);
+  m.def("swiglu_bwd", &swiglu_bwd);
+  m.def("rmsnorm_fwd", &rmsnorm_fwd);
+  m.def("rmsnorm_bwd", &rmsnorm_bwd);
+  m.def("rope_fwd", &rope_fwd);
+  m.def("ce_fwd", &ce_fwd);
+  m.def("ce_bwd_", &ce_bwd_);
+  m.def("attn_fwd", &attn_fwd);
+  m.def("attn_bwd", &attn_bwd);
+  m.def("mfma_probe_32x32x16", &mfma_probe_32x32x16);
+  m.def("mfma_probe_16x16x32", &mfma_probe_16x16x32);
+}

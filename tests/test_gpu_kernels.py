@@ -1,0 +1,235 @@
+"""GPU (MI355X) numerics tests: every HIP kernel vs a plain PyTorch fp32
+reference of the same op (the reference repo's golden-value pattern,
+parameter_server.rs:448-525)."""
+
+import math
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+if torch.cuda.is_available():
+    from hypha_amd import _C, ops
+    from hypha_amd.ops import reference as R
+
+DEV = "cuda:0"
+
+
+def rand_bf16(*shape, seed=0, scale=1.0):
+    g = torch.Generator(device="cpu").manual_seed(seed)
+    return (torch.randn(*shape, generator=g) * scale).bfloat16().to(DEV)
+
+
+# ---------------------------------------------------------------------------
+# MFMA fragment-layout probes (transpose-detecting: asymmetric B — guide G9)
+# ---------------------------------------------------------------------------
+
+def test_mfma_layout_32x32x16():
+    a = rand_bf16(32, 16, seed=1)
+    b = rand_bf16(16, 32, seed=2)
+    got = _C.mfma_probe_32x32x16(a, b)
+    want = a.float() @ b.float()
+    torch.testing.assert_close(got, want, rtol=2e-2, atol=2e-2)
+
+
+def test_mfma_layout_16x16x32():
+    a = rand_bf16(16, 32, seed=3)
+    b = rand_bf16(32, 16, seed=4)
+    got = _C.mfma_probe_16x16x32(a, b)
+    want = a.float() @ b.float()
+    torch.testing.assert_close(got, want, rtol=2e-2, atol=2e-2)
+
+
+# ---------------------------------------------------------------------------
+# Fused optimizers
+# ---------------------------------------------------------------------------
+
+def test_adamw_matches_reference():
+    n = 12345  # not a multiple of 8: exercises the tail path
+    torch.manual_seed(0)
+    master = torch.randn(n, device=DEV)
+    m = torch.rand(n, device=DEV) * 0.1
+    v = torch.rand(n, device=DEV) * 0.01
+    param = master.bfloat16()
+    cm, cmm, cv = master.cpu().clone(), m.cpu().clone(), v.cpu().clone()
+    cparam = cm.clone()
+    for step in (1, 2, 3):
+        g = torch.randn(n).bfloat16()
+        _C.adamw_step_(master, param, g.to(DEV), m, v, 1e-2, 0.9, 0.95, 1e-8, 0.1, step)
+        R.adamw_step(cm, cparam, g.float(), cmm, cv,
+                     lr=1e-2, beta1=0.9, beta2=0.95, eps=1e-8, weight_decay=0.1, step=step)
+    torch.testing.assert_close(master.cpu(), cm, rtol=1e-5, atol=1e-6)
+    torch.testing.assert_close(m.cpu(), cmm, rtol=1e-5, atol=1e-6)
+    torch.testing.assert_close(v.cpu(), cv, rtol=1e-5, atol=1e-7)
+    torch.testing.assert_close(param.cpu().float(), cm, rtol=1e-2, atol=1e-2)
+
+
+def test_nesterov_matches_torch_sgd():
+    n = 1001
+    torch.manual_seed(1)
+    theta = torch.randn(n, device=DEV)
+    mom = torch.zeros(n, device=DEV)
+    p_ref = theta.cpu().clone().requires_grad_(True)
+    opt = torch.optim.SGD([p_ref], lr=0.7, momentum=0.9, nesterov=True)
+    for i in range(4):
+        delta = torch.randn(n).bfloat16()
+        _C.nesterov_step_(theta, delta.to(DEV), mom, 0.7, 0.9)
+        p_ref.grad = -delta.float()
+        opt.step()
+    torch.testing.assert_close(theta.cpu(), p_ref.detach(), rtol=1e-4, atol=1e-4)
+
+
+def test_extract_delta():
+    n = 999
+    a = torch.randn(n, device=DEV)
+    b = torch.randn(n, device=DEV)
+    out = torch.empty(n, dtype=torch.bfloat16, device=DEV)
+    _C.extract_delta(a, b, out)
+    torch.testing.assert_close(out.cpu().float(), (a - b).cpu().bfloat16().float())
+
+
+# ---------------------------------------------------------------------------
+# SwiGLU
+# ---------------------------------------------------------------------------
+
+def test_swiglu_fwd_bwd():
+    g = rand_bf16(64, 256, seed=5, scale=2.0).requires_grad_(True)
+    u = rand_bf16(64, 256, seed=6, scale=2.0).requires_grad_(True)
+    out = ops.swiglu(g, u)
+    want = R.swiglu(g.detach().float(), u.detach().float())
+    torch.testing.assert_close(out.float(), want, rtol=2e-2, atol=2e-2)
+    dout = rand_bf16(64, 256, seed=7)
+    out.backward(dout)
+    gf = g.detach().float().requires_grad_(True)
+    uf = u.detach().float().requires_grad_(True)
+    R.swiglu(gf, uf).backward(dout.float())
+    torch.testing.assert_close(g.grad.float(), gf.grad, rtol=3e-2, atol=3e-2)
+    torch.testing.assert_close(u.grad.float(), uf.grad, rtol=3e-2, atol=3e-2)
+
+
+# ---------------------------------------------------------------------------
+# RMSNorm
+# ---------------------------------------------------------------------------
+
+def test_rmsnorm_fwd_bwd():
+    x = rand_bf16(128, 512, seed=8).requires_grad_(True)
+    w = (torch.randn(512) * 0.1 + 1.0).bfloat16().to(DEV).requires_grad_(True)
+    y = ops.rmsnorm(x, w, 1e-5)
+    want = R.rmsnorm(x.detach().float(), w.detach().float(), 1e-5)
+    torch.testing.assert_close(y.float(), want, rtol=2e-2, atol=2e-2)
+    dy = rand_bf16(128, 512, seed=9)
+    y.backward(dy)
+    xf = x.detach().float().requires_grad_(True)
+    wf = w.detach().float().requires_grad_(True)
+    R.rmsnorm(xf, wf, 1e-5).backward(dy.float())
+    torch.testing.assert_close(x.grad.float(), xf.grad, rtol=3e-2, atol=3e-2)
+    torch.testing.assert_close(w.grad.float(), wf.grad, rtol=3e-2, atol=3e-1)
+
+
+# ---------------------------------------------------------------------------
+# RoPE
+# ---------------------------------------------------------------------------
+
+def test_rope_fwd_bwd():
+    cos, sin = R.rope_cos_sin(128, 64, base=10000.0)
+    cos, sin = cos.to(DEV), sin.to(DEV)
+    q = rand_bf16(2, 4, 128, 64, seed=10).requires_grad_(True)
+    k = rand_bf16(2, 2, 128, 64, seed=11).requires_grad_(True)
+    qo, ko = ops.apply_rope_qk(q, k, cos, sin)
+    qw = R.apply_rope(q.detach().float(), cos.cpu(), sin.cpu())
+    kw = R.apply_rope(k.detach().float(), cos.cpu(), sin.cpu())
+    torch.testing.assert_close(qo.float().cpu(), qw, rtol=2e-2, atol=2e-2)
+    torch.testing.assert_close(ko.float().cpu(), kw, rtol=2e-2, atol=2e-2)
+    (qo.float().pow(2).sum() + ko.float().pow(2).sum()).backward()
+    # gradient of sum of squares through a rotation: dq = 2 * R^-1 R q = 2q
+    torch.testing.assert_close(q.grad.float(), 2 * q.detach().float(), rtol=3e-2, atol=3e-2)
+
+
+# ---------------------------------------------------------------------------
+# Cross-entropy
+# ---------------------------------------------------------------------------
+
+def test_ce_fwd_bwd():
+    torch.manual_seed(2)
+    N, V = 256, 1000
+    logits = rand_bf16(N, V, seed=12, scale=3.0).requires_grad_(True)
+    t = torch.randint(0, V, (N,))
+    t[5] = -100
+    t[77] = -100
+    tg = t.to(DEV)
+    loss = ops.cross_entropy_loss(logits, tg)
+    lf = logits.detach().float().requires_grad_(True)
+    want = torch.nn.functional.cross_entropy(lf, tg, ignore_index=-100)
+    assert abs(float(loss) - float(want)) < 2e-2 * max(1.0, abs(float(want)))
+    loss.backward()
+    want.backward()
+    torch.testing.assert_close(logits.grad.float(), lf.grad, rtol=5e-2, atol=1e-4)
+
+
+# ---------------------------------------------------------------------------
+# Flash attention
+# ---------------------------------------------------------------------------
+
+@pytest.mark.parametrize("hd,hq,hkv", [(128, 8, 2), (64, 4, 4)])
+def test_attention_fwd(hd, hq, hkv):
+    B, S = 2, 256
+    q = rand_bf16(B, hq, S, hd, seed=13)
+    k = rand_bf16(B, hkv, S, hd, seed=14)
+    v = rand_bf16(B, hkv, S, hd, seed=15)
+    o = ops.flash_attention(q, k, v, causal=True)
+    want = R.attention(q.float(), k.float(), v.float(), causal=True)
+    torch.testing.assert_close(o.float(), want, rtol=2e-2, atol=2e-2)
+
+
+@pytest.mark.parametrize("hd,hq,hkv", [(128, 8, 2), (64, 4, 4)])
+def test_attention_bwd(hd, hq, hkv):
+    B, S = 2, 256
+    q = rand_bf16(B, hq, S, hd, seed=16).requires_grad_(True)
+    k = rand_bf16(B, hkv, S, hd, seed=17).requires_grad_(True)
+    v = rand_bf16(B, hkv, S, hd, seed=18).requires_grad_(True)
+    o = ops.flash_attention(q, k, v, causal=True)
+    do = rand_bf16(B, hq, S, hd, seed=19)
+    o.backward(do)
+    qf = q.detach().float().requires_grad_(True)
+    kf = k.detach().float().requires_grad_(True)
+    vf = v.detach().float().requires_grad_(True)
+    R.attention(qf, kf, vf, causal=True).backward(do.float())
+    torch.testing.assert_close(q.grad.float(), qf.grad, rtol=5e-2, atol=5e-2)
+    torch.testing.assert_close(k.grad.float(), kf.grad, rtol=5e-2, atol=5e-2)
+    torch.testing.assert_close(v.grad.float(), vf.grad, rtol=5e-2, atol=5e-2)
+
+
+def test_attention_noncausal():
+    q = rand_bf16(1, 2, 128, 128, seed=20)
+    k = rand_bf16(1, 2, 128, 128, seed=21)
+    v = rand_bf16(1, 2, 128, 128, seed=22)
+    o = ops.flash_attention(q, k, v, causal=False)
+    want = R.attention(q.float(), k.float(), v.float(), causal=False)
+    torch.testing.assert_close(o.float(), want, rtol=2e-2, atol=2e-2)
+
+
+# ---------------------------------------------------------------------------
+# End-to-end: tiny model steps on GPU with native ops
+# ---------------------------------------------------------------------------
+
+def test_model_step_gpu():
+    from hypha_amd import models
+    from hypha_amd.data.synthetic import SyntheticTokens
+    from hypha_amd.parallel import Comm, DiLoCoConfig, DiLoCoWorker, InnerOptConfig
+
+    assert ops.has_native()
+    torch.manual_seed(0)
+    model = models.build("llama-tiny", max_seq_len=256, hidden_size=128, n_heads=2,
+                         n_kv_heads=2)  # head_dim 64
+    w = DiLoCoWorker(model, DiLoCoConfig(h=2, inner=InnerOptConfig(warmup_steps=0)),
+                     comm=Comm(), device=torch.device(DEV))
+    data = SyntheticTokens(512, 128, 2, seed=30)
+    ids, labels = data.next_batch()
+    first = w.train_step(ids, labels)
+    losses = [first]
+    for _ in range(8):
+        losses.append(w.train_step(ids.clone(), labels.clone()))
+        w.maybe_outer_sync()
+    assert all(math.isfinite(l) for l in losses), losses
+    assert losses[-1] < losses[0], losses
