@@ -20,7 +20,7 @@ __global__ void ce_fwd_kernel(const short* __restrict__ logits,
   const long tgt = targets[row];
 
   // online (max, sumexp) per thread
-  float m = -INFINITY, s = 0.f;
+  float m = -1e30f, s = 0.f;
   for (int i = threadIdx.x * 8; i < V; i += blockDim.x * 8) {
     if (i + 8 <= V) {
       s16x8 v8 = *reinterpret_cast<const s16x8*>(lr + i);

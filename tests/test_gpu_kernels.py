@@ -137,8 +137,8 @@ def test_rope_fwd_bwd():
     q = rand_bf16(2, 4, 128, 64, seed=10).requires_grad_(True)
     k = rand_bf16(2, 2, 128, 64, seed=11).requires_grad_(True)
     qo, ko = ops.apply_rope_qk(q, k, cos, sin)
-    qw = R.apply_rope(q.detach().float(), cos.cpu(), sin.cpu())
-    kw = R.apply_rope(k.detach().float(), cos.cpu(), sin.cpu())
+    qw = R.apply_rope(q.detach().float().cpu(), cos.cpu(), sin.cpu())
+    kw = R.apply_rope(k.detach().float().cpu(), cos.cpu(), sin.cpu())
     torch.testing.assert_close(qo.float().cpu(), qw, rtol=2e-2, atol=2e-2)
     torch.testing.assert_close(ko.float().cpu(), kw, rtol=2e-2, atol=2e-2)
     (qo.float().pow(2).sum() + ko.float().pow(2).sum()).backward()
@@ -161,7 +161,7 @@ def test_ce_fwd_bwd():
     loss = ops.cross_entropy_loss(logits, tg)
     lf = logits.detach().float().requires_grad_(True)
     want = torch.nn.functional.cross_entropy(lf, tg, ignore_index=-100)
-    assert abs(float(loss) - float(want)) < 2e-2 * max(1.0, abs(float(want)))
+    assert abs(float(loss.detach()) - float(want.detach())) < 2e-2 * max(1.0, abs(float(want)))
     loss.backward()
     want.backward()
     torch.testing.assert_close(logits.grad.float(), lf.grad, rtol=5e-2, atol=1e-4)
