@@ -56,8 +56,8 @@ PRESETS: dict[str, LlamaConfig] = {
     ),
     # small debug model for CPU tests
     "llama-tiny": LlamaConfig(
-        vocab_size=512, hidden_size=64, n_layers=2, n_heads=4, n_kv_heads=2,
-        ffn_hidden=128, max_seq_len=256, rope_base=10000.0,
+        vocab_size=512, hidden_size=128, n_layers=2, n_heads=2, n_kv_heads=2,
+        ffn_hidden=256, max_seq_len=256, rope_base=10000.0,
     ),
     # ~124M GPT-2-small-scale llama-style model for the CPU plumbing config
     "gpt2-small": LlamaConfig(

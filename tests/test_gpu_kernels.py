@@ -220,8 +220,7 @@ def test_model_step_gpu():
 
     assert ops.has_native()
     torch.manual_seed(0)
-    model = models.build("llama-tiny", max_seq_len=256, hidden_size=128, n_heads=2,
-                         n_kv_heads=2)  # head_dim 64
+    model = models.build("llama-tiny")  # head_dim 64
     w = DiLoCoWorker(model, DiLoCoConfig(h=2, inner=InnerOptConfig(warmup_steps=0)),
                      comm=Comm(), device=torch.device(DEV))
     data = SyntheticTokens(512, 128, 2, seed=30)
