@@ -105,7 +105,8 @@ class FlatParams:
         self.flat_grad.zero_()
 
     def grad_norm(self) -> torch.Tensor:
-        return self.flat_grad.float().norm()
+        # fp32 accumulation without materializing an fp32 copy of the grads
+        return torch.linalg.vector_norm(self.flat_grad, dtype=torch.float32)
 
 
 class DiLoCoWorker:
@@ -172,9 +173,9 @@ class DiLoCoWorker:
         c = self.cfg.inner
         if self.cfg.grad_clip > 0:
             gnorm = self.fp.grad_norm()
-            clip_coef = self.cfg.grad_clip / (gnorm + 1e-6)
-            if float(clip_coef) < 1.0:
-                self.fp.flat_grad.mul_(clip_coef.to(self.fp.flat_grad.dtype))
+            # clamp to 1 and always scale: no host sync, one fused mul
+            clip_coef = (self.cfg.grad_clip / (gnorm + 1e-6)).clamp_(max=1.0)
+            self.fp.flat_grad.mul_(clip_coef.to(self.fp.flat_grad.dtype))
         ops.fused_adamw(
             self.fp.master,
             self.fp.flat,
