@@ -1,0 +1,224 @@
+// pybind11 bindings for the C++ control-plane core (hypha_amd._core).
+// Exposes the scheduler/worker logic classes so the Python test-suite can
+// exercise them deterministically (injected clocks — the analogue of the
+// reference's paused-tokio-time unit tests).
+
+#include <pybind11/functional.h>
+#include <pybind11/pybind11.h>
+#include <pybind11/stl.h>
+
+#include "hypha/auction.h"
+#include "hypha/batch_scheduler.h"
+#include "hypha/json.h"
+#include "hypha/leases.h"
+#include "hypha/resources.h"
+#include "hypha/simulation.h"
+#include "hypha/trackers.h"
+
+namespace py = pybind11;
+using namespace hypha;
+
+PYBIND11_MODULE(_core, m) {
+  m.doc() = "hypha_amd C++ control-plane core";
+
+  // ---- json round-trip (used by tests to validate the wire format) ----
+  m.def("json_roundtrip", [](const std::string& s) { return Json::parse(s).dump(); });
+
+  // ---- resources ----
+  py::class_<Resources>(m, "Resources")
+      .def(py::init([](double gpu, double cpu, double memory, double storage) {
+             return Resources{gpu, cpu, memory, storage};
+           }),
+           py::arg("gpu") = 0, py::arg("cpu") = 0, py::arg("memory") = 0,
+           py::arg("storage") = 0)
+      .def_readwrite("gpu", &Resources::gpu)
+      .def_readwrite("cpu", &Resources::cpu)
+      .def_readwrite("memory", &Resources::memory)
+      .def_readwrite("storage", &Resources::storage)
+      .def("fits_in", &Resources::fits_in)
+      .def("partial_cmp",
+           [](const Resources& a, const Resources& b) -> py::object {
+             auto c = a.partial_cmp(b);
+             if (!c) return py::none();
+             return py::int_(*c);
+           })
+      .def("__add__", &Resources::operator+)
+      .def("__sub__", &Resources::operator-)
+      .def("__eq__", &Resources::operator==);
+
+  py::class_<WeightedResourceEvaluator>(m, "WeightedResourceEvaluator")
+      .def(py::init<>())
+      .def("weighted_units", &WeightedResourceEvaluator::weighted_units)
+      .def("score", &WeightedResourceEvaluator::score);
+
+  // ---- leases ----
+  using StrLedger = Ledger<std::string>;
+  py::class_<StrLedger>(m, "Ledger")
+      .def(py::init([](py::object clock) {
+             if (clock.is_none()) return StrLedger(system_clock_fn());
+             auto fn = clock.cast<std::function<double()>>();
+             return StrLedger(fn);
+           }),
+           py::arg("clock") = py::none())
+      .def("insert", &StrLedger::insert)
+      .def("get",
+           [](StrLedger& l, const std::string& id) -> py::object {
+             auto r = l.get(id);
+             if (!r) return py::none();
+             return py::make_tuple(r->id, r->leasable, r->timeout);
+           })
+      .def("remove", &StrLedger::remove)
+      .def("renew", &StrLedger::renew)
+      .def("list_expired",
+           [](StrLedger& l) {
+             std::vector<std::string> out;
+             for (auto& e : l.list_expired()) out.push_back(e.id);
+             return out;
+           })
+      .def("drain_expired",
+           [](StrLedger& l) {
+             std::vector<std::string> out;
+             for (auto& e : l.drain_expired()) out.push_back(e.id);
+             return out;
+           })
+      .def("size", &StrLedger::size)
+      .def("ids", &StrLedger::ids);
+
+  // ---- trackers ----
+  py::class_<SliceTracker>(m, "SliceTracker")
+      .def(py::init<const std::string&, int>())
+      .def("next",
+           [](SliceTracker& t, const std::string& peer) {
+             auto a = t.next(peer);
+             return py::make_tuple(a.index, a.epoch);
+           })
+      .def("set_statistic", &SliceTracker::set_statistic)
+      .def("remove_worker", &SliceTracker::remove_worker)
+      .def("available_count", &SliceTracker::available_count)
+      .def("owner", &SliceTracker::owner)
+      .def_property_readonly("epoch", &SliceTracker::epoch);
+
+  py::class_<ProgressTracker>(m, "ProgressTracker")
+      .def(py::init<int64_t, int64_t>())
+      .def("on_status", &ProgressTracker::on_status)
+      .def("next_round", &ProgressTracker::next_round)
+      .def("training_finished", &ProgressTracker::training_finished)
+      .def_property_readonly("counter", &ProgressTracker::counter)
+      .def_property_readonly("round", &ProgressTracker::round);
+
+  py::class_<RunningMean>(m, "RunningMean")
+      .def(py::init<>())
+      .def("record", &RunningMean::record)
+      .def_readonly("mean", &RunningMean::mean)
+      .def_readonly("count", &RunningMean::count);
+
+  // ---- simulation ----
+  py::class_<Projection>(m, "Projection")
+      .def_readonly("time_ms", &Projection::time_ms)
+      .def_readonly("remaining", &Projection::remaining)
+      .def_readonly("batches_per_worker", &Projection::batches_per_worker)
+      .def_readonly("capped", &Projection::capped);
+
+  // ---- batch scheduler FSM ----
+  py::enum_<WorkerState>(m, "WorkerState")
+      .value("Training", WorkerState::Training)
+      .value("UpdateScheduled", WorkerState::UpdateScheduled)
+      .value("Updating", WorkerState::Updating)
+      .value("Done", WorkerState::Done);
+
+  py::class_<BatchScheduler>(m, "BatchScheduler")
+      .def(py::init([](int64_t samples_per_round, int64_t rounds, py::object clock) {
+             if (clock.is_none())
+               return BatchScheduler(samples_per_round, rounds);
+             auto fn = clock.cast<std::function<double()>>();
+             return BatchScheduler(samples_per_round, rounds, fn);
+           }),
+           py::arg("samples_per_round"), py::arg("rounds"),
+           py::arg("clock") = py::none())
+      .def("add_worker", &BatchScheduler::add_worker)
+      .def("remove_worker", &BatchScheduler::remove_worker)
+      .def("worker_state", &BatchScheduler::worker_state)
+      .def_property_readonly("round", &BatchScheduler::round)
+      .def_property_readonly("counter", &BatchScheduler::counter)
+      .def("finished", &BatchScheduler::finished)
+      .def("handle",
+           [](BatchScheduler& s, const std::string& peer, const std::string& kind,
+              int64_t batch_size) {
+             Progress p{};
+             if (kind == "status") {
+               p.kind = Progress::Status;
+               p.batch_size = batch_size;
+             } else if (kind == "metrics") {
+               p.kind = Progress::Metrics;
+             } else if (kind == "update") {
+               p.kind = Progress::Update;
+             } else if (kind == "updated") {
+               p.kind = Progress::Updated;
+             } else if (kind == "update-received") {
+               p.kind = Progress::UpdateReceived;
+             } else {
+               throw std::runtime_error("bad kind " + kind);
+             }
+             auto r = s.handle(peer, p);
+             const char* names[] = {"ok", "continue", "schedule-update", "done",
+                                    "error"};
+             return py::make_tuple(names[(int)r.kind], r.counter);
+           },
+           py::arg("peer"), py::arg("kind"), py::arg("batch_size") = 0);
+
+  // ---- auction ----
+  py::class_<PriceRange>(m, "PriceRange")
+      .def(py::init([](double bid, double mx) { return PriceRange{bid, mx}; }),
+           py::arg("bid"), py::arg("max"))
+      .def_readwrite("bid", &PriceRange::bid)
+      .def_readwrite("max", &PriceRange::max);
+
+  py::class_<WorkerRequest>(m, "WorkerRequest")
+      .def(py::init([](std::string id, std::string scheduler, Resources r,
+                       std::vector<std::string> execs, double bid, double timeout) {
+             return WorkerRequest{std::move(id), std::move(scheduler), r,
+                                  std::move(execs), bid, timeout};
+           }),
+           py::arg("id"), py::arg("scheduler"), py::arg("resources"),
+           py::arg("executors"), py::arg("bid"), py::arg("timeout_s") = 5.0)
+      .def_readwrite("id", &WorkerRequest::id)
+      .def_readwrite("bid", &WorkerRequest::bid)
+      .def_readwrite("resources", &WorkerRequest::resources);
+
+  py::class_<WorkerOffer>(m, "WorkerOffer")
+      .def(py::init([](std::string id, std::string request_id, std::string worker,
+                       double price, Resources r, double expires_at) {
+             return WorkerOffer{std::move(id), std::move(request_id),
+                                std::move(worker), price, r, expires_at};
+           }),
+           py::arg("id"), py::arg("request_id"), py::arg("worker"),
+           py::arg("price"), py::arg("resources"), py::arg("expires_at"))
+      .def_readwrite("id", &WorkerOffer::id)
+      .def_readwrite("worker", &WorkerOffer::worker)
+      .def_readwrite("price", &WorkerOffer::price);
+
+  py::class_<GreedyOfferAggregator>(m, "GreedyOfferAggregator")
+      .def(py::init<size_t, PriceRange, double>())
+      .def("add", &GreedyOfferAggregator::add)
+      .def("finalize", &GreedyOfferAggregator::finalize)
+      .def_property_readonly("deadline", &GreedyOfferAggregator::deadline);
+
+  py::class_<OfferPolicy>(m, "OfferPolicy")
+      .def(py::init([](double price, double floor, std::vector<std::string> execs) {
+             return OfferPolicy{price, floor, std::move(execs)};
+           }),
+           py::arg("price"), py::arg("floor"), py::arg("supported_executors"));
+
+  py::class_<ArbiterDecision>(m, "ArbiterDecision")
+      .def_readonly("request", &ArbiterDecision::request)
+      .def_readonly("offer_price", &ArbiterDecision::offer_price);
+
+  m.def("select_requests", &select_requests);
+
+  py::class_<StaticResourceManager>(m, "StaticResourceManager")
+      .def(py::init<Resources>())
+      .def("reserve", &StaticResourceManager::reserve)
+      .def("release", &StaticResourceManager::release)
+      .def("available", &StaticResourceManager::available)
+      .def("total", &StaticResourceManager::total);
+}

@@ -1,0 +1,232 @@
+"""C++ control-plane core tests (hypha_amd._core) with injected clocks —
+the deterministic-time analogue of the reference's paused-tokio unit tests
+(batch_scheduler.rs:249-483, tracker/*.rs, simulation.rs:71-136,
+leases/src/lib.rs)."""
+
+import json
+
+import pytest
+
+core = pytest.importorskip("hypha_amd._core")
+
+
+class FakeClock:
+    def __init__(self):
+        self.t = 0.0
+
+    def __call__(self):
+        return self.t
+
+    def advance(self, dt):
+        self.t += dt
+
+
+# ---------------------------------------------------------------------------
+# json
+# ---------------------------------------------------------------------------
+
+def test_json_roundtrip():
+    payload = {"a": 1, "b": [1.5, "x", None, True], "c": {"d": "esc\"\n"}}
+    out = json.loads(core.json_roundtrip(json.dumps(payload)))
+    assert out == payload
+
+
+# ---------------------------------------------------------------------------
+# resources
+# ---------------------------------------------------------------------------
+
+def test_resources_partial_order():
+    a = core.Resources(1, 2, 3, 4)
+    b = core.Resources(2, 2, 3, 4)
+    mixed = core.Resources(2, 1, 3, 4)
+    assert a.partial_cmp(b) == -1
+    assert b.partial_cmp(a) == 1
+    assert a.partial_cmp(a) == 0
+    assert a.partial_cmp(mixed) is None  # incomparable (lib.rs:123-143)
+    assert a.fits_in(b)
+    assert not b.fits_in(a)
+
+
+def test_evaluator_default_weights():
+    ev = core.WeightedResourceEvaluator()
+    r = core.Resources(1, 1, 10, 100)
+    # gpu 25 + cpu 1 + mem 10*0.1 + storage 100*0.01 = 28
+    assert ev.weighted_units(r) == pytest.approx(28.0)
+    assert ev.score(56.0, r) == pytest.approx(2.0)
+
+
+# ---------------------------------------------------------------------------
+# leases
+# ---------------------------------------------------------------------------
+
+def test_ledger_lifecycle():
+    clk = FakeClock()
+    led = core.Ledger(clock=clk)
+    led.insert("l1", "job-a", 10.0)
+    assert led.get("l1")[1] == "job-a"
+    clk.advance(8)
+    assert led.renew("l1", 10.0)  # renew = now + duration
+    clk.advance(9)
+    assert led.list_expired() == []  # 8 + 10 = 18 > 17
+    clk.advance(2)
+    assert led.list_expired() == ["l1"]
+    assert not led.renew("l1", 10.0)  # expired leases cannot renew
+    assert led.drain_expired() == ["l1"]
+    assert led.size() == 0
+
+
+# ---------------------------------------------------------------------------
+# slice tracker
+# ---------------------------------------------------------------------------
+
+def test_slices_round_robin_and_epoch():
+    t = core.SliceTracker("mnist", 4)
+    seen = [t.next("w1")[0], t.next("w2")[0], t.next("w1")[0], t.next("w2")[0]]
+    assert sorted(seen) == [0, 1, 2, 3]
+    assert t.epoch == 0
+    # exhausted, equal (unknown) stats -> no steal target beats requester -> rollover
+    idx, epoch = t.next("w1")
+    assert epoch == 1
+
+
+def test_slices_cache_steal_from_slowest():
+    t = core.SliceTracker("ds", 2)
+    t.set_statistic("slow", 1000.0)
+    t.set_statistic("fast", 10.0)
+    a = t.next("slow")[0]
+    b = t.next("slow")[0]
+    # fast worker arrives with nothing left: steals slow's latest slice
+    idx, epoch = t.next("fast")
+    assert epoch == 0  # same epoch: stolen, not rolled over
+    assert idx == b
+    assert t.owner(idx) == "fast"
+
+
+def test_slices_remove_worker_reclaims():
+    t = core.SliceTracker("ds", 3)
+    t.next("w1")
+    t.next("w1")
+    assert t.available_count() == 1
+    t.remove_worker("w1")
+    assert t.available_count() == 3
+
+
+# ---------------------------------------------------------------------------
+# progress tracker
+# ---------------------------------------------------------------------------
+
+def test_progress_counter_and_rounds():
+    p = core.ProgressTracker(100, 2)
+    p.on_status(60)
+    p.on_status(60)
+    assert p.counter == -20
+    assert not p.training_finished()
+    p.next_round()
+    assert p.counter == 100 and p.round == 1
+    p.next_round()
+    assert p.training_finished()
+
+
+# ---------------------------------------------------------------------------
+# batch scheduler FSM (the reference's scripted multi-worker sequences)
+# ---------------------------------------------------------------------------
+
+def test_fsm_single_worker_round():
+    clk = FakeClock()
+    s = core.BatchScheduler(4, 1, clock=clk)
+    s.add_worker("w", 2)
+    # first status: no timing stats yet -> simulation can't fit -> continue
+    kind, _ = s.handle("w", "status", 2)
+    assert kind == "continue"
+    clk.advance(0.1)
+    # second status: mean known (100ms); remaining counter 0 after this
+    kind, counter = s.handle("w", "status", 2)
+    assert kind == "schedule-update"
+    assert s.worker_state("w") == core.WorkerState.UpdateScheduled
+    kind, _ = s.handle("w", "update", 0)
+    assert s.worker_state("w") == core.WorkerState.Updating
+    # aggregator applied the round
+    s.handle("aggregator", "updated", 0)
+    kind, _ = s.handle("w", "update-received", 0)
+    assert kind == "done"  # 1 round configured
+    assert s.worker_state("w") == core.WorkerState.Done
+    assert s.finished()
+
+
+def test_fsm_two_workers_heterogeneous():
+    clk = FakeClock()
+    s = core.BatchScheduler(12, 2, clock=clk)
+    s.add_worker("fast", 2)
+    s.add_worker("slow", 1)
+    # drive interleaved statuses (fast = 50ms/batch, slow = 200ms/batch)
+    # until each worker receives its personal ScheduleUpdate counter
+    sched = {}
+    for i in range(40):
+        clk.advance(0.05)
+        if "fast" not in sched:
+            k, c = s.handle("fast", "status", 2)
+            if k == "schedule-update":
+                sched["fast"] = c
+        if i % 4 == 3 and "slow" not in sched:
+            k, c = s.handle("slow", "status", 1)
+            if k == "schedule-update":
+                sched["slow"] = c
+        if len(sched) == 2:
+            break
+    assert set(sched) == {"fast", "slow"}
+    # heterogeneous shares: the faster worker is never assigned less work
+    assert sched["fast"] >= sched["slow"]
+    s.handle("fast", "update", 0)
+    assert s.worker_state("fast") == core.WorkerState.Updating
+    s.handle("slow", "update", 0)
+    s.handle("agg", "updated", 0)
+    assert s.handle("fast", "update-received", 0)[0] == "continue"
+    assert s.handle("slow", "update-received", 0)[0] == "continue"
+    assert s.round == 1
+    assert s.worker_state("fast") == core.WorkerState.Training
+
+
+# ---------------------------------------------------------------------------
+# auction
+# ---------------------------------------------------------------------------
+
+def test_aggregator_ranks_by_score_and_caps_price():
+    agg = core.GreedyOfferAggregator(2, core.PriceRange(bid=1.0, max=5.0), 100.0)
+    r = core.Resources(1, 8, 32, 100)
+    assert not agg.add(core.WorkerOffer("l3", "rq", "w-overpriced", 9.0, r, 50.0), 0.0)  # > max
+    assert not agg.add(core.WorkerOffer("l1", "rq", "w-expensive", 4.0, r, 50.0), 0.0)
+    full = agg.add(core.WorkerOffer("l2", "rq", "w-cheap", 1.0, r, 50.0), 0.0)
+    assert full  # capacity 2 reached -> early return (allocator.rs:316-419)
+    best = agg.finalize()
+    assert [o.worker for o in best] == ["w-cheap", "w-expensive"]  # score-ranked
+
+
+def test_aggregator_dedups_per_worker_and_shrinks_deadline():
+    agg = core.GreedyOfferAggregator(3, core.PriceRange(bid=1.0, max=5.0), 100.0)
+    r = core.Resources(1, 1, 1, 1)
+    agg.add(core.WorkerOffer("l1", "rq", "w1", 1.0, r, 7.0), 0.0)
+    assert agg.deadline == 7.0  # shrank to earliest offer expiry
+    agg.add(core.WorkerOffer("l2", "rq", "w1", 0.5, r, 9.0), 0.0)  # dup ignored
+    assert len(agg.finalize()) == 1
+
+
+def test_arbiter_select_requests():
+    policy = core.OfferPolicy(1.0, 0.5, ["diloco-transformer"])
+    avail = core.Resources(2, 16, 64, 200)
+    ads = [
+        core.WorkerRequest("r1", "s", core.Resources(1, 8, 32, 100), ["diloco-transformer"], 2.0),
+        core.WorkerRequest("r2", "s", core.Resources(1, 8, 32, 100), ["diloco-transformer"], 0.1),  # below floor
+        core.WorkerRequest("r3", "s", core.Resources(8, 8, 32, 100), ["diloco-transformer"], 5.0),  # doesn't fit
+        core.WorkerRequest("r4", "s", core.Resources(1, 8, 32, 100), ["exotic-executor"], 3.0),  # unsupported
+        core.WorkerRequest("r5", "s", core.Resources(1, 8, 32, 100), ["diloco-transformer"], 1.0),
+    ]
+    picked = core.select_requests(ads, policy, avail)
+    assert [d.request.id for d in picked] == ["r1", "r5"]  # best-paying first, greedy fit
+
+
+def test_static_resource_manager():
+    m = core.StaticResourceManager(core.Resources(2, 8, 32, 100))
+    assert m.reserve(core.Resources(1, 4, 16, 50))
+    assert not m.reserve(core.Resources(2, 1, 1, 1))  # double-checked
+    m.release(core.Resources(1, 4, 16, 50))
+    assert m.reserve(core.Resources(2, 8, 32, 100))
