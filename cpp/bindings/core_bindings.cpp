@@ -9,14 +9,57 @@
 
 #include "hypha/auction.h"
 #include "hypha/batch_scheduler.h"
+#include "hypha/gateway.h"
 #include "hypha/json.h"
 #include "hypha/leases.h"
+#include "hypha/net.h"
 #include "hypha/resources.h"
 #include "hypha/simulation.h"
 #include "hypha/trackers.h"
 
 namespace py = pybind11;
 using namespace hypha;
+
+// Json <-> Python object conversion for the net bindings
+static hypha::Json py_to_json(const py::object& o) {
+  if (o.is_none()) return hypha::Json(nullptr);
+  if (py::isinstance<py::bool_>(o)) return hypha::Json(o.cast<bool>());
+  if (py::isinstance<py::int_>(o)) return hypha::Json((int64_t)o.cast<int64_t>());
+  if (py::isinstance<py::float_>(o)) return hypha::Json(o.cast<double>());
+  if (py::isinstance<py::str>(o)) return hypha::Json(o.cast<std::string>());
+  if (py::isinstance<py::list>(o) || py::isinstance<py::tuple>(o)) {
+    hypha::JsonArray arr;
+    for (auto item : o.cast<py::sequence>()) arr.push_back(py_to_json(py::reinterpret_borrow<py::object>(item)));
+    return hypha::Json(std::move(arr));
+  }
+  if (py::isinstance<py::dict>(o)) {
+    hypha::JsonObject obj;
+    for (auto item : o.cast<py::dict>())
+      obj[item.first.cast<std::string>()] =
+          py_to_json(py::reinterpret_borrow<py::object>(item.second));
+    return hypha::Json(std::move(obj));
+  }
+  throw std::runtime_error("unsupported python type for json");
+}
+
+static py::object json_to_py(const hypha::Json& j) {
+  if (j.is_null()) return py::none();
+  if (j.is_bool()) return py::bool_(j.as_bool());
+  if (j.is_number()) {
+    double d = j.as_double();
+    if (d == (int64_t)d) return py::int_((int64_t)d);
+    return py::float_(d);
+  }
+  if (j.is_string()) return py::str(j.as_string());
+  if (j.is_array()) {
+    py::list l;
+    for (auto& e : j.as_array()) l.append(json_to_py(e));
+    return l;
+  }
+  py::dict d;
+  for (auto& [k, v] : j.as_object()) d[py::str(k)] = json_to_py(v);
+  return d;
+}
 
 PYBIND11_MODULE(_core, m) {
   m.doc() = "hypha_amd C++ control-plane core";
@@ -214,6 +257,92 @@ PYBIND11_MODULE(_core, m) {
       .def_readonly("offer_price", &ArbiterDecision::offer_price);
 
   m.def("select_requests", &select_requests);
+
+  // ---- in-process control-plane networking (multi-peer tests) ----
+  py::class_<Gateway>(m, "Gateway")
+      .def(py::init<>())
+      .def("start", &Gateway::start, py::arg("port") = 0,
+           py::call_guard<py::gil_scoped_release>())
+      .def("stop", &Gateway::stop, py::call_guard<py::gil_scoped_release>())
+      .def_property_readonly("port", &Gateway::port);
+
+  py::class_<Node>(m, "Node")
+      .def(py::init<std::string, std::string, int>(), py::arg("name"),
+           py::arg("gateway_host") = "127.0.0.1", py::arg("gateway_port") = 0)
+      .def("start", &Node::start, py::arg("port") = 0,
+           py::call_guard<py::gil_scoped_release>())
+      .def("stop", &Node::stop, py::call_guard<py::gil_scoped_release>())
+      .def_property_readonly("port", &Node::port)
+      .def("on",
+           [](Node& n, const std::string& type, py::function cb) {
+             // shared_ptr so std::function copies in server threads never
+             // touch Python refcounts without the GIL
+             auto cbp = std::make_shared<py::function>(std::move(cb));
+             n.on(type, [cbp](const std::string& from, const Json& body) -> Json {
+               py::gil_scoped_acquire gil;
+               try {
+                 py::object r = (*cbp)(from, json_to_py(body));
+                 return py_to_json(r);
+               } catch (py::error_already_set& e) {
+                 std::string msg = e.what();
+                 e.restore();
+                 PyErr_Clear();
+                 throw std::runtime_error(msg);
+               }
+             });
+           })
+      .def("request",
+           [](Node& n, const std::string& peer, const std::string& type, py::object body,
+              double timeout_s) {
+             Json b = py_to_json(body);
+             Json r;
+             {
+               py::gil_scoped_release rel;
+               r = n.request(peer, type, b, timeout_s);
+             }
+             return json_to_py(r);
+           },
+           py::arg("peer"), py::arg("type"), py::arg("body"), py::arg("timeout_s") = 10.0)
+      .def("publish",
+           [](Node& n, const std::string& topic, py::object data) {
+             Json d = py_to_json(data);
+             py::gil_scoped_release rel;
+             n.publish(topic, d);
+           })
+      .def("subscribe",
+           [](Node& n, const std::string& topic, py::function cb) {
+             auto cbp = std::make_shared<py::function>(std::move(cb));
+             n.subscribe(topic, [cbp](const std::string& from, const Json& data) {
+               py::gil_scoped_acquire gil;
+               try {
+                 (*cbp)(from, json_to_py(data));
+               } catch (py::error_already_set& e) {
+                 e.restore();
+                 PyErr_Clear();
+               }
+             });
+           })
+      .def("kv_put",
+           [](Node& n, const std::string& k, py::object v) {
+             Json j = py_to_json(v);
+             py::gil_scoped_release rel;
+             n.kv_put(k, j);
+           })
+      .def("kv_get",
+           [](Node& n, const std::string& k) -> py::object {
+             std::optional<Json> r;
+             {
+               py::gil_scoped_release rel;
+               r = n.kv_get(k);
+             }
+             if (!r) return py::none();
+             return json_to_py(*r);
+           })
+      .def("resolve", [](Node& n, const std::string& p) -> py::object {
+        auto r = n.resolve(p);
+        if (!r) return py::none();
+        return py::str(*r);
+      });
 
   py::class_<StaticResourceManager>(m, "StaticResourceManager")
       .def(py::init<Resources>())

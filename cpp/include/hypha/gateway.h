@@ -1,0 +1,159 @@
+// Gateway broker: peer registry (KV/discovery), topic pub/sub fan-out and
+// health — the role the reference's gateway plays for the swarm (relay +
+// DHT + gossipsub participant, crates/gateway/src/network.rs:41-49), with
+// DHT/gossip semantics collapsed into a broker as SURVEY.md §7 step 2 allows.
+#pragma once
+
+#include <sys/socket.h>
+#include <unistd.h>
+
+#include <atomic>
+#include <map>
+#include <memory>
+#include <mutex>
+#include <set>
+#include <string>
+#include <thread>
+#include <vector>
+
+#include "json.h"
+#include "net.h"
+
+namespace hypha {
+
+class Gateway {
+ public:
+  ~Gateway() { stop(); }
+
+  void start(int port = 0) {
+    listen_fd_ = tcp_listen(port);
+    if (listen_fd_ < 0) throw std::runtime_error("gateway: cannot listen");
+    port_ = listen_port(listen_fd_);
+    running_ = true;
+    accept_thread_ = std::thread([this] { accept_loop(); });
+  }
+
+  void stop() {
+    if (!running_.exchange(false)) return;
+    if (listen_fd_ >= 0) {
+      ::shutdown(listen_fd_, 2);
+    }
+    if (accept_thread_.joinable()) accept_thread_.join();
+    std::lock_guard<std::mutex> lk(mu_);
+    for (auto& [_, s] : peers_) s->close_now();
+    peers_.clear();
+    subs_.clear();
+  }
+
+  int port() const { return port_; }
+
+ private:
+  void accept_loop() {
+    while (running_) {
+      int fd = ::accept(listen_fd_, nullptr, nullptr);
+      if (fd < 0) break;
+      std::thread([this, fd] { handle_conn(fd); }).detach();
+    }
+    ::close(listen_fd_);
+    listen_fd_ = -1;
+  }
+
+  void handle_conn(int fd) {
+    auto sock = std::make_shared<MsgSocket>(fd);
+    std::string peer;  // set once registered (persistent connection)
+    while (running_) {
+      auto msg = sock->recv_json();
+      if (!msg) break;
+      std::string kind = msg->get_or("kind", Json("")).as_string();
+      if (kind == "register") {
+        peer = msg->at("peer").as_string();
+        std::lock_guard<std::mutex> lk(mu_);
+        peers_[peer] = sock;
+        kv_["addr:" + peer] = msg->at("addr");
+      } else if (kind == "subscribe") {
+        std::lock_guard<std::mutex> lk(mu_);
+        subs_[msg->at("topic").as_string()].insert(peer);
+      } else if (kind == "request") {
+        Json resp;
+        resp["kind"] = "response";
+        resp["ok"] = true;
+        resp["body"] = handle_request(msg->get_or("from", Json("")).as_string(),
+                                      msg->get_or("type", Json("")).as_string(),
+                                      msg->get_or("body", Json(JsonObject{})));
+        sock->send_json(resp);
+      } else {
+        break;
+      }
+    }
+    if (!peer.empty()) {
+      std::lock_guard<std::mutex> lk(mu_);
+      peers_.erase(peer);
+      for (auto& [_, set] : subs_) set.erase(peer);
+    }
+  }
+
+  Json handle_request(const std::string& from, const std::string& type, const Json& body) {
+    if (type == "publish") {
+      std::string topic = body.at("topic").as_string();
+      std::vector<std::shared_ptr<MsgSocket>> targets;
+      {
+        std::lock_guard<std::mutex> lk(mu_);
+        auto it = subs_.find(topic);
+        if (it != subs_.end())
+          for (auto& p : it->second) {
+            auto pit = peers_.find(p);
+            if (pit != peers_.end() && p != from) targets.push_back(pit->second);
+          }
+      }
+      Json ev;
+      ev["kind"] = "pub";
+      ev["topic"] = topic;
+      ev["from"] = from;
+      ev["data"] = body.at("data");
+      for (auto& t : targets) t->send_json(ev);
+      Json r;
+      r["delivered"] = (int64_t)targets.size();
+      return r;
+    }
+    if (type == "kv_put") {
+      std::lock_guard<std::mutex> lk(mu_);
+      kv_[body.at("key").as_string()] = body.at("value");
+      return Json(JsonObject{});
+    }
+    if (type == "kv_get") {
+      std::lock_guard<std::mutex> lk(mu_);
+      auto it = kv_.find(body.at("key").as_string());
+      Json r;
+      r["found"] = it != kv_.end();
+      if (it != kv_.end()) r["value"] = it->second;
+      return r;
+    }
+    if (type == "health") {
+      Json r;
+      r["healthy"] = true;
+      return r;
+    }
+    if (type == "peers") {
+      JsonArray arr;
+      std::lock_guard<std::mutex> lk(mu_);
+      for (auto& [p, _] : peers_) arr.push_back(Json(p));
+      Json r;
+      r["peers"] = arr;
+      return r;
+    }
+    Json r;
+    r["error"] = "unknown request " + type;
+    return r;
+  }
+
+  int listen_fd_ = -1;
+  int port_ = 0;
+  std::atomic<bool> running_{false};
+  std::thread accept_thread_;
+  std::mutex mu_;
+  std::map<std::string, std::shared_ptr<MsgSocket>> peers_;
+  std::map<std::string, std::set<std::string>> subs_;
+  std::map<std::string, Json> kv_;
+};
+
+}  // namespace hypha

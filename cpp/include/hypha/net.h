@@ -1,0 +1,115 @@
+// Control-plane networking: length-prefixed JSON messages over TCP.
+//
+// Replaces the reference's libp2p stack (crates/network) with a slim native
+// layer exposing the same five interaction patterns:
+//   * request/response with typed handler registration (request_response.rs)
+//   * topic pub/sub via the gateway broker (gossipsub.rs, topic hypha/worker)
+//   * KV/discovery registry hosted by the gateway (kad.rs DHT semantics)
+//   * raw byte streams for tensor files (stream_push.rs / stream_pull.rs)
+//   * health probes (messages/src/lib.rs health codec)
+// Bulk tensor movement on-node is RCCL over xGMI (hypha_amd.parallel.comm);
+// this layer carries control messages and the WAN/file tensor path.
+#pragma once
+
+#include <atomic>
+#include <functional>
+#include <map>
+#include <memory>
+#include <mutex>
+#include <optional>
+#include <string>
+#include <thread>
+#include <vector>
+
+#include "json.h"
+
+namespace hypha {
+
+// ---- low-level framed socket ----------------------------------------------
+
+class MsgSocket {
+ public:
+  explicit MsgSocket(int fd) : fd_(fd) {}
+  ~MsgSocket();
+  MsgSocket(const MsgSocket&) = delete;
+
+  bool send_json(const Json& j);
+  std::optional<Json> recv_json();  // nullopt on EOF/error
+  bool send_raw(const char* data, size_t n);
+  bool recv_raw(char* data, size_t n);
+  void close_now();
+  int fd() const { return fd_; }
+
+ private:
+  int fd_;
+  std::mutex write_mu_;
+};
+
+int tcp_connect(const std::string& host, int port, double timeout_s = 5.0);
+int tcp_listen(int port);  // returns listen fd; port 0 = ephemeral
+int listen_port(int listen_fd);
+
+// ---- node ------------------------------------------------------------------
+
+// A Node serves typed requests on its own TCP port and talks to peers either
+// directly (addresses from the gateway registry) or via the gateway broker
+// (pub/sub, KV). Handler threads are per-connection (control-plane rates).
+class Node {
+ public:
+  using Handler = std::function<Json(const std::string& from, const Json& body)>;
+  // Stream handler: gets header + the connection for raw payload IO.
+  using StreamHandler =
+      std::function<void(const std::string& from, const Json& header, MsgSocket& sock)>;
+
+  Node(std::string name, std::string gateway_host, int gateway_port);
+  ~Node();
+
+  // start serving on `port` (0 = ephemeral) and register with the gateway
+  void start(int port = 0);
+  void stop();
+  int port() const { return port_; }
+  const std::string& name() const { return name_; }
+
+  void on(const std::string& type, Handler h);
+  void on_stream(const std::string& type, StreamHandler h);
+
+  // direct request to a named peer (address resolved via gateway registry)
+  Json request(const std::string& peer, const std::string& type, const Json& body,
+               double timeout_s = 10.0);
+  // open a raw stream to a peer: send header, then return the socket for
+  // payload bytes (caller closes)
+  std::unique_ptr<MsgSocket> open_stream(const std::string& peer, const std::string& type,
+                                         const Json& header);
+
+  // gateway-brokered pub/sub + KV
+  void publish(const std::string& topic, const Json& data);
+  void subscribe(const std::string& topic, std::function<void(const std::string&, const Json&)> cb);
+  void kv_put(const std::string& key, const Json& value);
+  std::optional<Json> kv_get(const std::string& key);
+
+  std::optional<std::string> resolve(const std::string& peer);
+
+ private:
+  void accept_loop();
+  void handle_conn(int fd);
+  void gateway_listen_loop();
+  Json gateway_request(const std::string& type, const Json& body);
+
+  std::string name_, gw_host_;
+  int gw_port_;
+  int listen_fd_ = -1;
+  int port_ = 0;
+  std::atomic<bool> running_{false};
+  std::thread accept_thread_, gw_thread_;
+  std::mutex mu_;
+  std::map<std::string, Handler> handlers_;
+  std::map<std::string, StreamHandler> stream_handlers_;
+  std::map<std::string, std::function<void(const std::string&, const Json&)>> subs_;
+  std::map<std::string, std::string> addr_cache_;
+  std::unique_ptr<MsgSocket> gw_sock_;       // persistent broker connection
+  std::mutex gw_mu_;
+  std::map<int64_t, std::shared_ptr<std::pair<std::mutex, Json>>> pending_;
+  std::vector<std::thread> conn_threads_;
+};
+
+}  // namespace hypha

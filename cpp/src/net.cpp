@@ -1,0 +1,355 @@
+#include "hypha/net.h"
+
+#include <arpa/inet.h>
+#include <netdb.h>
+#include <netinet/in.h>
+#include <netinet/tcp.h>
+#include <string.h>
+#include <sys/socket.h>
+#include <sys/types.h>
+#include <unistd.h>
+
+#include <condition_variable>
+#include <stdexcept>
+
+namespace hypha {
+
+// ---- MsgSocket -------------------------------------------------------------
+
+MsgSocket::~MsgSocket() { close_now(); }
+
+void MsgSocket::close_now() {
+  if (fd_ >= 0) {
+    ::shutdown(fd_, SHUT_RDWR);
+    ::close(fd_);
+    fd_ = -1;
+  }
+}
+
+static bool write_all(int fd, const char* p, size_t n) {
+  while (n > 0) {
+    ssize_t w = ::send(fd, p, n, MSG_NOSIGNAL);
+    if (w <= 0) return false;
+    p += w;
+    n -= w;
+  }
+  return true;
+}
+
+static bool read_all(int fd, char* p, size_t n) {
+  while (n > 0) {
+    ssize_t r = ::recv(fd, p, n, 0);
+    if (r <= 0) return false;
+    p += r;
+    n -= r;
+  }
+  return true;
+}
+
+bool MsgSocket::send_json(const Json& j) {
+  std::string s = j.dump();
+  uint32_t len = htonl((uint32_t)s.size());
+  std::lock_guard<std::mutex> lk(write_mu_);
+  if (!write_all(fd_, (const char*)&len, 4)) return false;
+  return write_all(fd_, s.data(), s.size());
+}
+
+std::optional<Json> MsgSocket::recv_json() {
+  uint32_t len_be;
+  if (!read_all(fd_, (char*)&len_be, 4)) return std::nullopt;
+  uint32_t len = ntohl(len_be);
+  if (len > 256u * 1024u * 1024u) return std::nullopt;
+  std::string s(len, '\0');
+  if (!read_all(fd_, s.data(), len)) return std::nullopt;
+  try {
+    return Json::parse(s);
+  } catch (...) {
+    return std::nullopt;
+  }
+}
+
+bool MsgSocket::send_raw(const char* data, size_t n) { return write_all(fd_, data, n); }
+bool MsgSocket::recv_raw(char* data, size_t n) { return read_all(fd_, data, n); }
+
+int tcp_connect(const std::string& host, int port, double timeout_s) {
+  struct addrinfo hints = {}, *res = nullptr;
+  hints.ai_family = AF_INET;
+  hints.ai_socktype = SOCK_STREAM;
+  std::string ps = std::to_string(port);
+  if (getaddrinfo(host.c_str(), ps.c_str(), &hints, &res) != 0) return -1;
+  int fd = ::socket(res->ai_family, res->ai_socktype, res->ai_protocol);
+  if (fd < 0) {
+    freeaddrinfo(res);
+    return -1;
+  }
+  struct timeval tv;
+  tv.tv_sec = (long)timeout_s;
+  tv.tv_usec = (long)((timeout_s - tv.tv_sec) * 1e6);
+  setsockopt(fd, SOL_SOCKET, SO_RCVTIMEO, &tv, sizeof tv);
+  setsockopt(fd, SOL_SOCKET, SO_SNDTIMEO, &tv, sizeof tv);
+  int one = 1;
+  setsockopt(fd, IPPROTO_TCP, TCP_NODELAY, &one, sizeof one);
+  if (::connect(fd, res->ai_addr, res->ai_addrlen) != 0) {
+    ::close(fd);
+    freeaddrinfo(res);
+    return -1;
+  }
+  freeaddrinfo(res);
+  return fd;
+}
+
+int tcp_listen(int port) {
+  int fd = ::socket(AF_INET, SOCK_STREAM, 0);
+  if (fd < 0) return -1;
+  int one = 1;
+  setsockopt(fd, SOL_SOCKET, SO_REUSEADDR, &one, sizeof one);
+  struct sockaddr_in addr = {};
+  addr.sin_family = AF_INET;
+  addr.sin_addr.s_addr = htonl(INADDR_LOOPBACK);
+  addr.sin_port = htons((uint16_t)port);
+  if (::bind(fd, (struct sockaddr*)&addr, sizeof addr) != 0 || ::listen(fd, 64) != 0) {
+    ::close(fd);
+    return -1;
+  }
+  return fd;
+}
+
+int listen_port(int listen_fd) {
+  struct sockaddr_in addr = {};
+  socklen_t len = sizeof addr;
+  getsockname(listen_fd, (struct sockaddr*)&addr, &len);
+  return ntohs(addr.sin_port);
+}
+
+// ---- Node ------------------------------------------------------------------
+
+Node::Node(std::string name, std::string gateway_host, int gateway_port)
+    : name_(std::move(name)), gw_host_(std::move(gateway_host)), gw_port_(gateway_port) {}
+
+Node::~Node() { stop(); }
+
+void Node::start(int port) {
+  listen_fd_ = tcp_listen(port);
+  if (listen_fd_ < 0) throw std::runtime_error("node: cannot listen");
+  port_ = listen_port(listen_fd_);
+  running_ = true;
+  accept_thread_ = std::thread([this] { accept_loop(); });
+  if (gw_port_ > 0) {
+    // persistent broker connection: register + receive pub/sub events
+    int fd = tcp_connect(gw_host_, gw_port_, 10.0);
+    if (fd < 0) throw std::runtime_error("node: cannot reach gateway");
+    // long receive timeout on the event connection
+    struct timeval tv = {86400, 0};
+    setsockopt(fd, SOL_SOCKET, SO_RCVTIMEO, &tv, sizeof tv);
+    gw_sock_ = std::make_unique<MsgSocket>(fd);
+    Json reg;
+    reg["kind"] = "register";
+    reg["peer"] = name_;
+    reg["addr"] = std::string("127.0.0.1:") + std::to_string(port_);
+    gw_sock_->send_json(reg);
+    gw_thread_ = std::thread([this] { gateway_listen_loop(); });
+  }
+}
+
+void Node::stop() {
+  if (!running_.exchange(false)) return;
+  if (listen_fd_ >= 0) {
+    ::shutdown(listen_fd_, SHUT_RDWR);
+    ::close(listen_fd_);
+    listen_fd_ = -1;
+  }
+  if (gw_sock_) gw_sock_->close_now();
+  if (accept_thread_.joinable()) accept_thread_.join();
+  if (gw_thread_.joinable()) gw_thread_.join();
+}
+
+void Node::on(const std::string& type, Handler h) {
+  std::lock_guard<std::mutex> lk(mu_);
+  handlers_[type] = std::move(h);
+}
+
+void Node::on_stream(const std::string& type, StreamHandler h) {
+  std::lock_guard<std::mutex> lk(mu_);
+  stream_handlers_[type] = std::move(h);
+}
+
+void Node::accept_loop() {
+  while (running_) {
+    int fd = ::accept(listen_fd_, nullptr, nullptr);
+    if (fd < 0) break;
+    int one = 1;
+    setsockopt(fd, IPPROTO_TCP, TCP_NODELAY, &one, sizeof one);
+    std::thread([this, fd] { handle_conn(fd); }).detach();
+  }
+}
+
+void Node::handle_conn(int fd) {
+  MsgSocket sock(fd);
+  while (running_) {
+    auto msg = sock.recv_json();
+    if (!msg) break;
+    std::string kind = msg->get_or("kind", Json("")).as_string();
+    std::string type = msg->get_or("type", Json("")).as_string();
+    std::string from = msg->get_or("from", Json("")).as_string();
+    if (kind == "request") {
+      Handler h;
+      {
+        std::lock_guard<std::mutex> lk(mu_);
+        auto it = handlers_.find(type);
+        if (it != handlers_.end()) h = it->second;
+      }
+      Json resp;
+      resp["kind"] = "response";
+      if (h) {
+        try {
+          resp["body"] = h(from, msg->get_or("body", Json(JsonObject{})));
+          resp["ok"] = true;
+        } catch (const std::exception& e) {
+          resp["ok"] = false;
+          resp["error"] = std::string(e.what());
+        }
+      } else {
+        resp["ok"] = false;
+        resp["error"] = "no handler for " + type;
+      }
+      if (!sock.send_json(resp)) break;
+    } else if (kind == "stream") {
+      StreamHandler h;
+      {
+        std::lock_guard<std::mutex> lk(mu_);
+        auto it = stream_handlers_.find(type);
+        if (it != stream_handlers_.end()) h = it->second;
+      }
+      if (h) h(from, msg->get_or("header", Json(JsonObject{})), sock);
+      break;  // stream connections are single-use
+    } else {
+      break;
+    }
+  }
+}
+
+void Node::gateway_listen_loop() {
+  while (running_) {
+    auto msg = gw_sock_->recv_json();
+    if (!msg) break;
+    if (msg->get_or("kind", Json("")).as_string() == "pub") {
+      std::string topic = msg->at("topic").as_string();
+      std::string from = msg->get_or("from", Json("")).as_string();
+      std::function<void(const std::string&, const Json&)> cb;
+      {
+        std::lock_guard<std::mutex> lk(mu_);
+        auto it = subs_.find(topic);
+        if (it != subs_.end()) cb = it->second;
+      }
+      if (cb) cb(from, msg->get_or("data", Json(JsonObject{})));
+    }
+  }
+}
+
+Json Node::gateway_request(const std::string& type, const Json& body) {
+  int fd = tcp_connect(gw_host_, gw_port_, 10.0);
+  if (fd < 0) throw std::runtime_error("gateway unreachable");
+  MsgSocket sock(fd);
+  Json req;
+  req["kind"] = "request";
+  req["type"] = type;
+  req["from"] = name_;
+  req["body"] = body;
+  if (!sock.send_json(req)) throw std::runtime_error("gateway send failed");
+  auto resp = sock.recv_json();
+  if (!resp) throw std::runtime_error("gateway closed");
+  return resp->get_or("body", Json(JsonObject{}));
+}
+
+Json Node::request(const std::string& peer, const std::string& type, const Json& body,
+                   double timeout_s) {
+  auto addr = resolve(peer);
+  if (!addr) throw std::runtime_error("unknown peer " + peer);
+  size_t colon = addr->rfind(':');
+  int fd = tcp_connect(addr->substr(0, colon), std::stoi(addr->substr(colon + 1)), timeout_s);
+  if (fd < 0) throw std::runtime_error("peer unreachable: " + peer);
+  MsgSocket sock(fd);
+  Json req;
+  req["kind"] = "request";
+  req["type"] = type;
+  req["from"] = name_;
+  req["body"] = body;
+  if (!sock.send_json(req)) throw std::runtime_error("send failed");
+  auto resp = sock.recv_json();
+  if (!resp) throw std::runtime_error("peer closed: " + peer);
+  if (!resp->get_or("ok", Json(false)).as_bool())
+    throw std::runtime_error("rpc error from " + peer + ": " +
+                             resp->get_or("error", Json("unknown")).as_string());
+  return resp->get_or("body", Json(JsonObject{}));
+}
+
+std::unique_ptr<MsgSocket> Node::open_stream(const std::string& peer, const std::string& type,
+                                             const Json& header) {
+  auto addr = resolve(peer);
+  if (!addr) throw std::runtime_error("unknown peer " + peer);
+  size_t colon = addr->rfind(':');
+  int fd = tcp_connect(addr->substr(0, colon), std::stoi(addr->substr(colon + 1)), 30.0);
+  if (fd < 0) throw std::runtime_error("peer unreachable: " + peer);
+  struct timeval tv = {600, 0};
+  setsockopt(fd, SOL_SOCKET, SO_RCVTIMEO, &tv, sizeof tv);
+  setsockopt(fd, SOL_SOCKET, SO_SNDTIMEO, &tv, sizeof tv);
+  auto sock = std::make_unique<MsgSocket>(fd);
+  Json req;
+  req["kind"] = "stream";
+  req["type"] = type;
+  req["from"] = name_;
+  req["header"] = header;
+  if (!sock->send_json(req)) throw std::runtime_error("stream open failed");
+  return sock;
+}
+
+void Node::publish(const std::string& topic, const Json& data) {
+  Json b;
+  b["topic"] = topic;
+  b["data"] = data;
+  gateway_request("publish", b);
+}
+
+void Node::subscribe(const std::string& topic,
+                     std::function<void(const std::string&, const Json&)> cb) {
+  {
+    std::lock_guard<std::mutex> lk(mu_);
+    subs_[topic] = std::move(cb);
+  }
+  Json sub;
+  sub["kind"] = "subscribe";
+  sub["topic"] = topic;
+  std::lock_guard<std::mutex> lk(gw_mu_);
+  gw_sock_->send_json(sub);
+}
+
+void Node::kv_put(const std::string& key, const Json& value) {
+  Json b;
+  b["key"] = key;
+  b["value"] = value;
+  gateway_request("kv_put", b);
+}
+
+std::optional<Json> Node::kv_get(const std::string& key) {
+  Json b;
+  b["key"] = key;
+  Json r = gateway_request("kv_get", b);
+  if (r.get_or("found", Json(false)).as_bool()) return r.at("value");
+  return std::nullopt;
+}
+
+std::optional<std::string> Node::resolve(const std::string& peer) {
+  {
+    std::lock_guard<std::mutex> lk(mu_);
+    auto it = addr_cache_.find(peer);
+    if (it != addr_cache_.end()) return it->second;
+  }
+  auto v = kv_get("addr:" + peer);
+  if (!v) return std::nullopt;
+  std::string addr = v->as_string();
+  std::lock_guard<std::mutex> lk(mu_);
+  addr_cache_[peer] = addr;
+  return addr;
+}
+
+}  // namespace hypha

@@ -1,0 +1,120 @@
+"""In-process multi-peer tests of the C++ control-plane network layer —
+the analogue of the reference's libp2p-swarm-test integration suite
+(crates/network/tests/{request_response,kad,gossipsub}_test.rs), run with
+real TCP sockets on loopback inside one test process."""
+
+import threading
+import time
+
+import pytest
+
+core = pytest.importorskip("hypha_amd._core")
+
+
+@pytest.fixture
+def cluster():
+    gw = core.Gateway()
+    gw.start(0)
+    nodes = []
+
+    def make(name):
+        n = core.Node(name, "127.0.0.1", gw.port)
+        n.start(0)
+        nodes.append(n)
+        return n
+
+    yield make
+    for n in nodes:
+        n.stop()
+    gw.stop()
+
+
+def test_request_response(cluster):
+    a = cluster("alice")
+    b = cluster("bob")
+    b.on("echo", lambda frm, body: {"from_seen": frm, "payload": body["x"] * 2})
+    r = a.request("bob", "echo", {"x": 21})
+    assert r == {"from_seen": "alice", "payload": 42}
+
+
+def test_request_error_propagates(cluster):
+    a = cluster("alice")
+    b = cluster("bob")
+
+    def boom(frm, body):
+        raise ValueError("nope")
+
+    b.on("boom", boom)
+    with pytest.raises(RuntimeError, match="nope"):
+        a.request("bob", "boom", {})
+    with pytest.raises(RuntimeError, match="no handler"):
+        a.request("bob", "missing", {})
+
+
+def test_unknown_peer(cluster):
+    a = cluster("alice")
+    with pytest.raises(RuntimeError, match="unknown peer"):
+        a.request("ghost", "echo", {})
+
+
+def test_concurrent_requests(cluster):
+    a = cluster("alice")
+    b = cluster("bob")
+    b.on("slowmul", lambda frm, body: (time.sleep(0.05), {"y": body["x"] * 3})[1])
+    results = {}
+
+    def call(i):
+        results[i] = a.request("bob", "slowmul", {"x": i})["y"]
+
+    threads = [threading.Thread(target=call, args=(i,)) for i in range(8)]
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join()
+    assert results == {i: 3 * i for i in range(8)}
+
+
+def test_kv_store_and_discovery(cluster):
+    a = cluster("alice")
+    b = cluster("bob")
+    a.kv_put("dataset:mnist", {"num_slices": 12})
+    assert b.kv_get("dataset:mnist") == {"num_slices": 12}
+    assert b.kv_get("missing") is None
+    # peer registry doubles as discovery (kad.rs provider semantics)
+    assert b.resolve("alice").startswith("127.0.0.1:")
+
+
+def test_pubsub_topic_fanout(cluster):
+    a = cluster("alice")
+    b = cluster("bob")
+    c = cluster("carol")
+    got_b, got_c = [], []
+    ev = threading.Event()
+
+    def cb_b(frm, data):
+        got_b.append((frm, data))
+        if got_b and got_c:
+            ev.set()
+
+    def cb_c(frm, data):
+        got_c.append((frm, data))
+        if got_b and got_c:
+            ev.set()
+
+    b.subscribe("hypha/worker", cb_b)
+    c.subscribe("hypha/worker", cb_c)
+    time.sleep(0.05)  # subscription registration latency
+    a.publish("hypha/worker", {"id": "rq1", "bid": 2.5})
+    assert ev.wait(5.0)
+    assert got_b == [("alice", {"id": "rq1", "bid": 2.5})]
+    assert got_c == [("alice", {"id": "rq1", "bid": 2.5})]
+
+
+def test_publisher_not_echoed(cluster):
+    a = cluster("alice")
+    got = []
+    a.subscribe("t", lambda frm, d: got.append(d))
+    time.sleep(0.05)
+    a.publish("t", {"v": 1})
+    time.sleep(0.1)
+    assert got == []  # publisher doesn't receive its own message
