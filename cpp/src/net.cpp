@@ -179,7 +179,13 @@ void Node::accept_loop() {
     if (fd < 0) break;
     int one = 1;
     setsockopt(fd, IPPROTO_TCP, TCP_NODELAY, &one, sizeof one);
-    std::thread([this, fd] { handle_conn(fd); }).detach();
+    std::thread([this, fd] {
+      try {
+        handle_conn(fd);
+      } catch (const std::exception& e) {
+        fprintf(stderr, "[net:%s] conn handler error: %s\n", name_.c_str(), e.what());
+      }
+    }).detach();
   }
 }
 
@@ -232,6 +238,7 @@ void Node::gateway_listen_loop() {
   while (running_) {
     auto msg = gw_sock_->recv_json();
     if (!msg) break;
+    try {
     if (msg->get_or("kind", Json("")).as_string() == "pub") {
       std::string topic = msg->at("topic").as_string();
       std::string from = msg->get_or("from", Json("")).as_string();
@@ -242,6 +249,9 @@ void Node::gateway_listen_loop() {
         if (it != subs_.end()) cb = it->second;
       }
       if (cb) cb(from, msg->get_or("data", Json(JsonObject{})));
+    }
+    } catch (const std::exception& e) {
+      fprintf(stderr, "[net:%s] gateway event error: %s\n", name_.c_str(), e.what());
     }
   }
 }

@@ -101,9 +101,35 @@ def build_core_extension():
     print(f"built {out}")
 
 
+def build_daemons():
+    """C++ control-plane daemon binaries -> bin/ (in-tree, travel with snapshot)."""
+    srcs = sorted(CPP_DIR.glob("bin/*.cpp"))
+    if not srcs:
+        return
+    bindir = ROOT / "bin"
+    bindir.mkdir(exist_ok=True)
+    lib_srcs = [str(s) for s in sorted(CPP_DIR.glob("src/*.cpp"))]
+    headers = list(CPP_DIR.glob("include/hypha/*.h"))
+    procs = []
+    for src in srcs:
+        exe = bindir / src.stem.replace("_", "-")
+        if not _needs_rebuild(exe, [src] + lib_srcs + [str(h) for h in headers]):
+            continue
+        cmd = [
+            "g++", "-O2", "-std=c++17", "-pthread",
+            f"-I{CPP_DIR / 'include'}", str(src), *lib_srcs, "-o", str(exe),
+        ]
+        print("+", " ".join(cmd), flush=True)
+        procs.append(subprocess.Popen(cmd))
+    for p in procs:
+        if p.wait() != 0:
+            sys.exit(1)
+
+
 if __name__ == "__main__":
     if "build_ext" in sys.argv or len(sys.argv) == 1:
         build_core_extension()
+        build_daemons()
         build_hip_extension()
     else:
         print("usage: python setup.py build_ext --inplace")

@@ -1,0 +1,102 @@
+"""End-to-end cluster test — BASELINE config 1 at tiny scale: gateway +
+data node + 3 worker daemons (2 train + 1 parameter server) + scheduler,
+all real processes on loopback, training llama-tiny DiLoCo for 2 outer
+rounds on CPU. The reference's quickstart (docs/quickstart.md) is the
+manual version of this; here it is automated."""
+
+import os
+import signal
+import socket
+import subprocess
+import sys
+import time
+from pathlib import Path
+
+import pytest
+
+REPO = Path(__file__).resolve().parent.parent
+BIN = REPO / "bin"
+
+
+def free_port():
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    p = s.getsockname()[1]
+    s.close()
+    return p
+
+
+@pytest.fixture(scope="module")
+def binaries():
+    if not (BIN / "hypha-gateway").exists():
+        subprocess.run(
+            [sys.executable, "setup.py", "build_ext", "--inplace"], cwd=REPO, check=True
+        )
+    return BIN
+
+
+@pytest.mark.timeout(300)
+def test_cluster_diloco_round_trip(binaries, tmp_path):
+    from hypha_amd.data.synthetic import write_slice_files
+
+    data_dir = tmp_path / "slices"
+    write_slice_files(str(data_dir), "synth", num_slices=4, samples_per_slice=16,
+                      vocab_size=512, seq_len=128)
+
+    gw_port = free_port()
+    env = dict(os.environ, PYTHONPATH=str(REPO))
+    procs = []
+    logs = {}
+
+    def spawn(name, cmd):
+        log = open(tmp_path / f"{name}.log", "w")
+        logs[name] = tmp_path / f"{name}.log"
+        p = subprocess.Popen(cmd, cwd=REPO, env=env, stdout=log, stderr=log)
+        procs.append(p)
+        return p
+
+    try:
+        spawn("gateway", [str(BIN / "hypha-gateway"), "--port", str(gw_port)])
+        time.sleep(0.3)
+        spawn("data", [str(BIN / "hypha-data"), "--name", "data-node",
+                       "--gateway-host", "127.0.0.1", "--gateway-port", str(gw_port),
+                       "--dataset", "synth", "--dataset-path", str(data_dir)])
+        exec_cmd = (f"{sys.executable} -m hypha_amd.runtime.executor "
+                    "--socket {SOCKET_PATH} --work-dir {WORK_DIR} --job {JOB_JSON}")
+        for i in range(3):
+            spawn(f"worker{i}", [str(BIN / "hypha-worker"), "--name", f"worker-{i}",
+                                 "--gateway-host", "127.0.0.1", "--gateway-port", str(gw_port),
+                                 "--exec-cmd", exec_cmd,
+                                 "--work-root", str(tmp_path / f"work{i}")])
+        time.sleep(0.5)
+
+        cfg = tmp_path / "job.json"
+        cfg.write_text(
+            '{"model": "llama-tiny", "dataset": "synth", "num_workers": 2,'
+            ' "update_rounds": 2, "avg_samples_between_updates": 8,'
+            ' "batch_size": 2, "seq_len": 128, "inner_lr": 0.001}'
+        )
+        sched = subprocess.Popen(
+            [str(BIN / "hypha-scheduler"), "--name", "scheduler",
+             "--gateway-host", "127.0.0.1", "--gateway-port", str(gw_port),
+             "--config", str(cfg)],
+            cwd=REPO, env=env, stdout=subprocess.PIPE, stderr=open(tmp_path / "sched.log", "w"),
+            text=True,
+        )
+        procs.append(sched)
+        out, _ = sched.communicate(timeout=240)
+        assert "Job is completed." in out, (
+            out,
+            *[f"--- {n}: {p.read_text()[-2000:]}" for n, p in logs.items()],
+            (tmp_path / "sched.log").read_text()[-3000:],
+        )
+        assert sched.returncode == 0
+    finally:
+        for p in procs:
+            if p.poll() is None:
+                p.send_signal(signal.SIGKILL)
+        for p in procs:
+            try:
+                p.wait(timeout=10)
+            except Exception:
+                pass
