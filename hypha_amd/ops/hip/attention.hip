@@ -39,10 +39,11 @@ __device__ __forceinline__ unsigned pack_bf16x2(float lo, float hi) {
 }
 
 // Swizzled byte offset inside the K tile: row-major [KVB][HD] bf16 with
-// byte ^= (row & 7) << 4 (guide G4 XOR swizzle, <=2-way on ds_read_b128).
+// byte ^= (row & 15) << 4 (guide G4 XOR swizzle: conflict-free on
+// ds_read_b128 when the 16-lane group's rows are distinct mod 16 — ours are).
 template <int HD>
 __device__ __forceinline__ int k_lds_off(int row, int elem) {
-  return (row * HD + elem) * 2 ^ ((row & 7) << 4);
+  return (row * HD + elem) * 2 ^ ((row & 15) << 4);
 }
 
 template <int HD>
@@ -93,9 +94,17 @@ __global__ __launch_bounds__(256, 1) void attn_fwd_kernel(
       int row = c / (HD / 8), e0 = (c % (HD / 8)) * 8;
       s16x8 kv8 = *reinterpret_cast<const s16x8*>(kg + kvbase + (long long)(kv0 + row) * HD + e0);
       *reinterpret_cast<s16x8*>((char*)k_lds + k_lds_off<HD>(row, e0)) = kv8;
-      s16x8 vv8 = *reinterpret_cast<const s16x8*>(vg + kvbase + (long long)(kv0 + row) * HD + e0);
+    }
+    // V^T transpose staging: lane-per-kv so each scalar-store instruction's
+    // 64 lanes write one contiguous 128 B image row span (conflict-free;
+    // the d-major pattern was a 16-way bank conflict: 8*VT_PITCH*2 = 0 mod 128)
 #pragma unroll
-      for (int j = 0; j < 8; ++j) vt_lds[(e0 + j) * VT_PITCH + row] = vv8[j];
+    for (int i = 0; i < HD / 32; ++i) {
+      int kvr = tid & 63;
+      int e0 = (i * 4 + (tid >> 6)) * 8;
+      s16x8 vv8 = *reinterpret_cast<const s16x8*>(vg + kvbase + (long long)(kv0 + kvr) * HD + e0);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) vt_lds[(e0 + j) * VT_PITCH + kvr] = vv8[j];
     }
     __syncthreads();
 
@@ -314,8 +323,14 @@ __global__ __launch_bounds__(256, 1) void attn_bwd_kernel(
       *reinterpret_cast<s16x8*>((char*)k_img + k_lds_off<HD>(row, e0)) = k8;
       s16x8 v8 = *reinterpret_cast<const s16x8*>(vg + kvbase + (long long)(kv0 + row) * HD + e0);
       *reinterpret_cast<s16x8*>((char*)v_img + k_lds_off<HD>(row, e0)) = v8;
+    }
+    // K^T: lane-per-kv transpose staging (conflict-free scalar stores)
+    for (int i = 0; i < HD / 64; ++i) {
+      int kvr = tid & 31;
+      int e0 = (i * 8 + (tid >> 5)) * 8;
+      s16x8 k8 = *reinterpret_cast<const s16x8*>(kg + kvbase + (long long)(kv0 + kvr) * HD + e0);
 #pragma unroll
-      for (int j = 0; j < 8; ++j) kt_img[(e0 + j) * TP + row] = k8[j];
+      for (int j = 0; j < 8; ++j) kt_img[(e0 + j) * TP + kvr] = k8[j];
     }
   }
   __syncthreads();
@@ -338,17 +353,18 @@ __global__ __launch_bounds__(256, 1) void attn_bwd_kernel(
       const int q0 = t * QT;
       const int qrow = q0 + ln;  // this lane's q (for B-operand frags)
 
-      // ---- per-wave staging: Q^T and dO^T images (transpose scatter) ----
+      // ---- per-wave staging: Q^T and dO^T images. Lane-per-q: each scalar
+      // store's lanes write one contiguous q-row span of the image (the
+      // d-major order was 16-way bank-conflicted: 8*TP*2 = 0 mod 128 B).
 #pragma unroll
-      for (int it = 0; it < QT * HD / 8 / 64; ++it) {
-        int c = it * 64 + lane;
-        int row = c / (HD / 8), e0 = (c % (HD / 8)) * 8;
-        s16x8 q8 = *reinterpret_cast<const s16x8*>(qg + qbase + (long long)(q0 + row) * HD + e0);
-        s16x8 d8 = *reinterpret_cast<const s16x8*>(dog + qbase + (long long)(q0 + row) * HD + e0);
+      for (int it = 0; it < HD / 16; ++it) {
+        int d0 = 8 * hi + 16 * it;
+        s16x8 q8 = *reinterpret_cast<const s16x8*>(qg + qbase + (long long)(q0 + ln) * HD + d0);
+        s16x8 d8 = *reinterpret_cast<const s16x8*>(dog + qbase + (long long)(q0 + ln) * HD + d0);
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
-          qt[(e0 + j) * TP + row] = q8[j];
-          dot[(e0 + j) * TP + row] = d8[j];
+          qt[(d0 + j) * TP + ln] = q8[j];
+          dot[(d0 + j) * TP + ln] = d8[j];
         }
       }
 
