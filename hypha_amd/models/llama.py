@@ -91,12 +91,14 @@ class Attention(nn.Module):
         b, s, _ = x.shape
         cfg = self.cfg
         hd = cfg.head_dim
-        q = self.wq(x).view(b, s, cfg.n_heads, hd).transpose(1, 2)
-        k = self.wk(x).view(b, s, cfg.n_kv_heads, hd).transpose(1, 2)
-        v = self.wv(x).view(b, s, cfg.n_kv_heads, hd).transpose(1, 2)
-        q, k = ops.apply_rope_qk(q, k, cos, sin)
-        o = ops.flash_attention(q, k, v, causal=True)
-        return self.wo(o.transpose(1, 2).reshape(b, s, -1))
+        # BSHD layout throughout: the projections' natural layout, consumed
+        # directly by the stride-aware RoPE/attention kernels (no transposes)
+        q = self.wq(x).view(b, s, cfg.n_heads, hd)
+        k = self.wk(x).view(b, s, cfg.n_kv_heads, hd)
+        v = self.wv(x).view(b, s, cfg.n_kv_heads, hd)
+        q, k = ops.apply_rope_qk(q, k, cos, sin, layout="bshd")
+        o = ops.flash_attention(q, k, v, causal=True, layout="bshd")
+        return self.wo(o.reshape(b, s, -1))
 
 
 class MLP(nn.Module):

@@ -50,7 +50,9 @@ template <int HD>
 __global__ __launch_bounds__(256, 1) void attn_fwd_kernel(
     const short* __restrict__ qg, const short* __restrict__ kg,
     const short* __restrict__ vg, short* __restrict__ og, float* __restrict__ lseg,
-    int B, int Hq, int Hkv, int S, float scale, bool causal) {
+    int B, int Hq, int Hkv, int S, float scale, bool causal,
+    long long q_sb, long long q_sh, long long q_ss,  // q/o strides (elements)
+    long long kv_sb, long long kv_sh, long long kv_ss) {
   constexpr int KC = HD / 16;        // 16-wide k chunks over head dim
   constexpr int DBLK = HD / 32;      // 32-row d blocks
   constexpr int VT_PITCH = KVB + PADV;
@@ -68,8 +70,8 @@ __global__ __launch_bounds__(256, 1) void attn_fwd_kernel(
   const int b = bh / Hq;
   const int hq = bh % Hq;
   const int hkv = hq / (Hq / Hkv);
-  const long long qbase = ((long long)(b * Hq + hq) * S) * HD;
-  const long long kvbase = ((long long)(b * Hkv + hkv) * S) * HD;
+  const long long qbase = (long long)b * q_sb + (long long)hq * q_sh;
+  const long long kvbase = (long long)b * kv_sb + (long long)hkv * kv_sh;
 
   const int q0wg = blockIdx.x * WGQ;
   const int q0 = q0wg + wid * QB;   // this wave's q rows
@@ -79,7 +81,7 @@ __global__ __launch_bounds__(256, 1) void attn_fwd_kernel(
   s16x8 qfrag[KC];
 #pragma unroll
   for (int kc = 0; kc < KC; ++kc)
-    qfrag[kc] = *reinterpret_cast<const s16x8*>(qg + qbase + (long long)qrow * HD +
+    qfrag[kc] = *reinterpret_cast<const s16x8*>(qg + qbase + (long long)qrow * q_ss +
                                                 16 * kc + 8 * hi);
 
   f32x16 ot[DBLK] = {};
@@ -92,7 +94,7 @@ __global__ __launch_bounds__(256, 1) void attn_fwd_kernel(
     constexpr int CH = KVB * HD / 8;  // 16B chunks
     for (int c = tid; c < CH; c += 256) {
       int row = c / (HD / 8), e0 = (c % (HD / 8)) * 8;
-      s16x8 kv8 = *reinterpret_cast<const s16x8*>(kg + kvbase + (long long)(kv0 + row) * HD + e0);
+      s16x8 kv8 = *reinterpret_cast<const s16x8*>(kg + kvbase + (long long)(kv0 + row) * kv_ss + e0);
       *reinterpret_cast<s16x8*>((char*)k_lds + k_lds_off<HD>(row, e0)) = kv8;
     }
     // V^T transpose staging: lane-per-kv so each scalar-store instruction's
@@ -102,7 +104,7 @@ __global__ __launch_bounds__(256, 1) void attn_fwd_kernel(
     for (int i = 0; i < HD / 32; ++i) {
       int kvr = tid & 63;
       int e0 = (i * 4 + (tid >> 6)) * 8;
-      s16x8 vv8 = *reinterpret_cast<const s16x8*>(vg + kvbase + (long long)(kv0 + kvr) * HD + e0);
+      s16x8 vv8 = *reinterpret_cast<const s16x8*>(vg + kvbase + (long long)(kv0 + kvr) * kv_ss + e0);
 #pragma unroll
       for (int j = 0; j < 8; ++j) vt_lds[(e0 + j) * VT_PITCH + kvr] = vv8[j];
     }
@@ -204,17 +206,23 @@ __global__ __launch_bounds__(256, 1) void attn_fwd_kernel(
     int c = it * 64 + lane;
     int row = c / (HD / 8), e0 = (c % (HD / 8)) * 8;
     s16x8 o8 = *reinterpret_cast<const s16x8*>(ot_lds[wid] + row * OPITCH + e0);
-    *reinterpret_cast<s16x8*>(og + qbase + (long long)(q0 + row) * HD + e0) = o8;
+    *reinterpret_cast<s16x8*>(og + qbase + (long long)(q0 + row) * q_ss + e0) = o8;
   }
 }
 
 }  // namespace
 
-std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
-                                    bool causal) {
+// layout: "bhsd" = [B,H,S,D] contiguous; "bshd" = [B,S,H,D] contiguous
+// (the model's natural projection layout — no transpose copies needed).
+std::vector<torch::Tensor> attn_fwd_ex(torch::Tensor q, torch::Tensor k, torch::Tensor v,
+                                       bool causal, const std::string& layout) {
   TORCH_CHECK(q.dim() == 4 && q.dtype() == torch::kBFloat16 && q.is_contiguous());
-  const int B = q.size(0), Hq = q.size(1), S = q.size(2), HD = q.size(3);
-  const int Hkv = k.size(1);
+  const bool bshd = layout == "bshd";
+  const int B = q.size(0);
+  const int Hq = bshd ? q.size(2) : q.size(1);
+  const int S = bshd ? q.size(1) : q.size(2);
+  const int HD = q.size(3);
+  const int Hkv = bshd ? k.size(2) : k.size(1);
   TORCH_CHECK(Hq % Hkv == 0 && (HD == 64 || HD == 128));
   TORCH_CHECK(S % WGQ == 0, "seq len must be a multiple of 128");
   auto o = torch::empty_like(q);
@@ -222,17 +230,26 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k, torch::Ten
   const float scale = 1.0f / sqrtf((float)HD);
   dim3 grid(S / WGQ, B * Hq);
   hipStream_t stream = hypha_stream();
+  long long q_sb, q_sh, q_ss, kv_sb, kv_sh, kv_ss;
+  if (bshd) {
+    q_sh = HD; q_ss = (long long)Hq * HD; q_sb = (long long)S * Hq * HD;
+    kv_sh = HD; kv_ss = (long long)Hkv * HD; kv_sb = (long long)S * Hkv * HD;
+  } else {
+    q_ss = HD; q_sh = (long long)S * HD; q_sb = (long long)Hq * S * HD;
+    kv_ss = HD; kv_sh = (long long)S * HD; kv_sb = (long long)Hkv * S * HD;
+  }
+#define LAUNCH_FWD(HDV)                                                                  hipLaunchKernelGGL(attn_fwd_kernel<HDV>, grid, dim3(256), 0, stream,                                      (const short*)q.data_ptr(), (const short*)k.data_ptr(),                                (const short*)v.data_ptr(), (short*)o.data_ptr(),                                      lse.data_ptr<float>(), B, Hq, Hkv, S, scale, causal, q_sb, q_sh,                       q_ss, kv_sb, kv_sh, kv_ss)
   if (HD == 128)
-    hipLaunchKernelGGL(attn_fwd_kernel<128>, grid, dim3(256), 0, stream,
-                       (const short*)q.data_ptr(), (const short*)k.data_ptr(),
-                       (const short*)v.data_ptr(), (short*)o.data_ptr(),
-                       lse.data_ptr<float>(), B, Hq, Hkv, S, scale, causal);
+    LAUNCH_FWD(128);
   else
-    hipLaunchKernelGGL(attn_fwd_kernel<64>, grid, dim3(256), 0, stream,
-                       (const short*)q.data_ptr(), (const short*)k.data_ptr(),
-                       (const short*)v.data_ptr(), (short*)o.data_ptr(),
-                       lse.data_ptr<float>(), B, Hq, Hkv, S, scale, causal);
+    LAUNCH_FWD(64);
+#undef LAUNCH_FWD
   return {o, lse};
+}
+
+std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
+                                    bool causal) {
+  return attn_fwd_ex(q, k, v, causal, "bhsd");
 }
 
 // ===========================================================================
@@ -245,15 +262,20 @@ namespace {
 template <int HD>
 __global__ void attn_bwd_di_kernel(const short* __restrict__ dog,
                                    const short* __restrict__ og,
-                                   float* __restrict__ dig, long long nrows) {
+                                   float* __restrict__ dig, long long nrows, int Hq,
+                                   int S, long long sb, long long sh, long long ss) {
   const long long row = (long long)blockIdx.x * 4 + (threadIdx.x >> 6);
   if (row >= nrows) return;
+  const long long b = row / ((long long)Hq * S);
+  const long long h = (row / S) % Hq;
+  const long long sq = row % S;
+  const long long addr = b * sb + h * sh + sq * ss;
   const int lane = threadIdx.x & 63;
   float acc = 0.f;
 #pragma unroll
   for (int i = 0; i < HD / 64; ++i) {
     int d = lane + 64 * i;
-    acc += bf2f(dog[row * HD + d]) * bf2f(og[row * HD + d]);
+    acc += bf2f(dog[addr + d]) * bf2f(og[addr + d]);
   }
   acc = wave_reduce_sum(acc);
   if (lane == 0) dig[row] = acc;
@@ -286,7 +308,9 @@ __global__ __launch_bounds__(256, 1) void attn_bwd_kernel(
     const short* __restrict__ vg, const short* __restrict__ dog,
     const float* __restrict__ lseg, const float* __restrict__ dig,
     float* __restrict__ dqg, short* __restrict__ dkg, short* __restrict__ dvg,
-    int B, int Hq, int Hkv, int S, float scale, bool causal) {
+    int B, int Hq, int Hkv, int S, float scale, bool causal,
+    long long q_sb, long long q_sh, long long q_ss,
+    long long kv_sb, long long kv_sh, long long kv_ss) {
   constexpr int KVT = 32;            // kv rows per workgroup
   constexpr int QT = 32;             // q rows per tile
   constexpr int KC = HD / 16;        // chunks over head dim
@@ -312,23 +336,23 @@ __global__ __launch_bounds__(256, 1) void attn_bwd_kernel(
   const int hkv = bhkv % Hkv;
   const int G = Hq / Hkv;
   const int kv0 = blockIdx.x * KVT;
-  const long long kvbase = ((long long)(b * Hkv + hkv) * S) * HD;
+  const long long kvbase = (long long)b * kv_sb + (long long)hkv * kv_sh;
 
   // ---- stage K, V (swizzled row-major) and K^T ----
   {
     constexpr int CH = KVT * HD / 8;
     for (int c = tid; c < CH; c += 256) {
       int row = c / (HD / 8), e0 = (c % (HD / 8)) * 8;
-      s16x8 k8 = *reinterpret_cast<const s16x8*>(kg + kvbase + (long long)(kv0 + row) * HD + e0);
+      s16x8 k8 = *reinterpret_cast<const s16x8*>(kg + kvbase + (long long)(kv0 + row) * kv_ss + e0);
       *reinterpret_cast<s16x8*>((char*)k_img + k_lds_off<HD>(row, e0)) = k8;
-      s16x8 v8 = *reinterpret_cast<const s16x8*>(vg + kvbase + (long long)(kv0 + row) * HD + e0);
+      s16x8 v8 = *reinterpret_cast<const s16x8*>(vg + kvbase + (long long)(kv0 + row) * kv_ss + e0);
       *reinterpret_cast<s16x8*>((char*)v_img + k_lds_off<HD>(row, e0)) = v8;
     }
     // K^T: lane-per-kv transpose staging (conflict-free scalar stores)
     for (int i = 0; i < HD / 64; ++i) {
       int kvr = tid & 31;
       int e0 = (i * 8 + (tid >> 5)) * 8;
-      s16x8 k8 = *reinterpret_cast<const s16x8*>(kg + kvbase + (long long)(kv0 + kvr) * HD + e0);
+      s16x8 k8 = *reinterpret_cast<const s16x8*>(kg + kvbase + (long long)(kv0 + kvr) * kv_ss + e0);
 #pragma unroll
       for (int j = 0; j < 8; ++j) kt_img[(e0 + j) * TP + kvr] = k8[j];
     }
@@ -347,7 +371,7 @@ __global__ __launch_bounds__(256, 1) void attn_bwd_kernel(
   const int Tq = S / QT;
 
   for (int hq = hkv * G; hq < (hkv + 1) * G; ++hq) {
-    const long long qbase = ((long long)(b * Hq + hq) * S) * HD;
+    const long long qbase = (long long)b * q_sb + (long long)hq * q_sh;
     const long long lsebase = (long long)(b * Hq + hq) * S;
     for (int t = t0 + wid; t < Tq; t += NWAVE) {
       const int q0 = t * QT;
@@ -359,8 +383,8 @@ __global__ __launch_bounds__(256, 1) void attn_bwd_kernel(
 #pragma unroll
       for (int it = 0; it < HD / 16; ++it) {
         int d0 = 8 * hi + 16 * it;
-        s16x8 q8 = *reinterpret_cast<const s16x8*>(qg + qbase + (long long)(q0 + ln) * HD + d0);
-        s16x8 d8 = *reinterpret_cast<const s16x8*>(dog + qbase + (long long)(q0 + ln) * HD + d0);
+        s16x8 q8 = *reinterpret_cast<const s16x8*>(qg + qbase + (long long)(q0 + ln) * q_ss + d0);
+        s16x8 d8 = *reinterpret_cast<const s16x8*>(dog + qbase + (long long)(q0 + ln) * q_ss + d0);
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
           qt[(d0 + j) * TP + ln] = q8[j];
@@ -372,7 +396,7 @@ __global__ __launch_bounds__(256, 1) void attn_bwd_kernel(
       f32x16 st = {};
 #pragma unroll
       for (int kc = 0; kc < KC; ++kc) {
-        s16x8 qf = *reinterpret_cast<const s16x8*>(qg + qbase + (long long)qrow * HD +
+        s16x8 qf = *reinterpret_cast<const s16x8*>(qg + qbase + (long long)qrow * q_ss +
                                                    16 * kc + 8 * hi);
         s16x8 kf = *reinterpret_cast<const s16x8*>(
             (char*)k_img + k_lds_off<HD>(ln, 16 * kc + 8 * hi));
@@ -414,7 +438,7 @@ __global__ __launch_bounds__(256, 1) void attn_bwd_kernel(
       f32x16 dpt = {};
 #pragma unroll
       for (int kc = 0; kc < KC; ++kc) {
-        s16x8 df = *reinterpret_cast<const s16x8*>(dog + qbase + (long long)qrow * HD +
+        s16x8 df = *reinterpret_cast<const s16x8*>(dog + qbase + (long long)qrow * q_ss +
                                                    16 * kc + 8 * hi);
         s16x8 vf = *reinterpret_cast<const s16x8*>(
             (char*)v_img + k_lds_off<HD>(ln, 16 * kc + 8 * hi));
@@ -458,7 +482,7 @@ __global__ __launch_bounds__(256, 1) void attn_bwd_kernel(
 #pragma unroll
         for (int r = 0; r < 16; ++r) {
           int qi = q0 + (r & 3) + 8 * (r >> 2) + 4 * hi;
-          atomicAdd(dqg + (qbase + (long long)qi * HD + 32 * db + ln), dq[r]);
+          atomicAdd(dqg + (qbase + (long long)qi * q_ss + 32 * db + ln), dq[r]);
         }
       }
     }
@@ -487,7 +511,7 @@ __global__ __launch_bounds__(256, 1) void attn_bwd_kernel(
       s16x8 o8;
 #pragma unroll
       for (int j = 0; j < 8; ++j) o8[j] = f2bf(comb[row * HD + e0 + j]);
-      *reinterpret_cast<s16x8*>(outg + kvbase + (long long)(kv0 + row) * HD + e0) = o8;
+      *reinterpret_cast<s16x8*>(outg + kvbase + (long long)(kv0 + row) * kv_ss + e0) = o8;
     }
     __syncthreads();
   };
@@ -497,31 +521,46 @@ __global__ __launch_bounds__(256, 1) void attn_bwd_kernel(
 
 }  // namespace
 
-std::vector<torch::Tensor> attn_bwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
-                                    torch::Tensor o, torch::Tensor dout,
-                                    torch::Tensor lse, bool causal) {
-  const int B = q.size(0), Hq = q.size(1), S = q.size(2), HD = q.size(3);
-  const int Hkv = k.size(1);
+std::vector<torch::Tensor> attn_bwd_ex(torch::Tensor q, torch::Tensor k, torch::Tensor v,
+                                       torch::Tensor o, torch::Tensor dout,
+                                       torch::Tensor lse, bool causal,
+                                       const std::string& layout) {
+  const bool bshd = layout == "bshd";
+  const int B = q.size(0);
+  const int Hq = bshd ? q.size(2) : q.size(1);
+  const int S = bshd ? q.size(1) : q.size(2);
+  const int HD = q.size(3);
+  const int Hkv = bshd ? k.size(2) : k.size(1);
   TORCH_CHECK(S % 32 == 0 && (HD == 64 || HD == 128));
-  auto dq32 = torch::zeros({B, Hq, S, HD}, q.options().dtype(torch::kFloat32));
+  auto dq32 = torch::zeros(q.sizes(), q.options().dtype(torch::kFloat32));
   auto dk = torch::empty_like(k);
   auto dv = torch::empty_like(v);
   auto di = torch::empty({B, Hq, S}, q.options().dtype(torch::kFloat32));
   const float scale = 1.0f / sqrtf((float)HD);
   hipStream_t stream = hypha_stream();
   long long nrows = (long long)B * Hq * S;
+  long long q_sb, q_sh, q_ss, kv_sb, kv_sh, kv_ss;
+  if (bshd) {
+    q_sh = HD; q_ss = (long long)Hq * HD; q_sb = (long long)S * Hq * HD;
+    kv_sh = HD; kv_ss = (long long)Hkv * HD; kv_sb = (long long)S * Hkv * HD;
+  } else {
+    q_ss = HD; q_sh = (long long)S * HD; q_sb = (long long)Hq * S * HD;
+    kv_ss = HD; kv_sh = (long long)S * HD; kv_sb = (long long)Hkv * S * HD;
+  }
 
 #define DISPATCH(HDV)                                                                   \
   do {                                                                                  \
     hipLaunchKernelGGL(attn_bwd_di_kernel<HDV>, dim3((unsigned)((nrows + 3) / 4)),      \
                        dim3(256), 0, stream, (const short*)dout.data_ptr(),             \
-                       (const short*)o.data_ptr(), di.data_ptr<float>(), nrows);        \
+                       (const short*)o.data_ptr(), di.data_ptr<float>(), nrows, Hq, S,  \
+                       q_sb, q_sh, q_ss);                                               \
     hipLaunchKernelGGL(attn_bwd_kernel<HDV>, dim3(S / 32, B * Hkv), dim3(256), 0,       \
                        stream, (const short*)q.data_ptr(), (const short*)k.data_ptr(),  \
                        (const short*)v.data_ptr(), (const short*)dout.data_ptr(),       \
                        lse.data_ptr<float>(), di.data_ptr<float>(),                     \
                        dq32.data_ptr<float>(), (short*)dk.data_ptr(),                   \
-                       (short*)dv.data_ptr(), B, Hq, Hkv, S, scale, causal);            \
+                       (short*)dv.data_ptr(), B, Hq, Hkv, S, scale, causal, q_sb,       \
+                       q_sh, q_ss, kv_sb, kv_sh, kv_ss);                                \
   } while (0)
 
   if (HD == 128)
@@ -536,4 +575,10 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor q, torch::Tensor k, torch::Ten
                      dim3(256), 0, stream, dq32.data_ptr<float>(), (short*)dq.data_ptr(),
                      n);
   return {dq, dk, dv};
+}
+
+std::vector<torch::Tensor> attn_bwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
+                                    torch::Tensor o, torch::Tensor dout,
+                                    torch::Tensor lse, bool causal) {
+  return attn_bwd_ex(q, k, v, o, dout, lse, causal, "bhsd");
 }

@@ -21,6 +21,9 @@ std::vector<torch::Tensor> rmsnorm_bwd(torch::Tensor dy, torch::Tensor x, torch:
 // rope.hip
 std::vector<torch::Tensor> rope_fwd(torch::Tensor q, torch::Tensor k, torch::Tensor cos_t,
                                     torch::Tensor sin_t, bool inverse);
+std::vector<torch::Tensor> rope_fwd_ex(torch::Tensor q, torch::Tensor k,
+                                       torch::Tensor cos_t, torch::Tensor sin_t,
+                                       bool inverse, const std::string& layout);
 // cross_entropy.hip
 std::vector<torch::Tensor> ce_fwd(torch::Tensor logits, torch::Tensor targets);
 torch::Tensor ce_bwd_(torch::Tensor logits, torch::Tensor targets, torch::Tensor lse,
@@ -31,6 +34,12 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k, torch::Ten
 std::vector<torch::Tensor> attn_bwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
                                     torch::Tensor o, torch::Tensor dout, torch::Tensor lse,
                                     bool causal);
+std::vector<torch::Tensor> attn_fwd_ex(torch::Tensor q, torch::Tensor k, torch::Tensor v,
+                                       bool causal, const std::string& layout);
+std::vector<torch::Tensor> attn_bwd_ex(torch::Tensor q, torch::Tensor k, torch::Tensor v,
+                                       torch::Tensor o, torch::Tensor dout,
+                                       torch::Tensor lse, bool causal,
+                                       const std::string& layout);
 // probe.hip (MFMA fragment-layout verification)
 torch::Tensor mfma_probe_32x32x16(torch::Tensor a, torch::Tensor b);
 torch::Tensor mfma_probe_16x16x32(torch::Tensor a, torch::Tensor b);
@@ -44,10 +53,13 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm_fwd", &rmsnorm_fwd);
   m.def("rmsnorm_bwd", &rmsnorm_bwd);
   m.def("rope_fwd", &rope_fwd);
+  m.def("rope_fwd_ex", &rope_fwd_ex);
   m.def("ce_fwd", &ce_fwd);
   m.def("ce_bwd_", &ce_bwd_);
   m.def("attn_fwd", &attn_fwd);
   m.def("attn_bwd", &attn_bwd);
+  m.def("attn_fwd_ex", &attn_fwd_ex);
+  m.def("attn_bwd_ex", &attn_bwd_ex);
   m.def("mfma_probe_32x32x16", &mfma_probe_32x32x16);
   m.def("mfma_probe_16x16x32", &mfma_probe_16x16x32);
 }

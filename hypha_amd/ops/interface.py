@@ -65,22 +65,29 @@ def rmsnorm(x: torch.Tensor, weight: torch.Tensor, eps: float = 1e-5) -> torch.T
 
 class _RoPE(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, q, k, cos, sin):
-        qo, ko = _c().rope_fwd(q.contiguous(), k.contiguous(), cos, sin, False)
+    def forward(ctx, q, k, cos, sin, layout):
+        qo, ko = _c().rope_fwd_ex(q.contiguous(), k.contiguous(), cos, sin, False, layout)
         ctx.save_for_backward(cos, sin)
+        ctx.layout = layout
         return qo, ko
 
     @staticmethod
     def backward(ctx, dq, dk):
         cos, sin = ctx.saved_tensors
-        dqo, dko = _c().rope_fwd(dq.contiguous(), dk.contiguous(), cos, sin, True)
-        return dqo, dko, None, None
+        dqo, dko = _c().rope_fwd_ex(dq.contiguous(), dk.contiguous(), cos, sin, True,
+                                    ctx.layout)
+        return dqo, dko, None, None, None
 
 
-def apply_rope_qk(q, k, cos, sin):
-    """q [B,Hq,S,D], k [B,Hkv,S,D]; cos/sin [S_max, D/2] fp32."""
+def apply_rope_qk(q, k, cos, sin, layout: str = "bhsd"):
+    """layout 'bhsd': q [B,Hq,S,D]; 'bshd': q [B,S,Hq,D] (the projection's
+    natural layout — avoids transpose copies). cos/sin [S_max, D/2] fp32."""
     if use_native(q):
-        return _RoPE.apply(q, k, cos, sin)
+        return _RoPE.apply(q, k, cos, sin, layout)
+    if layout == "bshd":
+        qo = reference.apply_rope(q.transpose(1, 2), cos, sin).transpose(1, 2)
+        ko = reference.apply_rope(k.transpose(1, 2), cos, sin).transpose(1, 2)
+        return qo, ko
     return reference.apply_rope(q, cos, sin), reference.apply_rope(k, cos, sin)
 
 
@@ -117,24 +124,32 @@ def swiglu(gate: torch.Tensor, up: torch.Tensor) -> torch.Tensor:
 
 class _FlashAttention(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, q, k, v, causal):
+    def forward(ctx, q, k, v, causal, layout):
         q, k, v = q.contiguous(), k.contiguous(), v.contiguous()
-        o, lse = _c().attn_fwd(q, k, v, causal)
+        o, lse = _c().attn_fwd_ex(q, k, v, causal, layout)
         ctx.save_for_backward(q, k, v, o, lse)
         ctx.causal = causal
+        ctx.layout = layout
         return o
 
     @staticmethod
     def backward(ctx, do):
         q, k, v, o, lse = ctx.saved_tensors
-        dq, dk, dv = _c().attn_bwd(q, k, v, o, do.contiguous(), lse, ctx.causal)
-        return dq, dk, dv, None
+        dq, dk, dv = _c().attn_bwd_ex(q, k, v, o, do.contiguous(), lse, ctx.causal,
+                                      ctx.layout)
+        return dq, dk, dv, None, None
 
 
-def flash_attention(q, k, v, causal: bool = True) -> torch.Tensor:
-    """q [B,Hq,S,D] bf16, k/v [B,Hkv,S,D] bf16 -> [B,Hq,S,D]."""
+def flash_attention(q, k, v, causal: bool = True, layout: str = "bhsd") -> torch.Tensor:
+    """Causal GQA attention. layout 'bhsd': q [B,Hq,S,D]; 'bshd': q [B,S,Hq,D]
+    (no transpose copies around the projections)."""
     if use_native(q):
-        return _FlashAttention.apply(q, k, v, causal)
+        return _FlashAttention.apply(q, k, v, causal, layout)
+    if layout == "bshd":
+        o = reference.attention(
+            q.transpose(1, 2), k.transpose(1, 2), v.transpose(1, 2), causal
+        )
+        return o.transpose(1, 2)
     return reference.attention(q, k, v, causal)
 
 

@@ -232,3 +232,45 @@ def test_model_step_gpu():
         w.maybe_outer_sync()
     assert all(math.isfinite(l) for l in losses), losses
     assert losses[-1] < losses[0], losses
+
+
+def test_attention_bshd_matches_bhsd():
+    """The stride-aware BSHD path must equal the BHSD path (fwd + bwd)."""
+    B, S, Hq, Hkv, D = 2, 256, 8, 2, 128
+    qs = rand_bf16(B, S, Hq, D, seed=40).requires_grad_(True)
+    ks = rand_bf16(B, S, Hkv, D, seed=41).requires_grad_(True)
+    vs = rand_bf16(B, S, Hkv, D, seed=42).requires_grad_(True)
+    o_bshd = ops.flash_attention(qs, ks, vs, causal=True, layout="bshd")
+
+    qb = qs.detach().transpose(1, 2).contiguous().requires_grad_(True)
+    kb = ks.detach().transpose(1, 2).contiguous().requires_grad_(True)
+    vb = vs.detach().transpose(1, 2).contiguous().requires_grad_(True)
+    o_bhsd = ops.flash_attention(qb, kb, vb, causal=True)
+    torch.testing.assert_close(
+        o_bshd.transpose(1, 2).contiguous(), o_bhsd, rtol=1e-3, atol=1e-3
+    )
+    do = rand_bf16(B, S, Hq, D, seed=43)
+    o_bshd.backward(do)
+    o_bhsd.backward(do.transpose(1, 2).contiguous())
+    torch.testing.assert_close(
+        qs.grad.transpose(1, 2).contiguous(), qb.grad, rtol=1e-3, atol=1e-3
+    )
+    torch.testing.assert_close(
+        ks.grad.transpose(1, 2).contiguous(), kb.grad, rtol=1e-3, atol=1e-3
+    )
+    torch.testing.assert_close(
+        vs.grad.transpose(1, 2).contiguous(), vb.grad, rtol=1e-3, atol=1e-3
+    )
+
+
+def test_rope_bshd_matches_bhsd():
+    cos, sin = R.rope_cos_sin(256, 128, base=10000.0)
+    cos, sin = cos.to(DEV), sin.to(DEV)
+    q = rand_bf16(2, 256, 4, 128, seed=44)
+    k = rand_bf16(2, 256, 2, 128, seed=45)
+    qo, ko = ops.apply_rope_qk(q, k, cos, sin, layout="bshd")
+    qb, kb = ops.apply_rope_qk(
+        q.transpose(1, 2).contiguous(), k.transpose(1, 2).contiguous(), cos, sin
+    )
+    torch.testing.assert_close(qo.transpose(1, 2).contiguous(), qb)
+    torch.testing.assert_close(ko.transpose(1, 2).contiguous(), kb)
