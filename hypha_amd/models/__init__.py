@@ -13,7 +13,7 @@ from typing import Callable
 
 import torch.nn as nn
 
-from . import gpt2, llama
+from . import gpt2, llama, moe
 
 _REGISTRY: dict[str, Callable[..., nn.Module]] = {}
 
@@ -40,6 +40,9 @@ for _name in llama.PRESETS:
     register(_name, (lambda n: (lambda **o: llama.build_model(n, **o)))(_name))
 for _name in gpt2.PRESETS:
     register(_name, (lambda n: (lambda **o: gpt2.build_model(n))) (_name))
+for _name in moe.PRESETS:
+    register(_name, (lambda n: (lambda **o: moe.build_model(n, **o)))(_name))
 
 from .llama import LlamaConfig, LlamaForCausalLM  # noqa: E402,F401
 from .gpt2 import GPT2Config, GPT2ForCausalLM  # noqa: E402,F401
+from .moe import MoEConfig, MoEForCausalLM  # noqa: E402,F401

@@ -82,6 +82,18 @@ def main() -> int:
     batch_iter = batches()
     print(f"[executor] model={cfg['model']} device={device} bs={batch_size}", flush=True)
 
+    if cfg.get("join"):
+        # replacement worker (kill/rejoin path): catch up to the current
+        # global weights via the PS's cumulative offset before training
+        print("[executor] joining: waiting for global offset", flush=True)
+        ev = updates_q.get(timeout=600)
+        offset = load_file(ev["path"])
+        if "delta" in offset:
+            worker.fp.theta0.add_(offset["delta"].to(worker.fp.theta0.device))
+            worker.fp.master.copy_(worker.fp.theta0)
+            worker.fp.flat.copy_(worker.fp.master.to(worker.fp.work_dtype))
+        print("[executor] joined at current global weights", flush=True)
+
     done = False
     round_idx = 0
     while not done:

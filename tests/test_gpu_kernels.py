@@ -274,3 +274,20 @@ def test_rope_bshd_matches_bhsd():
     )
     torch.testing.assert_close(qo.transpose(1, 2).contiguous(), qb)
     torch.testing.assert_close(ko.transpose(1, 2).contiguous(), kb)
+
+
+def test_moe_model_step_gpu():
+    from hypha_amd import models
+    from hypha_amd.data.synthetic import SyntheticTokens
+    from hypha_amd.parallel import Comm, DiLoCoConfig, DiLoCoWorker, InnerOptConfig
+
+    torch.manual_seed(0)
+    model = models.build("moe-tiny")
+    w = DiLoCoWorker(model, DiLoCoConfig(h=2, inner=InnerOptConfig(warmup_steps=0)),
+                     comm=Comm(), device=torch.device(DEV))
+    data = SyntheticTokens(512, 128, 2, seed=31)
+    ids, labels = data.next_batch()
+    first = w.train_step(ids, labels)
+    for _ in range(6):
+        last = w.train_step(ids.clone(), labels.clone())
+    assert math.isfinite(last) and last < first, (first, last)

@@ -44,3 +44,31 @@ def test_registry_lists_flagships():
     av = models.available()
     for name in ("llama3-8b", "llama3-70b", "gpt2-small", "llama-tiny"):
         assert name in av
+
+
+def test_moe_tiny_forward_backward():
+    torch.manual_seed(0)
+    m = models.build("moe-tiny")
+    ids = torch.randint(0, 512, (2, 32))
+    loss = m(ids, labels=ids)
+    assert torch.isfinite(loss)
+    loss.backward()
+    # every expert and the router must receive gradient
+    blk = m.blocks[0].mlp
+    assert blk.router.weight.grad is not None
+    assert blk.w_gate.grad is not None and torch.isfinite(blk.w_gate.grad).all()
+
+
+def test_moe_routing_weights_normalized():
+    torch.manual_seed(1)
+    m = models.build("moe-tiny")
+    m.eval()
+    ids = torch.randint(0, 512, (1, 16))
+    out = m(ids)
+    assert out.shape == (1, 16, 512)
+
+
+def test_mixtral_param_count():
+    cfg = models.moe.PRESETS["mixtral-8x7b"]
+    n = cfg.num_params()
+    assert 45e9 < n < 48e9, n  # Mixtral-8x7B ~46.7B params
