@@ -67,9 +67,14 @@ class Gateway {
       std::string kind = msg->get_or("kind", Json("")).as_string();
       if (kind == "register") {
         peer = msg->at("peer").as_string();
-        std::lock_guard<std::mutex> lk(mu_);
-        peers_[peer] = sock;
-        kv_["addr:" + peer] = msg->at("addr");
+        {
+          std::lock_guard<std::mutex> lk(mu_);
+          peers_[peer] = sock;
+          kv_["addr:" + peer] = msg->at("addr");
+        }
+        Json ack;
+        ack["kind"] = "registered";
+        sock->send_json(ack);
       } else if (kind == "subscribe") {
         std::lock_guard<std::mutex> lk(mu_);
         subs_[msg->at("topic").as_string()].insert(peer);

@@ -147,6 +147,10 @@ void Node::start(int port) {
     reg["peer"] = name_;
     reg["addr"] = std::string("127.0.0.1:") + std::to_string(port_);
     gw_sock_->send_json(reg);
+    // synchronous ack: the registry entry is visible before start() returns
+    auto ack = gw_sock_->recv_json();
+    if (!ack || ack->get_or("kind", Json("")).as_string() != "registered")
+      throw std::runtime_error("node: gateway registration failed");
     gw_thread_ = std::thread([this] { gateway_listen_loop(); });
   }
 }
@@ -354,12 +358,19 @@ std::optional<std::string> Node::resolve(const std::string& peer) {
     auto it = addr_cache_.find(peer);
     if (it != addr_cache_.end()) return it->second;
   }
-  auto v = kv_get("addr:" + peer);
-  if (!v) return std::nullopt;
-  std::string addr = v->as_string();
-  std::lock_guard<std::mutex> lk(mu_);
-  addr_cache_[peer] = addr;
-  return addr;
+  // brief retry: peers may register concurrently with the first lookup
+  for (int attempt = 0; attempt < 20; ++attempt) {
+    auto v = kv_get("addr:" + peer);
+    if (v) {
+      std::string addr = v->as_string();
+      std::lock_guard<std::mutex> lk(mu_);
+      addr_cache_[peer] = addr;
+      return addr;
+    }
+    struct timespec ts = {0, 50 * 1000 * 1000};
+    nanosleep(&ts, nullptr);
+  }
+  return std::nullopt;
 }
 
 }  // namespace hypha
