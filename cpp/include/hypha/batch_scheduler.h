@@ -66,9 +66,11 @@ class BatchScheduler {
       }
       case Progress::Updated: {
         // aggregate applied: advance the round; workers go back to Training
-        // when their UpdateReceived arrives
+        // when their UpdateReceived arrives. Done tells the aggregator to
+        // exit once the configured rounds are exhausted.
         progress_.next_round();
         round_scheduled_ = false;
+        if (progress_.training_finished()) return {ProgressResponse::Done};
         return {ProgressResponse::Ok};
       }
       case Progress::UpdateReceived: {

@@ -141,7 +141,7 @@ def test_cluster_worker_kill_and_rejoin(binaries, tmp_path):
         cfg = tmp_path / "job.json"
         cfg.write_text(
             '{"model": "llama-tiny", "dataset": "synth", "num_workers": 2,'
-            ' "update_rounds": 3, "avg_samples_between_updates": 8,'
+            ' "update_rounds": 6, "avg_samples_between_updates": 8,'
             ' "batch_size": 2, "seq_len": 128, "inner_lr": 0.001}'
         )
         sched = subprocess.Popen(
