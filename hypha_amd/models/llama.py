@@ -87,7 +87,7 @@ class Attention(nn.Module):
         self.wv = nn.Linear(cfg.hidden_size, cfg.n_kv_heads * hd, bias=False)
         self.wo = nn.Linear(cfg.n_heads * hd, cfg.hidden_size, bias=False)
 
-    def forward(self, x, cos, sin):
+    def forward(self, x, cos, sin, cache=None, pos: int = 0):
         b, s, _ = x.shape
         cfg = self.cfg
         hd = cfg.head_dim
@@ -96,8 +96,29 @@ class Attention(nn.Module):
         q = self.wq(x).view(b, s, cfg.n_heads, hd)
         k = self.wk(x).view(b, s, cfg.n_kv_heads, hd)
         v = self.wv(x).view(b, s, cfg.n_kv_heads, hd)
-        q, k = ops.apply_rope_qk(q, k, cos, sin, layout="bshd")
-        o = ops.flash_attention(q, k, v, causal=True, layout="bshd")
+        if cache is None:
+            q, k = ops.apply_rope_qk(q, k, cos, sin, layout="bshd")
+            o = ops.flash_attention(q, k, v, causal=True, layout="bshd")
+            return self.wo(o.reshape(b, s, -1))
+        # inference with KV cache: rotate at absolute positions, append, and
+        # attend over the full prefix (decode = batched matvec via hipBLASLt)
+        q, k = ops.apply_rope_qk(q, k, cos[pos:], sin[pos:], layout="bshd")
+        cache["k"] = k if "k" not in cache else torch.cat([cache["k"], k], dim=1)
+        cache["v"] = v if "v" not in cache else torch.cat([cache["v"], v], dim=1)
+        kc, vc = cache["k"], cache["v"]
+        rep = cfg.n_heads // cfg.n_kv_heads
+        qh = q.transpose(1, 2)  # [b, hq, s, d]
+        kh = kc.transpose(1, 2).repeat_interleave(rep, dim=1)
+        vh = vc.transpose(1, 2).repeat_interleave(rep, dim=1)
+        scores = (qh.float() @ kh.float().transpose(-1, -2)) / math.sqrt(hd)
+        t = kc.shape[1]
+        if s > 1:  # prefill chunk: causal mask inside the chunk
+            mask = torch.arange(t, device=x.device)[None, :] > (
+                pos + torch.arange(s, device=x.device)[:, None]
+            )
+            scores = scores.masked_fill(mask, float("-inf"))
+        p = torch.softmax(scores, dim=-1)
+        o = (p @ vh.float()).to(x.dtype).transpose(1, 2)
         return self.wo(o.reshape(b, s, -1))
 
 
@@ -120,8 +141,8 @@ class Block(nn.Module):
         self.mlp_norm = RMSNorm(cfg.hidden_size, cfg.norm_eps)
         self.mlp = MLP(cfg)
 
-    def forward(self, x, cos, sin):
-        x = x + self.attn(self.attn_norm(x), cos, sin)
+    def forward(self, x, cos, sin, cache=None, pos: int = 0):
+        x = x + self.attn(self.attn_norm(x), cos, sin, cache=cache, pos=pos)
         x = x + self.mlp(self.mlp_norm(x))
         return x
 
@@ -170,6 +191,48 @@ class LlamaForCausalLM(nn.Module):
         # next-token prediction: shift
         loss = ops.cross_entropy_loss(logits[:, :-1], labels[:, 1:])
         return loss
+
+
+@torch.no_grad()
+def _generate(self, input_ids: torch.Tensor, max_new_tokens: int = 32,
+              temperature: float = 0.0, top_k: int = 0,
+              seed: int | None = None) -> torch.Tensor:
+    """Autoregressive generation with a per-layer KV cache (inference parity:
+    the reference serves inference through the same executor surface)."""
+    self.eval()
+    b, s = input_ids.shape
+    caches = [dict() for _ in self.blocks]
+    gen = None
+    if seed is not None:
+        gen = torch.Generator(device=input_ids.device).manual_seed(seed)
+    tokens = input_ids
+    x_in = input_ids
+    pos = 0
+    for _ in range(max_new_tokens):
+        x = self.embed(x_in)
+        for blk, cache in zip(self.blocks, caches):
+            x = blk(x, self.rope_cos, self.rope_sin, cache=cache, pos=pos)
+        x = self.norm(x[:, -1:])
+        if self.lm_head is not None:
+            logits = self.lm_head(x)[:, 0]
+        else:
+            logits = torch.nn.functional.linear(x, self.embed.weight)[:, 0]
+        if temperature <= 0:
+            nxt = logits.argmax(-1, keepdim=True)
+        else:
+            logits = logits / temperature
+            if top_k > 0:
+                kth = logits.topk(top_k, dim=-1).values[..., -1, None]
+                logits = logits.masked_fill(logits < kth, float("-inf"))
+            probs = torch.softmax(logits.float(), dim=-1)
+            nxt = torch.multinomial(probs, 1, generator=gen)
+        pos += x_in.shape[1]
+        tokens = torch.cat([tokens, nxt], dim=1)
+        x_in = nxt
+    return tokens
+
+
+LlamaForCausalLM.generate = _generate
 
 
 def build_model(name: str, **overrides) -> LlamaForCausalLM:

@@ -1,0 +1,110 @@
+"""Config layering/validation + telemetry registry tests (SURVEY.md §5:
+config crate and telemetry crate parity)."""
+
+import json
+
+import pytest
+
+from hypha_amd.config import ConfigError, JobConfig, example_config, load_job_config
+from hypha_amd.telemetry import Metrics
+
+
+def test_defaults_validate():
+    cfg = load_job_config()
+    assert cfg.model == "llama3-8b" and cfg.num_workers == 8
+
+
+def test_file_layer(tmp_path):
+    p = tmp_path / "job.json"
+    p.write_text(json.dumps({"model": "gpt2-small", "num_workers": 2}))
+    cfg = load_job_config(str(p))
+    assert cfg.model == "gpt2-small" and cfg.num_workers == 2
+    assert cfg.seq_len == 2048  # default survives
+
+
+def test_toml_layer(tmp_path):
+    p = tmp_path / "job.toml"
+    p.write_text('model = "llama3-70b"\nbatch_size = 2\n')
+    cfg = load_job_config(str(p))
+    assert cfg.model == "llama3-70b" and cfg.batch_size == 2
+
+
+def test_env_overrides_file(tmp_path):
+    p = tmp_path / "job.json"
+    p.write_text(json.dumps({"num_workers": 2}))
+    cfg = load_job_config(str(p), env={"HYPHA_NUM_WORKERS": "4"})
+    assert cfg.num_workers == 4
+
+
+def test_explicit_overrides_env():
+    cfg = load_job_config(env={"HYPHA_SEQ_LEN": "512"}, seq_len=1024)
+    assert cfg.seq_len == 1024
+
+
+def test_unknown_key_rejected(tmp_path):
+    p = tmp_path / "job.json"
+    p.write_text(json.dumps({"no_such_option": 1}))
+    with pytest.raises(ConfigError, match="no_such_option"):
+        load_job_config(str(p))
+
+
+def test_validation_errors():
+    with pytest.raises(ConfigError, match="num_workers"):
+        JobConfig(num_workers=0).validate()
+    with pytest.raises(ConfigError, match="lr_schedule"):
+        JobConfig(lr_schedule="exotic").validate()
+
+
+def test_example_config_mentions_every_field():
+    text = example_config()
+    for f in ("model", "outer_lr", "update_rounds", "seq_len"):
+        assert f'"{f}"' in text
+
+
+def test_metrics_counters_and_gauges():
+    m = Metrics()
+    m.counter_add("a.b", 2)
+    m.counter_add("a.b", 3)
+    m.gauge_set("g", 7.5, rank=0)
+    snap = m.snapshot()
+    assert snap["counters"]["a.b"] == 5
+    assert snap["gauges"]["g{rank=0}"] == 7.5
+
+
+def test_comm_instrumentation():
+    import torch
+
+    from hypha_amd.parallel import Comm
+    from hypha_amd.telemetry import METRICS, instrument_comm
+
+    c = Comm()
+    instrument_comm(c)
+    before = METRICS.snapshot()["counters"].get(
+        "hypha.bandwidth.collective.payload_bytes", 0
+    )
+    c.all_reduce_mean_flat(torch.zeros(1000))
+    after = METRICS.snapshot()["counters"]["hypha.bandwidth.collective.payload_bytes"]
+    assert after - before == 4000
+
+
+def test_certutil_pki_roundtrip(tmp_path):
+    """3-tier PKI generation + chain verification (certutil crate parity)."""
+    import subprocess
+    import sys
+    from pathlib import Path
+
+    tool = Path(__file__).resolve().parent.parent / "tools" / "hypha_certutil.py"
+    out = tmp_path / "pki"
+    subprocess.run([sys.executable, str(tool), "root", "--out", str(out)], check=True)
+    subprocess.run([sys.executable, str(tool), "org", "--out", str(out),
+                    "--name", "org1"], check=True)
+    r = subprocess.run([sys.executable, str(tool), "node", "--out", str(out),
+                        "--org", "org1", "--name", "worker-0"],
+                       check=True, capture_output=True, text=True)
+    assert "peer-id: peer-" in r.stdout
+    # openssl verifies the chain root -> org -> node
+    v = subprocess.run(
+        ["openssl", "verify", "-CAfile", str(out / "root.crt"),
+         "-untrusted", str(out / "org1.crt"), str(out / "worker-0.crt")],
+        capture_output=True, text=True)
+    assert v.returncode == 0, v.stderr
