@@ -11,6 +11,11 @@ void adamw_step_(torch::Tensor master, torch::Tensor param, torch::Tensor grad,
 void nesterov_step_(torch::Tensor theta, torch::Tensor delta, torch::Tensor mom, double lr,
                     double mu);
 void extract_delta(torch::Tensor master, torch::Tensor theta0, torch::Tensor out);
+// adamw8.hip
+void adamw8_step_(torch::Tensor master, torch::Tensor param, torch::Tensor grad,
+                  torch::Tensor m8, torch::Tensor v8, torch::Tensor m_scale,
+                  torch::Tensor v_scale, double lr, double beta1, double beta2,
+                  double eps, double wd, long step);
 torch::Tensor swiglu_fwd(torch::Tensor gate, torch::Tensor up);
 std::vector<torch::Tensor> swiglu_bwd(torch::Tensor dout, torch::Tensor gate,
                                       torch::Tensor up);
@@ -48,6 +53,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("adamw_step_", &adamw_step_);
   m.def("nesterov_step_", &nesterov_step_);
   m.def("extract_delta", &extract_delta);
+  m.def("adamw8_step_", &adamw8_step_);
   m.def("swiglu_fwd", &swiglu_fwd);
   m.def("swiglu_bwd", &swiglu_bwd);
   m.def("rmsnorm_fwd", &rmsnorm_fwd);
