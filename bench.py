@@ -31,6 +31,8 @@ def main() -> int:
     p.add_argument("--seq-len", type=int, default=2048)
     p.add_argument("--h", type=int, default=100, help="DiLoCo inner steps per outer sync")
     p.add_argument("--device", type=str, default=None)
+    p.add_argument("--fp8", action="store_true", help="fp8 e4m3 GEMMs (config 5 path)")
+    p.add_argument("--state-bits", type=int, default=32, choices=(32, 8))
     args = p.parse_args()
 
     from hypha_amd import models, ops
@@ -48,9 +50,16 @@ def main() -> int:
 
     torch.manual_seed(1234)  # same init on every rank pre-broadcast
     model = models.build(args.model)
+    if args.fp8 and on_gpu:
+        from hypha_amd.ops.fp8 import convert_linears_to_fp8
+
+        n_conv = convert_linears_to_fp8(model)
+        if rank == 0:
+            print(f"# fp8: converted {n_conv} linears", file=sys.stderr)
     cfg = DiLoCoConfig(
         h=args.h,
-        inner=InnerOptConfig(lr=4e-4, warmup_steps=10, schedule="constant"),
+        inner=InnerOptConfig(lr=4e-4, warmup_steps=10, schedule="constant",
+                             state_bits=args.state_bits),
     )
     worker = DiLoCoWorker(model, cfg, comm=comm, device=device)
     data = SyntheticTokens(
@@ -106,7 +115,7 @@ def main() -> int:
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,
-            "dtype": "bf16" if on_gpu else "fp32",
+            "dtype": ("fp8-gemm/bf16" if args.fp8 else "bf16") if on_gpu else "fp32",
             "data": "synthetic",
             "config": {
                 "model": args.model,

@@ -23,15 +23,16 @@ def save_checkpoint(worker, out_dir: str) -> None:
     os.makedirs(out_dir, exist_ok=True)
     fp = worker.fp
     save_file({"theta_global": fp.theta0.cpu()}, os.path.join(out_dir, "0_global_weights.safetensors"))
-    save_file(
-        {
-            "outer_momentum": fp.outer_momentum.cpu(),
-            "exp_avg": fp.exp_avg.cpu(),
-            "exp_avg_sq": fp.exp_avg_sq.cpu(),
-            "master": fp.master.cpu(),
-        },
-        os.path.join(out_dir, "optimizer_state.safetensors"),
-    )
+    state = {
+        "outer_momentum": fp.outer_momentum.cpu(),
+        "master": fp.master.cpu(),
+    }
+    if getattr(fp, "state_bits", 32) == 8:
+        state.update(m8=fp.m8.cpu(), v8=fp.v8.cpu(),
+                     m_scale=fp.m_scale.cpu(), v_scale=fp.v_scale.cpu())
+    else:
+        state.update(exp_avg=fp.exp_avg.cpu(), exp_avg_sq=fp.exp_avg_sq.cpu())
+    save_file(state, os.path.join(out_dir, "optimizer_state.safetensors"))
     manifest = {
         "format": "hypha_amd.checkpoint.v1",
         "numel": fp.numel,
@@ -59,8 +60,14 @@ def load_checkpoint(worker, ckpt_dir: str) -> dict:
     fp.theta0.copy_(gw["theta_global"].to(fp.theta0.device))
     opt = load_file(os.path.join(ckpt_dir, "optimizer_state.safetensors"))
     fp.outer_momentum.copy_(opt["outer_momentum"].to(fp.master.device))
-    fp.exp_avg.copy_(opt["exp_avg"].to(fp.master.device))
-    fp.exp_avg_sq.copy_(opt["exp_avg_sq"].to(fp.master.device))
+    if "m8" in opt:
+        fp.m8.copy_(opt["m8"].to(fp.master.device))
+        fp.v8.copy_(opt["v8"].to(fp.master.device))
+        fp.m_scale.copy_(opt["m_scale"].to(fp.master.device))
+        fp.v_scale.copy_(opt["v_scale"].to(fp.master.device))
+    else:
+        fp.exp_avg.copy_(opt["exp_avg"].to(fp.master.device))
+        fp.exp_avg_sq.copy_(opt["exp_avg_sq"].to(fp.master.device))
     fp.master.copy_(opt["master"].to(fp.master.device))
     fp.flat.copy_(fp.master.to(fp.work_dtype))
     worker.inner_step_count = manifest["inner_step_count"]

@@ -37,6 +37,7 @@ class InnerOptConfig:
     schedule: str = "cosine"  # constant | cosine | linear | wsd
     total_steps: int = 10000
     min_lr_frac: float = 0.1
+    state_bits: int = 32  # 32 = fp32 m/v; 8 = blockwise-uint8 m/v (config 5)
 
 
 @dataclass
@@ -79,7 +80,10 @@ class FlatParams:
     """All trainable params flattened into one contiguous working buffer,
     with fp32 master/optimizer/outer state as matching flat buffers."""
 
-    def __init__(self, model: nn.Module, device: torch.device, work_dtype: torch.dtype):
+    QBLOCK = 2048  # must match the adamw8 kernel's quant block
+
+    def __init__(self, model: nn.Module, device: torch.device, work_dtype: torch.dtype,
+                 state_bits: int = 32):
         self.params = [p for p in model.parameters() if p.requires_grad]
         self.numel = sum(p.numel() for p in self.params)
         self.device = device
@@ -96,8 +100,18 @@ class FlatParams:
             offset += n
 
         self.master = self.flat.float()  # fp32 master weights
-        self.exp_avg = torch.zeros_like(self.master)
-        self.exp_avg_sq = torch.zeros_like(self.master)
+        self.state_bits = state_bits
+        if state_bits == 8 and device.type == "cuda":
+            nblocks = (self.numel + self.QBLOCK - 1) // self.QBLOCK
+            self.m8 = torch.full((self.numel,), 127, dtype=torch.uint8, device=device)
+            self.v8 = torch.zeros(self.numel, dtype=torch.uint8, device=device)
+            self.m_scale = torch.full((nblocks,), 1e-12, device=device)
+            self.v_scale = torch.full((nblocks,), 1e-12, device=device)
+            self.exp_avg = self.exp_avg_sq = None
+        else:
+            self.state_bits = 32
+            self.exp_avg = torch.zeros_like(self.master)
+            self.exp_avg_sq = torch.zeros_like(self.master)
         self.theta0 = self.master.clone()  # global weights at round start
         self.outer_momentum = torch.zeros_like(self.master)
 
@@ -136,7 +150,7 @@ class DiLoCoWorker:
         for buf in model.buffers():
             if buf.dtype in (torch.bfloat16, torch.float16):
                 buf.data = buf.data.float()
-        self.fp = FlatParams(model, device, work_dtype)
+        self.fp = FlatParams(model, device, work_dtype, state_bits=cfg.inner.state_bits)
         # make every rank start from identical weights (rank0's init wins)
         if self.comm.is_distributed:
             self.comm.broadcast_flat(self.fp.master, src=0)
@@ -176,19 +190,29 @@ class DiLoCoWorker:
             # clamp to 1 and always scale: no host sync, one fused mul
             clip_coef = (self.cfg.grad_clip / (gnorm + 1e-6)).clamp_(max=1.0)
             self.fp.flat_grad.mul_(clip_coef.to(self.fp.flat_grad.dtype))
-        ops.fused_adamw(
-            self.fp.master,
-            self.fp.flat,
-            self.fp.flat_grad,
-            self.fp.exp_avg,
-            self.fp.exp_avg_sq,
-            lr=lr_at(c, self.inner_step_count - 1),
-            beta1=c.beta1,
-            beta2=c.beta2,
-            eps=c.eps,
-            weight_decay=c.weight_decay,
-            step=self.inner_step_count,
-        )
+        if self.fp.state_bits == 8:
+            from hypha_amd import _C
+
+            _C.adamw8_step_(
+                self.fp.master, self.fp.flat, self.fp.flat_grad,
+                self.fp.m8, self.fp.v8, self.fp.m_scale, self.fp.v_scale,
+                lr_at(c, self.inner_step_count - 1), c.beta1, c.beta2, c.eps,
+                c.weight_decay, self.inner_step_count,
+            )
+        else:
+            ops.fused_adamw(
+                self.fp.master,
+                self.fp.flat,
+                self.fp.flat_grad,
+                self.fp.exp_avg,
+                self.fp.exp_avg_sq,
+                lr=lr_at(c, self.inner_step_count - 1),
+                beta1=c.beta1,
+                beta2=c.beta2,
+                eps=c.eps,
+                weight_decay=c.weight_decay,
+                step=self.inner_step_count,
+            )
         self.fp.zero_grad()
 
     # -- outer ------------------------------------------------------------
