@@ -77,15 +77,18 @@ class Fp8Linear(nn.Module):
         return out.reshape(*shape[:-1], -1)
 
 
-def convert_linears_to_fp8(model: nn.Module, min_features: int = 1024) -> int:
+def convert_linears_to_fp8(model: nn.Module, min_features: int = 1024,
+                           max_features: int = 65536) -> int:
     """Swap every large nn.Linear (bias-free) for Fp8Linear. Returns count.
-    Small projections (routers, tiny models) stay bf16."""
+    Small projections (routers, tiny models) stay bf16; so does the vocab
+    projection (> max_features): its backward would materialize transposed
+    fp8 copies of the [tokens, vocab] gradient (gigabytes)."""
     n = 0
     for parent in model.modules():
         for name, child in list(parent.named_children()):
             if (isinstance(child, nn.Linear) and child.bias is None
-                    and child.in_features >= min_features
-                    and child.out_features >= min_features):
+                    and min_features <= child.in_features <= max_features
+                    and min_features <= child.out_features <= max_features):
                 setattr(parent, name, Fp8Linear.from_linear(child))
                 n += 1
     return n
