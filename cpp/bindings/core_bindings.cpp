@@ -13,6 +13,7 @@
 #include "hypha/json.h"
 #include "hypha/leases.h"
 #include "hypha/net.h"
+#include "hypha/ps_math.h"
 #include "hypha/resources.h"
 #include "hypha/simulation.h"
 #include "hypha/trackers.h"
@@ -257,6 +258,24 @@ PYBIND11_MODULE(_core, m) {
       .def_readonly("offer_price", &ArbiterDecision::offer_price);
 
   m.def("select_requests", &select_requests);
+
+  // parameter-server file pipeline (the production code path, for golden tests)
+  m.def("ps_aggregate_files",
+        [](const std::vector<std::string>& delta_files, const std::string& momentum_io,
+           const std::string& update_out, double lr, double mu) {
+          SafeTensors avg = ps_average(delta_files);
+          SafeTensors momentum;
+          FILE* f = fopen(momentum_io.c_str(), "rb");
+          if (f) {
+            fclose(f);
+            momentum = SafeTensors::load(momentum_io);
+          } else {
+            momentum = ps_zeros_like(avg);
+          }
+          SafeTensors update = ps_nesterov(avg, momentum, lr, mu);
+          momentum.save(momentum_io);
+          update.save(update_out);
+        });
 
   // ---- in-process control-plane networking (multi-peer tests) ----
   py::class_<Gateway>(m, "Gateway")

@@ -230,3 +230,50 @@ def test_static_resource_manager():
     assert not m.reserve(core.Resources(2, 1, 1, 1))  # double-checked
     m.release(core.Resources(1, 4, 16, 50))
     assert m.reserve(core.Resources(2, 8, 32, 100))
+
+
+def test_ps_nesterov_golden_vs_torch(tmp_path):
+    """The C++ parameter-server file pipeline (safetensors average + outer
+    Nesterov with persistent momentum) must match torch SGD(nesterov=True)
+    fed the negated average delta — the reference's golden-value test
+    (parameter_server.rs:448-525) applied to OUR production code path."""
+    import torch
+    from safetensors.torch import load_file, save_file
+
+    torch.manual_seed(5)
+    n = 64
+    theta = torch.randn(n)
+    p_ref = theta.clone().requires_grad_(True)
+    opt = torch.optim.SGD([p_ref], lr=0.7, momentum=0.9, nesterov=True)
+    mom_path = str(tmp_path / "momentum.safetensors")
+    for rnd in range(3):
+        deltas = [torch.randn(n) for _ in range(3)]
+        files = []
+        for i, d in enumerate(deltas):
+            fp = str(tmp_path / f"d{rnd}_{i}.safetensors")
+            save_file({"w": d}, fp)
+            files.append(fp)
+        upd_path = str(tmp_path / f"u{rnd}.safetensors")
+        core.ps_aggregate_files(files, mom_path, upd_path, 0.7, 0.9)
+        update = load_file(upd_path)["w"]
+        theta = theta + update  # worker merge: theta <- theta_0 + U
+        avg = torch.stack(deltas).mean(0)
+        p_ref.grad = -avg
+        opt.step()
+    torch.testing.assert_close(theta, p_ref.detach(), rtol=1e-4, atol=1e-5)
+
+
+def test_ps_average_bf16_and_f32(tmp_path):
+    import torch
+    from safetensors.torch import load_file, save_file
+
+    a = torch.randn(10, dtype=torch.bfloat16)
+    b = torch.randn(10, dtype=torch.bfloat16)
+    save_file({"t": a}, str(tmp_path / "a.safetensors"))
+    save_file({"t": b}, str(tmp_path / "b.safetensors"))
+    core.ps_aggregate_files(
+        [str(tmp_path / "a.safetensors"), str(tmp_path / "b.safetensors")],
+        str(tmp_path / "m.safetensors"), str(tmp_path / "u.safetensors"), 1.0, 0.0)
+    u = load_file(str(tmp_path / "u.safetensors"))["t"]
+    want = ((a.float() + b.float()) / 2).bfloat16()
+    torch.testing.assert_close(u.float(), want.float(), rtol=2e-2, atol=2e-2)
