@@ -69,7 +69,7 @@ def load_checkpoint(worker, ckpt_dir: str) -> dict:
         fp.exp_avg.copy_(opt["exp_avg"].to(fp.master.device))
         fp.exp_avg_sq.copy_(opt["exp_avg_sq"].to(fp.master.device))
     fp.master.copy_(opt["master"].to(fp.master.device))
-    fp.flat.copy_(fp.master.to(fp.work_dtype))
+    fp.flat.copy_(fp.master)
     worker.inner_step_count = manifest["inner_step_count"]
     worker.round = manifest["round"]
     worker.steps_in_round = manifest["steps_in_round"]

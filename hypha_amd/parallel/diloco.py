@@ -154,7 +154,7 @@ class DiLoCoWorker:
         # make every rank start from identical weights (rank0's init wins)
         if self.comm.is_distributed:
             self.comm.broadcast_flat(self.fp.master, src=0)
-            self.fp.flat.copy_(self.fp.master.to(work_dtype))
+            self.fp.flat.copy_(self.fp.master)
             self.fp.theta0.copy_(self.fp.master)
         self.inner_step_count = 0  # global inner step counter (for LR/bias corr)
         self.round = 0  # outer rounds completed
@@ -245,7 +245,7 @@ class DiLoCoWorker:
                 mu=self.cfg.outer.momentum,
             )
         fp.master.copy_(fp.theta0)
-        fp.flat.copy_(fp.master.to(fp.work_dtype))
+        fp.flat.copy_(fp.master)
         self.outer_sync_payload_bytes += n * self._delta_buf.element_size()
         self.round += 1
         self.steps_in_round = 0

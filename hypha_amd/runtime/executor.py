@@ -91,7 +91,7 @@ def main() -> int:
         if "delta" in offset:
             worker.fp.theta0.add_(offset["delta"].to(worker.fp.theta0.device))
             worker.fp.master.copy_(worker.fp.theta0)
-            worker.fp.flat.copy_(worker.fp.master.to(worker.fp.work_dtype))
+            worker.fp.flat.copy_(worker.fp.master)
         print("[executor] joined at current global weights", flush=True)
 
     done = False
@@ -123,7 +123,7 @@ def main() -> int:
         u = load_file(ev["path"])["delta"].to(worker.fp.theta0.device)
         worker.fp.theta0.add_(u)
         worker.fp.master.copy_(worker.fp.theta0)
-        worker.fp.flat.copy_(worker.fp.master.to(worker.fp.work_dtype))
+        worker.fp.flat.copy_(worker.fp.master)
 
         resp = session.send_status({"kind": "update-received"})
         done = resp.get("kind") == "done"
