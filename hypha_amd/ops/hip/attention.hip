@@ -87,28 +87,51 @@ __global__ __launch_bounds__(256, 1) void attn_fwd_kernel(
   f32x16 ot[DBLK] = {};
   float m_run = -INFINITY, l_run = 0.f;
 
-  const int kv_end = causal ? (q0wg + WGQ) : S;  // exclusive upper bound
-  for (int kv0 = 0; kv0 < kv_end; kv0 += KVB) {
-    // ---- stage K [KVB][HD] (swizzled) and V^T [HD][KVB] ----
-    __syncthreads();
-    constexpr int CH = KVB * HD / 8;  // 16B chunks
-    for (int c = tid; c < CH; c += 256) {
-      int row = c / (HD / 8), e0 = (c % (HD / 8)) * 8;
-      s16x8 kv8 = *reinterpret_cast<const s16x8*>(kg + kvbase + (long long)(kv0 + row) * kv_ss + e0);
-      *reinterpret_cast<s16x8*>((char*)k_lds + k_lds_off<HD>(row, e0)) = kv8;
+  // ---- register-staged K/V prefetch (guide T14: issue the next tile's
+  // global loads before computing the current one; HBM latency hides under
+  // the MFMA phase instead of stalling every wave at the staging barrier) --
+  constexpr int KCH = KVB * HD / 8 / 256;  // 16B K chunks per thread
+  s16x8 kreg[KCH];
+  s16x8 vreg[HD / 32];
+  auto load_tile = [&](int kv0) {
+#pragma unroll
+    for (int c = 0; c < KCH; ++c) {
+      int cc = c * 256 + tid;
+      int row = cc / (HD / 8), e0 = (cc % (HD / 8)) * 8;
+      kreg[c] = *reinterpret_cast<const s16x8*>(kg + kvbase + (long long)(kv0 + row) * kv_ss + e0);
     }
-    // V^T transpose staging: lane-per-kv so each scalar-store instruction's
-    // 64 lanes write one contiguous 128 B image row span (conflict-free;
-    // the d-major pattern was a 16-way bank conflict: 8*VT_PITCH*2 = 0 mod 128)
 #pragma unroll
     for (int i = 0; i < HD / 32; ++i) {
       int kvr = tid & 63;
       int e0 = (i * 4 + (tid >> 6)) * 8;
-      s16x8 vv8 = *reinterpret_cast<const s16x8*>(vg + kvbase + (long long)(kv0 + kvr) * kv_ss + e0);
-#pragma unroll
-      for (int j = 0; j < 8; ++j) vt_lds[(e0 + j) * VT_PITCH + kvr] = vv8[j];
+      vreg[i] = *reinterpret_cast<const s16x8*>(vg + kvbase + (long long)(kv0 + kvr) * kv_ss + e0);
     }
-    __syncthreads();
+  };
+  auto write_tile = [&]() {
+#pragma unroll
+    for (int c = 0; c < KCH; ++c) {
+      int cc = c * 256 + tid;
+      int row = cc / (HD / 8), e0 = (cc % (HD / 8)) * 8;
+      *reinterpret_cast<s16x8*>((char*)k_lds + k_lds_off<HD>(row, e0)) = kreg[c];
+    }
+    // V^T transpose: lane-per-kv stores walk the contiguous image row
+    // (conflict-free; the d-major pattern was a 16-way bank conflict)
+#pragma unroll
+    for (int i = 0; i < HD / 32; ++i) {
+      int kvr = tid & 63;
+      int e0 = (i * 4 + (tid >> 6)) * 8;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) vt_lds[(e0 + j) * VT_PITCH + kvr] = vreg[i][j];
+    }
+  };
+
+  const int kv_end = causal ? (q0wg + WGQ) : S;  // exclusive upper bound
+  load_tile(0);
+  for (int kv0 = 0; kv0 < kv_end; kv0 += KVB) {
+    __syncthreads();  // previous tile's LDS consumers done
+    write_tile();
+    if (kv0 + KVB < kv_end) load_tile(kv0 + KVB);  // prefetch next tile
+    __syncthreads();  // this tile's LDS image ready
 
     if (causal && kv0 > q0 + QB - 1) continue;  // wave fully above the diagonal
 
