@@ -31,7 +31,7 @@ __global__ void adamw8_kernel(float* __restrict__ master, short* __restrict__ pa
     const float ms = m_scale[blk];
     const float vs = v_scale[blk];
 
-    float mv[8], vv[8], wv[8];
+    float mv[8], vv[8];
     float local_am = 0.f, local_av = 0.f;
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
