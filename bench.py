@@ -23,12 +23,14 @@ os.environ.setdefault("PYTORCH_ALLOC_CONF", "expandable_segments:True")
 
 # Pre-tuned hipBLASLt algorithm selections for the flagship shapes (generated
 # once with PYTORCH_TUNABLEOP_TUNING=1 on an MI355X; +2% step time).
-_TUNED = os.path.join(os.path.dirname(os.path.abspath(__file__)),
-                      "tunableop_gfx950_llama8b.csv")
+# torch inserts the device ordinal before .csv: rank N reads ...llama8b{N}.csv
+_TUNED_BASE = os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                           "tunableop_gfx950_llama8b.csv")
+_TUNED = _TUNED_BASE.replace(".csv", "0.csv")
 if os.path.exists(_TUNED) and "PYTORCH_TUNABLEOP_ENABLED" not in os.environ:
     os.environ["PYTORCH_TUNABLEOP_ENABLED"] = "1"
     os.environ["PYTORCH_TUNABLEOP_TUNING"] = "0"
-    os.environ["PYTORCH_TUNABLEOP_FILENAME"] = _TUNED
+    os.environ["PYTORCH_TUNABLEOP_FILENAME"] = _TUNED_BASE
 
 import torch
 
