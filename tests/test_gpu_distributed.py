@@ -1,7 +1,8 @@
-"""Real-RCCL distributed test on a single MI355X: two ranks share GPU 0 with
-backend "nccl" (= RCCL on ROCm) and run DiLoCo outer syncs — exercising the
-exact collective path the 8-GPU scaling bench uses, without needing 8 GPUs.
-(HSA_ENABLE_IPC_MODE_LEGACY=0 must be in the env — it is, per the image.)"""
+"""Real-RCCL distributed test: two ranks on two GPUs with backend "nccl"
+(= RCCL on ROCm) running DiLoCo outer syncs — the exact collective path the
+8-GPU scaling bench uses. NCCL/RCCL forbids two ranks sharing one device, so
+this skips on single-GPU boxes (the gloo world-2 tests cover the protocol
+there; HSA_ENABLE_IPC_MODE_LEGACY=0 must stay in the env for RCCL IPC)."""
 
 import os
 
@@ -9,12 +10,18 @@ import pytest
 import torch
 import torch.multiprocessing as mp
 
-pytestmark = pytest.mark.gpu
+pytestmark = [
+    pytest.mark.gpu,
+    pytest.mark.skipif(
+        not torch.cuda.is_available() or torch.cuda.device_count() < 2,
+        reason="needs >= 2 GPUs (RCCL forbids rank-sharing one device)",
+    ),
+]
 
 
 def _rank_main(rank, world, port, q):
     os.environ.update(
-        RANK=str(rank), WORLD_SIZE=str(world), LOCAL_RANK="0",
+        RANK=str(rank), WORLD_SIZE=str(world), LOCAL_RANK=str(rank),
         MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port),
     )
     try:
@@ -30,7 +37,7 @@ def _rank_main(rank, world, port, q):
             DiLoCoConfig(h=2, inner=InnerOptConfig(lr=1e-3, warmup_steps=0,
                                                    schedule="constant")),
             comm=comm,
-            device=torch.device("cuda", 0),
+            device=torch.device("cuda", rank),
         )
         start = float(w.fp.master.sum())
         data = SyntheticTokens(512, 128, 2, seed=21, rank=rank)
