@@ -17,6 +17,7 @@ void adamw8_step_(torch::Tensor master, torch::Tensor param, torch::Tensor grad,
                   torch::Tensor v_scale, double lr, double beta1, double beta2,
                   double eps, double wd, long step);
 torch::Tensor swiglu_fwd(torch::Tensor gate, torch::Tensor up);
+torch::Tensor grad_norm_sq(torch::Tensor x);
 std::vector<torch::Tensor> swiglu_bwd(torch::Tensor dout, torch::Tensor gate,
                                       torch::Tensor up);
 // rmsnorm.hip
@@ -55,6 +56,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("extract_delta", &extract_delta);
   m.def("adamw8_step_", &adamw8_step_);
   m.def("swiglu_fwd", &swiglu_fwd);
+  m.def("grad_norm_sq", &grad_norm_sq);
   m.def("swiglu_bwd", &swiglu_bwd);
   m.def("rmsnorm_fwd", &rmsnorm_fwd);
   m.def("rmsnorm_bwd", &rmsnorm_bwd);

@@ -120,6 +120,10 @@ class FlatParams:
 
     def grad_norm(self) -> torch.Tensor:
         # fp32 accumulation without materializing an fp32 copy of the grads
+        if self.flat_grad.is_cuda and self.flat_grad.dtype == torch.bfloat16:
+            from hypha_amd import _C
+
+            return _C.grad_norm_sq(self.flat_grad).sqrt_()
         return torch.linalg.vector_norm(self.flat_grad, dtype=torch.float32)
 
 

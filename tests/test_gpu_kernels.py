@@ -366,3 +366,10 @@ def test_fp8_convert_model_step():
     for _ in range(6):
         last = w.train_step(ids.clone(), labels.clone())
     assert math.isfinite(last) and last < first, (first, last)
+
+
+def test_grad_norm_sq():
+    x = rand_bf16(123457, seed=60)
+    got = _C.grad_norm_sq(x).sqrt()
+    want = torch.linalg.vector_norm(x, dtype=torch.float32)
+    torch.testing.assert_close(got, want, rtol=1e-3, atol=1e-3)
