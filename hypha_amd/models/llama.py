@@ -82,12 +82,9 @@ class Attention(nn.Module):
         super().__init__()
         self.cfg = cfg
         hd = cfg.head_dim
-        # fused QKV projection: one wide GEMM instead of one square + two
-        # tall-skinny GEMMs (the KV projections at Hkv=8 are N=1024 — poor
-        # hipBLASLt shapes on their own)
-        self.wqkv = nn.Linear(
-            cfg.hidden_size, (cfg.n_heads + 2 * cfg.n_kv_heads) * hd, bias=False
-        )
+        self.wq = nn.Linear(cfg.hidden_size, cfg.n_heads * hd, bias=False)
+        self.wk = nn.Linear(cfg.hidden_size, cfg.n_kv_heads * hd, bias=False)
+        self.wv = nn.Linear(cfg.hidden_size, cfg.n_kv_heads * hd, bias=False)
         self.wo = nn.Linear(cfg.n_heads * hd, cfg.hidden_size, bias=False)
 
     def forward(self, x, cos, sin, cache=None, pos: int = 0):
@@ -96,12 +93,9 @@ class Attention(nn.Module):
         hd = cfg.head_dim
         # BSHD layout throughout: the projections' natural layout, consumed
         # directly by the stride-aware RoPE/attention kernels (no transposes)
-        qkv = self.wqkv(x)
-        nq = cfg.n_heads * hd
-        nkv = cfg.n_kv_heads * hd
-        q = qkv[..., :nq].view(b, s, cfg.n_heads, hd)
-        k = qkv[..., nq : nq + nkv].view(b, s, cfg.n_kv_heads, hd)
-        v = qkv[..., nq + nkv :].view(b, s, cfg.n_kv_heads, hd)
+        q = self.wq(x).view(b, s, cfg.n_heads, hd)
+        k = self.wk(x).view(b, s, cfg.n_kv_heads, hd)
+        v = self.wv(x).view(b, s, cfg.n_kv_heads, hd)
         if cache is None:
             q, k = ops.apply_rope_qk(q, k, cos, sin, layout="bshd")
             o = ops.flash_attention(q, k, v, causal=True, layout="bshd")
@@ -131,14 +125,12 @@ class Attention(nn.Module):
 class MLP(nn.Module):
     def __init__(self, cfg: LlamaConfig):
         super().__init__()
-        # fused gate+up projection (one GEMM)
-        self.w_gateup = nn.Linear(cfg.hidden_size, 2 * cfg.ffn_hidden, bias=False)
+        self.w_gate = nn.Linear(cfg.hidden_size, cfg.ffn_hidden, bias=False)
+        self.w_up = nn.Linear(cfg.hidden_size, cfg.ffn_hidden, bias=False)
         self.w_down = nn.Linear(cfg.ffn_hidden, cfg.hidden_size, bias=False)
 
     def forward(self, x):
-        gu = self.w_gateup(x)
-        f = gu.shape[-1] // 2
-        return self.w_down(ops.swiglu(gu[..., :f], gu[..., f:]))
+        return self.w_down(ops.swiglu(self.w_gate(x), self.w_up(x)))
 
 
 class Block(nn.Module):

@@ -33,6 +33,7 @@ constexpr int QB = 32;    // q rows per wave
 constexpr int NWAVE = 4;  // waves per workgroup
 constexpr int WGQ = QB * NWAVE;
 constexpr int PADV = 8;   // V^T pitch pad (72 elems -> conflict-free b128)
+constexpr int KVT_COMB = 32;  // bwd combine-buffer kv rows
 
 __device__ __forceinline__ unsigned pack_bf16x2(float lo, float hi) {
   return (unsigned)(unsigned short)f2bf(lo) | ((unsigned)(unsigned short)f2bf(hi) << 16);
@@ -321,10 +322,14 @@ __global__ void cast_f32_to_bf16_kernel(const float* __restrict__ in,
   }
 }
 
-// Main backward: one workgroup = 4 waves sharing ONE 32-row kv tile; q-tiles
-// (and the GQA q-head group) are walked by all waves with a round-robin
-// split; dK/dV accumulate in per-wave MFMA accumulators and are combined
-// through LDS at the end; dQ accumulates via global f32 atomics (FA2-style).
+// Main backward: one workgroup owns TWO 32-row kv tiles of one (b, hkv);
+// q-tiles (and the GQA q-head group) are walked by all 4 waves round-robin.
+// Per staged q-tile the wave processes BOTH kv tiles (staging, lse/Di loads
+// and dQ atomics amortize over 2x the MFMA work; dQ accumulates across both
+// tiles in registers so the global f32 atomics halve). dK/dV accumulate in
+// per-wave, per-tile MFMA accumulators (AGPR-backed at 1 wave/SIMD) and are
+// combined through LDS at the end. P/dS are recomputed from q/k/lse
+// (FA2-style).
 template <int HD>
 __global__ __launch_bounds__(256, 1) void attn_bwd_kernel(
     const short* __restrict__ qg, const short* __restrict__ kg,
@@ -334,19 +339,23 @@ __global__ __launch_bounds__(256, 1) void attn_bwd_kernel(
     int B, int Hq, int Hkv, int S, float scale, bool causal,
     long long q_sb, long long q_sh, long long q_ss,
     long long kv_sb, long long kv_sh, long long kv_ss) {
-  constexpr int KVT = 32;            // kv rows per workgroup
-  constexpr int QT = 32;             // q rows per tile
-  constexpr int KC = HD / 16;        // chunks over head dim
-  constexpr int DBLK = HD / 32;      // d blocks
-  constexpr int TP = 40;             // transposed-image pitch (16B-aligned rows)
-  __shared__ __attribute__((aligned(16))) short k_img[KVT * HD];        // [kv][d], XOR-swizzled
-  __shared__ __attribute__((aligned(16))) short v_img[KVT * HD];        // [kv][d], XOR-swizzled
-  __shared__ __attribute__((aligned(16))) short kt_img[HD * TP];        // [d][kv]
-  __shared__ __attribute__((aligned(16))) short qt_img[NWAVE][HD * TP];   // per wave: [d][q]
-  __shared__ __attribute__((aligned(16))) short dot_img[NWAVE][HD * TP];  // per wave: [d][q]
-  __shared__ __attribute__((aligned(16))) short ptds_img[NWAVE][KVT * TP];  // per wave: ptT then ds reuse
-  __shared__ __attribute__((aligned(16))) short dst_img[NWAVE][KVT * TP];   // per wave: dsT
-  __shared__ __attribute__((aligned(16))) float comb[KVT * HD];         // cross-wave f32 combine
+  constexpr int KVT = 32;        // kv rows per tile
+  constexpr int NT = 2;          // kv tiles per workgroup
+  constexpr int QT = 32;         // q rows per tile
+  constexpr int KC = HD / 16;    // chunks over head dim
+  constexpr int DBLK = HD / 32;  // d blocks
+  constexpr int TP = 40;         // ptds/dst pitch (16B-aligned rows)
+  constexpr int DSP = NT * KVT + 8;  // ds image pitch (72: 16B-aligned)
+  constexpr int KTP = NT * KVT + 8;  // kt image pitch
+  // q-image: [HD][32] bf16 with an 8-elem group swizzle (conflict-free b128)
+  __shared__ __attribute__((aligned(16))) short k_img[NT * KVT * HD];   // [kv][d] swizzled
+  __shared__ __attribute__((aligned(16))) short v_img[NT * KVT * HD];   // [kv][d] swizzled
+  __shared__ __attribute__((aligned(16))) short kt_img[HD * KTP];       // [d][kv]
+  __shared__ __attribute__((aligned(16))) short qt_img[NWAVE][HD * 32];   // [d][q] swizzled
+  __shared__ __attribute__((aligned(16))) short dot_img[NWAVE][HD * 32];  // [d][q] swizzled
+  __shared__ __attribute__((aligned(16))) short ptds_img[NWAVE][KVT * TP];  // ptT per tile
+  __shared__ __attribute__((aligned(16))) short dst_img[NWAVE][KVT * TP];   // dsT per tile
+  __shared__ __attribute__((aligned(16))) short ds_img[NWAVE][QT * DSP];    // dS [q][kv 0..63]
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -358,39 +367,45 @@ __global__ __launch_bounds__(256, 1) void attn_bwd_kernel(
   const int b = bhkv / Hkv;
   const int hkv = bhkv % Hkv;
   const int G = Hq / Hkv;
-  const int kv0 = blockIdx.x * KVT;
+  const int kv_base = blockIdx.x * (NT * KVT);
   const long long kvbase = (long long)b * kv_sb + (long long)hkv * kv_sh;
 
-  // ---- stage K, V (swizzled row-major) and K^T ----
+  // swizzled q-image offset: 8-elem groups XORed by (d>>2)&3
+  auto qimg_off = [](int d, int qe) {
+    return d * 32 + ((((qe >> 3) ^ ((d >> 2) & 3)) << 3) | (qe & 7));
+  };
+
+  // ---- stage K, V (swizzled row-major, both tiles) and K^T ----
   {
-    constexpr int CH = KVT * HD / 8;
+    constexpr int CH = NT * KVT * HD / 8;
     for (int c = tid; c < CH; c += 256) {
       int row = c / (HD / 8), e0 = (c % (HD / 8)) * 8;
-      s16x8 k8 = *reinterpret_cast<const s16x8*>(kg + kvbase + (long long)(kv0 + row) * kv_ss + e0);
+      s16x8 k8 = *reinterpret_cast<const s16x8*>(kg + kvbase + (long long)(kv_base + row) * kv_ss + e0);
       *reinterpret_cast<s16x8*>((char*)k_img + k_lds_off<HD>(row, e0)) = k8;
-      s16x8 v8 = *reinterpret_cast<const s16x8*>(vg + kvbase + (long long)(kv0 + row) * kv_ss + e0);
+      s16x8 v8 = *reinterpret_cast<const s16x8*>(vg + kvbase + (long long)(kv_base + row) * kv_ss + e0);
       *reinterpret_cast<s16x8*>((char*)v_img + k_lds_off<HD>(row, e0)) = v8;
     }
     // K^T: lane-per-kv transpose staging (conflict-free scalar stores)
-    for (int i = 0; i < HD / 64; ++i) {
-      int kvr = tid & 31;
-      int e0 = (i * 8 + (tid >> 5)) * 8;
-      s16x8 k8 = *reinterpret_cast<const s16x8*>(kg + kvbase + (long long)(kv0 + kvr) * kv_ss + e0);
+    for (int i = 0; i < HD / 32; ++i) {
+      int kvr = tid & 63;
+      int e0 = (i * 4 + (tid >> 6)) * 8;
+      s16x8 k8 = *reinterpret_cast<const s16x8*>(kg + kvbase + (long long)(kv_base + kvr) * kv_ss + e0);
 #pragma unroll
-      for (int j = 0; j < 8; ++j) kt_img[(e0 + j) * TP + kvr] = k8[j];
+      for (int j = 0; j < 8; ++j) kt_img[(e0 + j) * KTP + kvr] = k8[j];
     }
   }
   __syncthreads();
 
-  f32x16 dk_acc[DBLK] = {};  // D[m=kv][n=d]
-  f32x16 dv_acc[DBLK] = {};  // D[m=d][n=kv]
+  f32x16 dk_acc[NT][DBLK] = {};  // D[m=kv][n=d] per tile
+  f32x16 dv_acc[NT][DBLK] = {};  // D[m=d][n=kv] per tile
 
   short* qt = qt_img[wid];
   short* dot = dot_img[wid];
   short* ptds = ptds_img[wid];
   short* dst = dst_img[wid];
+  short* dsw = ds_img[wid];
 
-  const int t0 = causal ? kv0 / QT : 0;
+  const int t0 = causal ? kv_base / QT : 0;
   const int Tq = S / QT;
 
   for (int hq = hkv * G; hq < (hkv + 1) * G; ++hq) {
@@ -400,9 +415,7 @@ __global__ __launch_bounds__(256, 1) void attn_bwd_kernel(
       const int q0 = t * QT;
       const int qrow = q0 + ln;  // this lane's q (for B-operand frags)
 
-      // ---- per-wave staging: Q^T and dO^T images. Lane-per-q: each scalar
-      // store's lanes write one contiguous q-row span of the image (the
-      // d-major order was 16-way bank-conflicted: 8*TP*2 = 0 mod 128 B).
+      // ---- per-wave staging: swizzled Q^T / dO^T images (lane-per-q) ----
 #pragma unroll
       for (int it = 0; it < HD / 16; ++it) {
         int d0 = 8 * hi + 16 * it;
@@ -410,95 +423,100 @@ __global__ __launch_bounds__(256, 1) void attn_bwd_kernel(
         s16x8 d8 = *reinterpret_cast<const s16x8*>(dog + qbase + (long long)(q0 + ln) * q_ss + d0);
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
-          qt[(d0 + j) * TP + ln] = q8[j];
-          dot[(d0 + j) * TP + ln] = d8[j];
+          qt[qimg_off(d0 + j, ln)] = q8[j];
+          dot[qimg_off(d0 + j, ln)] = d8[j];
         }
       }
-
-      // ---- S^T = K Q^T (Q frags straight from global) ----
-      f32x16 st = {};
-#pragma unroll
-      for (int kc = 0; kc < KC; ++kc) {
-        s16x8 qf = *reinterpret_cast<const s16x8*>(qg + qbase + (long long)qrow * q_ss +
-                                                   16 * kc + 8 * hi);
-        s16x8 kf = *reinterpret_cast<const s16x8*>(
-            (char*)k_img + k_lds_off<HD>(ln, 16 * kc + 8 * hi));
-        st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qf, st, 0, 0, 0);
-      }
-
-      // ---- P^T = exp(scale*S^T - lse[q]) with causal mask ----
       const float lse = lseg[lsebase + qrow];
       const float di = dig[lsebase + qrow];
-      f32x16 pt;
-#pragma unroll
-      for (int r = 0; r < 16; ++r) {
-        int kv = kv0 + (r & 3) + 8 * (r >> 2) + 4 * hi;
-        float p = __expf(st[r] * scale - lse);
-        if (causal && kv > qrow) p = 0.f;
-        pt[r] = p;
-      }
-      // write ptT image [kv][q]
-#pragma unroll
-      for (int r = 0; r < 16; ++r) {
-        int kv = (r & 3) + 8 * (r >> 2) + 4 * hi;
-        ptds[kv * TP + ln] = f2bf(pt[r]);
-      }
 
-      // ---- dV^T += dO^T P  (A = dO^T image, B = ptT image) ----
 #pragma unroll
-      for (int db = 0; db < DBLK; ++db) {
+      for (int tile = 0; tile < NT; ++tile) {
+        const int kv0 = kv_base + tile * KVT;
+        if (causal && kv0 > q0 + QT - 1) continue;  // tile above the diagonal
+
+        // ---- S^T = K Q^T (Q frags straight from global) ----
+        f32x16 st = {};
 #pragma unroll
-        for (int kcq = 0; kcq < QT / 16; ++kcq) {
-          s16x8 af = *reinterpret_cast<const s16x8*>(dot + (32 * db + ln) * TP +
-                                                     16 * kcq + 8 * hi);
-          s16x8 bf = *reinterpret_cast<const s16x8*>(ptds + ln * TP + 16 * kcq + 8 * hi);
-          dv_acc[db] =
-              __builtin_amdgcn_mfma_f32_32x32x16_bf16(af, bf, dv_acc[db], 0, 0, 0);
+        for (int kc = 0; kc < KC; ++kc) {
+          s16x8 qf = *reinterpret_cast<const s16x8*>(qg + qbase + (long long)qrow * q_ss +
+                                                     16 * kc + 8 * hi);
+          s16x8 kf = *reinterpret_cast<const s16x8*>(
+              (char*)k_img + k_lds_off<HD>(tile * KVT + ln, 16 * kc + 8 * hi));
+          st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qf, st, 0, 0, 0);
+        }
+
+        // ---- P^T = exp(scale*S^T - lse) with causal mask; write ptT ----
+        f32x16 pt;
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          int kv = kv0 + (r & 3) + 8 * (r >> 2) + 4 * hi;
+          float pv = __expf(st[r] * scale - lse);
+          if (causal && kv > qrow) pv = 0.f;
+          pt[r] = pv;
+        }
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          int kv = (r & 3) + 8 * (r >> 2) + 4 * hi;
+          ptds[kv * TP + ln] = f2bf(pt[r]);
+        }
+
+        // ---- dV^T += dO^T P  (A = dO^T image, B = ptT image) ----
+#pragma unroll
+        for (int db = 0; db < DBLK; ++db) {
+#pragma unroll
+          for (int kcq = 0; kcq < QT / 16; ++kcq) {
+            s16x8 af = *reinterpret_cast<const s16x8*>(
+                dot + qimg_off(32 * db + ln, 16 * kcq + 8 * hi));
+            s16x8 bf = *reinterpret_cast<const s16x8*>(ptds + ln * TP + 16 * kcq + 8 * hi);
+            dv_acc[tile][db] =
+                __builtin_amdgcn_mfma_f32_32x32x16_bf16(af, bf, dv_acc[tile][db], 0, 0, 0);
+          }
+        }
+
+        // ---- dP^T = V dO^T (A = V image, B = dO frags from global) ----
+        f32x16 dpt = {};
+#pragma unroll
+        for (int kc = 0; kc < KC; ++kc) {
+          s16x8 df = *reinterpret_cast<const s16x8*>(dog + qbase + (long long)qrow * q_ss +
+                                                     16 * kc + 8 * hi);
+          s16x8 vf = *reinterpret_cast<const s16x8*>(
+              (char*)v_img + k_lds_off<HD>(tile * KVT + ln, 16 * kc + 8 * hi));
+          dpt = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vf, df, dpt, 0, 0, 0);
+        }
+
+        // ---- dS^T = P^T * (dP^T - Di) * scale; write dsT + ds columns ----
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          int kv = (r & 3) + 8 * (r >> 2) + 4 * hi;
+          float v = pt[r] * (dpt[r] - di) * scale;
+          short vb = f2bf(v);
+          dst[kv * TP + ln] = vb;
+          dsw[ln * DSP + tile * KVT + kv] = vb;
+        }
+
+        // ---- dK += dS^T Q (A = dsT image, B = swizzled Q^T image) ----
+#pragma unroll
+        for (int db = 0; db < DBLK; ++db) {
+#pragma unroll
+          for (int kcq = 0; kcq < QT / 16; ++kcq) {
+            s16x8 af = *reinterpret_cast<const s16x8*>(dst + ln * TP + 16 * kcq + 8 * hi);
+            s16x8 bf = *reinterpret_cast<const s16x8*>(
+                qt + qimg_off(32 * db + ln, 16 * kcq + 8 * hi));
+            dk_acc[tile][db] =
+                __builtin_amdgcn_mfma_f32_32x32x16_bf16(af, bf, dk_acc[tile][db], 0, 0, 0);
+          }
         }
       }
 
-      // ---- dP^T = V dO^T (A = V image, B = dO frags from global) ----
-      f32x16 dpt = {};
-#pragma unroll
-      for (int kc = 0; kc < KC; ++kc) {
-        s16x8 df = *reinterpret_cast<const s16x8*>(dog + qbase + (long long)qrow * q_ss +
-                                                   16 * kc + 8 * hi);
-        s16x8 vf = *reinterpret_cast<const s16x8*>(
-            (char*)v_img + k_lds_off<HD>(ln, 16 * kc + 8 * hi));
-        dpt = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vf, df, dpt, 0, 0, 0);
-      }
-
-      // ---- dS^T = P^T * (dP^T - Di) * scale; write dsT [kv][q] + ds [q][kv]
-#pragma unroll
-      for (int r = 0; r < 16; ++r) {
-        int kv = (r & 3) + 8 * (r >> 2) + 4 * hi;
-        float v = pt[r] * (dpt[r] - di) * scale;
-        short vb = f2bf(v);
-        dst[kv * TP + ln] = vb;
-        ptds[ln * TP + kv] = vb;  // ds image reuses the ptT buffer (pt consumed)
-      }
-
-      // ---- dK += dS^T Q (A = dsT image, B = Q^T image) ----
-#pragma unroll
-      for (int db = 0; db < DBLK; ++db) {
-#pragma unroll
-        for (int kcq = 0; kcq < QT / 16; ++kcq) {
-          s16x8 af = *reinterpret_cast<const s16x8*>(dst + ln * TP + 16 * kcq + 8 * hi);
-          s16x8 bf = *reinterpret_cast<const s16x8*>(qt + (32 * db + ln) * TP +
-                                                     16 * kcq + 8 * hi);
-          dk_acc[db] =
-              __builtin_amdgcn_mfma_f32_32x32x16_bf16(af, bf, dk_acc[db], 0, 0, 0);
-        }
-      }
-
-      // ---- dQ(tile) = dS K (A = ds image, B = K^T image); global atomics --
+      // ---- dQ(tile-pair) = dS K over kv 0..63; one atomic pass ----
+      const int kck_hi = (causal && kv_base + KVT > q0 + QT - 1) ? NT : 2 * NT;
 #pragma unroll
       for (int db = 0; db < DBLK; ++db) {
         f32x16 dq = {};
-#pragma unroll
-        for (int kck = 0; kck < KVT / 16; ++kck) {
-          s16x8 af = *reinterpret_cast<const s16x8*>(ptds + ln * TP + 16 * kck + 8 * hi);
-          s16x8 bf = *reinterpret_cast<const s16x8*>(kt_img + (32 * db + ln) * TP +
+        for (int kck = 0; kck < kck_hi && kck < 2 * NT; ++kck) {
+          s16x8 af = *reinterpret_cast<const s16x8*>(dsw + ln * DSP + 16 * kck + 8 * hi);
+          s16x8 bf = *reinterpret_cast<const s16x8*>(kt_img + (32 * db + ln) * KTP +
                                                      16 * kck + 8 * hi);
           dq = __builtin_amdgcn_mfma_f32_32x32x16_bf16(af, bf, dq, 0, 0, 0);
         }
@@ -511,9 +529,14 @@ __global__ __launch_bounds__(256, 1) void attn_bwd_kernel(
     }
   }
 
-  // ---- combine dK across waves (phased, no atomics), store bf16 ----
-  auto combine_store = [&](const f32x16* acc, short* outg, bool transposed) {
-    for (int c = tid; c < KVT * HD; c += 256) comb[c] = 0.f;
+  // ---- combine dK/dV across waves (phased, no atomics), store bf16 ----
+  // the combine buffer reuses the (now dead) qt image area: 16 KB f32 within
+  // the 64 KB qt arena; every use is bracketed by __syncthreads()
+  float* comb2 = reinterpret_cast<float*>(&qt_img[0][0]);
+  __syncthreads();
+  auto combine_store = [&](int tile, const f32x16* acc, short* outg, bool transposed) {
+    constexpr int KVT = 32;
+    for (int c = tid; c < KVT * HD; c += 256) comb2[c] = 0.f;
     __syncthreads();
     for (int w = 0; w < NWAVE; ++w) {
       if (w == wid) {
@@ -524,22 +547,26 @@ __global__ __launch_bounds__(256, 1) void attn_bwd_kernel(
             int m = (r & 3) + 8 * (r >> 2) + 4 * hi;
             int kv = transposed ? ln : m;
             int d = transposed ? (32 * db + m) : (32 * db + ln);
-            comb[kv * HD + d] += acc[db][r];
+            comb2[kv * HD + d] += acc[db][r];
           }
       }
       __syncthreads();
     }
+    const int kv0 = kv_base + tile * KVT;
     for (int c = tid; c < KVT * HD / 8; c += 256) {
       int row = c / (HD / 8), e0 = (c % (HD / 8)) * 8;
       s16x8 o8;
 #pragma unroll
-      for (int j = 0; j < 8; ++j) o8[j] = f2bf(comb[row * HD + e0 + j]);
+      for (int j = 0; j < 8; ++j) o8[j] = f2bf(comb2[row * HD + e0 + j]);
       *reinterpret_cast<s16x8*>(outg + kvbase + (long long)(kv0 + row) * kv_ss + e0) = o8;
     }
     __syncthreads();
   };
-  combine_store(dk_acc, dkg, false);
-  combine_store(dv_acc, dvg, true);
+#pragma unroll
+  for (int tile = 0; tile < NT; ++tile) {
+    combine_store(tile, dk_acc[tile], dkg, false);
+    combine_store(tile, dv_acc[tile], dvg, true);
+  }
 }
 
 }  // namespace
@@ -554,7 +581,7 @@ std::vector<torch::Tensor> attn_bwd_ex(torch::Tensor q, torch::Tensor k, torch::
   const int S = bshd ? q.size(1) : q.size(2);
   const int HD = q.size(3);
   const int Hkv = bshd ? k.size(2) : k.size(1);
-  TORCH_CHECK(S % 32 == 0 && (HD == 64 || HD == 128));
+  TORCH_CHECK(S % 64 == 0 && (HD == 64 || HD == 128));
   auto dq32 = torch::zeros(q.sizes(), q.options().dtype(torch::kFloat32));
   auto dk = torch::empty_like(k);
   auto dv = torch::empty_like(v);
@@ -577,7 +604,7 @@ std::vector<torch::Tensor> attn_bwd_ex(torch::Tensor q, torch::Tensor k, torch::
                        dim3(256), 0, stream, (const short*)dout.data_ptr(),             \
                        (const short*)o.data_ptr(), di.data_ptr<float>(), nrows, Hq, S,  \
                        q_sb, q_sh, q_ss);                                               \
-    hipLaunchKernelGGL(attn_bwd_kernel<HDV>, dim3(S / 32, B * Hkv), dim3(256), 0,       \
+    hipLaunchKernelGGL(attn_bwd_kernel<HDV>, dim3(S / 64, B * Hkv), dim3(256), 0,       \
                        stream, (const short*)q.data_ptr(), (const short*)k.data_ptr(),  \
                        (const short*)v.data_ptr(), (const short*)dout.data_ptr(),       \
                        lse.data_ptr<float>(), di.data_ptr<float>(),                     \
