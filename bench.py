@@ -62,8 +62,15 @@ def main() -> int:
     if on_gpu:
         ops.native_available_or_raise()
 
-    torch.manual_seed(1234)  # same init on every rank pre-broadcast
-    model = models.build(args.model)
+    torch.manual_seed(1234)
+    # build directly on the GPU: 8 ranks x 32 GB of fp32 CPU-side init would
+    # strain host RAM and add ~a minute per rank; rank-0's broadcast makes
+    # every rank's weights identical regardless of init device RNG
+    import contextlib
+
+    build_ctx = torch.device(device) if on_gpu else contextlib.nullcontext()
+    with build_ctx:
+        model = models.build(args.model)
     if args.fp8 and on_gpu:
         from hypha_amd.ops.fp8 import convert_linears_to_fp8
 

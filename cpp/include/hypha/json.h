@@ -215,14 +215,34 @@ class Json {
             if (pos + 4 > s.size()) throw std::runtime_error("json: bad \\u");
             unsigned code = std::stoul(s.substr(pos, 4), nullptr, 16);
             pos += 4;
-            // encode as UTF-8 (BMP only; surrogate pairs unsupported)
+            // UTF-16 surrogate pair -> full code point
+            if (code >= 0xD800 && code <= 0xDBFF) {
+              if (pos + 6 <= s.size() && s[pos] == '\\' && s[pos + 1] == 'u') {
+                unsigned low = std::stoul(s.substr(pos + 2, 4), nullptr, 16);
+                if (low >= 0xDC00 && low <= 0xDFFF) {
+                  code = 0x10000 + ((code - 0xD800) << 10) + (low - 0xDC00);
+                  pos += 6;
+                } else {
+                  code = 0xFFFD;  // lone high surrogate
+                }
+              } else {
+                code = 0xFFFD;
+              }
+            } else if (code >= 0xDC00 && code <= 0xDFFF) {
+              code = 0xFFFD;  // lone low surrogate
+            }
             if (code < 0x80) {
               out += (char)code;
             } else if (code < 0x800) {
               out += (char)(0xC0 | (code >> 6));
               out += (char)(0x80 | (code & 0x3F));
-            } else {
+            } else if (code < 0x10000) {
               out += (char)(0xE0 | (code >> 12));
+              out += (char)(0x80 | ((code >> 6) & 0x3F));
+              out += (char)(0x80 | (code & 0x3F));
+            } else {
+              out += (char)(0xF0 | (code >> 18));
+              out += (char)(0x80 | ((code >> 12) & 0x3F));
               out += (char)(0x80 | ((code >> 6) & 0x3F));
               out += (char)(0x80 | (code & 0x3F));
             }

@@ -277,3 +277,44 @@ def test_ps_average_bf16_and_f32(tmp_path):
     u = load_file(str(tmp_path / "u.safetensors"))["t"]
     want = ((a.float() + b.float()) / 2).bfloat16()
     torch.testing.assert_close(u.float(), want.float(), rtol=2e-2, atol=2e-2)
+
+
+def test_json_parser_fuzz_roundtrip():
+    """Property fuzz: the C++ JSON wire codec must round-trip anything the
+    Python side can produce (the control plane crosses this boundary on
+    every message)."""
+    import json as pyjson
+
+    from hypothesis import given, settings, strategies as st
+
+    scalars = st.one_of(
+        st.none(), st.booleans(),
+        st.integers(min_value=-(2**53) + 1, max_value=2**53 - 1),
+        st.floats(allow_nan=False, allow_infinity=False, width=32),
+        st.text(max_size=40),
+    )
+    values = st.recursive(
+        scalars,
+        lambda children: st.one_of(
+            st.lists(children, max_size=4),
+            st.dictionaries(st.text(max_size=8), children, max_size=4),
+        ),
+        max_leaves=20,
+    )
+
+    @settings(max_examples=200, deadline=None)
+    @given(values)
+    def check(v):
+        out = pyjson.loads(core.json_roundtrip(pyjson.dumps(v)))
+        if isinstance(v, float):
+            assert out == pytest.approx(v, rel=1e-6, abs=1e-9)
+        else:
+            assert out == v
+
+    check()
+
+
+def test_json_parser_rejects_garbage():
+    for bad in ("", "{", "[1,", '{"a"}', "tru", "nul", '"\\u12', "1e999x", "{}{}"):
+        with pytest.raises(Exception):
+            core.json_roundtrip(bad)
