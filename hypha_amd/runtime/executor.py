@@ -56,6 +56,17 @@ def main() -> int:
     seq_len = int(cfg.get("seq_len", 128))
     data_ref = cfg["data"]
     updates_ref = cfg["updates"]
+    ckpt_dir = cfg.get("checkpoint_dir") or os.path.join(args.work_dir, "checkpoint")
+    ckpt_every = int(cfg.get("checkpoint_every_rounds", 0))
+
+    # resume from a previous run's checkpoint if one exists
+    resume_round = 0
+    if os.path.exists(os.path.join(ckpt_dir, "manifest.json")):
+        from hypha_amd import checkpoint as ckpt_mod
+
+        manifest = ckpt_mod.load_checkpoint(worker, ckpt_dir)
+        resume_round = int(manifest["round"])
+        print(f"[executor] resumed from round {resume_round}", flush=True)
 
     # background SSE listener for aggregated updates
     updates_q: "queue.Queue[dict]" = queue.Queue()
@@ -95,7 +106,7 @@ def main() -> int:
         print("[executor] joined at current global weights", flush=True)
 
     done = False
-    round_idx = 0
+    round_idx = resume_round
     while not done:
         # ---- inner loop: train until the scheduler's counter is exhausted ----
         remaining = None
@@ -128,6 +139,12 @@ def main() -> int:
         resp = session.send_status({"kind": "update-received"})
         done = resp.get("kind") == "done"
         round_idx += 1
+        worker.round = round_idx
+        if ckpt_every and (round_idx % ckpt_every == 0 or done):
+            from hypha_amd import checkpoint as ckpt_mod
+
+            ckpt_mod.save_checkpoint(worker, ckpt_dir)
+            print(f"[executor] checkpoint saved at round {round_idx}", flush=True)
         print(f"[executor] round {round_idx} merged, loss={loss:.4f} done={done}",
               flush=True)
 

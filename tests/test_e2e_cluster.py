@@ -179,3 +179,69 @@ def test_cluster_worker_kill_and_rejoin(binaries, tmp_path):
                 p.wait(timeout=10)
             except Exception:
                 pass
+
+
+@pytest.mark.timeout(300)
+def test_cluster_checkpointing(binaries, tmp_path):
+    """Train jobs write periodic checkpoints when the job config asks for
+    them (checkpoint_every_rounds), in the SafeTensors+manifest format."""
+    import json as pyjson
+
+    from hypha_amd.data.synthetic import write_slice_files
+
+    data_dir = tmp_path / "slices"
+    write_slice_files(str(data_dir), "synth", 4, 16, 512, 128)
+    gw_port = free_port()
+    env = dict(os.environ, PYTHONPATH=str(REPO))
+    procs = []
+
+    def spawn(name, cmd):
+        log = open(tmp_path / f"{name}.log", "w")
+        p = subprocess.Popen(cmd, cwd=REPO, env=env, stdout=log, stderr=log)
+        procs.append(p)
+        return p
+
+    try:
+        spawn("gateway", [str(BIN / "hypha-gateway"), "--port", str(gw_port)])
+        time.sleep(0.3)
+        spawn("data", [str(BIN / "hypha-data"), "--name", "data-node",
+                       "--gateway-host", "127.0.0.1", "--gateway-port", str(gw_port),
+                       "--dataset", "synth", "--dataset-path", str(data_dir)])
+        exec_cmd = (f"{sys.executable} -m hypha_amd.runtime.executor "
+                    "--socket {SOCKET_PATH} --work-dir {WORK_DIR} --job {JOB_JSON}")
+        for i in range(2):
+            spawn(f"worker{i}", [str(BIN / "hypha-worker"), "--name", f"worker-{i}",
+                                 "--gateway-host", "127.0.0.1", "--gateway-port", str(gw_port),
+                                 "--exec-cmd", exec_cmd,
+                                 "--work-root", str(tmp_path / f"work{i}")])
+        time.sleep(0.5)
+        ckpt_root = tmp_path / "ckpt"
+        cfg = tmp_path / "job.json"
+        cfg.write_text(pyjson.dumps({
+            "model": "llama-tiny", "dataset": "synth", "num_workers": 1,
+            "update_rounds": 2, "avg_samples_between_updates": 8,
+            "batch_size": 2, "seq_len": 128, "inner_lr": 0.001,
+            "checkpoint_every_rounds": 1, "checkpoint_dir": str(ckpt_root),
+        }))
+        sched = subprocess.Popen(
+            [str(BIN / "hypha-scheduler"), "--name", "scheduler",
+             "--gateway-host", "127.0.0.1", "--gateway-port", str(gw_port),
+             "--config", str(cfg)],
+            cwd=REPO, env=env, stdout=subprocess.PIPE,
+            stderr=open(tmp_path / "sched.log", "w"), text=True)
+        procs.append(sched)
+        out, _ = sched.communicate(timeout=200)
+        assert "Job is completed." in out, (tmp_path / "sched.log").read_text()[-2000:]
+        manifest = pyjson.loads((ckpt_root / "manifest.json").read_text())
+        assert manifest["round"] == 2
+        assert (ckpt_root / "0_global_weights.safetensors").exists()
+        assert (ckpt_root / "optimizer_state.safetensors").exists()
+    finally:
+        for p in procs:
+            if p.poll() is None:
+                p.send_signal(signal.SIGKILL)
+        for p in procs:
+            try:
+                p.wait(timeout=10)
+            except Exception:
+                pass
