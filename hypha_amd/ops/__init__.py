@@ -54,6 +54,7 @@ def use_native(*tensors: torch.Tensor) -> bool:
 from .interface import (  # noqa: E402
     apply_rope_qk,
     cross_entropy_loss,
+    extract_delta,
     flash_attention,
     fused_adamw,
     fused_nesterov,
@@ -73,4 +74,5 @@ __all__ = [
     "cross_entropy_loss",
     "fused_adamw",
     "fused_nesterov",
+    "extract_delta",
 ]

@@ -4,16 +4,17 @@ PyTorch reference path (CPU).
 The HIP extension ABI (hypha_amd._C):
   rmsnorm_fwd(x2d, w, eps) -> (y, rstd)
   rmsnorm_bwd(dy, x2d, w, rstd) -> (dx, dw)
-  rope_fwd(q, k, cos, sin, inverse) -> (q_out, k_out)      # [B,H,S,D]
-  swiglu_fwd(gate, up) -> out
-  swiglu_bwd(dout, gate, up) -> (dgate, dup)
+  rope_fwd_ex(q, k, cos, sin, inverse, layout) -> (q_out, k_out)
+  swiglu_fwd(gate, up) -> out ; swiglu_bwd(dout, gate, up) -> (dgate, dup)
   ce_fwd(logits2d, targets) -> (loss_sum, lse, n_valid)    # fp32 scalars/rows
   ce_bwd_(logits2d, targets, lse, scale) -> dlogits        # overwrites logits
-  attn_fwd(q, k, v, causal) -> (o, lse)
-  attn_bwd(q, k, v, o, do, lse, causal) -> (dq, dk, dv)
+  attn_fwd_ex(q, k, v, causal, layout) -> (o, lse)         # layout bhsd|bshd
+  attn_bwd_ex(q, k, v, o, do, lse, causal, layout) -> (dq, dk, dv)
   adamw_step_(master, param, grad, m, v, lr, b1, b2, eps, wd, step)
-  nesterov_step_(master, delta, momentum, lr, mu)
+  adamw8_step_(master, param, grad, m8, v8, m_scale, v_scale, ...)  # 8-bit state
+  nesterov_step_(master, delta_bf16, momentum, lr, mu)
   extract_delta(master, theta0, out_bf16)
+  grad_norm_sq(flat_bf16) -> fp32 scalar
 """
 
 from __future__ import annotations
