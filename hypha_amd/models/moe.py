@@ -89,19 +89,35 @@ class MoEMLP(nn.Module):
         order = torch.argsort(flat_expert, stable=True)
         counts = torch.bincount(flat_expert, minlength=cfg.n_experts).tolist()
         gathered = xf[flat_tok[order]]  # [T*K, h] grouped by expert
-        out_groups = []
-        start = 0
-        for e in range(cfg.n_experts):
-            n = counts[e]
-            if n == 0:
-                continue
-            xe = gathered[start : start + n]
-            ge = xe @ self.w_gate[e].t()
-            ue = xe @ self.w_up[e].t()
+        use_native_grouped = (
+            not torch.is_grad_enabled() and gathered.is_cuda
+            and gathered.dtype == torch.bfloat16 and ops.has_native()
+            and cfg.ffn_hidden % 128 == 0 and cfg.hidden_size % 128 == 0
+        )
+        if use_native_grouped:
+            # inference path: one native grouped-GEMM kernel per projection
+            from hypha_amd import _C
+
+            off = torch.zeros(cfg.n_experts + 1, dtype=torch.int32)
+            off[1:] = torch.tensor(counts, dtype=torch.int32).cumsum(0)
+            ge = _C.grouped_gemm(gathered.contiguous(), self.w_gate.contiguous(), off)
+            ue = _C.grouped_gemm(gathered.contiguous(), self.w_up.contiguous(), off)
             he = ops.swiglu(ge, ue)
-            out_groups.append(he @ self.w_down[e].t())
-            start += n
-        grouped_out = torch.cat(out_groups, dim=0) if out_groups else gathered
+            grouped_out = _C.grouped_gemm(he.contiguous(), self.w_down.contiguous(), off)
+        else:
+            out_groups = []
+            start = 0
+            for e in range(cfg.n_experts):
+                n = counts[e]
+                if n == 0:
+                    continue
+                xe = gathered[start : start + n]
+                ge = xe @ self.w_gate[e].t()
+                ue = xe @ self.w_up[e].t()
+                he = ops.swiglu(ge, ue)
+                out_groups.append(he @ self.w_down[e].t())
+                start += n
+            grouped_out = torch.cat(out_groups, dim=0) if out_groups else gathered
         # scatter-add back with routing weights
         weights = topv.reshape(-1)[order].unsqueeze(1)
         out = torch.zeros_like(xf)

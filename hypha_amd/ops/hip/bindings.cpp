@@ -18,6 +18,8 @@ void adamw8_step_(torch::Tensor master, torch::Tensor param, torch::Tensor grad,
                   double eps, double wd, long step);
 torch::Tensor swiglu_fwd(torch::Tensor gate, torch::Tensor up);
 torch::Tensor grad_norm_sq(torch::Tensor x);
+// grouped_gemm.hip
+torch::Tensor grouped_gemm(torch::Tensor x, torch::Tensor w, torch::Tensor group_offsets);
 std::vector<torch::Tensor> swiglu_bwd(torch::Tensor dout, torch::Tensor gate,
                                       torch::Tensor up);
 // rmsnorm.hip
@@ -57,6 +59,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("adamw8_step_", &adamw8_step_);
   m.def("swiglu_fwd", &swiglu_fwd);
   m.def("grad_norm_sq", &grad_norm_sq);
+  m.def("grouped_gemm", &grouped_gemm);
   m.def("swiglu_bwd", &swiglu_bwd);
   m.def("rmsnorm_fwd", &rmsnorm_fwd);
   m.def("rmsnorm_bwd", &rmsnorm_bwd);
