@@ -108,3 +108,23 @@ def test_certutil_pki_roundtrip(tmp_path):
          "-untrusted", str(out / "org1.crt"), str(out / "worker-0.crt")],
         capture_output=True, text=True)
     assert v.returncode == 0, v.stderr
+
+
+def test_aim_driver_endpoint(tmp_path, monkeypatch):
+    """drivers/aim_driver.py must accept the scheduler's metric posts
+    (metrics_bridge.rs -> aim-driver/main.py parity)."""
+    import importlib
+    import json as pyjson
+
+    monkeypatch.setenv("HYPHA_AIM_LOG", str(tmp_path / "run.jsonl"))
+    import drivers.aim_driver as ad
+
+    importlib.reload(ad)
+    from fastapi.testclient import TestClient
+
+    client = TestClient(ad.app)
+    r = client.post("/status", json={"worker_id": "w0", "round": 3,
+                                     "metric_name": "loss", "value": 1.25})
+    assert r.status_code == 200 and r.json() == {"ok": True}
+    rec = pyjson.loads((tmp_path / "run.jsonl").read_text().splitlines()[0])
+    assert rec["worker_id"] == "w0" and rec["value"] == 1.25
