@@ -1,0 +1,83 @@
+// hypha-data: dataset node. Scans a directory of SafeTensors slice files
+// (one file = one slice), announces {dataset: num_slices} in the registry,
+// and serves pull streams by slice index.
+// Parity with /root/reference/crates/data/src/bin/hypha-data.rs:150-209 and
+// tensor_data.rs:8-16 (files served verbatim).
+
+#include <dirent.h>
+#include <signal.h>
+#include <sys/stat.h>
+
+#include <algorithm>
+#include <string>
+#include <thread>
+#include <vector>
+
+#include "hypha/json.h"
+#include "hypha/net.h"
+
+using namespace hypha;
+
+int main(int argc, char** argv) {
+  std::string name = "data", gw_host = "127.0.0.1", dataset = "dataset", dir = ".";
+  int gw_port = 0, port = 0;
+  for (int i = 1; i < argc; ++i) {
+    std::string a = argv[i];
+    auto next = [&] { return std::string(argv[++i]); };
+    if (a == "--name") name = next();
+    else if (a == "--gateway-host") gw_host = next();
+    else if (a == "--gateway-port") gw_port = std::stoi(next());
+    else if (a == "--port") port = std::stoi(next());
+    else if (a == "--dataset") dataset = next();
+    else if (a == "--dataset-path") dir = next();
+  }
+  signal(SIGPIPE, SIG_IGN);
+
+  std::vector<std::string> files;
+  DIR* d = opendir(dir.c_str());
+  if (!d) {
+    fprintf(stderr, "hypha-data: cannot open %s\n", dir.c_str());
+    return 1;
+  }
+  struct dirent* ent;
+  while ((ent = readdir(d)) != nullptr) {
+    std::string fn = ent->d_name;
+    if (fn.size() > 4 && fn[0] != '.') files.push_back(dir + "/" + fn);
+  }
+  closedir(d);
+  std::sort(files.begin(), files.end());
+
+  Node node(name, gw_host, gw_port);
+  node.on_stream("pull_slice", [&](const std::string& from, const Json& header,
+                                   MsgSocket& sock) {
+    int64_t index = header.at("index").as_int();
+    if (index < 0 || index >= (int64_t)files.size()) return;
+    FILE* f = fopen(files[index].c_str(), "rb");
+    if (!f) return;
+    fseek(f, 0, SEEK_END);
+    long size = ftell(f);
+    fseek(f, 0, SEEK_SET);
+    Json szmsg;
+    szmsg["size"] = (int64_t)size;
+    sock.send_json(szmsg);
+    std::vector<char> buf(1 << 20);
+    long left = size;
+    while (left > 0) {
+      size_t chunk = std::min((long)buf.size(), left);
+      if (fread(buf.data(), 1, chunk, f) != chunk) break;
+      if (!sock.send_raw(buf.data(), chunk)) break;
+      left -= chunk;
+    }
+    fclose(f);
+  });
+  node.start(port);
+  // announce the dataset record (kad put, hypha-data.rs:176-185)
+  Json rec;
+  rec["num_slices"] = (int64_t)files.size();
+  rec["provider"] = name;
+  node.kv_put("dataset:" + dataset, rec);
+  printf("hypha-data %s serving %zu slices of %s on port %d\n", name.c_str(),
+         files.size(), dataset.c_str(), node.port());
+  fflush(stdout);
+  while (true) std::this_thread::sleep_for(std::chrono::seconds(3600));
+}

@@ -1,0 +1,532 @@
+// hypha-scheduler: auction-based worker allocation, lease renewal, DiLoCo
+// job dispatch, data-slice scheduling and the synchronization-point FSM.
+// Native redesign of /root/reference/crates/scheduler
+// (bin/hypha-scheduler.rs run() :54-432, allocator.rs, worker.rs renewal
+// loop :100-116, task.rs dispatch, scheduling/{batch_scheduler,
+// data_scheduler}.rs, tracker/*).
+
+#include <signal.h>
+#include <unistd.h>
+
+#include <atomic>
+#include <chrono>
+#include <condition_variable>
+#include <fstream>
+#include <map>
+#include <mutex>
+#include <sstream>
+#include <thread>
+#include <vector>
+
+#include <netdb.h>
+
+#include "hypha/auction.h"
+#include "hypha/batch_scheduler.h"
+#include "hypha/json.h"
+#include "hypha/net.h"
+#include "hypha/trackers.h"
+
+using namespace hypha;
+
+static double now_s() {
+  return std::chrono::duration<double>(
+             std::chrono::steady_clock::now().time_since_epoch())
+      .count();
+}
+
+struct AllocatedWorker {
+  std::string peer;
+  std::string lease_id;
+  Resources resources;
+  std::atomic<bool> alive{true};
+};
+
+// Minimal HTTP POST (the metrics bridge to the AIM driver,
+// metrics_bridge.rs:126-146). Fire-and-forget; bridge form "host:port".
+static void http_post_json(const std::string& bridge, const std::string& path,
+                           const std::string& body) {
+  size_t colon = bridge.rfind(':');
+  if (colon == std::string::npos) return;
+  int fd = tcp_connect(bridge.substr(0, colon), std::stoi(bridge.substr(colon + 1)), 2.0);
+  if (fd < 0) return;
+  char hdr[256];
+  int n = snprintf(hdr, sizeof hdr,
+                   "POST %s HTTP/1.1\r\nHost: bridge\r\nContent-Type: application/json\r\n"
+                   "Content-Length: %zu\r\nConnection: close\r\n\r\n",
+                   path.c_str(), body.size());
+  ::send(fd, hdr, n, MSG_NOSIGNAL);
+  ::send(fd, body.data(), body.size(), MSG_NOSIGNAL);
+  char buf[256];
+  ::recv(fd, buf, sizeof buf, 0);
+  ::close(fd);
+}
+
+int main(int argc, char** argv) {
+  std::string name = "scheduler", gw_host = "127.0.0.1", config_path, status_bridge;
+  int gw_port = 0, port = 0;
+  bool probe = false;
+  for (int i = 1; i < argc; ++i) {
+    std::string a = argv[i];
+    auto next = [&] { return std::string(argv[++i]); };
+    if (a == "--name") name = next();
+    else if (a == "--gateway-host") gw_host = next();
+    else if (a == "--gateway-port") gw_port = std::stoi(next());
+    else if (a == "--port") port = std::stoi(next());
+    else if (a == "--config") config_path = next();
+    else if (a == "--status-bridge") status_bridge = next();
+    else if (a == "probe") probe = true;
+  }
+  signal(SIGPIPE, SIG_IGN);
+  if (probe) {  // readiness check: gateway health RR (hypha-worker.rs probe)
+    int fd = tcp_connect(gw_host, gw_port, 3.0);
+    if (fd < 0) {
+      fprintf(stderr, "probe: gateway unreachable\n");
+      return 1;
+    }
+    MsgSocket sock(fd);
+    Json req;
+    req["kind"] = "request";
+    req["type"] = "health";
+    req["from"] = name;
+    req["body"] = Json(JsonObject{});
+    sock.send_json(req);
+    auto resp = sock.recv_json();
+    bool ok = resp && resp->get_or("body", Json(JsonObject{}))
+                          .get_or("healthy", Json(false)).as_bool();
+    printf("probe: %s\n", ok ? "healthy" : "unhealthy");
+    return ok ? 0 : 1;
+  }
+
+  // job config (scheduler_config.rs analogue; JSON instead of TOML)
+  std::ifstream cf(config_path);
+  std::stringstream ss;
+  ss << cf.rdbuf();
+  Json cfg = Json::parse(ss.str());
+  const std::string model = cfg.at("model").as_string();
+  const std::string dataset = cfg.at("dataset").as_string();
+  const int64_t num_workers = cfg.at("num_workers").as_int();
+  const int64_t update_rounds = cfg.at("update_rounds").as_int();
+  const int64_t samples_between = cfg.at("avg_samples_between_updates").as_int();
+  const int64_t batch_size = cfg.get_or("batch_size", Json(4)).as_int();
+  const int64_t seq_len = cfg.get_or("seq_len", Json(128)).as_int();
+  const double worker_bid = cfg.get_or("worker_bid", Json(1.0)).as_double();
+  const double worker_max = cfg.get_or("worker_max_price", Json(10.0)).as_double();
+
+  Node node(name, gw_host, gw_port);
+
+  // --- offer collection state (allocator.rs) ---
+  std::mutex offer_mu;
+  std::condition_variable offer_cv;
+  std::vector<WorkerOffer> offers;
+  node.on("worker_offer", [&](const std::string& from, const Json& body) {
+    WorkerOffer o;
+    o.id = body.at("id").as_string();
+    o.request_id = body.at("request_id").as_string();
+    o.worker = from;
+    o.price = body.at("price").as_double();
+    o.resources = Resources::from_json(body.at("resources"));
+    o.expires_at = now_s() + body.get_or("timeout_s", Json(0.5)).as_double();
+    {
+      std::lock_guard<std::mutex> lk(offer_mu);
+      offers.push_back(o);
+    }
+    offer_cv.notify_all();
+    Json r;
+    r["accepted"] = true;
+    return r;
+  });
+
+  // --- trackers + FSM ---
+  SliceTracker slices(dataset, 1);  // real count filled after discovery
+  std::string data_provider = "data";
+  BatchScheduler fsm(samples_between * num_workers, update_rounds);
+  std::mutex fsm_mu;
+  std::map<std::string, std::string> job_status;  // job id -> status
+  std::mutex status_mu;
+  std::condition_variable status_cv;
+
+  node.on("data", [&](const std::string& from, const Json& body) {
+    std::lock_guard<std::mutex> lk(fsm_mu);
+    auto a = slices.next(from);
+    Json r;
+    r["data_provider"] = data_provider;
+    r["index"] = (int64_t)a.index;
+    r["epoch"] = (int64_t)a.epoch;
+    return r;
+  });
+
+  node.on("progress", [&](const std::string& from, const Json& body) {
+    std::string kind = body.get_or("kind", Json("status")).as_string();
+    Progress p{};
+    if (kind == "status") {
+      p.kind = Progress::Status;
+      p.batch_size = body.get_or("batch_size", Json((int64_t)1)).as_int();
+    } else if (kind == "metrics") {
+      p.kind = Progress::Metrics;
+    } else if (kind == "update") {
+      p.kind = Progress::Update;
+    } else if (kind == "updated") {
+      p.kind = Progress::Updated;
+    } else if (kind == "update-received") {
+      p.kind = Progress::UpdateReceived;
+    }
+    ProgressResponse r;
+    {
+      std::lock_guard<std::mutex> lk(fsm_mu);
+      r = fsm.handle(from, p);
+    }
+    const char* names[] = {"ok", "continue", "schedule-update", "done", "error"};
+    Json out;
+    out["kind"] = names[(int)r.kind];
+    out["counter"] = r.counter;
+    if (kind == "metrics" && body.has("metrics")) {
+      // metrics bridge: log + forward to the AIM driver (drivers/aim_driver.py)
+      fprintf(stderr, "[metrics] %s %s\n", from.c_str(), body.at("metrics").dump().c_str());
+      if (!status_bridge.empty()) {
+        for (auto& [mname, mval] : body.at("metrics").as_object()) {
+          Json post;
+          post["worker_id"] = from;
+          post["round"] = body.get_or("round", Json((int64_t)0));
+          post["metric_name"] = mname;
+          post["value"] = mval;
+          http_post_json(status_bridge, "/status", post.dump());
+        }
+      }
+    }
+    return out;
+  });
+
+  node.on("job_status", [&](const std::string& from, const Json& body) {
+    {
+      std::lock_guard<std::mutex> lk(status_mu);
+      job_status[body.at("id").as_string()] = body.at("status").as_string();
+    }
+    status_cv.notify_all();
+    Json r;
+    r["ok"] = true;
+    return r;
+  });
+
+  node.start(port);
+  fprintf(stderr, "[scheduler] up on port %d\n", node.port());
+
+  // --- discover the data provider via the registry (DHT get, kad.rs) ---
+  int64_t num_slices = 0;
+  for (int i = 0; i < 100; ++i) {
+    auto rec = node.kv_get("dataset:" + dataset);
+    if (rec) {
+      num_slices = rec->at("num_slices").as_int();
+      data_provider = rec->get_or("provider", Json(std::string("data"))).as_string();
+      break;
+    }
+    usleep(200000);
+  }
+  if (num_slices == 0) {
+    fprintf(stderr, "[scheduler] dataset %s not found\n", dataset.c_str());
+    return 1;
+  }
+  {
+    std::lock_guard<std::mutex> lk(fsm_mu);
+    slices = SliceTracker(dataset, (int)num_slices);
+  }
+
+  // --- allocation + immediate lease renewal -----------------------------
+  std::atomic<bool> running{true};
+  std::vector<std::thread> renewers;
+  std::atomic<long> rq_seq{0};
+  // lease renewal at 2/3 of the 10 s timeout (worker.rs:100-116); the first
+  // renewal fires immediately to upgrade the 500 ms offer lease
+  auto start_renewal = [&](std::shared_ptr<AllocatedWorker> w) {
+    renewers.emplace_back([&, w] {
+      while (running && w->alive) {
+        Json b;
+        b["id"] = w->lease_id;
+        try {
+          Json r = node.request(w->peer, "renew_lease", b, 5.0);
+          if (!r.get_or("granted", Json(false)).as_bool()) {
+            w->alive = false;
+            break;
+          }
+        } catch (...) {
+          w->alive = false;
+          break;
+        }
+        for (int i = 0; i < 66 && running; ++i)
+          std::this_thread::sleep_for(std::chrono::milliseconds(100));
+      }
+      if (!w->alive) fprintf(stderr, "[scheduler] lost worker %s\n", w->peer.c_str());
+    });
+  };
+
+  auto allocate = [&](size_t count, const std::vector<std::string>& execs,
+                      Resources req) -> std::vector<WorkerOffer> {
+    GreedyOfferAggregator agg(count, PriceRange{worker_bid, worker_max},
+                              now_s() + 5.0);  // 5 s deadline (allocator.rs:25)
+    std::string rqid = name + "-rq-" + std::to_string(rq_seq.fetch_add(1));
+    {
+      std::lock_guard<std::mutex> lk(offer_mu);
+      offers.clear();
+    }
+    Json ad;
+    ad["id"] = rqid;
+    ad["resources"] = req.to_json();
+    JsonArray ex;
+    for (auto& e : execs) ex.push_back(Json(e));
+    ad["executors"] = ex;
+    ad["bid"] = worker_bid;
+    ad["timeout_s"] = 5.0;
+    node.publish("hypha/worker", ad);
+    size_t cursor = 0;
+    double last_pub = now_s();
+    std::unique_lock<std::mutex> lk(offer_mu);
+    while (now_s() < agg.deadline()) {
+      offer_cv.wait_for(lk, std::chrono::milliseconds(100));
+      bool full = false;
+      while (cursor < offers.size()) {
+        if (offers[cursor].request_id == rqid) full = agg.add(offers[cursor], now_s());
+        ++cursor;
+      }
+      if (full) break;
+      if (now_s() - last_pub > 1.0) {  // republish: capacity may have freed up
+        last_pub = now_s();
+        lk.unlock();
+        node.publish("hypha/worker", ad);
+        lk.lock();
+      }
+    }
+    return agg.finalize();
+  };
+
+  Resources worker_req = Resources::from_json(
+      cfg.get_or("worker_resources", Json::parse("{\"gpu\":1,\"cpu\":1,\"memory\":1,\"storage\":1}")));
+  auto train_offers = allocate(num_workers, {"diloco-transformer"}, worker_req);
+  if ((int64_t)train_offers.size() < num_workers) {
+    fprintf(stderr, "[scheduler] insufficient workers: %zu/%lld\n", train_offers.size(),
+            (long long)num_workers);
+    return 1;
+  }
+  auto shutdown_fail = [&](const char* why) {
+    fprintf(stderr, "[scheduler] %s\n", why);
+    running = false;
+    for (auto& t : renewers) t.join();
+    node.stop();
+    return 1;
+  };
+  std::vector<std::shared_ptr<AllocatedWorker>> train_workers;
+  for (auto& o : train_offers) {
+    auto w = std::make_shared<AllocatedWorker>();
+    w->peer = o.worker;
+    w->lease_id = o.id;
+    w->resources = o.resources;
+    start_renewal(w);
+    train_workers.push_back(w);
+  }
+  auto ps_offers = allocate(1, {"parameter-server"}, worker_req);
+  if (ps_offers.empty()) return shutdown_fail("no parameter server offer");
+  std::string ps_peer = ps_offers[0].worker;
+
+  // (train workers already renewing)
+  std::vector<std::shared_ptr<AllocatedWorker>> workers;
+  auto psw = std::make_shared<AllocatedWorker>();
+  psw->peer = ps_peer;
+  psw->lease_id = ps_offers[0].id;
+  start_renewal(psw);
+  workers.push_back(psw);
+  // give renewers a moment to upgrade the offer leases
+  std::this_thread::sleep_for(std::chrono::milliseconds(200));
+
+  // --- dispatch jobs (task.rs / hypha-scheduler.rs:328-370) ---
+  JsonArray train_peer_names;
+  for (auto& w : train_workers) train_peer_names.push_back(Json(w->peer));
+
+  // the aggregate job first, so the PS is listening before updates flow
+  std::string ps_job_id = "job-ps";
+  {
+    Json agg_cfg;
+    agg_cfg["num_workers"] = num_workers;
+    Json nesterov;
+    nesterov["learning_rate"] = cfg.get_or("outer_lr", Json(0.7)).as_double();
+    nesterov["momentum"] = cfg.get_or("outer_momentum", Json(0.9)).as_double();
+    Json opt;
+    opt["nesterov"] = nesterov;
+    agg_cfg["optimizer"] = opt;
+    Json results;
+    Json peers;
+    peers["peers"] = train_peer_names;
+    peers["strategy"] = std::string("all");
+    results["peers"] = peers;
+    agg_cfg["results"] = results;
+    // PS broadcasts carry the PS job id; workers route them to their
+    // unique local train job (see hypha_worker.cpp receive_resource)
+    Json ex;
+    ex["aggregate"] = agg_cfg;
+    Json job;
+    job["id"] = ps_job_id;
+    job["executor"] = ex;
+    Json d;
+    d["job"] = job;
+    d["lease"] = psw->lease_id;
+    node.request(ps_peer, "dispatch_job", d, 10.0);
+  }
+
+  auto dispatch_train = [&](std::shared_ptr<AllocatedWorker> w, const std::string& jid,
+                            bool join) {
+    Json tr;
+    tr["model"] = model;
+    Json fetch;
+    Json sref;
+    sref["peer"] = name;
+    sref["dataset"] = dataset;
+    fetch["scheduler"] = sref;
+    tr["data"] = fetch;
+    Json updates;
+    Json upeers;
+    upeers["peers"] = JsonArray{Json(ps_peer)};
+    upeers["strategy"] = std::string("all");
+    updates["peers"] = upeers;
+    updates["job"] = ps_job_id;  // tag pushes for the PS job
+    tr["updates"] = updates;
+    Json adam;
+    adam["learning_rate"] = cfg.get_or("inner_lr", Json(4e-4)).as_double();
+    Json opt;
+    opt["adam"] = adam;
+    tr["optimizer"] = opt;
+    tr["batch_size"] = batch_size;
+    tr["seq_len"] = seq_len;
+    if (cfg.has("checkpoint_every_rounds"))
+      tr["checkpoint_every_rounds"] = cfg.at("checkpoint_every_rounds");
+    if (cfg.has("checkpoint_dir")) tr["checkpoint_dir"] = cfg.at("checkpoint_dir");
+    if (join) tr["join"] = true;
+    Json ex;
+    ex["train"] = tr;
+    Json job;
+    job["id"] = jid;
+    job["executor"] = ex;
+    Json d;
+    d["job"] = job;
+    d["lease"] = w->lease_id;
+    node.request(w->peer, "dispatch_job", d, 10.0);
+    std::lock_guard<std::mutex> lk(fsm_mu);
+    fsm.add_worker(w->peer, batch_size);
+  };
+
+  for (size_t i = 0; i < train_workers.size(); ++i)
+    dispatch_train(train_workers[i], std::string("job-train-") + std::to_string(i),
+                   false);
+  fprintf(stderr, "[scheduler] dispatched %lld train jobs + 1 aggregate\n",
+          (long long)num_workers);
+
+  // --- fault tolerance: detect lost workers (lease renewal failure) and
+  //     allocate + join a replacement (BASELINE config 3; the reference only
+  //     detects the loss — rejoin is this framework's addition) ---
+  std::mutex tw_mu;
+  std::atomic<long long> live_workers{num_workers};
+  std::atomic<int> replace_seq{0};
+  std::thread monitor([&] {
+    std::set<std::string> handled;
+    while (running) {
+      std::this_thread::sleep_for(std::chrono::milliseconds(300));
+      std::vector<std::shared_ptr<AllocatedWorker>> lost;
+      {
+        std::lock_guard<std::mutex> lk(tw_mu);
+        for (auto& w : train_workers)
+          if (!w->alive && !handled.count(w->peer)) {
+            handled.insert(w->peer);
+            lost.push_back(w);
+          }
+      }
+      for (auto& w : lost) {
+        {
+          std::lock_guard<std::mutex> lk(fsm_mu);
+          if (fsm.finished()) break;  // training over: no point replacing
+        }
+        fprintf(stderr, "[scheduler] worker %s lost: shrinking round to %lld\n",
+                w->peer.c_str(), (long long)(live_workers - 1));
+        live_workers -= 1;
+        {
+          std::lock_guard<std::mutex> lk(fsm_mu);
+          fsm.remove_worker(w->peer);
+          slices.remove_worker(w->peer);
+        }
+        Json pc;
+        pc["job"] = ps_job_id;
+        pc["remove_peer"] = w->peer;
+        try {
+          node.request(ps_peer, "ps_control", pc, 10.0);
+        } catch (...) {
+        }
+        // replacement: re-run the auction for one worker
+        auto offers2 = allocate(1, {"diloco-transformer"}, worker_req);
+        if (offers2.empty()) {
+          fprintf(stderr, "[scheduler] no replacement worker available\n");
+          continue;
+        }
+        auto nw = std::make_shared<AllocatedWorker>();
+        nw->peer = offers2[0].worker;
+        nw->lease_id = offers2[0].id;
+        start_renewal(nw);
+        std::this_thread::sleep_for(std::chrono::milliseconds(200));
+        std::string jid = "job-train-r" + std::to_string(replace_seq.fetch_add(1));
+        try {
+          // order matters: the PS must require the joiner's update BEFORE the
+          // joiner can start pushing, or a round can close without it
+          Json pc1;
+          pc1["job"] = ps_job_id;
+          pc1["add_member"] = nw->peer;
+          node.request(ps_peer, "ps_control", pc1, 10.0);
+          dispatch_train(nw, jid, true);  // join: catch up via PS offset
+          Json pc2;
+          pc2["job"] = ps_job_id;
+          pc2["sync_to"] = nw->peer;  // sends the cumulative offset
+          node.request(ps_peer, "ps_control", pc2, 10.0);
+          live_workers += 1;
+          {
+            std::lock_guard<std::mutex> lk(tw_mu);
+            train_workers.push_back(nw);
+          }
+          fprintf(stderr, "[scheduler] replacement %s joined as %s\n",
+                  nw->peer.c_str(), jid.c_str());
+        } catch (const std::exception& e) {
+          // benign when training finished while the replacement was joining
+          fprintf(stderr, "[scheduler] replacement dispatch failed: %s\n", e.what());
+        }
+      }
+    }
+  });
+
+  // --- wait for completion ---
+  {
+    // done when every live train job completed, or (kill/rejoin edge: a
+    // replacement still catching up when the final round closed) the FSM is
+    // finished and at least one worker completed
+    std::unique_lock<std::mutex> lk(status_mu);
+    while (true) {
+      long long completed = 0;
+      for (auto& [id, st] : job_status)
+        if (st == "completed" && id != ps_job_id) ++completed;
+      if (completed >= live_workers) break;
+      bool fsm_done;
+      {
+        std::lock_guard<std::mutex> flk(fsm_mu);
+        fsm_done = fsm.finished();
+      }
+      if (fsm_done && completed >= 1) {
+        // brief grace period for stragglers, then finish
+        status_cv.wait_for(lk, std::chrono::seconds(3));
+        break;
+      }
+      status_cv.wait_for(lk, std::chrono::milliseconds(500));
+    }
+  }
+  {
+    std::lock_guard<std::mutex> lk(fsm_mu);
+    fprintf(stderr, "[scheduler] rounds completed: %lld\n", (long long)fsm.round());
+  }
+  printf("Job is completed.\n");
+  fflush(stdout);
+  running = false;
+  monitor.join();
+  for (auto& t : renewers) t.join();
+  node.stop();
+  return 0;
+}

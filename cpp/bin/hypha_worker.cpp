@@ -1,0 +1,657 @@
+// hypha-worker: the worker-node daemon.
+// Native redesign of /root/reference/crates/worker (bin/hypha-worker.rs,
+// arbiter.rs, lease_manager.rs, job_manager.rs, executor/{process,bridge,
+// parameter_server}.rs, connector/mod.rs): sells capacity in the gossip
+// auction, maintains lease-based fault tolerance (10 s TTL, 250 ms prune,
+// orphan-job cancellation), runs Train jobs as bridge-isolated subprocesses
+// and Aggregate jobs as the built-in parameter-server executor (safetensors
+// average + Nesterov on CPU — the WAN path; on-node GPU workers use RCCL).
+
+#include <signal.h>
+#include <sys/stat.h>
+#include <sys/types.h>
+#include <sys/wait.h>
+#include <unistd.h>
+
+#include <algorithm>
+#include <atomic>
+#include <chrono>
+#include <condition_variable>
+#include <cstring>
+#include <deque>
+#include <map>
+#include <mutex>
+#include <random>
+#include <set>
+#include <thread>
+#include <vector>
+
+#include "hypha/auction.h"
+#include "hypha/bridge.h"
+#include "hypha/gateway.h"
+#include "hypha/json.h"
+#include "hypha/leases.h"
+#include "hypha/net.h"
+#include "hypha/ps_math.h"
+#include "hypha/safetensors.h"
+
+using namespace hypha;
+
+static double now_s() {
+  return std::chrono::duration<double>(
+             std::chrono::steady_clock::now().time_since_epoch())
+      .count();
+}
+
+struct LeaseInfo {
+  std::string scheduler;
+  Resources resources;
+  std::string job_id;  // empty until a job is dispatched on this lease
+};
+
+struct Job {
+  std::string id;
+  std::string lease_id;
+  std::string scheduler;
+  std::string work_dir;
+  pid_t pid = -1;
+  std::shared_ptr<Bridge> bridge;
+  std::thread runner;
+  std::atomic<bool> cancelled{false};
+  // aggregate executor state
+  bool is_aggregate = false;
+  std::mutex agg_mu;
+  std::condition_variable agg_cv;
+  std::set<std::string> members;                // peers whose update closes a round
+  std::map<std::string, std::string> round_files;  // peer -> pushed file
+  std::vector<std::string> pending_sync_peers;  // joiners awaiting the offset
+};
+
+struct WorkerDaemon {
+  std::string name;
+  Node node;
+  StaticResourceManager resman;
+  OfferPolicy policy;
+  std::string exec_cmd;  // template with {SOCKET_PATH} {WORK_DIR} {JOB_JSON}
+  std::string work_root;
+
+  std::mutex mu;
+  Ledger<LeaseInfo> leases;
+  std::map<std::string, std::shared_ptr<Job>> jobs;  // job id -> job
+  struct BufferedAd {
+    std::string from;
+    Json ad;
+    double expires_at;
+  };
+  std::deque<BufferedAd> ad_buffer;
+  std::atomic<bool> running{true};
+  std::atomic<long> lease_seq{0};
+
+  WorkerDaemon(std::string nm, const std::string& gw_host, int gw_port, Resources total,
+               OfferPolicy pol, std::string cmd, std::string root)
+      : name(std::move(nm)),
+        node(name, gw_host, gw_port),
+        resman(total),
+        policy(std::move(pol)),
+        exec_cmd(std::move(cmd)),
+        work_root(std::move(root)) {}
+
+  void start(int port) {
+    node.on("health", [&](const std::string&, const Json&) {
+      Json r;
+      r["healthy"] = true;
+      return r;
+    });
+    node.on("renew_lease", [&](const std::string& from, const Json& body) {
+      std::lock_guard<std::mutex> lk(mu);
+      std::string id = body.at("id").as_string();
+      auto l = leases.get(id);
+      Json r;
+      if (!l || l->leasable.scheduler != from) {  // owner-validated (arbiter.rs:155-199)
+        r["granted"] = false;
+      } else {
+        r["granted"] = leases.renew(id, 10.0);  // lease TTL 10 s (arbiter.rs:29)
+      }
+      return r;
+    });
+    node.on("dispatch_job", [&](const std::string& from, const Json& body) {
+      return dispatch_job(from, body);
+    });
+    // PS membership control: worker kill/rejoin (BASELINE config 3).
+    // The aggregate job closes a round when EVERY current member has pushed
+    // (set semantics, not a count — immune to membership/round races).
+    // {job, remove_peer?} | {job, sync_to?} (sync_to also adds the member).
+    node.on("ps_control", [&](const std::string& from, const Json& body) {
+      std::shared_ptr<Job> job;
+      {
+        std::lock_guard<std::mutex> lk(mu);
+        auto it = jobs.find(body.at("job").as_string());
+        if (it == jobs.end()) throw std::runtime_error("no such job");
+        job = it->second;
+      }
+      if (!job->is_aggregate) throw std::runtime_error("not an aggregate job");
+      {
+        std::lock_guard<std::mutex> lk(job->agg_mu);
+        if (body.has("add_member")) job->members.insert(body.at("add_member").as_string());
+        if (body.has("sync_to")) {
+          job->pending_sync_peers.push_back(body.at("sync_to").as_string());
+          job->members.insert(body.at("sync_to").as_string());
+        }
+        if (body.has("remove_peer")) job->members.erase(body.at("remove_peer").as_string());
+      }
+      job->agg_cv.notify_all();
+      Json r;
+      r["ok"] = true;
+      return r;
+    });
+    node.on_stream("push_resource",
+                   [&](const std::string& from, const Json& header, MsgSocket& sock) {
+                     receive_resource(from, header, sock);
+                   });
+    node.start(port);
+    node.subscribe("hypha/worker", [&](const std::string& from, const Json& ad) {
+      double ttl = ad.get_or("timeout_s", Json(5.0)).as_double();
+      std::lock_guard<std::mutex> lk(mu);
+      ad_buffer.push_back(BufferedAd{from, ad, now_s() + ttl});
+    });
+    std::thread([this] {
+      while (running) {
+        try {
+          arbiter_loop();
+        } catch (const std::exception& e) {
+          fprintf(stderr, "[%s] arbiter error: %s\n", name.c_str(), e.what());
+        }
+      }
+    }).detach();
+    std::thread([this] {
+      while (running) {
+        try {
+          prune_loop();
+        } catch (const std::exception& e) {
+          fprintf(stderr, "[%s] pruner error: %s\n", name.c_str(), e.what());
+        }
+      }
+    }).detach();
+  }
+
+  // --- dRAP sell side (arbiter.rs:88-437) --------------------------------
+
+  void arbiter_loop() {
+    while (running) {
+      std::this_thread::sleep_for(std::chrono::milliseconds(200));  // ad batch window
+      std::vector<BufferedAd> batch;
+      {
+        std::lock_guard<std::mutex> lk(mu);
+        while (!ad_buffer.empty() && batch.size() < 100) {
+          if (ad_buffer.front().expires_at > now_s()) batch.push_back(ad_buffer.front());
+          ad_buffer.pop_front();
+        }
+      }
+      if (batch.empty()) continue;
+      std::vector<WorkerRequest> ads;
+      std::map<std::string, BufferedAd*> by_id;
+      for (auto& b : batch) {
+        const Json& j = b.ad;
+        WorkerRequest rq;
+        rq.id = j.at("id").as_string();
+        rq.scheduler = b.from;
+        rq.resources = Resources::from_json(j.at("resources"));
+        Json execs = j.get_or("executors", Json(JsonArray{}));  // bind: get_or returns by value
+        for (auto& e : execs.as_array()) rq.executors.push_back(e.as_string());
+        rq.bid = j.get_or("bid", Json(0.0)).as_double();
+        ads.push_back(rq);
+        by_id[rq.id] = &b;
+      }
+      auto picked = select_requests(ads, policy, resman.available());
+      std::set<std::string> answered;
+      for (auto& d : picked) {
+        if (!resman.reserve(d.request.resources)) continue;
+        answered.insert(d.request.id);
+        std::string lease_id =
+            name + "-lease-" + std::to_string(lease_seq.fetch_add(1));
+        {
+          std::lock_guard<std::mutex> lk(mu);
+          // temporary offer lease: 500 ms TTL (arbiter.rs:383-435)
+          leases.insert(lease_id, LeaseInfo{d.request.scheduler, d.request.resources, ""},
+                        0.5);
+        }
+        Json offer;
+        offer["id"] = lease_id;
+        offer["request_id"] = d.request.id;
+        offer["price"] = d.offer_price;
+        offer["resources"] = d.request.resources.to_json();
+        offer["timeout_s"] = 0.5;
+        try {
+          node.request(d.request.scheduler, "worker_offer", offer, 5.0);
+        } catch (const std::exception& e) {
+          fprintf(stderr, "[%s] offer to %s failed: %s\n", name.c_str(),
+                  d.request.scheduler.c_str(), e.what());
+        }
+      }
+      // requeue unanswered, unexpired ads (resources may free up before the
+      // scheduler's offer deadline passes)
+      {
+        std::lock_guard<std::mutex> lk(mu);
+        for (auto& b : batch) {
+          std::string id = b.ad.at("id").as_string();
+          if (!answered.count(id) && b.expires_at > now_s()) ad_buffer.push_back(b);
+        }
+      }
+    }
+  }
+
+  // --- lease pruning (arbiter.rs:98-141, 250 ms tick) --------------------
+
+  void prune_loop() {
+    while (running) {
+      std::this_thread::sleep_for(std::chrono::milliseconds(250));
+      std::vector<Ledger<LeaseInfo>::Lease> expired;
+      {
+        std::lock_guard<std::mutex> lk(mu);
+        expired = leases.drain_expired();
+      }
+      for (auto& l : expired) {
+        resman.release(l.leasable.resources);
+        if (!l.leasable.job_id.empty()) cancel_job(l.leasable.job_id);
+      }
+    }
+  }
+
+  void cancel_job(const std::string& job_id) {
+    std::shared_ptr<Job> job;
+    {
+      std::lock_guard<std::mutex> lk(mu);
+      auto it = jobs.find(job_id);
+      if (it == jobs.end()) return;
+      job = it->second;
+    }
+    job->cancelled = true;
+    if (job->pid > 0) {
+      kill(job->pid, SIGTERM);
+      // SIGKILL escalation after 5 s (process.rs:146-187)
+      std::thread([job] {
+        std::this_thread::sleep_for(std::chrono::seconds(5));
+        pid_t p = job->pid;
+        if (p > 0) kill(p, SIGKILL);
+      }).detach();
+    }
+    job->agg_cv.notify_all();
+    fprintf(stderr, "[%s] cancelled job %s (lease expired)\n", name.c_str(),
+            job_id.c_str());
+  }
+
+  // --- job dispatch (job_manager.rs:85-158) ------------------------------
+
+  Json dispatch_job(const std::string& from, const Json& body) {
+    std::string lease_id = body.at("lease").as_string();
+    Json jobj = body.at("job");
+    std::string job_id = jobj.at("id").as_string();
+    {
+      std::lock_guard<std::mutex> lk(mu);
+      auto l = leases.get(lease_id);
+      if (!l || l->leasable.scheduler != from)
+        throw std::runtime_error("invalid lease " + lease_id);
+      // dispatching upgrades the offer lease to the working TTL
+      leases.renew(lease_id, 10.0);
+      auto info = l->leasable;
+      info.job_id = job_id;
+      leases.remove(lease_id);
+      leases.insert(lease_id, info, 10.0);
+    }
+    auto job = std::make_shared<Job>();
+    job->id = job_id;
+    job->lease_id = lease_id;
+    job->scheduler = from;
+    job->work_dir = work_root + "/hypha-" + job_id;
+    mkdir(work_root.c_str(), 0755);
+    mkdir(job->work_dir.c_str(), 0755);
+    {
+      std::lock_guard<std::mutex> lk(mu);
+      jobs[job_id] = job;
+    }
+    Json ex = jobj.at("executor");
+    if (ex.has("train")) {
+      job->runner = std::thread([this, job, ex] { run_process_job(job, ex.at("train")); });
+    } else if (ex.has("aggregate")) {
+      job->is_aggregate = true;
+      job->runner = std::thread(
+          [this, job, ex] { run_aggregate_job(job, ex.at("aggregate")); });
+    } else {
+      throw std::runtime_error("unknown executor kind");
+    }
+    job->runner.detach();
+    Json r;
+    r["accepted"] = true;
+    return r;
+  }
+
+  // --- process executor (executor/process.rs:78-198) ---------------------
+
+  void run_process_job(std::shared_ptr<Job> job, Json config) {
+    std::string sock_path = job->work_dir + "/bridge.sock";
+    std::string job_json = job->work_dir + "/job.json";
+    {
+      FILE* f = fopen(job_json.c_str(), "w");
+      if (!f) {
+        fprintf(stderr, "[%s] cannot write %s\n", name.c_str(), job_json.c_str());
+        report_job_status(job, "failed");
+        return;
+      }
+      std::string s = config.dump();
+      fwrite(s.data(), 1, s.size(), f);
+      fclose(f);
+    }
+    auto bridge = std::make_shared<Bridge>(sock_path, job->work_dir);
+    job->bridge = bridge;
+    bridge->fetch_cb = [this, job](const Json& ref) { return connector_fetch(job, ref); };
+    bridge->send_cb = [this, job](const Json& ref, const std::string& path) {
+      return connector_send(job, ref, path);
+    };
+    bridge->status_cb = [this, job](const Json& progress) {
+      Json body = progress;
+      body["job"] = job->id;
+      return node.request(job->scheduler, "progress", body, 30.0);
+    };
+    bridge->start();
+
+    // {SOCKET_PATH}/{WORK_DIR}/{JOB_JSON} substitution (process.rs:201-205)
+    std::string cmd = exec_cmd;
+    auto subst = [&](const std::string& key, const std::string& val) {
+      size_t p;
+      while ((p = cmd.find(key)) != std::string::npos) cmd.replace(p, key.size(), val);
+    };
+    subst("{SOCKET_PATH}", sock_path);
+    subst("{WORK_DIR}", job->work_dir);
+    subst("{JOB_JSON}", job_json);
+
+    pid_t pid = fork();
+    if (pid == 0) {
+      execl("/bin/sh", "sh", "-c", cmd.c_str(), (char*)nullptr);
+      _exit(127);
+    }
+    job->pid = pid;
+    int status = 0;
+    waitpid(pid, &status, 0);
+    job->pid = -1;
+    bridge->stop();
+    bool ok = WIFEXITED(status) && WEXITSTATUS(status) == 0;
+    report_job_status(job, job->cancelled ? "cancelled" : (ok ? "completed" : "failed"));
+  }
+
+  void report_job_status(std::shared_ptr<Job> job, const std::string& status) {
+    Json st;
+    st["id"] = job->id;
+    st["status"] = status;
+    try {
+      node.request(job->scheduler, "job_status", st, 5.0);
+    } catch (...) {
+    }
+    std::lock_guard<std::mutex> lk(mu);
+    jobs.erase(job->id);
+  }
+
+  // --- connector (connector/mod.rs) --------------------------------------
+
+  Json connector_fetch(std::shared_ptr<Job> job, const Json& ref) {
+    // {"scheduler": {"peer": ..., "dataset": ...}} -> ask scheduler for a
+    // slice index, then pull from the data provider (mod.rs:436-507)
+    if (ref.has("scheduler")) {
+      const Json& sc = ref.at("scheduler");
+      Json q;
+      q["dataset"] = sc.at("dataset");
+      Json a = node.request(sc.at("peer").as_string(), "data", q, 30.0);
+      std::string provider = a.at("data_provider").as_string();
+      int64_t index = a.at("index").as_int();
+      Json hdr;
+      hdr["dataset"] = sc.at("dataset");
+      hdr["index"] = index;
+      auto stream = node.open_stream(provider, "pull_slice", hdr);
+      auto szmsg = stream->recv_json();
+      if (!szmsg) throw std::runtime_error("pull: no size header");
+      size_t size = (size_t)szmsg->at("size").as_int();
+      std::string out = job->work_dir + "/slice-" + std::to_string(index) + ".safetensors";
+      FILE* f = fopen(out.c_str(), "wb");
+      std::vector<char> buf(1 << 20);
+      size_t left = size;
+      while (left > 0) {
+        size_t chunk = std::min(left, buf.size());
+        if (!stream->recv_raw(buf.data(), chunk)) break;
+        fwrite(buf.data(), 1, chunk, f);
+        left -= chunk;
+      }
+      fclose(f);
+      if (left != 0) throw std::runtime_error("pull: short read");
+      Json r;
+      r["files"] = JsonArray{Json(out)};
+      return r;
+    }
+    throw std::runtime_error("unsupported fetch reference");
+  }
+
+  Json connector_send(std::shared_ptr<Job> job, const Json& ref, const std::string& path) {
+    // {"peers": {"peers": [...], "strategy": "all"}} push (mod.rs:305-433)
+    FILE* f = fopen(path.c_str(), "rb");
+    if (!f) throw std::runtime_error("send: no such file " + path);
+    fseek(f, 0, SEEK_END);
+    long size = ftell(f);
+    fseek(f, 0, SEEK_SET);
+    std::vector<char> data(size);
+    if (size > 0 && fread(data.data(), 1, size, f) != (size_t)size) {
+      fclose(f);
+      throw std::runtime_error("send: short file read");
+    }
+    fclose(f);
+    std::string base = path.substr(path.rfind('/') + 1);
+    for (auto& peer : ref.at("peers").at("peers").as_array()) {
+      Json hdr;
+      hdr["name"] = base;
+      hdr["size"] = (int64_t)size;
+      hdr["job"] = ref.get_or("job", Json(job->id));
+      auto stream = node.open_stream(peer.as_string(), "push_resource", hdr);
+      if (!stream->send_raw(data.data(), size))
+        throw std::runtime_error("send: stream write failed to " + peer.as_string());
+    }
+    Json r;
+    r["sent"] = true;
+    return r;
+  }
+
+  void receive_resource(const std::string& from, const Json& header, MsgSocket& sock) {
+    std::string jid = header.get_or("job", Json("")).as_string();
+    std::shared_ptr<Job> job;
+    {
+      std::lock_guard<std::mutex> lk(mu);
+      auto it = jobs.find(jid);
+      if (it != jobs.end()) {
+        job = it->second;
+      } else {
+        // fall back to the unique local train job (PS broadcasts carry the
+        // PS job id; each worker runs at most one train job per scheduler)
+        for (auto& [id, j] : jobs)
+          if (!j->is_aggregate) {
+            job = j;
+            break;
+          }
+      }
+    }
+    size_t size = (size_t)header.at("size").as_int();
+    std::string dir = job ? job->work_dir : work_root + "/orphan";
+    mkdir(dir.c_str(), 0755);
+    std::string out =
+        dir + "/recv-" + from + "-" + header.at("name").as_string();
+    FILE* f = fopen(out.c_str(), "wb");
+    std::vector<char> buf(1 << 20);
+    size_t left = size;
+    while (left > 0) {
+      size_t chunk = std::min(left, buf.size());
+      if (!sock.recv_raw(buf.data(), chunk)) break;
+      fwrite(buf.data(), 1, chunk, f);
+      left -= chunk;
+    }
+    fclose(f);
+    if (left != 0 || !job) return;
+    if (job->is_aggregate) {
+      std::lock_guard<std::mutex> lk(job->agg_mu);
+      job->round_files[from] = out;  // one update per peer per round
+      job->agg_cv.notify_all();
+    } else if (job->bridge) {
+      Json ev;
+      ev["path"] = out;
+      ev["size"] = (int64_t)size;
+      ev["from_peer"] = from;
+      job->bridge->push_event(ev);
+    }
+  }
+
+  // --- built-in aggregate executor (parameter_server.rs:44-304) ----------
+
+  void run_aggregate_job(std::shared_ptr<Job> job, Json config) {
+    {
+      std::lock_guard<std::mutex> lk(job->agg_mu);
+      for (auto& pj : config.at("results").at("peers").at("peers").as_array())
+        job->members.insert(pj.as_string());
+    }
+    double lr = config.at("optimizer").at("nesterov").at("learning_rate").as_double();
+    double mu_ = config.at("optimizer").at("nesterov").at("momentum").as_double();
+    SafeTensors momentum;  // persists across rounds (parameter_server.rs:393-398)
+    SafeTensors cum;       // cumulative sum of updates: joiner catch-up state
+    bool have_momentum = false;
+
+    auto serve_joiners = [&]() {
+      std::vector<std::string> peers;
+      {
+        std::lock_guard<std::mutex> lk(job->agg_mu);
+        peers.swap(job->pending_sync_peers);
+      }
+      if (peers.empty()) return;
+      std::string cpath = job->work_dir + "/global_offset.safetensors";
+      if (!have_momentum) {
+        SafeTensors empty;  // no rounds yet: zero offset (deterministic init)
+        empty.save(cpath);
+      } else {
+        cum.save(cpath);
+      }
+      for (auto& p : peers) {
+        Json ref;
+        Json jp;
+        jp["peers"] = JsonArray{Json(p)};
+        jp["strategy"] = std::string("all");
+        ref["peers"] = jp;
+        ref["offset"] = true;
+        try {
+          connector_send(job, ref, cpath);
+        } catch (const std::exception& e) {
+          fprintf(stderr, "[%s] PS joiner sync to %s failed: %s\n", name.c_str(),
+                  p.c_str(), e.what());
+        }
+      }
+    };
+
+    while (!job->cancelled) {
+      // wait until every current member has pushed this round's update
+      std::vector<std::string> files;
+      std::vector<std::string> targets;
+      {
+        std::unique_lock<std::mutex> lk(job->agg_mu);
+        auto complete = [&] {
+          if (job->members.empty()) return false;
+          for (auto& m : job->members)
+            if (!job->round_files.count(m)) return false;
+          return true;
+        };
+        job->agg_cv.wait(lk, [&] {
+          return job->cancelled || !job->pending_sync_peers.empty() || complete();
+        });
+        if (job->cancelled) break;
+        if (!complete()) {
+          lk.unlock();
+          serve_joiners();
+          continue;
+        }
+        for (auto& m : job->members) files.push_back(job->round_files.at(m));
+        targets.assign(job->members.begin(), job->members.end());
+        job->round_files.clear();  // stale non-member leftovers dropped too
+      }
+      // running average + outer Nesterov (shared ps_math.h; golden-tested
+      // vs torch SGD(nesterov=True) in tests/test_control_plane.py)
+      SafeTensors avg = ps_average(files);
+      if (!have_momentum) {
+        momentum = ps_zeros_like(avg);
+        cum = momentum;  // zeros, same shapes
+        have_momentum = true;
+      }
+      SafeTensors update = ps_nesterov(avg, momentum, lr, mu_);
+      // cumulative offset for joiners: theta_global = theta_init + cum
+      for (auto& [nm, tm] : update.tensors) {
+        auto& ctm = cum.tensors.at(nm);
+        for (int64_t e = 0; e < tm.numel(); ++e)
+          cum.set_elem(ctm, e, cum.get_elem(ctm, e) + update.get_elem(tm, e));
+      }
+      std::string upath = job->work_dir + "/update.safetensors";
+      update.save(upath);
+      // broadcast to every CURRENT member (parameter_server.rs:232-269)
+      Json send_ref;
+      Json jp;
+      JsonArray names;
+      for (auto& t : targets) names.push_back(Json(t));
+      jp["peers"] = names;
+      jp["strategy"] = std::string("all");
+      send_ref["peers"] = jp;
+      connector_send(job, send_ref, upath);
+      // notify scheduler (Progress::Updated, parameter_server.rs:274-282)
+      Json prog;
+      prog["kind"] = "updated";
+      prog["job"] = job->id;
+      try {
+        Json resp = node.request(job->scheduler, "progress", prog, 30.0);
+        if (resp.get_or("kind", Json("")).as_string() == "done") break;
+      } catch (const std::exception& e) {
+        fprintf(stderr, "[%s] PS progress failed: %s\n", name.c_str(), e.what());
+        break;
+      }
+      serve_joiners();
+    }
+    report_job_status(job, job->cancelled ? "cancelled" : "completed");
+  }
+};
+
+int main(int argc, char** argv) {
+  std::string name = "worker", gw_host = "127.0.0.1", cmd, work_root = "/tmp/hypha-work";
+  int gw_port = 0, port = 0;
+  bool probe = false;
+  Resources total{1, 4, 16, 100};
+  OfferPolicy policy{1.0, 0.0, {"diloco-transformer", "parameter-server"}};
+  for (int i = 1; i < argc; ++i) {
+    std::string a = argv[i];
+    auto next = [&] { return std::string(argv[++i]); };
+    if (a == "--name") name = next();
+    else if (a == "--gateway-host") gw_host = next();
+    else if (a == "--gateway-port") gw_port = std::stoi(next());
+    else if (a == "--port") port = std::stoi(next());
+    else if (a == "--gpu") total.gpu = std::stod(next());
+    else if (a == "--cpu") total.cpu = std::stod(next());
+    else if (a == "--memory") total.memory = std::stod(next());
+    else if (a == "--storage") total.storage = std::stod(next());
+    else if (a == "--price") policy.price = std::stod(next());
+    else if (a == "--floor") policy.floor = std::stod(next());
+    else if (a == "--exec-cmd") cmd = next();
+    else if (a == "--work-root") work_root = next();
+    else if (a == "probe") probe = true;
+  }
+  signal(SIGPIPE, SIG_IGN);
+  if (probe) {
+    int fd = tcp_connect(gw_host, gw_port, 3.0);
+    if (fd < 0) {
+      fprintf(stderr, "probe: gateway unreachable\n");
+      return 1;
+    }
+    ::close(fd);
+    printf("probe: healthy\n");
+    return 0;
+  }
+  WorkerDaemon daemon(name, gw_host, gw_port, total, policy, cmd, work_root);
+  daemon.start(port);
+  printf("hypha-worker %s ready on port %d\n", name.c_str(), daemon.node.port());
+  fflush(stdout);
+  while (true) std::this_thread::sleep_for(std::chrono::seconds(3600));
+}
