@@ -21,6 +21,7 @@ using namespace hypha;
 int main(int argc, char** argv) {
   std::string name = "data", gw_host = "127.0.0.1", dataset = "dataset", dir = ".";
   int gw_port = 0, port = 0;
+  TlsConfig tls;
   for (int i = 1; i < argc; ++i) {
     std::string a = argv[i];
     auto next = [&] { return std::string(argv[++i]); };
@@ -30,6 +31,9 @@ int main(int argc, char** argv) {
     else if (a == "--port") port = std::stoi(next());
     else if (a == "--dataset") dataset = next();
     else if (a == "--dataset-path") dir = next();
+    else if (a == "--tls-cert") tls.cert_path = next();
+    else if (a == "--tls-key") tls.key_path = next();
+    else if (a == "--tls-ca") tls.ca_path = next();
   }
   signal(SIGPIPE, SIG_IGN);
 
@@ -47,7 +51,7 @@ int main(int argc, char** argv) {
   closedir(d);
   std::sort(files.begin(), files.end());
 
-  Node node(name, gw_host, gw_port);
+  Node node(name, gw_host, gw_port, tls);
   node.on_stream("pull_slice", [&](const std::string& from, const Json& header,
                                    MsgSocket& sock) {
     int64_t index = header.at("index").as_int();

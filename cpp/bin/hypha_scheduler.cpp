@@ -65,6 +65,7 @@ int main(int argc, char** argv) {
   std::string name = "scheduler", gw_host = "127.0.0.1", config_path, status_bridge;
   int gw_port = 0, port = 0;
   bool probe = false;
+  TlsConfig tls;
   for (int i = 1; i < argc; ++i) {
     std::string a = argv[i];
     auto next = [&] { return std::string(argv[++i]); };
@@ -74,6 +75,9 @@ int main(int argc, char** argv) {
     else if (a == "--port") port = std::stoi(next());
     else if (a == "--config") config_path = next();
     else if (a == "--status-bridge") status_bridge = next();
+    else if (a == "--tls-cert") tls.cert_path = next();
+    else if (a == "--tls-key") tls.key_path = next();
+    else if (a == "--tls-ca") tls.ca_path = next();
     else if (a == "probe") probe = true;
   }
   signal(SIGPIPE, SIG_IGN);
@@ -112,7 +116,7 @@ int main(int argc, char** argv) {
   const double worker_bid = cfg.get_or("worker_bid", Json(1.0)).as_double();
   const double worker_max = cfg.get_or("worker_max_price", Json(10.0)).as_double();
 
-  Node node(name, gw_host, gw_port);
+  Node node(name, gw_host, gw_port, tls);
 
   // --- offer collection state (allocator.rs) ---
   std::mutex offer_mu;

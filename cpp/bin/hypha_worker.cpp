@@ -88,9 +88,9 @@ struct WorkerDaemon {
   std::atomic<long> lease_seq{0};
 
   WorkerDaemon(std::string nm, const std::string& gw_host, int gw_port, Resources total,
-               OfferPolicy pol, std::string cmd, std::string root)
+               OfferPolicy pol, std::string cmd, std::string root, TlsConfig tls = {})
       : name(std::move(nm)),
-        node(name, gw_host, gw_port),
+        node(name, gw_host, gw_port, std::move(tls)),
         resman(total),
         policy(std::move(pol)),
         exec_cmd(std::move(cmd)),
@@ -489,11 +489,18 @@ struct WorkerDaemon {
       left -= chunk;
     }
     fclose(f);
-    if (left != 0 || !job) return;
+    if (left != 0 || !job) {
+      fprintf(stderr, "[%s] recv incomplete from %s: left=%zu job=%d\n", name.c_str(),
+              from.c_str(), left, job != nullptr);
+      return;
+    }
     if (job->is_aggregate) {
-      std::lock_guard<std::mutex> lk(job->agg_mu);
-      job->round_files[from] = out;  // one update per peer per round
+      {
+        std::lock_guard<std::mutex> lk(job->agg_mu);
+        job->round_files[from] = out;  // one update per peer per round
+      }
       job->agg_cv.notify_all();
+      fprintf(stderr, "[%s] PS got file from %s\n", name.c_str(), from.c_str());
     } else if (job->bridge) {
       Json ev;
       ev["path"] = out;
@@ -510,6 +517,7 @@ struct WorkerDaemon {
       std::lock_guard<std::mutex> lk(job->agg_mu);
       for (auto& pj : config.at("results").at("peers").at("peers").as_array())
         job->members.insert(pj.as_string());
+      fprintf(stderr, "[%s] PS start: %zu members\n", name.c_str(), job->members.size());
     }
     double lr = config.at("optimizer").at("nesterov").at("learning_rate").as_double();
     double mu_ = config.at("optimizer").at("nesterov").at("momentum").as_double();
@@ -572,6 +580,7 @@ struct WorkerDaemon {
         targets.assign(job->members.begin(), job->members.end());
         job->round_files.clear();  // stale non-member leftovers dropped too
       }
+      fprintf(stderr, "[%s] PS round: %zu files\n", name.c_str(), files.size());
       // running average + outer Nesterov (shared ps_math.h; golden-tested
       // vs torch SGD(nesterov=True) in tests/test_control_plane.py)
       SafeTensors avg = ps_average(files);
@@ -589,6 +598,7 @@ struct WorkerDaemon {
       }
       std::string upath = job->work_dir + "/update.safetensors";
       update.save(upath);
+      fprintf(stderr, "[%s] PS broadcasting to %zu targets\n", name.c_str(), targets.size());
       // broadcast to every CURRENT member (parameter_server.rs:232-269)
       Json send_ref;
       Json jp;
@@ -619,6 +629,7 @@ int main(int argc, char** argv) {
   std::string name = "worker", gw_host = "127.0.0.1", cmd, work_root = "/tmp/hypha-work";
   int gw_port = 0, port = 0;
   bool probe = false;
+  TlsConfig tls;
   Resources total{1, 4, 16, 100};
   OfferPolicy policy{1.0, 0.0, {"diloco-transformer", "parameter-server"}};
   for (int i = 1; i < argc; ++i) {
@@ -636,6 +647,9 @@ int main(int argc, char** argv) {
     else if (a == "--floor") policy.floor = std::stod(next());
     else if (a == "--exec-cmd") cmd = next();
     else if (a == "--work-root") work_root = next();
+    else if (a == "--tls-cert") tls.cert_path = next();
+    else if (a == "--tls-key") tls.key_path = next();
+    else if (a == "--tls-ca") tls.ca_path = next();
     else if (a == "probe") probe = true;
   }
   signal(SIGPIPE, SIG_IGN);
@@ -649,7 +663,7 @@ int main(int argc, char** argv) {
     printf("probe: healthy\n");
     return 0;
   }
-  WorkerDaemon daemon(name, gw_host, gw_port, total, policy, cmd, work_root);
+  WorkerDaemon daemon(name, gw_host, gw_port, total, policy, cmd, work_root, tls);
   daemon.start(port);
   printf("hypha-worker %s ready on port %d\n", name.c_str(), daemon.node.port());
   fflush(stdout);

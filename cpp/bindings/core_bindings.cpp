@@ -280,14 +280,25 @@ PYBIND11_MODULE(_core, m) {
   // ---- in-process control-plane networking (multi-peer tests) ----
   py::class_<Gateway>(m, "Gateway")
       .def(py::init<>())
+      .def(py::init([](const std::string& cert, const std::string& key,
+                       const std::string& ca) {
+             return new Gateway(TlsConfig{cert, key, ca});
+           }),
+           py::arg("tls_cert"), py::arg("tls_key"), py::arg("tls_ca"))
       .def("start", &Gateway::start, py::arg("port") = 0,
            py::call_guard<py::gil_scoped_release>())
       .def("stop", &Gateway::stop, py::call_guard<py::gil_scoped_release>())
       .def_property_readonly("port", &Gateway::port);
 
   py::class_<Node>(m, "Node")
-      .def(py::init<std::string, std::string, int>(), py::arg("name"),
-           py::arg("gateway_host") = "127.0.0.1", py::arg("gateway_port") = 0)
+      .def(py::init([](std::string name, std::string gh, int gp, std::string cert,
+                       std::string key, std::string ca) {
+             return new Node(std::move(name), std::move(gh), gp,
+                             TlsConfig{cert, key, ca});
+           }),
+           py::arg("name"), py::arg("gateway_host") = "127.0.0.1",
+           py::arg("gateway_port") = 0, py::arg("tls_cert") = "",
+           py::arg("tls_key") = "", py::arg("tls_ca") = "")
       .def("start", &Node::start, py::arg("port") = 0,
            py::call_guard<py::gil_scoped_release>())
       .def("stop", &Node::stop, py::call_guard<py::gil_scoped_release>())
@@ -357,11 +368,41 @@ PYBIND11_MODULE(_core, m) {
              if (!r) return py::none();
              return json_to_py(*r);
            })
-      .def("resolve", [](Node& n, const std::string& p) -> py::object {
-        auto r = n.resolve(p);
-        if (!r) return py::none();
-        return py::str(*r);
-      });
+      .def("resolve",
+           [](Node& n, const std::string& p) -> py::object {
+             auto r = n.resolve(p);
+             if (!r) return py::none();
+             return py::str(*r);
+           })
+      .def("push_blob",
+           [](Node& n, const std::string& peer, const std::string& type,
+              py::object header, py::bytes payload) {
+             Json h = py_to_json(header);
+             std::string data = payload;
+             h["size"] = (int64_t)data.size();
+             py::gil_scoped_release rel;
+             auto stream = n.open_stream(peer, type, h);
+             if (!stream->send_raw(data.data(), data.size()))
+               throw std::runtime_error("stream write failed");
+           })
+      .def("on_blob",
+           [](Node& n, const std::string& type, py::function cb) {
+             auto cbp = std::make_shared<py::function>(std::move(cb));
+             n.on_stream(type, [cbp](const std::string& from, const Json& header,
+                                     MsgSocket& sock) {
+               size_t size = (size_t)header.get_or("size", Json((int64_t)0)).as_int();
+               std::string data(size, '\0');
+               bool ok = size == 0 || sock.recv_raw(data.data(), size);
+               py::gil_scoped_acquire gil;
+               py::object blob = ok ? py::object(py::bytes(data)) : py::object(py::none());
+               try {
+                 (*cbp)(from, json_to_py(header), blob);
+               } catch (py::error_already_set& e) {
+                 e.restore();
+                 PyErr_Clear();
+               }
+             });
+           });
 
   py::class_<StaticResourceManager>(m, "StaticResourceManager")
       .def(py::init<Resources>())

@@ -23,6 +23,10 @@ namespace hypha {
 
 class Gateway {
  public:
+  Gateway() = default;
+  explicit Gateway(TlsConfig tls) {
+    if (tls.enabled()) tls_ = std::make_unique<TlsContext>(tls);
+  }
   ~Gateway() { stop(); }
 
   void start(int port = 0) {
@@ -52,14 +56,24 @@ class Gateway {
     while (running_) {
       int fd = ::accept(listen_fd_, nullptr, nullptr);
       if (fd < 0) break;
-      std::thread([this, fd] { handle_conn(fd); }).detach();
+      std::thread([this, fd] {
+        SSL* ssl = nullptr;
+        if (tls_) {
+          ssl = tls_->wrap(fd, true);
+          if (!ssl) {
+            ::close(fd);
+            return;  // unauthenticated peer rejected at handshake
+          }
+        }
+        handle_conn(fd, ssl);
+      }).detach();
     }
     ::close(listen_fd_);
     listen_fd_ = -1;
   }
 
-  void handle_conn(int fd) {
-    auto sock = std::make_shared<MsgSocket>(fd);
+  void handle_conn(int fd, SSL* ssl = nullptr) {
+    auto sock = std::make_shared<MsgSocket>(fd, ssl);
     std::string peer;  // set once registered (persistent connection)
     while (running_) {
       auto msg = sock->recv_json();
@@ -159,6 +173,7 @@ class Gateway {
   std::map<std::string, std::shared_ptr<MsgSocket>> peers_;
   std::map<std::string, std::set<std::string>> subs_;
   std::map<std::string, Json> kv_;
+  std::unique_ptr<TlsContext> tls_;
 };
 
 }  // namespace hypha

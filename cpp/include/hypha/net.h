@@ -22,6 +22,7 @@
 #include <vector>
 
 #include "json.h"
+#include "tls.h"
 
 namespace hypha {
 
@@ -29,7 +30,7 @@ namespace hypha {
 
 class MsgSocket {
  public:
-  explicit MsgSocket(int fd) : fd_(fd) {}
+  explicit MsgSocket(int fd, SSL* ssl = nullptr) : fd_(fd), ssl_(ssl) {}
   ~MsgSocket();
   MsgSocket(const MsgSocket&) = delete;
 
@@ -39,10 +40,18 @@ class MsgSocket {
   bool recv_raw(char* data, size_t n);
   void close_now();
   int fd() const { return fd_; }
+  std::string peer_identity() const { return TlsContext::peer_cn(ssl_); }
 
  private:
+  bool ssl_wait_readable();  // poll until data (or error) without the io lock
+
   int fd_;
-  std::mutex write_mu_;
+  SSL* ssl_ = nullptr;  // owned; non-null when the link is mutually-TLS
+  std::mutex write_mu_;  // framing lock for plain sockets
+  // OpenSSL forbids concurrent SSL_read/SSL_write on one SSL: all TLS io is
+  // serialized by io_mu_, with readers polling for data BEFORE taking it so
+  // writers are never starved by a blocked read.
+  std::mutex io_mu_;
 };
 
 int tcp_connect(const std::string& host, int port, double timeout_s = 5.0);
@@ -61,7 +70,8 @@ class Node {
   using StreamHandler =
       std::function<void(const std::string& from, const Json& header, MsgSocket& sock)>;
 
-  Node(std::string name, std::string gateway_host, int gateway_port);
+  Node(std::string name, std::string gateway_host, int gateway_port,
+       TlsConfig tls = {});
   ~Node();
 
   // start serving on `port` (0 = ephemeral) and register with the gateway
@@ -91,12 +101,13 @@ class Node {
 
  private:
   void accept_loop();
-  void handle_conn(int fd);
+  void handle_conn(int fd, SSL* ssl = nullptr);
   void gateway_listen_loop();
   Json gateway_request(const std::string& type, const Json& body);
 
   std::string name_, gw_host_;
   int gw_port_;
+  std::unique_ptr<TlsContext> tls_;
   int listen_fd_ = -1;
   int port_ = 0;
   std::atomic<bool> running_{false};

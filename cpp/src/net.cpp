@@ -4,6 +4,7 @@
 #include <netdb.h>
 #include <netinet/in.h>
 #include <netinet/tcp.h>
+#include <poll.h>
 #include <string.h>
 #include <sys/socket.h>
 #include <sys/types.h>
@@ -19,14 +20,26 @@ namespace hypha {
 MsgSocket::~MsgSocket() { close_now(); }
 
 void MsgSocket::close_now() {
+  if (ssl_) {
+    {
+      // graceful close_notify first (if no other thread is mid-io), so the
+      // peer sees an orderly TLS shutdown instead of an unexpected EOF
+      std::unique_lock<std::mutex> lk(io_mu_, std::try_to_lock);
+      if (lk.owns_lock() && ssl_) SSL_shutdown(ssl_);
+    }
+    if (fd_ >= 0) ::shutdown(fd_, SHUT_RDWR);  // wake any blocked SSL io
+    std::lock_guard<std::mutex> lk(io_mu_);
+    if (ssl_) SSL_free(ssl_);
+    ssl_ = nullptr;
+  }
+  if (fd_ >= 0) ::shutdown(fd_, SHUT_RDWR);
   if (fd_ >= 0) {
-    ::shutdown(fd_, SHUT_RDWR);
     ::close(fd_);
     fd_ = -1;
   }
 }
 
-static bool write_all(int fd, const char* p, size_t n) {
+static bool write_all_fd(int fd, const char* p, size_t n) {
   while (n > 0) {
     ssize_t w = ::send(fd, p, n, MSG_NOSIGNAL);
     if (w <= 0) return false;
@@ -36,7 +49,7 @@ static bool write_all(int fd, const char* p, size_t n) {
   return true;
 }
 
-static bool read_all(int fd, char* p, size_t n) {
+static bool read_all_fd(int fd, char* p, size_t n) {
   while (n > 0) {
     ssize_t r = ::recv(fd, p, n, 0);
     if (r <= 0) return false;
@@ -46,21 +59,74 @@ static bool read_all(int fd, char* p, size_t n) {
   return true;
 }
 
+static bool write_all_ssl(SSL* ssl, const char* p, size_t n) {
+  while (n > 0) {
+    int w = SSL_write(ssl, p, (int)n);
+    if (w <= 0) return false;
+    p += w;
+    n -= w;
+  }
+  return true;
+}
+
+static bool read_all_ssl(SSL* ssl, char* p, size_t n) {
+  while (n > 0) {
+    int r = SSL_read(ssl, p, (int)n);
+    if (r <= 0) return false;
+    p += r;
+    n -= r;
+  }
+  return true;
+}
+
+bool MsgSocket::ssl_wait_readable() {
+  while (true) {
+    {
+      std::lock_guard<std::mutex> lk(io_mu_);
+      if (!ssl_) return false;
+      if (SSL_pending(ssl_) > 0) return true;  // buffered record bytes
+    }
+    struct pollfd p = {fd_, POLLIN, 0};
+    int rc = ::poll(&p, 1, 200);
+    if (fd_ < 0) return false;
+    if (rc < 0) return false;
+    if (rc > 0) return true;  // readable (or HUP/ERR: the read will report it)
+  }
+}
+
 bool MsgSocket::send_json(const Json& j) {
   std::string s = j.dump();
   uint32_t len = htonl((uint32_t)s.size());
+  if (ssl_) {
+    std::lock_guard<std::mutex> lk(io_mu_);
+    if (!ssl_) return false;
+    if (!write_all_ssl(ssl_, (const char*)&len, 4)) return false;
+    return write_all_ssl(ssl_, s.data(), s.size());
+  }
   std::lock_guard<std::mutex> lk(write_mu_);
-  if (!write_all(fd_, (const char*)&len, 4)) return false;
-  return write_all(fd_, s.data(), s.size());
+  if (!write_all_fd(fd_, (const char*)&len, 4)) return false;
+  return write_all_fd(fd_, s.data(), s.size());
 }
 
 std::optional<Json> MsgSocket::recv_json() {
   uint32_t len_be;
-  if (!read_all(fd_, (char*)&len_be, 4)) return std::nullopt;
-  uint32_t len = ntohl(len_be);
-  if (len > 256u * 1024u * 1024u) return std::nullopt;
-  std::string s(len, '\0');
-  if (!read_all(fd_, s.data(), len)) return std::nullopt;
+  std::string s;
+  if (ssl_) {
+    if (!ssl_wait_readable()) return std::nullopt;
+    std::lock_guard<std::mutex> lk(io_mu_);
+    if (!ssl_) return std::nullopt;
+    if (!read_all_ssl(ssl_, (char*)&len_be, 4)) return std::nullopt;
+    uint32_t len = ntohl(len_be);
+    if (len > 256u * 1024u * 1024u) return std::nullopt;
+    s.assign(len, '\0');
+    if (!read_all_ssl(ssl_, s.data(), len)) return std::nullopt;
+  } else {
+    if (!read_all_fd(fd_, (char*)&len_be, 4)) return std::nullopt;
+    uint32_t len = ntohl(len_be);
+    if (len > 256u * 1024u * 1024u) return std::nullopt;
+    s.assign(len, '\0');
+    if (!read_all_fd(fd_, s.data(), len)) return std::nullopt;
+  }
   try {
     return Json::parse(s);
   } catch (...) {
@@ -68,8 +134,23 @@ std::optional<Json> MsgSocket::recv_json() {
   }
 }
 
-bool MsgSocket::send_raw(const char* data, size_t n) { return write_all(fd_, data, n); }
-bool MsgSocket::recv_raw(char* data, size_t n) { return read_all(fd_, data, n); }
+bool MsgSocket::send_raw(const char* data, size_t n) {
+  if (ssl_) {
+    std::lock_guard<std::mutex> lk(io_mu_);
+    if (!ssl_) return false;
+    return write_all_ssl(ssl_, data, n);
+  }
+  return write_all_fd(fd_, data, n);
+}
+bool MsgSocket::recv_raw(char* data, size_t n) {
+  if (ssl_) {
+    if (!ssl_wait_readable()) return false;
+    std::lock_guard<std::mutex> lk(io_mu_);
+    if (!ssl_) return false;
+    return read_all_ssl(ssl_, data, n);
+  }
+  return read_all_fd(fd_, data, n);
+}
 
 int tcp_connect(const std::string& host, int port, double timeout_s) {
   struct addrinfo hints = {}, *res = nullptr;
@@ -123,8 +204,10 @@ int listen_port(int listen_fd) {
 
 // ---- Node ------------------------------------------------------------------
 
-Node::Node(std::string name, std::string gateway_host, int gateway_port)
-    : name_(std::move(name)), gw_host_(std::move(gateway_host)), gw_port_(gateway_port) {}
+Node::Node(std::string name, std::string gateway_host, int gateway_port, TlsConfig tls)
+    : name_(std::move(name)), gw_host_(std::move(gateway_host)), gw_port_(gateway_port) {
+  if (tls.enabled()) tls_ = std::make_unique<TlsContext>(tls);
+}
 
 Node::~Node() { stop(); }
 
@@ -141,7 +224,15 @@ void Node::start(int port) {
     // long receive timeout on the event connection
     struct timeval tv = {86400, 0};
     setsockopt(fd, SOL_SOCKET, SO_RCVTIMEO, &tv, sizeof tv);
-    gw_sock_ = std::make_unique<MsgSocket>(fd);
+    SSL* gssl = nullptr;
+    if (tls_) {
+      gssl = tls_->wrap(fd, false);
+      if (!gssl) {
+        ::close(fd);
+        throw std::runtime_error("node: gateway TLS handshake failed");
+      }
+    }
+    gw_sock_ = std::make_unique<MsgSocket>(fd, gssl);
     Json reg;
     reg["kind"] = "register";
     reg["peer"] = name_;
@@ -185,7 +276,15 @@ void Node::accept_loop() {
     setsockopt(fd, IPPROTO_TCP, TCP_NODELAY, &one, sizeof one);
     std::thread([this, fd] {
       try {
-        handle_conn(fd);
+        SSL* ssl = nullptr;
+        if (tls_) {
+          ssl = tls_->wrap(fd, true);
+          if (!ssl) {
+            ::close(fd);
+            return;  // unauthenticated peer rejected at handshake
+          }
+        }
+        handle_conn(fd, ssl);
       } catch (const std::exception& e) {
         fprintf(stderr, "[net:%s] conn handler error: %s\n", name_.c_str(), e.what());
       }
@@ -193,8 +292,8 @@ void Node::accept_loop() {
   }
 }
 
-void Node::handle_conn(int fd) {
-  MsgSocket sock(fd);
+void Node::handle_conn(int fd, SSL* ssl) {
+  MsgSocket sock(fd, ssl);
   while (running_) {
     auto msg = sock.recv_json();
     if (!msg) break;
@@ -263,7 +362,15 @@ void Node::gateway_listen_loop() {
 Json Node::gateway_request(const std::string& type, const Json& body) {
   int fd = tcp_connect(gw_host_, gw_port_, 10.0);
   if (fd < 0) throw std::runtime_error("gateway unreachable");
-  MsgSocket sock(fd);
+  SSL* ssl = nullptr;
+  if (tls_) {
+    ssl = tls_->wrap(fd, false);
+    if (!ssl) {
+      ::close(fd);
+      throw std::runtime_error("gateway TLS handshake failed");
+    }
+  }
+  MsgSocket sock(fd, ssl);
   Json req;
   req["kind"] = "request";
   req["type"] = type;
@@ -282,7 +389,15 @@ Json Node::request(const std::string& peer, const std::string& type, const Json&
   size_t colon = addr->rfind(':');
   int fd = tcp_connect(addr->substr(0, colon), std::stoi(addr->substr(colon + 1)), timeout_s);
   if (fd < 0) throw std::runtime_error("peer unreachable: " + peer);
-  MsgSocket sock(fd);
+  SSL* pssl = nullptr;
+  if (tls_) {
+    pssl = tls_->wrap(fd, false);
+    if (!pssl) {
+      ::close(fd);
+      throw std::runtime_error("TLS handshake failed with " + peer);
+    }
+  }
+  MsgSocket sock(fd, pssl);
   Json req;
   req["kind"] = "request";
   req["type"] = type;
@@ -307,7 +422,15 @@ std::unique_ptr<MsgSocket> Node::open_stream(const std::string& peer, const std:
   struct timeval tv = {600, 0};
   setsockopt(fd, SOL_SOCKET, SO_RCVTIMEO, &tv, sizeof tv);
   setsockopt(fd, SOL_SOCKET, SO_SNDTIMEO, &tv, sizeof tv);
-  auto sock = std::make_unique<MsgSocket>(fd);
+  SSL* sssl = nullptr;
+  if (tls_) {
+    sssl = tls_->wrap(fd, false);
+    if (!sssl) {
+      ::close(fd);
+      throw std::runtime_error("stream TLS handshake failed with " + peer);
+    }
+  }
+  auto sock = std::make_unique<MsgSocket>(fd, sssl);
   Json req;
   req["kind"] = "stream";
   req["type"] = type;

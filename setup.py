@@ -95,7 +95,7 @@ def build_core_extension():
         "g++", "-O2", "-std=c++17", "-fPIC", "-shared", "-pthread",
         "-DTORCH_EXTENSION_NAME=_core",
         f"-I{CPP_DIR / 'include'}", f"-I{pybind11.get_include()}", f"-I{py_inc}",
-        *[str(s) for s in srcs], "-o", str(out),
+        *[str(s) for s in srcs], "-lssl", "-lcrypto", "-o", str(out),
     ]
     _run(cmd)
     print(f"built {out}")
@@ -117,7 +117,8 @@ def build_daemons():
             continue
         cmd = [
             "g++", "-O2", "-std=c++17", "-pthread",
-            f"-I{CPP_DIR / 'include'}", str(src), *lib_srcs, "-o", str(exe),
+            f"-I{CPP_DIR / 'include'}", str(src), *lib_srcs,
+            "-lssl", "-lcrypto", "-o", str(exe),
         ]
         print("+", " ".join(cmd), flush=True)
         procs.append(subprocess.Popen(cmd))

@@ -245,3 +245,76 @@ def test_cluster_checkpointing(binaries, tmp_path):
                 p.wait(timeout=10)
             except Exception:
                 pass
+
+
+@pytest.mark.timeout(300)
+def test_cluster_diloco_over_mtls(binaries, tmp_path):
+    """The full cluster flow with every control-plane connection mutually
+    authenticated (certutil PKI -> --tls-* flags on all daemons)."""
+    from hypha_amd.data.synthetic import write_slice_files
+
+    pki = tmp_path / "pki"
+    tool = REPO / "tools" / "hypha_certutil.py"
+    run = lambda *a: subprocess.run([sys.executable, str(tool), *a], check=True)
+    run("root", "--out", str(pki))
+    run("org", "--out", str(pki), "--name", "org1")
+    for n in ("gw", "data-node", "scheduler", "worker-0", "worker-1", "worker-2"):
+        run("node", "--out", str(pki), "--org", "org1", "--name", n)
+
+    def tls(n):
+        return ["--tls-cert", str(pki / f"{n}.chain.pem"),
+                "--tls-key", str(pki / f"{n}.key"),
+                "--tls-ca", str(pki / "root.crt")]
+
+    data_dir = tmp_path / "slices"
+    write_slice_files(str(data_dir), "synth", 4, 16, 512, 128)
+    gw_port = free_port()
+    env = dict(os.environ, PYTHONPATH=str(REPO))
+    procs = []
+
+    def spawn(name, cmd):
+        log = open(tmp_path / f"{name}.log", "w")
+        p = subprocess.Popen(cmd, cwd=REPO, env=env, stdout=log, stderr=log)
+        procs.append(p)
+        return p
+
+    try:
+        spawn("gateway", [str(BIN / "hypha-gateway"), "--port", str(gw_port), *tls("gw")])
+        time.sleep(0.3)
+        spawn("data", [str(BIN / "hypha-data"), "--name", "data-node",
+                       "--gateway-host", "127.0.0.1", "--gateway-port", str(gw_port),
+                       "--dataset", "synth", "--dataset-path", str(data_dir),
+                       *tls("data-node")])
+        exec_cmd = (f"{sys.executable} -m hypha_amd.runtime.executor "
+                    "--socket {SOCKET_PATH} --work-dir {WORK_DIR} --job {JOB_JSON}")
+        for i in range(3):
+            spawn(f"worker{i}", [str(BIN / "hypha-worker"), "--name", f"worker-{i}",
+                                 "--gateway-host", "127.0.0.1", "--gateway-port", str(gw_port),
+                                 "--exec-cmd", exec_cmd,
+                                 "--work-root", str(tmp_path / f"work{i}"),
+                                 *tls(f"worker-{i}")])
+        time.sleep(0.5)
+        cfg = tmp_path / "job.json"
+        cfg.write_text(
+            '{"model": "llama-tiny", "dataset": "synth", "num_workers": 2,'
+            ' "update_rounds": 2, "avg_samples_between_updates": 8,'
+            ' "batch_size": 2, "seq_len": 128, "inner_lr": 0.001}'
+        )
+        sched = subprocess.Popen(
+            [str(BIN / "hypha-scheduler"), "--name", "scheduler",
+             "--gateway-host", "127.0.0.1", "--gateway-port", str(gw_port),
+             "--config", str(cfg), *tls("scheduler")],
+            cwd=REPO, env=env, stdout=subprocess.PIPE,
+            stderr=open(tmp_path / "sched.log", "w"), text=True)
+        procs.append(sched)
+        out, _ = sched.communicate(timeout=240)
+        assert "Job is completed." in out, (tmp_path / "sched.log").read_text()[-3000:]
+    finally:
+        for p in procs:
+            if p.poll() is None:
+                p.send_signal(signal.SIGKILL)
+        for p in procs:
+            try:
+                p.wait(timeout=10)
+            except Exception:
+                pass
