@@ -47,6 +47,9 @@ def main() -> int:
     p.add_argument("--device", type=str, default=None)
     p.add_argument("--fp8", action="store_true", help="fp8 e4m3 GEMMs (config 5 path)")
     p.add_argument("--state-bits", type=int, default=32, choices=(32, 8))
+    p.add_argument("--memory-mode", default="full", choices=("full", "lean"),
+                   help="lean: bf16-SR master-free params, 8-bit state, streamed "
+                        "grads, host theta0 (configs 4/5 sizing)")
     args = p.parse_args()
 
     from hypha_amd import models, ops
@@ -69,8 +72,12 @@ def main() -> int:
     import contextlib
 
     build_ctx = torch.device(device) if on_gpu else contextlib.nullcontext()
+    if args.memory_mode == "lean":
+        torch.set_default_dtype(torch.bfloat16)  # build 46B+ models without a
+        # transient fp32 copy (187 GB for Mixtral); rope tables stay fp32
     with build_ctx:
         model = models.build(args.model)
+    torch.set_default_dtype(torch.float32)
     if args.fp8 and on_gpu:
         from hypha_amd.ops.fp8 import convert_linears_to_fp8
 
@@ -82,7 +89,12 @@ def main() -> int:
         inner=InnerOptConfig(lr=4e-4, warmup_steps=10, schedule="constant",
                              state_bits=args.state_bits),
     )
-    worker = DiLoCoWorker(model, cfg, comm=comm, device=device)
+    if args.memory_mode == "lean":
+        from hypha_amd.parallel import LeanDiLoCoWorker
+
+        worker = LeanDiLoCoWorker(model, cfg, comm=comm, device=device)
+    else:
+        worker = DiLoCoWorker(model, cfg, comm=comm, device=device)
     data = SyntheticTokens(
         model.cfg.vocab_size, args.seq_len, args.batch, seed=77, rank=rank
     )
@@ -122,7 +134,8 @@ def main() -> int:
     value = tokens_per_step_job * args.steps / elapsed
     ms_per_step = elapsed / args.steps * 1000.0
 
-    payload_bytes_per_sync = worker.fp.numel * 2  # bf16 comm dtype
+    model_numel = worker.numel if args.memory_mode == "lean" else worker.fp.numel
+    payload_bytes_per_sync = model_numel * 2  # bf16 comm dtype
     synced_in_window = worker.round - rounds_before
     if rank == 0:
         result = {
@@ -144,7 +157,8 @@ def main() -> int:
                 "seq_len": args.seq_len,
                 "parallelism": f"diloco-dp{n_gpus}",
                 "h": args.h,
-                "model_params": worker.fp.numel,
+                "model_params": model_numel,
+                "memory_mode": args.memory_mode,
                 "outer_syncs_in_timed_window": synced_in_window,
                 "outer_sync_payload_bytes": payload_bytes_per_sync,
                 "outer_sync_wire_bytes_per_rank": comm.wire_bytes_per_rank(payload_bytes_per_sync),

@@ -20,6 +20,14 @@ torch::Tensor swiglu_fwd(torch::Tensor gate, torch::Tensor up);
 torch::Tensor grad_norm_sq(torch::Tensor x);
 // grouped_gemm.hip
 torch::Tensor grouped_gemm(torch::Tensor x, torch::Tensor w, torch::Tensor group_offsets);
+// lean_opt.hip
+void adamw8_lean_(torch::Tensor param, torch::Tensor grad, torch::Tensor m8,
+                  torch::Tensor v8, torch::Tensor m_scale, torch::Tensor v_scale,
+                  double lr, double beta1, double beta2, double eps, double wd,
+                  long step, long seed);
+void extract_delta_bf16(torch::Tensor theta_t, torch::Tensor theta0, torch::Tensor out);
+void nesterov_bf16_(torch::Tensor theta, torch::Tensor delta, torch::Tensor mom,
+                    double lr, double mu, long seed);
 std::vector<torch::Tensor> swiglu_bwd(torch::Tensor dout, torch::Tensor gate,
                                       torch::Tensor up);
 // rmsnorm.hip
@@ -60,6 +68,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("swiglu_fwd", &swiglu_fwd);
   m.def("grad_norm_sq", &grad_norm_sq);
   m.def("grouped_gemm", &grouped_gemm);
+  m.def("adamw8_lean_", &adamw8_lean_);
+  m.def("extract_delta_bf16", &extract_delta_bf16);
+  m.def("nesterov_bf16_", &nesterov_bf16_);
   m.def("swiglu_bwd", &swiglu_bwd);
   m.def("rmsnorm_fwd", &rmsnorm_fwd);
   m.def("rmsnorm_bwd", &rmsnorm_bwd);

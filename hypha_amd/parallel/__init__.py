@@ -1,5 +1,6 @@
 from .comm import Comm
 from .diloco import DiLoCoConfig, DiLoCoWorker, InnerOptConfig, OuterOptConfig, lr_at
+from .lean import LeanDiLoCoWorker
 
 __all__ = [
     "Comm",
@@ -8,4 +9,5 @@ __all__ = [
     "InnerOptConfig",
     "OuterOptConfig",
     "lr_at",
+    "LeanDiLoCoWorker",
 ]
