@@ -47,6 +47,8 @@ def main() -> int:
     p.add_argument("--device", type=str, default=None)
     p.add_argument("--fp8", action="store_true", help="fp8 e4m3 GEMMs (config 5 path)")
     p.add_argument("--state-bits", type=int, default=32, choices=(32, 8))
+    p.add_argument("--grad-checkpoint", action="store_true",
+                   help="per-block activation checkpointing (70B-class fits)")
     p.add_argument("--memory-mode", default="full", choices=("full", "lean"),
                    help="lean: bf16-SR master-free params, 8-bit state, streamed "
                         "grads, host theta0 (configs 4/5 sizing)")
@@ -76,7 +78,13 @@ def main() -> int:
         torch.set_default_dtype(torch.bfloat16)  # build 46B+ models without a
         # transient fp32 copy (187 GB for Mixtral); rope tables stay fp32
     with build_ctx:
-        model = models.build(args.model)
+        overrides = {}
+        if args.grad_checkpoint:
+            overrides["gradient_checkpointing"] = True
+        try:
+            model = models.build(args.model, **overrides)
+        except TypeError:  # model family without the flag
+            model = models.build(args.model)
     torch.set_default_dtype(torch.float32)
     if args.fp8 and on_gpu:
         from hypha_amd.ops.fp8 import convert_linears_to_fp8
