@@ -19,8 +19,10 @@ MANIFEST = "manifest.json"
 
 
 def save_checkpoint(worker, out_dir: str) -> None:
-    """Save a DiLoCoWorker's full training state."""
+    """Save a DiLoCoWorker's (or LeanDiLoCoWorker's) full training state."""
     os.makedirs(out_dir, exist_ok=True)
+    if not hasattr(worker, "fp"):  # lean engine
+        return _save_lean(worker, out_dir)
     fp = worker.fp
     save_file({"theta_global": fp.theta0.cpu()}, os.path.join(out_dir, "0_global_weights.safetensors"))
     state = {
@@ -47,10 +49,40 @@ def save_checkpoint(worker, out_dir: str) -> None:
         json.dump(manifest, f, indent=2)
 
 
+def _save_lean(worker, out_dir: str) -> None:
+    save_file({"theta_global": worker.theta0_host.clone()},
+              os.path.join(out_dir, "0_global_weights.safetensors"))
+    save_file(
+        {
+            "outer_momentum": worker.outer_m_host.clone(),
+            "params": worker.flat.cpu(),
+            "m8": worker.m8.cpu(),
+            "v8": worker.v8.cpu(),
+            "m_scale": worker.m_scale.cpu(),
+            "v_scale": worker.v_scale.cpu(),
+        },
+        os.path.join(out_dir, "optimizer_state.safetensors"),
+    )
+    manifest = {
+        "format": "hypha_amd.checkpoint.lean.v1",
+        "numel": worker.numel,
+        "inner_step_count": worker.inner_step_count,
+        "round": worker.round,
+        "steps_in_round": worker.steps_in_round,
+        "h": worker.cfg.h,
+        "outer_lr": worker.cfg.outer.lr,
+        "outer_momentum": worker.cfg.outer.momentum,
+    }
+    with open(os.path.join(out_dir, MANIFEST), "w") as f:
+        json.dump(manifest, f, indent=2)
+
+
 def load_checkpoint(worker, ckpt_dir: str) -> dict:
     """Restore a DiLoCoWorker's state in place; returns the manifest."""
     with open(os.path.join(ckpt_dir, MANIFEST)) as f:
         manifest = json.load(f)
+    if manifest.get("format", "").startswith("hypha_amd.checkpoint.lean"):
+        return _load_lean(worker, ckpt_dir, manifest)
     if manifest["numel"] != worker.fp.numel:
         raise ValueError(
             f"checkpoint numel {manifest['numel']} != model numel {worker.fp.numel}"
@@ -70,6 +102,24 @@ def load_checkpoint(worker, ckpt_dir: str) -> dict:
         fp.exp_avg_sq.copy_(opt["exp_avg_sq"].to(fp.master.device))
     fp.master.copy_(opt["master"].to(fp.master.device))
     fp.flat.copy_(fp.master)
+    worker.inner_step_count = manifest["inner_step_count"]
+    worker.round = manifest["round"]
+    worker.steps_in_round = manifest["steps_in_round"]
+    return manifest
+
+
+def _load_lean(worker, ckpt_dir: str, manifest: dict) -> dict:
+    if manifest["numel"] != worker.numel:
+        raise ValueError("checkpoint numel mismatch")
+    gw = load_file(os.path.join(ckpt_dir, "0_global_weights.safetensors"))
+    worker.theta0_host.copy_(gw["theta_global"])
+    opt = load_file(os.path.join(ckpt_dir, "optimizer_state.safetensors"))
+    worker.outer_m_host.copy_(opt["outer_momentum"])
+    worker.flat.copy_(opt["params"].to(worker.device))
+    worker.m8.copy_(opt["m8"].to(worker.device))
+    worker.v8.copy_(opt["v8"].to(worker.device))
+    worker.m_scale.copy_(opt["m_scale"].to(worker.device))
+    worker.v_scale.copy_(opt["v_scale"].to(worker.device))
     worker.inner_step_count = manifest["inner_step_count"]
     worker.round = manifest["round"]
     worker.steps_in_round = manifest["steps_in_round"]

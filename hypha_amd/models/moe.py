@@ -89,8 +89,16 @@ class MoEMLP(nn.Module):
         order = torch.argsort(flat_expert, stable=True)
         counts = torch.bincount(flat_expert, minlength=cfg.n_experts).tolist()
         gathered = xf[flat_tok[order]]  # [T*K, h] grouped by expert
+        # The native grouped kernel measured SLOWER than per-expert hipBLASLt
+        # at every tested size (tools/moe_gemm_bench.py: 205-687 vs 250-1154
+        # TF/s), so the library loop is the default; HYPHA_NATIVE_GROUPED=1
+        # routes inference through the native kernel (building block for the
+        # round-2 fused MoE path).
+        import os
+
         use_native_grouped = (
-            not torch.is_grad_enabled() and gathered.is_cuda
+            os.environ.get("HYPHA_NATIVE_GROUPED", "0") == "1"
+            and not torch.is_grad_enabled() and gathered.is_cuda
             and gathered.dtype == torch.bfloat16 and ops.has_native()
             and cfg.ffn_hidden % 128 == 0 and cfg.hidden_size % 128 == 0
         )
