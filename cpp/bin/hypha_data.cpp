@@ -34,6 +34,7 @@ int main(int argc, char** argv) {
     else if (a == "--tls-cert") tls.cert_path = next();
     else if (a == "--tls-key") tls.key_path = next();
     else if (a == "--tls-ca") tls.ca_path = next();
+    else if (a == "--tls-crl") tls.crl_path = next();
   }
   signal(SIGPIPE, SIG_IGN);
 

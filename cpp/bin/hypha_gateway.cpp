@@ -17,6 +17,7 @@ int main(int argc, char** argv) {
     else if (a == "--tls-cert") tls.cert_path = argv[++i];
     else if (a == "--tls-key") tls.key_path = argv[++i];
     else if (a == "--tls-ca") tls.ca_path = argv[++i];
+    else if (a == "--tls-crl") tls.crl_path = argv[++i];
   }
   signal(SIGPIPE, SIG_IGN);
   hypha::Gateway gw(tls);

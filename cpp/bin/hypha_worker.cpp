@@ -86,6 +86,11 @@ struct WorkerDaemon {
   std::deque<BufferedAd> ad_buffer;
   std::atomic<bool> running{true};
   std::atomic<long> lease_seq{0};
+  // Versioned parameter KV (messages lib.rs:698-739 parameter_pull/push —
+  // declared in the reference protocol; here a working store):
+  // "<job>/<key>" -> version -> blob. Monotonic versions, latest by default.
+  std::mutex kv_mu;
+  std::map<std::string, std::map<uint64_t, std::string>> param_store;
 
   WorkerDaemon(std::string nm, const std::string& gw_host, int gw_port, Resources total,
                OfferPolicy pol, std::string cmd, std::string root, TlsConfig tls = {})
@@ -147,6 +152,60 @@ struct WorkerDaemon {
     node.on_stream("push_resource",
                    [&](const std::string& from, const Json& header, MsgSocket& sock) {
                      receive_resource(from, header, sock);
+                   });
+    // parameter_push (lib.rs:720-739): header {job, key, version?, size} +
+    // payload bytes -> ack {ok, version}. Version defaults to latest+1.
+    node.on_stream("param_push",
+                   [&](const std::string&, const Json& header, MsgSocket& sock) {
+                     int64_t size = header.at("size").as_int();
+                     std::string data((size_t)size, '\0');
+                     if (size > 0 && !sock.recv_raw(data.data(), (size_t)size)) return;
+                     Json r;
+                     {
+                       std::lock_guard<std::mutex> lk(kv_mu);
+                       auto& versions = param_store[header.at("job").as_string() + "/" +
+                                                    header.at("key").as_string()];
+                       uint64_t v = header.has("version")
+                                        ? (uint64_t)header.at("version").as_int()
+                                        : (versions.empty() ? 1
+                                                            : versions.rbegin()->first + 1);
+                       versions[v] = std::move(data);
+                       r["ok"] = true;
+                       r["version"] = (int64_t)v;
+                     }
+                     sock.send_json(r);
+                   });
+    // parameter_pull (lib.rs:698-717): header {job, key, version?} ->
+    // {found, version, size} + payload (NotFound -> found=false).
+    node.on_stream("param_pull",
+                   [&](const std::string&, const Json& header, MsgSocket& sock) {
+                     std::string blob;
+                     Json r;
+                     r["found"] = false;
+                     {
+                       std::lock_guard<std::mutex> lk(kv_mu);
+                       auto it = param_store.find(header.at("job").as_string() + "/" +
+                                                  header.at("key").as_string());
+                       if (it != param_store.end() && !it->second.empty()) {
+                         if (header.has("version")) {
+                           auto vit =
+                               it->second.find((uint64_t)header.at("version").as_int());
+                           if (vit != it->second.end()) {
+                             r["found"] = true;
+                             r["version"] = (int64_t)vit->first;
+                             blob = vit->second;
+                           }
+                         } else {
+                           auto& last = *it->second.rbegin();
+                           r["found"] = true;
+                           r["version"] = (int64_t)last.first;
+                           blob = last.second;
+                         }
+                       }
+                     }
+                     r["size"] = (int64_t)blob.size();
+                     if (!sock.send_json(r)) return;
+                     if (!blob.empty()) sock.send_raw(blob.data(), blob.size());
                    });
     node.start(port);
     node.subscribe("hypha/worker", [&](const std::string& from, const Json& ad) {
@@ -442,7 +501,26 @@ struct WorkerDaemon {
     }
     fclose(f);
     std::string base = path.substr(path.rfind('/') + 1);
-    for (auto& peer : ref.at("peers").at("peers").as_array()) {
+    // SelectionStrategy (connector/mod.rs:330-380): "all" -> every peer;
+    // "one" -> first; "random" -> uniform pick (the reference collapses
+    // Random to first — we honor the declared semantics).
+    const Json& pj = ref.at("peers");
+    std::vector<Json> targets;
+    {
+      const JsonArray& peers = pj.at("peers").as_array();
+      if (peers.empty()) throw std::runtime_error("send: no peers provided");
+      std::string strategy = pj.get_or("strategy", Json("all")).as_string();
+      if (strategy == "all") {
+        targets.assign(peers.begin(), peers.end());
+      } else if (strategy == "one") {
+        targets.push_back(peers.front());
+      } else if (strategy == "random") {
+        targets.push_back(peers[(size_t)rand() % peers.size()]);
+      } else {
+        throw std::runtime_error("send: unknown strategy " + strategy);
+      }
+    }
+    for (auto& peer : targets) {
       Json hdr;
       hdr["name"] = base;
       hdr["size"] = (int64_t)size;
@@ -650,6 +728,7 @@ int main(int argc, char** argv) {
     else if (a == "--tls-cert") tls.cert_path = next();
     else if (a == "--tls-key") tls.key_path = next();
     else if (a == "--tls-ca") tls.ca_path = next();
+    else if (a == "--tls-crl") tls.crl_path = next();
     else if (a == "probe") probe = true;
   }
   signal(SIGPIPE, SIG_IGN);

@@ -281,10 +281,11 @@ PYBIND11_MODULE(_core, m) {
   py::class_<Gateway>(m, "Gateway")
       .def(py::init<>())
       .def(py::init([](const std::string& cert, const std::string& key,
-                       const std::string& ca) {
-             return new Gateway(TlsConfig{cert, key, ca});
+                       const std::string& ca, const std::string& crl) {
+             return new Gateway(TlsConfig{cert, key, ca, crl});
            }),
-           py::arg("tls_cert"), py::arg("tls_key"), py::arg("tls_ca"))
+           py::arg("tls_cert"), py::arg("tls_key"), py::arg("tls_ca"),
+           py::arg("tls_crl") = "")
       .def("start", &Gateway::start, py::arg("port") = 0,
            py::call_guard<py::gil_scoped_release>())
       .def("stop", &Gateway::stop, py::call_guard<py::gil_scoped_release>())
@@ -292,13 +293,14 @@ PYBIND11_MODULE(_core, m) {
 
   py::class_<Node>(m, "Node")
       .def(py::init([](std::string name, std::string gh, int gp, std::string cert,
-                       std::string key, std::string ca) {
+                       std::string key, std::string ca, std::string crl) {
              return new Node(std::move(name), std::move(gh), gp,
-                             TlsConfig{cert, key, ca});
+                             TlsConfig{cert, key, ca, crl});
            }),
            py::arg("name"), py::arg("gateway_host") = "127.0.0.1",
            py::arg("gateway_port") = 0, py::arg("tls_cert") = "",
-           py::arg("tls_key") = "", py::arg("tls_ca") = "")
+           py::arg("tls_key") = "", py::arg("tls_ca") = "",
+           py::arg("tls_crl") = "")
       .def("start", &Node::start, py::arg("port") = 0,
            py::call_guard<py::gil_scoped_release>())
       .def("stop", &Node::stop, py::call_guard<py::gil_scoped_release>())
@@ -385,6 +387,41 @@ PYBIND11_MODULE(_core, m) {
              if (!stream->send_raw(data.data(), data.size()))
                throw std::runtime_error("stream write failed");
            })
+      .def("stream_call",
+           // bidirectional stream RPC: send header (+payload), read a JSON
+           // reply and an optional `size`-byte body. Client side of the
+           // param_push/param_pull versioned-KV protocol (lib.rs:698-739).
+           [](Node& n, const std::string& peer, const std::string& type,
+              py::object header, py::object payload) {
+             Json h = py_to_json(header);
+             std::string data;
+             if (!payload.is_none()) {
+               data = std::string(py::cast<py::bytes>(payload));
+               h["size"] = (int64_t)data.size();
+             }
+             Json reply;
+             std::string body;
+             {
+               py::gil_scoped_release rel;
+               auto s = n.open_stream(peer, type, h);
+               if (!data.empty() && !s->send_raw(data.data(), data.size()))
+                 throw std::runtime_error("stream_call: write failed");
+               auto r = s->recv_json();
+               if (!r) throw std::runtime_error("stream_call: no reply");
+               reply = *r;
+               int64_t sz = reply.get_or("size", Json((int64_t)0)).as_int();
+               if (sz > 0) {
+                 body.resize((size_t)sz);
+                 if (!s->recv_raw(body.data(), (size_t)sz))
+                   throw std::runtime_error("stream_call: short body");
+               }
+             }
+             py::object b = body.empty() ? py::object(py::none())
+                                         : py::object(py::bytes(body));
+             return py::make_tuple(json_to_py(reply), b);
+           },
+           py::arg("peer"), py::arg("type"), py::arg("header"),
+           py::arg("payload") = py::none())
       .def("on_blob",
            [](Node& n, const std::string& type, py::function cb) {
              auto cbp = std::make_shared<py::function>(std::move(cb));

@@ -6,6 +6,7 @@
 #pragma once
 
 #include <openssl/err.h>
+#include <openssl/pem.h>
 #include <openssl/ssl.h>
 #include <openssl/x509.h>
 
@@ -19,6 +20,7 @@ struct TlsConfig {
   std::string cert_path;  // node certificate (PEM; may be a chain bundle)
   std::string key_path;   // node private key (PEM)
   std::string ca_path;    // trust anchor bundle (root + org CAs)
+  std::string crl_path;   // optional CRL bundle (PEM; one CRL per issuing CA)
   bool enabled() const { return !cert_path.empty(); }
 };
 
@@ -39,6 +41,25 @@ class TlsContext {
     // mutual authentication: both sides must present a CA-signed cert
     SSL_CTX_set_verify(ctx_, SSL_VERIFY_PEER | SSL_VERIFY_FAIL_IF_NO_PEER_CERT,
                        nullptr);
+    // Revocation: the reference's forked libp2p checks CRLs via rustls
+    // WebPkiClientVerifier (rfc/2025-05-30_mtls.md). When a CRL bundle is
+    // configured, every peer leaf cert is checked against its issuer's CRL
+    // (X509_V_FLAG_CRL_CHECK: leaf-only, so only issuing org CAs need CRLs).
+    if (!cfg.crl_path.empty()) {
+      X509_STORE* store = SSL_CTX_get_cert_store(ctx_);
+      FILE* fp = fopen(cfg.crl_path.c_str(), "r");
+      if (!fp) throw std::runtime_error("tls: cannot open CRL " + cfg.crl_path);
+      int loaded = 0;
+      while (X509_CRL* crl = PEM_read_X509_CRL(fp, nullptr, nullptr, nullptr)) {
+        X509_STORE_add_crl(store, crl);
+        X509_CRL_free(crl);
+        ++loaded;
+      }
+      fclose(fp);
+      if (loaded == 0)
+        throw std::runtime_error("tls: no CRLs found in " + cfg.crl_path);
+      X509_STORE_set_flags(store, X509_V_FLAG_CRL_CHECK);
+    }
     // No TLS1.3 session tickets: a client that only WRITES (tensor push
     // streams) would otherwise leave the server's post-handshake ticket
     // records unread, and closing a socket with unread data sends TCP RST —

@@ -74,6 +74,45 @@ def test_mtls_rejects_foreign_ca(pki):
         gw.stop()
 
 
+def test_crl_revoked_peer_rejected(pki, tmp_path_factory):
+    """Revocation parity (rfc/2025-05-30_mtls.md CRL checking): after
+    `certutil revoke`, a gateway loading the org CRL refuses the revoked
+    node's handshake while still accepting unrevoked peers."""
+    out = tmp_path_factory.mktemp("crlpki")
+    run = lambda *a: subprocess.run([sys.executable, str(TOOL), *a], check=True)
+    run("root", "--out", str(out))
+    run("org", "--out", str(out), "--name", "org1")
+    for n in ("gw", "carol", "dave"):
+        run("node", "--out", str(out), "--org", "org1", "--name", n)
+    run("crl", "--out", str(out), "--org", "org1")  # empty CRL first
+    crl = str(out / "org1.crl.pem")
+
+    gw = core.Gateway(str(out / "gw.chain.pem"), str(out / "gw.key"),
+                      str(out / "root.crt"), crl)
+    gw.start(0)
+    carol = core.Node("carol", "127.0.0.1", gw.port, **_tls_args(out, "carol"))
+    try:
+        carol.start(0)  # accepted with an empty CRL
+        carol.stop()
+        run("revoke", "--out", str(out), "--org", "org1", "--name", "dave")
+        gw2 = core.Gateway(str(out / "gw.chain.pem"), str(out / "gw.key"),
+                           str(out / "root.crt"), crl)
+        gw2.start(0)
+        try:
+            dave = core.Node("dave", "127.0.0.1", gw2.port, **_tls_args(out, "dave"))
+            with pytest.raises(RuntimeError, match="TLS|registration"):
+                dave.start(0)
+            dave.stop()
+            carol2 = core.Node("carol", "127.0.0.1", gw2.port,
+                               **_tls_args(out, "carol"))
+            carol2.start(0)  # unrevoked peer still accepted
+            carol2.stop()
+        finally:
+            gw2.stop()
+    finally:
+        gw.stop()
+
+
 def test_plaintext_client_rejected_by_tls_gateway(pki):
     out, _ = pki
     gw = core.Gateway(str(out / "gw.chain.pem"), str(out / "gw.key"),
