@@ -21,6 +21,7 @@ using namespace hypha;
 int main(int argc, char** argv) {
   std::string name = "data", gw_host = "127.0.0.1", dataset = "dataset", dir = ".";
   int gw_port = 0, port = 0;
+  bool probe = false, init = false;
   TlsConfig tls;
   for (int i = 1; i < argc; ++i) {
     std::string a = argv[i];
@@ -35,8 +36,29 @@ int main(int argc, char** argv) {
     else if (a == "--tls-key") tls.key_path = next();
     else if (a == "--tls-ca") tls.ca_path = next();
     else if (a == "--tls-crl") tls.crl_path = next();
+    else if (a == "probe") probe = true;
+    else if (a == "init") init = true;
   }
   signal(SIGPIPE, SIG_IGN);
+  if (init) {  // reference CLI Init subcommand: emit a commented config
+    printf("# hypha-data configuration (flags)\n"
+           "# --name data-0                 node name in the registry\n"
+           "# --gateway-host/--gateway-port gateway broker address\n"
+           "# --dataset synth               dataset name to announce\n"
+           "# --dataset-path /data/slices   dir of SafeTensors slice files\n"
+           "# --tls-cert/--tls-key/--tls-ca [--tls-crl]  mTLS identity\n");
+    return 0;
+  }
+  if (probe) {  // readiness = gateway reachable (hypha-data.rs probe analogue)
+    int fd = tcp_connect(gw_host, gw_port, 2.0);
+    if (fd < 0) {
+      fprintf(stderr, "probe: gateway unreachable\n");
+      return 1;
+    }
+    close(fd);
+    printf("probe: healthy\n");
+    return 0;
+  }
 
   std::vector<std::string> files;
   DIR* d = opendir(dir.c_str());

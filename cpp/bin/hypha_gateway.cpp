@@ -10,6 +10,7 @@
 
 int main(int argc, char** argv) {
   int port = 41000;
+  bool probe = false, init = false;
   hypha::TlsConfig tls;
   for (int i = 1; i < argc; ++i) {
     std::string a = argv[i];
@@ -18,8 +19,26 @@ int main(int argc, char** argv) {
     else if (a == "--tls-key") tls.key_path = argv[++i];
     else if (a == "--tls-ca") tls.ca_path = argv[++i];
     else if (a == "--tls-crl") tls.crl_path = argv[++i];
+    else if (a == "probe") probe = true;
+    else if (a == "init") init = true;
   }
   signal(SIGPIPE, SIG_IGN);
+  if (init) {  // reference CLI Init subcommand: emit a commented config
+    printf("# hypha-gateway configuration (flags)\n"
+           "# --port 41000        broker listen port\n"
+           "# --tls-cert/--tls-key/--tls-ca [--tls-crl]  mTLS identity\n");
+    return 0;
+  }
+  if (probe) {  // readiness: is a gateway listening on the port?
+    int fd = hypha::tcp_connect("127.0.0.1", port, 2.0);
+    if (fd < 0) {
+      fprintf(stderr, "probe: gateway unreachable on port %d\n", port);
+      return 1;
+    }
+    close(fd);
+    printf("probe: healthy\n");
+    return 0;
+  }
   hypha::Gateway gw(tls);
   gw.start(port);
   printf("hypha-gateway on port %d\n", gw.port());

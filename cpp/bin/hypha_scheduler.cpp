@@ -64,7 +64,7 @@ static void http_post_json(const std::string& bridge, const std::string& path,
 int main(int argc, char** argv) {
   std::string name = "scheduler", gw_host = "127.0.0.1", config_path, status_bridge;
   int gw_port = 0, port = 0;
-  bool probe = false;
+  bool probe = false, init = false;
   TlsConfig tls;
   for (int i = 1; i < argc; ++i) {
     std::string a = argv[i];
@@ -80,8 +80,32 @@ int main(int argc, char** argv) {
     else if (a == "--tls-ca") tls.ca_path = next();
     else if (a == "--tls-crl") tls.crl_path = next();
     else if (a == "probe") probe = true;
+    else if (a == "init") init = true;
   }
   signal(SIGPIPE, SIG_IGN);
+  if (init) {  // reference CLI Init subcommand: emit a commented job config
+    printf("# hypha-scheduler configuration\n"
+           "# --name scheduler --gateway-host H --gateway-port P\n"
+           "# --config job.json|job.toml    job spec (see below)\n"
+           "# --status-bridge 127.0.0.1:53800  AIM metrics forwarding\n"
+           "# --tls-cert/--tls-key/--tls-ca [--tls-crl]  mTLS identity\n#\n"
+           "# job config (scheduler_config.rs analogue); JSON or TOML:\n"
+           "{\n"
+           "  \"model\": \"llama3-8b\",\n"
+           "  \"dataset\": \"synth\",\n"
+           "  \"num_workers\": 2,\n"
+           "  \"update_rounds\": 100,\n"
+           "  \"avg_samples_between_updates\": 1200,\n"
+           "  \"batch_size\": 6,\n"
+           "  \"seq_len\": 2048,\n"
+           "  \"inner_lr\": 4e-4,\n"
+           "  \"outer_lr\": 0.7,\n"
+           "  \"outer_momentum\": 0.9,\n"
+           "  \"worker_price\": 1.0,\n"
+           "  \"checkpoint_every\": 0\n"
+           "}\n");
+    return 0;
+  }
   if (probe) {  // readiness check: gateway health RR (hypha-worker.rs probe)
     int fd = tcp_connect(gw_host, gw_port, 3.0);
     if (fd < 0) {

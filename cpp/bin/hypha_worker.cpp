@@ -706,7 +706,7 @@ struct WorkerDaemon {
 int main(int argc, char** argv) {
   std::string name = "worker", gw_host = "127.0.0.1", cmd, work_root = "/tmp/hypha-work";
   int gw_port = 0, port = 0;
-  bool probe = false;
+  bool probe = false, init = false;
   TlsConfig tls;
   Resources total{1, 4, 16, 100};
   OfferPolicy policy{1.0, 0.0, {"diloco-transformer", "parameter-server"}};
@@ -730,8 +730,21 @@ int main(int argc, char** argv) {
     else if (a == "--tls-ca") tls.ca_path = next();
     else if (a == "--tls-crl") tls.crl_path = next();
     else if (a == "probe") probe = true;
+    else if (a == "init") init = true;
   }
   signal(SIGPIPE, SIG_IGN);
+  if (init) {  // reference CLI Init subcommand: emit a commented config
+    printf("# hypha-worker configuration (flags)\n"
+           "# --name worker-0               node name in the registry\n"
+           "# --gateway-host/--gateway-port gateway broker address\n"
+           "# --gpu 1 --cpu 4 --memory 16 --storage 100   sellable resources\n"
+           "# --price 1.0 --floor 0.0       auction offer policy\n"
+           "# --exec-cmd 'python -m hypha_amd.runtime.executor --socket "
+           "{SOCKET_PATH} --work-dir {WORK_DIR} --job {JOB_JSON}'\n"
+           "# --work-root /tmp/hypha-work   per-job working directories\n"
+           "# --tls-cert/--tls-key/--tls-ca [--tls-crl]  mTLS identity\n");
+    return 0;
+  }
   if (probe) {
     int fd = tcp_connect(gw_host, gw_port, 3.0);
     if (fd < 0) {

@@ -128,3 +128,25 @@ def test_aim_driver_endpoint(tmp_path, monkeypatch):
     assert r.status_code == 200 and r.json() == {"ok": True}
     rec = pyjson.loads((tmp_path / "run.jsonl").read_text().splitlines()[0])
     assert rec["worker_id"] == "w0" and rec["value"] == 1.25
+
+
+def test_daemon_init_and_probe_subcommands():
+    """Reference CLI parity (scheduler/src/bin/hypha-scheduler.rs:459-548
+    Init/Probe/Run): every daemon emits a commented config on `init` and
+    fails `probe` cleanly when no gateway is listening."""
+    import subprocess
+    from pathlib import Path
+
+    bins = Path(__file__).resolve().parent.parent / "bin"
+    if not (bins / "hypha-gateway").exists():
+        pytest.skip("daemon binaries not built")
+    for d in ("hypha-gateway", "hypha-worker", "hypha-scheduler", "hypha-data"):
+        r = subprocess.run([str(bins / d), "init"], capture_output=True, text=True)
+        assert r.returncode == 0 and r.stdout.startswith("#"), d
+    for d, args in (("hypha-gateway", ["--port", "59997"]),
+                    ("hypha-worker", ["--gateway-port", "59997"]),
+                    ("hypha-scheduler", ["--gateway-port", "59997"]),
+                    ("hypha-data", ["--gateway-port", "59997"])):
+        r = subprocess.run([str(bins / d), "probe", *args],
+                           capture_output=True, text=True, timeout=30)
+        assert r.returncode == 1 and "unreachable" in r.stderr, d
