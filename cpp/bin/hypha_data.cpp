@@ -22,6 +22,7 @@ int main(int argc, char** argv) {
   std::string name = "data", gw_host = "127.0.0.1", dataset = "dataset", dir = ".";
   int gw_port = 0, port = 0;
   bool probe = false, init = false;
+  std::vector<std::string> exclude_cidrs;
   TlsConfig tls;
   for (int i = 1; i < argc; ++i) {
     std::string a = argv[i];
@@ -36,6 +37,7 @@ int main(int argc, char** argv) {
     else if (a == "--tls-key") tls.key_path = next();
     else if (a == "--tls-ca") tls.ca_path = next();
     else if (a == "--tls-crl") tls.crl_path = next();
+    else if (a == "--exclude-cidr") exclude_cidrs.push_back(next());
     else if (a == "probe") probe = true;
     else if (a == "init") init = true;
   }

@@ -65,6 +65,7 @@ int main(int argc, char** argv) {
   std::string name = "scheduler", gw_host = "127.0.0.1", config_path, status_bridge;
   int gw_port = 0, port = 0;
   bool probe = false, init = false;
+  std::vector<std::string> exclude_cidrs;
   TlsConfig tls;
   for (int i = 1; i < argc; ++i) {
     std::string a = argv[i];
@@ -79,6 +80,7 @@ int main(int argc, char** argv) {
     else if (a == "--tls-key") tls.key_path = next();
     else if (a == "--tls-ca") tls.ca_path = next();
     else if (a == "--tls-crl") tls.crl_path = next();
+    else if (a == "--exclude-cidr") exclude_cidrs.push_back(next());
     else if (a == "probe") probe = true;
     else if (a == "init") init = true;
   }
@@ -142,6 +144,7 @@ int main(int argc, char** argv) {
   const double worker_max = cfg.get_or("worker_max_price", Json(10.0)).as_double();
 
   Node node(name, gw_host, gw_port, tls);
+  node.set_exclude_cidrs(exclude_cidrs);
 
   // --- offer collection state (allocator.rs) ---
   std::mutex offer_mu;

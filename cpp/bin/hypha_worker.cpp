@@ -707,6 +707,7 @@ int main(int argc, char** argv) {
   std::string name = "worker", gw_host = "127.0.0.1", cmd, work_root = "/tmp/hypha-work";
   int gw_port = 0, port = 0;
   bool probe = false, init = false;
+  std::vector<std::string> exclude_cidrs;
   TlsConfig tls;
   Resources total{1, 4, 16, 100};
   OfferPolicy policy{1.0, 0.0, {"diloco-transformer", "parameter-server"}};
@@ -729,6 +730,7 @@ int main(int argc, char** argv) {
     else if (a == "--tls-key") tls.key_path = next();
     else if (a == "--tls-ca") tls.ca_path = next();
     else if (a == "--tls-crl") tls.crl_path = next();
+    else if (a == "--exclude-cidr") exclude_cidrs.push_back(next());
     else if (a == "probe") probe = true;
     else if (a == "init") init = true;
   }
@@ -756,6 +758,7 @@ int main(int argc, char** argv) {
     return 0;
   }
   WorkerDaemon daemon(name, gw_host, gw_port, total, policy, cmd, work_root, tls);
+  daemon.node.set_exclude_cidrs(exclude_cidrs);
   daemon.start(port);
   printf("hypha-worker %s ready on port %d\n", name.c_str(), daemon.node.port());
   fflush(stdout);

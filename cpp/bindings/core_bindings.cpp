@@ -67,6 +67,7 @@ PYBIND11_MODULE(_core, m) {
 
   // ---- json round-trip (used by tests to validate the wire format) ----
   m.def("json_roundtrip", [](const std::string& s) { return Json::parse(s).dump(); });
+  m.def("cidr_contains", &cidr_contains, py::arg("cidr"), py::arg("host"));
 
   // ---- resources ----
   py::class_<Resources>(m, "Resources")
@@ -387,6 +388,7 @@ PYBIND11_MODULE(_core, m) {
              if (!stream->send_raw(data.data(), data.size()))
                throw std::runtime_error("stream write failed");
            })
+      .def("set_exclude_cidrs", &Node::set_exclude_cidrs, py::arg("cidrs"))
       .def("stream_call",
            // bidirectional stream RPC: send header (+payload), read a JSON
            // reply and an optional `size`-byte body. Client side of the

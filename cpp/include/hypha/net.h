@@ -58,6 +58,11 @@ int tcp_connect(const std::string& host, int port, double timeout_s = 5.0);
 int tcp_listen(int port);  // returns listen fd; port 0 = ephemeral
 int listen_port(int listen_fd);
 
+// "10.0.0.0/8" contains "10.1.2.3"? (reference utils.rs:18
+// find_containing_cidr + dial.rs CIDR exclusion). IPv4 dotted-quad only;
+// non-numeric hosts are never matched.
+bool cidr_contains(const std::string& cidr, const std::string& host);
+
 // ---- node ------------------------------------------------------------------
 
 // A Node serves typed requests on its own TCP port and talks to peers either
@@ -99,7 +104,16 @@ class Node {
 
   std::optional<std::string> resolve(const std::string& peer);
 
+  // Dial exclusion (dial.rs CIDR exclusion): peers whose resolved address
+  // falls in an excluded CIDR are refused before connect.
+  void set_exclude_cidrs(std::vector<std::string> cidrs) {
+    exclude_cidrs_ = std::move(cidrs);
+  }
+
  private:
+  void check_dialable(const std::string& peer, const std::string& host) const;
+  std::vector<std::string> exclude_cidrs_;
+
   void accept_loop();
   void handle_conn(int fd, SSL* ssl = nullptr);
   void gateway_listen_loop();

@@ -382,11 +382,32 @@ Json Node::gateway_request(const std::string& type, const Json& body) {
   return resp->get_or("body", Json(JsonObject{}));
 }
 
+bool cidr_contains(const std::string& cidr, const std::string& host) {
+  size_t slash = cidr.find('/');
+  if (slash == std::string::npos) return false;
+  struct in_addr net_a, host_a;
+  if (inet_pton(AF_INET, cidr.substr(0, slash).c_str(), &net_a) != 1) return false;
+  if (inet_pton(AF_INET, host.c_str(), &host_a) != 1) return false;
+  int bits = std::stoi(cidr.substr(slash + 1));
+  if (bits <= 0) return true;
+  if (bits > 32) return false;
+  uint32_t mask = bits == 32 ? 0xFFFFFFFFu : ~(0xFFFFFFFFu >> bits);
+  return (ntohl(net_a.s_addr) & mask) == (ntohl(host_a.s_addr) & mask);
+}
+
+void Node::check_dialable(const std::string& peer, const std::string& host) const {
+  for (const auto& c : exclude_cidrs_)
+    if (cidr_contains(c, host))
+      throw std::runtime_error("dial refused: peer " + peer + " at " + host +
+                               " is in excluded CIDR " + c);
+}
+
 Json Node::request(const std::string& peer, const std::string& type, const Json& body,
                    double timeout_s) {
   auto addr = resolve(peer);
   if (!addr) throw std::runtime_error("unknown peer " + peer);
   size_t colon = addr->rfind(':');
+  check_dialable(peer, addr->substr(0, colon));
   int fd = tcp_connect(addr->substr(0, colon), std::stoi(addr->substr(colon + 1)), timeout_s);
   if (fd < 0) throw std::runtime_error("peer unreachable: " + peer);
   SSL* pssl = nullptr;
@@ -417,6 +438,7 @@ std::unique_ptr<MsgSocket> Node::open_stream(const std::string& peer, const std:
   auto addr = resolve(peer);
   if (!addr) throw std::runtime_error("unknown peer " + peer);
   size_t colon = addr->rfind(':');
+  check_dialable(peer, addr->substr(0, colon));
   int fd = tcp_connect(addr->substr(0, colon), std::stoi(addr->substr(colon + 1)), 30.0);
   if (fd < 0) throw std::runtime_error("peer unreachable: " + peer);
   struct timeval tv = {600, 0};

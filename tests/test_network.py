@@ -118,3 +118,30 @@ def test_publisher_not_echoed(cluster):
     a.publish("t", {"v": 1})
     time.sleep(0.1)
     assert got == []  # publisher doesn't receive its own message
+
+
+def test_cidr_helper():
+    """find_containing_cidr parity (reference network utils.rs:18)."""
+    assert core.cidr_contains("10.0.0.0/8", "10.1.2.3")
+    assert not core.cidr_contains("10.0.0.0/8", "11.0.0.1")
+    assert core.cidr_contains("127.0.0.0/8", "127.0.0.1")
+    assert core.cidr_contains("192.168.1.0/24", "192.168.1.200")
+    assert not core.cidr_contains("192.168.1.0/24", "192.168.2.1")
+    assert core.cidr_contains("1.2.3.4/32", "1.2.3.4")
+    assert not core.cidr_contains("1.2.3.4/32", "1.2.3.5")
+    assert core.cidr_contains("0.0.0.0/0", "8.8.8.8")
+    assert not core.cidr_contains("10.0.0.0/8", "not-an-ip")
+
+
+def test_dial_exclusion_refuses_excluded_peer(cluster):
+    """Dial-time CIDR exclusion (reference dial.rs): a peer whose resolved
+    address lands in an excluded CIDR is refused before connect."""
+    a = cluster("exa")
+    b = cluster("exb")
+    b.on("echo", lambda frm, body: {"ok": True})
+    assert a.request("exb", "echo", {}) == {"ok": True}
+    a.set_exclude_cidrs(["127.0.0.0/8"])
+    with pytest.raises(RuntimeError, match="excluded CIDR"):
+        a.request("exb", "echo", {})
+    a.set_exclude_cidrs([])
+    assert a.request("exb", "echo", {}) == {"ok": True}
