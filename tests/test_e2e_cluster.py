@@ -139,9 +139,11 @@ def test_cluster_worker_kill_and_rejoin(binaries, tmp_path):
                  "--exec-cmd", exec_cmd, "--work-root", str(tmp_path / f"work{i}")])
         time.sleep(0.5)
         cfg = tmp_path / "job.json"
+        # enough rounds that the 10 s lease TTL always expires (and the
+        # replacement joins) well before training can complete on its own
         cfg.write_text(
             '{"model": "llama-tiny", "dataset": "synth", "num_workers": 2,'
-            ' "update_rounds": 6, "avg_samples_between_updates": 8,'
+            ' "update_rounds": 12, "avg_samples_between_updates": 8,'
             ' "batch_size": 2, "seq_len": 128, "inner_lr": 0.001}'
         )
         sched = subprocess.Popen(
