@@ -68,6 +68,13 @@ PYBIND11_MODULE(_core, m) {
   // ---- json round-trip (used by tests to validate the wire format) ----
   m.def("json_roundtrip", [](const std::string& s) { return Json::parse(s).dump(); });
   m.def("cidr_contains", &cidr_contains, py::arg("cidr"), py::arg("host"));
+  m.def("bandwidth_stats", [] {
+    auto s = bandwidth_stats();
+    py::dict d;
+    d["inbound_bytes"] = s.inbound_bytes;
+    d["outbound_bytes"] = s.outbound_bytes;
+    return d;
+  });
 
   // ---- resources ----
   py::class_<Resources>(m, "Resources")

@@ -58,6 +58,16 @@ int tcp_connect(const std::string& host, int port, double timeout_s = 5.0);
 int tcp_listen(int port);  // returns listen fd; port 0 = ephemeral
 int listen_port(int listen_fd);
 
+// Process-global transport byte counters (reference telemetry/src/
+// bandwidth.rs:33-60: bandwidth::Transport wraps every connection with
+// inbound/outbound counters). Counted at the MsgSocket framing layer, so
+// requests, pub/sub, KV and tensor streams are all included.
+struct BandwidthStats {
+  unsigned long long inbound_bytes;
+  unsigned long long outbound_bytes;
+};
+BandwidthStats bandwidth_stats();
+
 // "10.0.0.0/8" contains "10.1.2.3"? (reference utils.rs:18
 // find_containing_cidr + dial.rs CIDR exclusion). IPv4 dotted-quad only;
 // non-numeric hosts are never matched.

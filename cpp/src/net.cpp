@@ -10,6 +10,7 @@
 #include <sys/types.h>
 #include <unistd.h>
 
+#include <atomic>
 #include <condition_variable>
 #include <stdexcept>
 
@@ -79,6 +80,15 @@ static bool read_all_ssl(SSL* ssl, char* p, size_t n) {
   return true;
 }
 
+// Transport byte accounting (reference telemetry/src/bandwidth.rs:33-60:
+// every connection counts inbound/outbound bytes). Process-global, lock-free.
+static std::atomic<unsigned long long> g_bytes_in{0}, g_bytes_out{0};
+
+BandwidthStats bandwidth_stats() {
+  return {g_bytes_in.load(std::memory_order_relaxed),
+          g_bytes_out.load(std::memory_order_relaxed)};
+}
+
 bool MsgSocket::ssl_wait_readable() {
   while (true) {
     {
@@ -97,15 +107,19 @@ bool MsgSocket::ssl_wait_readable() {
 bool MsgSocket::send_json(const Json& j) {
   std::string s = j.dump();
   uint32_t len = htonl((uint32_t)s.size());
+  bool ok;
   if (ssl_) {
     std::lock_guard<std::mutex> lk(io_mu_);
     if (!ssl_) return false;
-    if (!write_all_ssl(ssl_, (const char*)&len, 4)) return false;
-    return write_all_ssl(ssl_, s.data(), s.size());
+    ok = write_all_ssl(ssl_, (const char*)&len, 4) &&
+         write_all_ssl(ssl_, s.data(), s.size());
+  } else {
+    std::lock_guard<std::mutex> lk(write_mu_);
+    ok = write_all_fd(fd_, (const char*)&len, 4) &&
+         write_all_fd(fd_, s.data(), s.size());
   }
-  std::lock_guard<std::mutex> lk(write_mu_);
-  if (!write_all_fd(fd_, (const char*)&len, 4)) return false;
-  return write_all_fd(fd_, s.data(), s.size());
+  if (ok) g_bytes_out.fetch_add(4 + s.size(), std::memory_order_relaxed);
+  return ok;
 }
 
 std::optional<Json> MsgSocket::recv_json() {
@@ -127,6 +141,7 @@ std::optional<Json> MsgSocket::recv_json() {
     s.assign(len, '\0');
     if (!read_all_fd(fd_, s.data(), len)) return std::nullopt;
   }
+  g_bytes_in.fetch_add(4 + s.size(), std::memory_order_relaxed);
   try {
     return Json::parse(s);
   } catch (...) {
@@ -135,21 +150,29 @@ std::optional<Json> MsgSocket::recv_json() {
 }
 
 bool MsgSocket::send_raw(const char* data, size_t n) {
+  bool ok;
   if (ssl_) {
     std::lock_guard<std::mutex> lk(io_mu_);
     if (!ssl_) return false;
-    return write_all_ssl(ssl_, data, n);
+    ok = write_all_ssl(ssl_, data, n);
+  } else {
+    ok = write_all_fd(fd_, data, n);
   }
-  return write_all_fd(fd_, data, n);
+  if (ok) g_bytes_out.fetch_add(n, std::memory_order_relaxed);
+  return ok;
 }
 bool MsgSocket::recv_raw(char* data, size_t n) {
+  bool ok;
   if (ssl_) {
     if (!ssl_wait_readable()) return false;
     std::lock_guard<std::mutex> lk(io_mu_);
     if (!ssl_) return false;
-    return read_all_ssl(ssl_, data, n);
+    ok = read_all_ssl(ssl_, data, n);
+  } else {
+    ok = read_all_fd(fd_, data, n);
   }
-  return read_all_fd(fd_, data, n);
+  if (ok) g_bytes_in.fetch_add(n, std::memory_order_relaxed);
+  return ok;
 }
 
 int tcp_connect(const std::string& host, int port, double timeout_s) {

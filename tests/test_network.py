@@ -145,3 +145,19 @@ def test_dial_exclusion_refuses_excluded_peer(cluster):
         a.request("exb", "echo", {})
     a.set_exclude_cidrs([])
     assert a.request("exb", "echo", {}) == {"ok": True}
+
+
+def test_bandwidth_accounting(cluster):
+    """Transport byte counters (reference telemetry bandwidth.rs:33-60):
+    every framed send/receive is counted process-wide."""
+    before = core.bandwidth_stats()
+    a = cluster("bwa")
+    b = cluster("bwb")
+    b.on("echo", lambda frm, body: {"pong": body["ping"]})
+    payload = "x" * 10_000
+    assert a.request("bwb", "echo", {"ping": payload})["pong"] == payload
+    after = core.bandwidth_stats()
+    # both request and response cross the counter (>= 2x payload, in and out
+    # both grow because client and server share the process)
+    assert after["outbound_bytes"] - before["outbound_bytes"] > 2 * len(payload)
+    assert after["inbound_bytes"] - before["inbound_bytes"] > 2 * len(payload)
