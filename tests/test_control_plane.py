@@ -318,3 +318,101 @@ def test_json_parser_rejects_garbage():
     for bad in ("", "{", "[1,", '{"a"}', "tru", "nul", '"\\u12', "1e999x", "{}{}"):
         with pytest.raises(Exception):
             core.json_roundtrip(bad)
+
+
+def test_resources_partial_order_properties():
+    """Property test (reference worker dev-deps use proptest; resources
+    lib.rs:123-143 partial order): reflexive, antisymmetric, consistent
+    with fits_in, and incomparable when dimensions disagree in sign."""
+    from hypothesis import given, settings, strategies as st
+
+    dim = st.floats(min_value=0, max_value=1e6, allow_nan=False)
+    quad = st.tuples(dim, dim, dim, dim)
+
+    @settings(max_examples=200, deadline=None)
+    @given(quad, quad)
+    def check(a4, b4):
+        a = core.Resources(*a4)
+        b = core.Resources(*b4)
+        assert a.partial_cmp(a) == 0  # reflexive
+        c_ab = a.partial_cmp(b)
+        c_ba = b.partial_cmp(a)
+        if c_ab is None:
+            assert c_ba is None  # incomparability is symmetric
+        else:
+            assert c_ba == -c_ab  # antisymmetry
+        # fits_in consistency: a fits in b iff a <= b on every dimension
+        fits = all(x <= y for x, y in zip(a4, b4))
+        assert a.fits_in(b) == fits
+        if fits:
+            assert c_ab in (-1, 0)
+        # arithmetic round-trip: (a + b) - b == a (within float error)
+        rt = (a + b) - b
+        for name in ("gpu", "cpu", "memory", "storage"):
+            assert abs(getattr(rt, name) - getattr(a, name)) <= 1e-6 * max(
+                1.0, getattr(a, name))
+
+    check()
+
+
+def test_ledger_properties():
+    """Property test: leases expire exactly by the injected clock; renew
+    extends from NOW (leases lib.rs:103-114), never shortens relative to
+    the new timeout; drain removes precisely the expired set."""
+    from hypothesis import given, settings, strategies as st
+
+    @settings(max_examples=100, deadline=None)
+    @given(st.lists(st.tuples(st.floats(0.1, 50.0), st.floats(0.0, 100.0)),
+                    min_size=1, max_size=20),
+           st.floats(0.0, 100.0))
+    def check(leases, probe_time):
+        now = [0.0]
+        led = core.Ledger(clock=lambda: now[0])
+        ids = []
+        for i, (ttl, _) in enumerate(leases):
+            led.insert(f"lease{i}", f"v{i}", ttl)
+            ids.append(f"lease{i}")
+        now[0] = probe_time
+        expired = set(led.list_expired())
+        for lease_id, (ttl, _) in zip(ids, leases):
+            assert (lease_id in expired) == (probe_time > ttl)
+        drained = set(led.drain_expired())
+        assert drained == expired
+        assert led.size() == len(ids) - len(expired)
+        # renew survivors: timeout becomes now + d, so nothing is expired
+        # at any probe <= now + min_renewal
+        for lease_id in led.ids():
+            assert led.renew(lease_id, 5.0)
+        now[0] = probe_time + 4.999
+        assert led.list_expired() == []
+
+    check()
+
+
+def test_slice_tracker_properties():
+    """Property test: every slice index handed out is valid; each epoch
+    serves every slice at least once across workers; removing a worker
+    never loses slices (slice.rs:92-114 semantics)."""
+    from hypothesis import given, settings, strategies as st
+
+    @settings(max_examples=50, deadline=None)
+    @given(st.integers(1, 12), st.integers(1, 4), st.integers(5, 60))
+    def check(num_slices, num_workers, pulls):
+        t = core.SliceTracker("ds", num_slices)
+        served = {}
+        for i in range(pulls):
+            peer = f"w{i % num_workers}"
+            idx, epoch = t.next(peer)
+            assert 0 <= idx < num_slices
+            served.setdefault(epoch, set()).add(idx)
+        # every completed epoch covered all slices
+        max_epoch = max(served)
+        for e, s in served.items():
+            if e < max_epoch:
+                assert s == set(range(num_slices))
+        t.remove_worker("w0")
+        # after reclaim the tracker still serves valid indices
+        idx, _ = t.next("w1")
+        assert 0 <= idx < num_slices
+
+    check()
