@@ -62,3 +62,37 @@ def test_two_process_gloo_diloco():
     assert results[0][2] == pytest.approx(results[1][2], rel=1e-6)
     assert results[0][3] == pytest.approx(results[1][3], rel=1e-6)
     assert results[0][5] == 2  # two outer rounds completed
+
+
+@pytest.mark.timeout(300)
+def test_bench_contract_world2_gloo(tmp_path):
+    """The driver's scale run (torch.distributed.run, one rank per GPU)
+    exercised end-to-end on CPU/gloo: rendezvous on 127.0.0.1, bucketed
+    all-reduce outer sync inside the timed window, max-over-ranks timing,
+    and exactly ONE JSON line from rank 0 honoring the bench contract."""
+    import json
+    import subprocess
+    import sys
+    from pathlib import Path
+
+    repo = Path(__file__).resolve().parent.parent
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29613", str(repo / "bench.py"), "--gpus", "2",
+         "--steps", "4", "--warmup", "1", "--model", "llama-tiny",
+         "--batch", "2", "--seq-len", "128", "--h", "3"],
+        cwd=repo, capture_output=True, text=True, timeout=280,
+    )
+    assert r.returncode == 0, r.stderr[-2000:]
+    json_lines = [l for l in r.stdout.splitlines() if l.startswith("{")]
+    assert len(json_lines) == 1, r.stdout  # rank 0 only
+    out = json.loads(json_lines[0])
+    assert out["n_gpus"] == 2 and out["steps"] == 4
+    assert out["value"] > 0 and out["higher_is_better"] is True
+    assert out["scaling"] == "weak"
+    cfg = out["config"]
+    assert cfg["parallelism"] == "diloco-dp2"
+    assert cfg["global_batch"] == 4  # whole-job aggregate
+    assert cfg["outer_syncs_in_timed_window"] == 1  # h=3 < steps+warmup
+    assert cfg["outer_sync_wire_bytes_per_rank"] > 0  # ring wire accounting
