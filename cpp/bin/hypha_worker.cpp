@@ -493,12 +493,6 @@ struct WorkerDaemon {
     if (!f) throw std::runtime_error("send: no such file " + path);
     fseek(f, 0, SEEK_END);
     long size = ftell(f);
-    fseek(f, 0, SEEK_SET);
-    std::vector<char> data(size);
-    if (size > 0 && fread(data.data(), 1, size, f) != (size_t)size) {
-      fclose(f);
-      throw std::runtime_error("send: short file read");
-    }
     fclose(f);
     std::string base = path.substr(path.rfind('/') + 1);
     // SelectionStrategy (connector/mod.rs:330-380): "all" -> every peer;
@@ -520,14 +514,28 @@ struct WorkerDaemon {
         throw std::runtime_error("send: unknown strategy " + strategy);
       }
     }
+    // stream in 1 MB chunks: pseudo-gradient files are model-sized (16 GB
+    // for 8B) — never buffer the whole payload in host memory
+    std::vector<char> buf(1 << 20);
     for (auto& peer : targets) {
       Json hdr;
       hdr["name"] = base;
       hdr["size"] = (int64_t)size;
       hdr["job"] = ref.get_or("job", Json(job->id));
       auto stream = node.open_stream(peer.as_string(), "push_resource", hdr);
-      if (!stream->send_raw(data.data(), size))
-        throw std::runtime_error("send: stream write failed to " + peer.as_string());
+      FILE* fp = fopen(path.c_str(), "rb");
+      if (!fp) throw std::runtime_error("send: cannot reopen " + path);
+      long left = size;
+      while (left > 0) {
+        size_t chunk = std::min((long)buf.size(), left);
+        if (fread(buf.data(), 1, chunk, fp) != chunk ||
+            !stream->send_raw(buf.data(), chunk)) {
+          fclose(fp);
+          throw std::runtime_error("send: stream write failed to " + peer.as_string());
+        }
+        left -= chunk;
+      }
+      fclose(fp);
     }
     Json r;
     r["sent"] = true;
@@ -555,9 +563,22 @@ struct WorkerDaemon {
     size_t size = (size_t)header.at("size").as_int();
     std::string dir = job ? job->work_dir : work_root + "/orphan";
     mkdir(dir.c_str(), 0755);
-    std::string out =
-        dir + "/recv-" + from + "-" + header.at("name").as_string();
+    // peer-supplied names must stay inside the work dir (the safe_join rule
+    // the bridge enforces, bridge.rs:330-346): keep only the basename and
+    // drop dot-leading components
+    auto sanitize = [](std::string s) {
+      size_t slash = s.rfind('/');
+      if (slash != std::string::npos) s = s.substr(slash + 1);
+      if (s.empty() || s[0] == '.') s = "unnamed";
+      return s;
+    };
+    std::string out = dir + "/recv-" + sanitize(from) + "-" +
+                      sanitize(header.at("name").as_string());
     FILE* f = fopen(out.c_str(), "wb");
+    if (!f) {
+      fprintf(stderr, "[%s] recv: cannot open %s\n", name.c_str(), out.c_str());
+      return;
+    }
     std::vector<char> buf(1 << 20);
     size_t left = size;
     while (left > 0) {

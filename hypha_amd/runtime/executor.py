@@ -75,8 +75,9 @@ def main() -> int:
         try:
             for ev in session.receive():
                 updates_q.put(ev)
-        except Exception:
-            pass
+        except Exception as e:  # bridge socket closes at job teardown;
+            # anything else is worth a trace on stderr (thread would die silently)
+            print(f"# sse listener ended: {type(e).__name__}: {e}", file=sys.stderr)
 
     threading.Thread(target=listen, daemon=True).start()
 
