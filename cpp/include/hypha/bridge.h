@@ -128,16 +128,27 @@ class Bridge {
     }
     // request line
     size_t sp1 = head.find(' ');
-    size_t sp2 = head.find(' ', sp1 + 1);
+    size_t sp2 = sp1 == std::string::npos ? std::string::npos : head.find(' ', sp1 + 1);
+    if (sp2 == std::string::npos) {
+      respond(fd, 400, "{\"error\":\"malformed request line\"}");
+      ::close(fd);
+      return;
+    }
     std::string method = head.substr(0, sp1);
     std::string path = head.substr(sp1 + 1, sp2 - sp1 - 1);
-    // content-length
+    // content-length (capped: job configs / status posts are small; file
+    // payloads travel over the peer streams, never through the bridge body)
     size_t clen = 0;
     {
       std::string lower;
       for (char ch : head) lower += (char)tolower((unsigned char)ch);
       size_t p = lower.find("content-length:");
       if (p != std::string::npos) clen = strtoul(lower.c_str() + p + 15, nullptr, 10);
+    }
+    if (clen > (256u << 20)) {
+      respond(fd, 400, "{\"error\":\"body too large\"}");
+      ::close(fd);
+      return;
     }
     std::string body(clen, '\0');
     size_t got = 0;
