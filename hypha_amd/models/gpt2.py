@@ -37,14 +37,22 @@ PRESETS = {
 }
 
 
+class LayerNorm(nn.LayerNorm):
+    """nn.LayerNorm parameters, dispatched to the native HIP kernel on GPU
+    (ops/hip/layernorm.hip) and torch on CPU."""
+
+    def forward(self, x):
+        return ops.layernorm(x, self.weight, self.bias, self.eps)
+
+
 class GPT2Block(nn.Module):
     def __init__(self, cfg: GPT2Config):
         super().__init__()
         h = cfg.hidden_size
-        self.ln1 = nn.LayerNorm(h, eps=cfg.norm_eps)
+        self.ln1 = LayerNorm(h, eps=cfg.norm_eps)
         self.attn_qkv = nn.Linear(h, 3 * h)
         self.attn_out = nn.Linear(h, h)
-        self.ln2 = nn.LayerNorm(h, eps=cfg.norm_eps)
+        self.ln2 = LayerNorm(h, eps=cfg.norm_eps)
         self.mlp_fc = nn.Linear(h, 4 * h)
         self.mlp_proj = nn.Linear(4 * h, h)
         self.n_heads = cfg.n_heads
@@ -58,7 +66,7 @@ class GPT2Block(nn.Module):
         v = v.view(b, s, self.n_heads, hd).transpose(1, 2)
         o = ops.flash_attention(q, k, v, causal=True)
         x = x + self.attn_out(o.transpose(1, 2).reshape(b, s, h))
-        x = x + self.mlp_proj(F.gelu(self.mlp_fc(self.ln2(x)), approximate="tanh"))
+        x = x + self.mlp_proj(ops.gelu(self.mlp_fc(self.ln2(x))))
         return x
 
 
@@ -69,7 +77,7 @@ class GPT2ForCausalLM(nn.Module):
         self.wte = nn.Embedding(cfg.vocab_size, cfg.hidden_size)
         self.wpe = nn.Embedding(cfg.n_positions, cfg.hidden_size)
         self.blocks = nn.ModuleList(GPT2Block(cfg) for _ in range(cfg.n_layers))
-        self.ln_f = nn.LayerNorm(cfg.hidden_size, eps=cfg.norm_eps)
+        self.ln_f = LayerNorm(cfg.hidden_size, eps=cfg.norm_eps)
         self.apply(self._init)
         for blk in self.blocks:
             for lin in (blk.attn_out, blk.mlp_proj):

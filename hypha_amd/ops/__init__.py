@@ -58,6 +58,8 @@ from .interface import (  # noqa: E402
     flash_attention,
     fused_adamw,
     fused_nesterov,
+    gelu,
+    layernorm,
     rmsnorm,
     swiglu,
 )
@@ -68,6 +70,8 @@ __all__ = [
     "use_native",
     "reference",
     "rmsnorm",
+    "layernorm",
+    "gelu",
     "apply_rope_qk",
     "swiglu",
     "flash_attention",

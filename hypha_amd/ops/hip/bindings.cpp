@@ -31,6 +31,13 @@ void nesterov_bf16_(torch::Tensor theta, torch::Tensor delta, torch::Tensor mom,
 std::vector<torch::Tensor> swiglu_bwd(torch::Tensor dout, torch::Tensor gate,
                                       torch::Tensor up);
 // rmsnorm.hip
+std::vector<torch::Tensor> layernorm_fwd(torch::Tensor x, torch::Tensor w,
+                                         torch::Tensor b, double eps);
+std::vector<torch::Tensor> layernorm_bwd(torch::Tensor dy, torch::Tensor x,
+                                         torch::Tensor w, torch::Tensor mean,
+                                         torch::Tensor rstd);
+torch::Tensor gelu_fwd(torch::Tensor x);
+torch::Tensor gelu_bwd(torch::Tensor dy, torch::Tensor x);
 std::vector<torch::Tensor> rmsnorm_fwd(torch::Tensor x, torch::Tensor w, double eps);
 std::vector<torch::Tensor> rmsnorm_bwd(torch::Tensor dy, torch::Tensor x, torch::Tensor w,
                                        torch::Tensor rstd);
@@ -74,6 +81,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("swiglu_bwd", &swiglu_bwd);
   m.def("rmsnorm_fwd", &rmsnorm_fwd);
   m.def("rmsnorm_bwd", &rmsnorm_bwd);
+  m.def("layernorm_fwd", &layernorm_fwd);
+  m.def("layernorm_bwd", &layernorm_bwd);
+  m.def("gelu_fwd", &gelu_fwd);
+  m.def("gelu_bwd", &gelu_bwd);
   m.def("rope_fwd", &rope_fwd);
   m.def("rope_fwd_ex", &rope_fwd_ex);
   m.def("ce_fwd", &ce_fwd);

@@ -93,6 +93,56 @@ def apply_rope_qk(q, k, cos, sin, layout: str = "bhsd"):
 
 
 # ---------------------------------------------------------------------------
+# LayerNorm + tanh-GELU (the GPT-2 block's norm/activation)
+# ---------------------------------------------------------------------------
+
+
+class _LayerNorm(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, weight, bias, eps):
+        shape = x.shape
+        x2d = x.reshape(-1, shape[-1]).contiguous()
+        y, mean, rstd = _c().layernorm_fwd(x2d, weight, bias, eps)
+        ctx.save_for_backward(x2d, weight, mean, rstd)
+        ctx.shape = shape
+        return y.view(shape)
+
+    @staticmethod
+    def backward(ctx, dy):
+        x2d, weight, mean, rstd = ctx.saved_tensors
+        dx, dw, db = _c().layernorm_bwd(dy.reshape(x2d.shape).contiguous(), x2d,
+                                        weight, mean, rstd)
+        return dx.view(ctx.shape), dw.to(weight.dtype), db.to(weight.dtype), None
+
+
+def layernorm(x: torch.Tensor, weight: torch.Tensor, bias: torch.Tensor,
+              eps: float = 1e-5) -> torch.Tensor:
+    if use_native(x):
+        return _LayerNorm.apply(x, weight, bias, eps)
+    return torch.nn.functional.layer_norm(x, (x.shape[-1],), weight, bias, eps)
+
+
+class _Gelu(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x):
+        x = x.contiguous()
+        ctx.save_for_backward(x)
+        return _c().gelu_fwd(x)
+
+    @staticmethod
+    def backward(ctx, dy):
+        (x,) = ctx.saved_tensors
+        return _c().gelu_bwd(dy.contiguous(), x)
+
+
+def gelu(x: torch.Tensor) -> torch.Tensor:
+    """tanh-approximation GELU (GPT-2's activation)."""
+    if use_native(x):
+        return _Gelu.apply(x)
+    return torch.nn.functional.gelu(x, approximate="tanh")
+
+
+# ---------------------------------------------------------------------------
 # SwiGLU
 # ---------------------------------------------------------------------------
 

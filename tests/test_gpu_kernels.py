@@ -128,6 +128,64 @@ def test_rmsnorm_fwd_bwd():
 
 
 # ---------------------------------------------------------------------------
+# LayerNorm + GELU (GPT-2 block ops, ops/hip/layernorm.hip)
+# ---------------------------------------------------------------------------
+
+def test_layernorm_fwd_bwd():
+    import torch.nn.functional as F
+
+    x = rand_bf16(128, 512, seed=18).requires_grad_(True)
+    w = (torch.randn(512) * 0.1 + 1.0).bfloat16().to(DEV).requires_grad_(True)
+    b = (torch.randn(512) * 0.1).bfloat16().to(DEV).requires_grad_(True)
+    y = ops.layernorm(x, w, b, 1e-5)
+    want = F.layer_norm(x.detach().float(), (512,), w.detach().float(),
+                        b.detach().float(), 1e-5)
+    torch.testing.assert_close(y.float(), want, rtol=2e-2, atol=2e-2)
+    dy = rand_bf16(128, 512, seed=19)
+    y.backward(dy)
+    xf = x.detach().float().requires_grad_(True)
+    wf = w.detach().float().requires_grad_(True)
+    bf = b.detach().float().requires_grad_(True)
+    F.layer_norm(xf, (512,), wf, bf, 1e-5).backward(dy.float())
+    torch.testing.assert_close(x.grad.float(), xf.grad, rtol=3e-2, atol=3e-2)
+    torch.testing.assert_close(w.grad.float(), wf.grad, rtol=3e-2, atol=3e-1)
+    torch.testing.assert_close(b.grad.float(), bf.grad, rtol=3e-2, atol=3e-1)
+
+
+def test_gelu_fwd_bwd():
+    import torch.nn.functional as F
+
+    x = rand_bf16(1000, 256, seed=20).requires_grad_(True)
+    y = ops.gelu(x)
+    want = F.gelu(x.detach().float(), approximate="tanh")
+    torch.testing.assert_close(y.float(), want, rtol=2e-2, atol=2e-2)
+    dy = rand_bf16(1000, 256, seed=21)
+    y.backward(dy)
+    xf = x.detach().float().requires_grad_(True)
+    F.gelu(xf, approximate="tanh").backward(dy.float())
+    torch.testing.assert_close(x.grad.float(), xf.grad, rtol=3e-2, atol=3e-2)
+
+
+def test_gpt2_block_native_matches_cpu():
+    """The GPT-2 model forward+backward on GPU (native LayerNorm/GELU/attention)
+    vs the same weights on CPU (torch reference path)."""
+    from hypha_amd import models
+
+    torch.manual_seed(5)
+    # gpt2-small: head_dim 64 and seq >= 128 satisfy the attention kernel
+    m_cpu = models.build("gpt2-small")
+    m_gpu = models.build("gpt2-small")
+    m_gpu.load_state_dict(m_cpu.state_dict())
+    m_gpu = m_gpu.to(DEV, torch.bfloat16)
+    ids = torch.randint(0, m_cpu.cfg.vocab_size, (2, 128))
+    loss_cpu = m_cpu(ids, labels=ids)
+    loss_gpu = m_gpu(ids.to(DEV), labels=ids.to(DEV))
+    assert abs(float(loss_cpu) - float(loss_gpu)) < 0.05 * max(1.0, float(loss_cpu))
+    loss_gpu.backward()  # exercises the native backward path end-to-end
+    assert all(p.grad is not None for p in m_gpu.parameters() if p.requires_grad)
+
+
+# ---------------------------------------------------------------------------
 # RoPE
 # ---------------------------------------------------------------------------
 
