@@ -569,6 +569,223 @@ __global__ __launch_bounds__(256, 1) void attn_bwd_kernel(
   }
 }
 
+// Backward v3 (experimental, HYPHA_ATTN_BWD_V3=1): occupancy-first variant.
+// All 4 waves co-process ONE q-tile per iteration (v2 gives each wave its own
+// q-tile), splitting work by (kv-tile = wid&1, d-block subset = wid>>1):
+//   * qt/dO^T staging is cooperative (4x cheaper per wave);
+//   * each (tile, db) accumulator is owned by exactly one wave, so dK/dV
+//     store directly with NO cross-wave combine phase;
+//   * the dS [q][kv] image for dQ reuses the ptds buffers (per tile, [32][40]
+//     halves split at the kv-32 boundary) behind an explicit barrier;
+//   * LDS = 76 KB -> TWO blocks/CU (v2: 152 KB -> one), 8 waves in flight.
+// Cost: S^T and dP^T are computed twice (once per wave pair sharing a tile),
+// +40% MFMA issue — the bet is that the kernel is latency- not MFMA-bound
+// (PMC: SQ_WAIT_ANY ~3x SQ_BUSY on v2). A/B via tools/attn_bench.py.
+template <int HD>
+__global__ __launch_bounds__(256, 2) void attn_bwd_v3_kernel(
+    const short* __restrict__ qg, const short* __restrict__ kg,
+    const short* __restrict__ vg, const short* __restrict__ dog,
+    const float* __restrict__ lseg, const float* __restrict__ dig,
+    float* __restrict__ dqg, short* __restrict__ dkg, short* __restrict__ dvg,
+    int B, int Hq, int Hkv, int S, float scale, bool causal,
+    long long q_sb, long long q_sh, long long q_ss,
+    long long kv_sb, long long kv_sh, long long kv_ss) {
+  constexpr int KVT = 32;
+  constexpr int NT = 2;
+  constexpr int QT = 32;
+  constexpr int KC = HD / 16;
+  constexpr int DBLK = HD / 32;
+  constexpr int TP = 40;
+  constexpr int KTP = NT * KVT + 8;
+  constexpr int NACC = DBLK >= 2 ? DBLK / 2 : 1;
+  __shared__ __attribute__((aligned(16))) short k_img[NT * KVT * HD];
+  __shared__ __attribute__((aligned(16))) short v_img[NT * KVT * HD];
+  __shared__ __attribute__((aligned(16))) short kt_img[HD * KTP];
+  __shared__ __attribute__((aligned(16))) short qt_img[HD * 32];       // shared, 1 copy
+  __shared__ __attribute__((aligned(16))) short dot_img[HD * 32];      // shared, 1 copy
+  __shared__ __attribute__((aligned(16))) short pd_img[NT][KVT * TP];  // ptT, then dS [q][kv]
+  __shared__ __attribute__((aligned(16))) short dst_img[NT][KVT * TP];
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int hi = lane >> 5;
+  const int ln = lane & 31;
+  const int tile = wid & 1;   // this wave's kv tile
+  const int dbb = wid >> 1;   // this wave's d-block base (stride 2)
+
+  const int bhkv = blockIdx.y;
+  const int b = bhkv / Hkv;
+  const int hkv = bhkv % Hkv;
+  const int G = Hq / Hkv;
+  const int kv_base = blockIdx.x * (NT * KVT);
+  const long long kvbase = (long long)b * kv_sb + (long long)hkv * kv_sh;
+
+  auto qimg_off = [](int d, int qe) {
+    return d * 32 + ((((qe >> 3) ^ ((d >> 2) & 3)) << 3) | (qe & 7));
+  };
+
+  {  // stage K, V (swizzled) and K^T — identical to v2
+    constexpr int CH = NT * KVT * HD / 8;
+    for (int c = tid; c < CH; c += 256) {
+      int row = c / (HD / 8), e0 = (c % (HD / 8)) * 8;
+      s16x8 k8 = *reinterpret_cast<const s16x8*>(kg + kvbase + (long long)(kv_base + row) * kv_ss + e0);
+      *reinterpret_cast<s16x8*>((char*)k_img + k_lds_off<HD>(row, e0)) = k8;
+      s16x8 v8 = *reinterpret_cast<const s16x8*>(vg + kvbase + (long long)(kv_base + row) * kv_ss + e0);
+      *reinterpret_cast<s16x8*>((char*)v_img + k_lds_off<HD>(row, e0)) = v8;
+    }
+    for (int i = 0; i < HD / 32; ++i) {
+      int kvr = tid & 63;
+      int e0 = (i * 4 + (tid >> 6)) * 8;
+      s16x8 k8 = *reinterpret_cast<const s16x8*>(kg + kvbase + (long long)(kv_base + kvr) * kv_ss + e0);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) kt_img[(e0 + j) * KTP + kvr] = k8[j];
+    }
+  }
+  __syncthreads();
+
+  f32x16 dk_acc[NACC] = {};
+  f32x16 dv_acc[NACC] = {};
+
+  const int t0 = causal ? kv_base / QT : 0;
+  const int Tq = S / QT;
+  const int kv0 = kv_base + tile * KVT;
+
+  for (int hq = hkv * G; hq < (hkv + 1) * G; ++hq) {
+    const long long qbase = (long long)b * q_sb + (long long)hq * q_sh;
+    const long long lsebase = (long long)(b * Hq + hq) * S;
+    for (int t = t0; t < Tq; ++t) {
+      const int q0 = t * QT;
+      const int qrow = q0 + ln;
+
+      // cooperative Q^T / dO^T staging: the HD/16 column-chunks split 4 ways
+      for (int it = wid; it < HD / 16; it += NWAVE) {
+        int d0 = 8 * hi + 16 * it;
+        s16x8 q8 = *reinterpret_cast<const s16x8*>(qg + qbase + (long long)(q0 + ln) * q_ss + d0);
+        s16x8 d8 = *reinterpret_cast<const s16x8*>(dog + qbase + (long long)(q0 + ln) * q_ss + d0);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          qt_img[qimg_off(d0 + j, ln)] = q8[j];
+          dot_img[qimg_off(d0 + j, ln)] = d8[j];
+        }
+      }
+      const float lse = lseg[lsebase + qrow];
+      const float di = dig[lsebase + qrow];
+      __syncthreads();  // staged images ready
+
+      const bool active = !(causal && kv0 > q0 + QT - 1);
+      f32x16 pt;
+      f32x16 dpt = {};
+      if (active) {
+        // S^T = K Q^T
+        f32x16 st = {};
+#pragma unroll
+        for (int kc = 0; kc < KC; ++kc) {
+          s16x8 qf = *reinterpret_cast<const s16x8*>(qg + qbase + (long long)qrow * q_ss +
+                                                     16 * kc + 8 * hi);
+          s16x8 kf = *reinterpret_cast<const s16x8*>(
+              (char*)k_img + k_lds_off<HD>(tile * KVT + ln, 16 * kc + 8 * hi));
+          st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qf, st, 0, 0, 0);
+        }
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          int kv = kv0 + (r & 3) + 8 * (r >> 2) + 4 * hi;
+          float pv = __expf(st[r] * scale - lse);
+          if (causal && kv > qrow) pv = 0.f;
+          pt[r] = pv;
+        }
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          int kv = (r & 3) + 8 * (r >> 2) + 4 * hi;
+          pd_img[tile][kv * TP + ln] = f2bf(pt[r]);
+        }
+        // dV^T += dO^T P for this wave's d-blocks
+#pragma unroll
+        for (int i = 0; i < NACC; ++i) {
+          int db = dbb + 2 * i;
+#pragma unroll
+          for (int kcq = 0; kcq < QT / 16; ++kcq) {
+            s16x8 af = *reinterpret_cast<const s16x8*>(
+                dot_img + qimg_off(32 * db + ln, 16 * kcq + 8 * hi));
+            s16x8 bf = *reinterpret_cast<const s16x8*>(pd_img[tile] + ln * TP +
+                                                       16 * kcq + 8 * hi);
+            dv_acc[i] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(af, bf, dv_acc[i], 0, 0, 0);
+          }
+        }
+        // dP^T = V dO^T
+#pragma unroll
+        for (int kc = 0; kc < KC; ++kc) {
+          s16x8 df = *reinterpret_cast<const s16x8*>(dog + qbase + (long long)qrow * q_ss +
+                                                     16 * kc + 8 * hi);
+          s16x8 vf = *reinterpret_cast<const s16x8*>(
+              (char*)v_img + k_lds_off<HD>(tile * KVT + ln, 16 * kc + 8 * hi));
+          dpt = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vf, df, dpt, 0, 0, 0);
+        }
+      }
+      __syncthreads();  // both tile-waves done reading ptds (dS reuses it)
+
+      if (active) {
+        // dS^T; write dsT (for dK) and the dS [q][kv] half (for dQ, aliased)
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          int kv = (r & 3) + 8 * (r >> 2) + 4 * hi;
+          float v = pt[r] * (dpt[r] - di) * scale;
+          short vb = f2bf(v);
+          dst_img[tile][kv * TP + ln] = vb;
+          pd_img[tile][ln * TP + kv] = vb;
+        }
+        // dK += dS^T Q for this wave's d-blocks
+#pragma unroll
+        for (int i = 0; i < NACC; ++i) {
+          int db = dbb + 2 * i;
+#pragma unroll
+          for (int kcq = 0; kcq < QT / 16; ++kcq) {
+            s16x8 af = *reinterpret_cast<const s16x8*>(dst_img[tile] + ln * TP +
+                                                       16 * kcq + 8 * hi);
+            s16x8 bf = *reinterpret_cast<const s16x8*>(
+                qt_img + qimg_off(32 * db + ln, 16 * kcq + 8 * hi));
+            dk_acc[i] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(af, bf, dk_acc[i], 0, 0, 0);
+          }
+        }
+      }
+      __syncthreads();  // dS halves visible to every wave
+
+      // dQ = dS K over kv 0..63; d-blocks split across waves
+      const int kck_hi = (causal && kv_base + KVT > q0 + QT - 1) ? NT : 2 * NT;
+      for (int db = wid; db < DBLK; db += NWAVE) {
+        f32x16 dq = {};
+        for (int kck = 0; kck < kck_hi; ++kck) {
+          s16x8 af = *reinterpret_cast<const s16x8*>(pd_img[kck >> 1] + ln * TP +
+                                                     16 * (kck & 1) + 8 * hi);
+          s16x8 bf = *reinterpret_cast<const s16x8*>(kt_img + (32 * db + ln) * KTP +
+                                                     16 * kck + 8 * hi);
+          dq = __builtin_amdgcn_mfma_f32_32x32x16_bf16(af, bf, dq, 0, 0, 0);
+        }
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          int qi = q0 + (r & 3) + 8 * (r >> 2) + 4 * hi;
+          atomicAdd(dqg + (qbase + (long long)qi * q_ss + 32 * db + ln), dq[r]);
+        }
+      }
+      __syncthreads();  // before the next q-tile restages qt/dot/pd
+    }
+  }
+
+  // Direct bf16 stores: each (tile, db) accumulator has exactly one owner
+#pragma unroll
+  for (int i = 0; i < NACC; ++i) {
+    int db = dbb + 2 * i;
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      int m = (r & 3) + 8 * (r >> 2) + 4 * hi;
+      // dK: D[m=kv][n=d]
+      dkg[kvbase + (long long)(kv0 + m) * kv_ss + 32 * db + ln] = f2bf(dk_acc[i][r]);
+      // dV: transposed D[m=d][n=kv]
+      dvg[kvbase + (long long)(kv0 + ln) * kv_ss + 32 * db + m] = f2bf(dv_acc[i][r]);
+    }
+  }
+}
+
 }  // namespace
 
 std::vector<torch::Tensor> attn_bwd_ex(torch::Tensor q, torch::Tensor k, torch::Tensor v,
@@ -598,19 +815,34 @@ std::vector<torch::Tensor> attn_bwd_ex(torch::Tensor q, torch::Tensor k, torch::
     kv_ss = HD; kv_sh = (long long)S * HD; kv_sb = (long long)Hkv * S * HD;
   }
 
+  // experimental occupancy-first backward (see attn_bwd_v3_kernel docstring)
+  static const bool use_v3 = [] {
+    const char* e = getenv("HYPHA_ATTN_BWD_V3");
+    return e && e[0] == '1';
+  }();
+
 #define DISPATCH(HDV)                                                                   \
   do {                                                                                  \
     hipLaunchKernelGGL(attn_bwd_di_kernel<HDV>, dim3((unsigned)((nrows + 3) / 4)),      \
                        dim3(256), 0, stream, (const short*)dout.data_ptr(),             \
                        (const short*)o.data_ptr(), di.data_ptr<float>(), nrows, Hq, S,  \
                        q_sb, q_sh, q_ss);                                               \
-    hipLaunchKernelGGL(attn_bwd_kernel<HDV>, dim3(S / 64, B * Hkv), dim3(256), 0,       \
-                       stream, (const short*)q.data_ptr(), (const short*)k.data_ptr(),  \
-                       (const short*)v.data_ptr(), (const short*)dout.data_ptr(),       \
-                       lse.data_ptr<float>(), di.data_ptr<float>(),                     \
-                       dq32.data_ptr<float>(), (short*)dk.data_ptr(),                   \
-                       (short*)dv.data_ptr(), B, Hq, Hkv, S, scale, causal, q_sb,       \
-                       q_sh, q_ss, kv_sb, kv_sh, kv_ss);                                \
+    if (use_v3)                                                                         \
+      hipLaunchKernelGGL(attn_bwd_v3_kernel<HDV>, dim3(S / 64, B * Hkv), dim3(256), 0,  \
+                         stream, (const short*)q.data_ptr(), (const short*)k.data_ptr(),\
+                         (const short*)v.data_ptr(), (const short*)dout.data_ptr(),     \
+                         lse.data_ptr<float>(), di.data_ptr<float>(),                   \
+                         dq32.data_ptr<float>(), (short*)dk.data_ptr(),                 \
+                         (short*)dv.data_ptr(), B, Hq, Hkv, S, scale, causal, q_sb,     \
+                         q_sh, q_ss, kv_sb, kv_sh, kv_ss);                              \
+    else                                                                                \
+      hipLaunchKernelGGL(attn_bwd_kernel<HDV>, dim3(S / 64, B * Hkv), dim3(256), 0,     \
+                         stream, (const short*)q.data_ptr(), (const short*)k.data_ptr(),\
+                         (const short*)v.data_ptr(), (const short*)dout.data_ptr(),     \
+                         lse.data_ptr<float>(), di.data_ptr<float>(),                   \
+                         dq32.data_ptr<float>(), (short*)dk.data_ptr(),                 \
+                         (short*)dv.data_ptr(), B, Hq, Hkv, S, scale, causal, q_sb,     \
+                         q_sh, q_ss, kv_sb, kv_sh, kv_ss);                              \
   } while (0)
 
   if (HD == 128)
