@@ -95,6 +95,37 @@ class Comm:
             return 0
         return int(2 * (w - 1) / w * payload_bytes)
 
+    def reform(self, rank: int, world_size: int, master_addr: str = "127.0.0.1",
+               master_port: int = 29532, timeout_s: float = 600.0) -> None:
+        """Rebuild the process group with a NEW membership (elastic DiLoCo:
+        worker kill/rejoin, BASELINE config 3 on the RCCL path).
+
+        RCCL communicators cannot shrink/grow in place; because DiLoCo only
+        communicates every H inner steps, tearing the group down and
+        re-initialising on a fresh rendezvous at the next outer sync is cheap
+        (one bootstrap per membership change, amortized over H steps). The
+        scheduler names the surviving/joining ranks and a fresh port; every
+        member calls reform() with its new rank before the next outer sync.
+        (The reference's star-topology PS achieves the same via its mutable
+        member set — parameter_server.rs round semantics.)
+        """
+        if dist.is_initialized():
+            dist.destroy_process_group()
+        self.rank = rank
+        self.world_size = world_size
+        if world_size > 1:
+            backend = "nccl" if torch.cuda.is_available() else "gloo"
+            dist.init_process_group(
+                backend=backend,
+                rank=rank,
+                world_size=world_size,
+                init_method=f"tcp://{master_addr}:{master_port}",
+                timeout=datetime.timedelta(seconds=timeout_s),
+            )
+            self.backend = backend
+        else:
+            self.backend = "none"
+
     def shutdown(self) -> None:
         if dist.is_initialized():
             dist.destroy_process_group()

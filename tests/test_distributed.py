@@ -96,3 +96,68 @@ def test_bench_contract_world2_gloo(tmp_path):
     assert cfg["global_batch"] == 4  # whole-job aggregate
     assert cfg["outer_syncs_in_timed_window"] == 1  # h=3 < steps+warmup
     assert cfg["outer_sync_wire_bytes_per_rank"] > 0  # ring wire accounting
+
+
+def _elastic_worker(rank, port, q):
+    """World starts at 3; rank 2 dies after round 1; survivors reform to
+    world 2 on a fresh rendezvous and complete another round."""
+    os.environ["RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = "3"
+    os.environ["LOCAL_RANK"] = str(rank)
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    try:
+        from hypha_amd import models
+        from hypha_amd.data.synthetic import SyntheticTokens
+        from hypha_amd.parallel import Comm, DiLoCoConfig, DiLoCoWorker, InnerOptConfig
+
+        torch.manual_seed(200 + rank)
+        model = models.build("llama-tiny")
+        cfg = DiLoCoConfig(h=2, inner=InnerOptConfig(lr=1e-3, warmup_steps=0,
+                                                     schedule="constant"))
+        comm = Comm(backend="gloo")
+        w = DiLoCoWorker(model, cfg, comm=comm, device=torch.device("cpu"))
+        data = SyntheticTokens(512, 32, 2, seed=13, rank=rank)
+        for _ in range(2):  # round 1
+            ids, labels = data.next_batch()
+            w.train_step(ids, labels)
+            w.maybe_outer_sync()
+
+        if rank == 2:  # this worker "dies" after round 1
+            q.put(("dead", rank, 0.0, 0))
+            comm.shutdown()
+            return
+
+        # survivors: scheduler would assign new ranks {0,1} + a fresh port
+        comm.reform(rank=rank, world_size=2, master_port=port + 1)
+        for _ in range(2):  # round 2 with the shrunken world
+            ids, labels = data.next_batch()
+            w.train_step(ids, labels)
+            w.maybe_outer_sync()
+        q.put(("ok", rank, float(w.fp.master.sum()), w.round))
+        comm.shutdown()
+    except Exception:  # pragma: no cover
+        import traceback
+
+        q.put(("err", rank, traceback.format_exc(), 0))
+
+
+@pytest.mark.timeout(180)
+def test_elastic_reform_after_worker_death():
+    """Config-3 semantics on the collective path: the communicator is
+    re-formed on membership change at an outer-sync boundary (Comm.reform —
+    RCCL groups cannot shrink in place; re-bootstrap is amortized over H)."""
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    port = 29551
+    procs = [ctx.Process(target=_elastic_worker, args=(r, port, q)) for r in range(3)]
+    for p in procs:
+        p.start()
+    results = [q.get() for _ in range(3)]
+    for p in procs:
+        p.join(timeout=120)
+    ok = sorted(r for r in results if r[0] == "ok")
+    assert len(ok) == 2 and any(r[0] == "dead" for r in results), results
+    # both survivors completed 2 rounds and hold identical global weights
+    assert ok[0][3] == 2 and ok[1][3] == 2
+    assert ok[0][2] == pytest.approx(ok[1][2], rel=1e-6)
