@@ -20,6 +20,7 @@
 #include <functional>
 #include <mutex>
 #include <string>
+#include <chrono>
 #include <thread>
 #include <vector>
 
@@ -191,13 +192,20 @@ class Bridge {
         size_t cursor = 0;
         while (running_) {
           Json ev;
+          bool have = false;
           {
             std::unique_lock<std::mutex> lk(ev_mu_);
-            ev_cv_.wait(lk, [&] { return closed_ || events_.size() > cursor; });
+            // 30 s keepalive comments: a DiLoCo round can be quiet for many
+            // minutes and executor HTTP clients enforce read timeouts
+            ev_cv_.wait_for(lk, std::chrono::seconds(30),
+                            [&] { return closed_ || events_.size() > cursor; });
             if (closed_ && events_.size() <= cursor) break;
-            ev = events_[cursor++];
+            if (events_.size() > cursor) {
+              ev = events_[cursor++];
+              have = true;
+            }
           }
-          std::string line = "data: " + ev.dump() + "\n\n";
+          std::string line = have ? "data: " + ev.dump() + "\n\n" : ": keepalive\n\n";
           if (::send(fd, line.data(), line.size(), MSG_NOSIGNAL) <= 0) break;
         }
       } else {
