@@ -93,6 +93,20 @@ class Metrics:
 METRICS = Metrics()
 
 
+def collect_transport_bytes() -> dict | None:
+    """Pull the C++ control-plane transport byte counters (net.cpp
+    bandwidth_stats — every framed request/stream byte) into the registry.
+    Returns the raw stats, or None when the native core is not built."""
+    try:
+        from hypha_amd import _core
+    except ImportError:
+        return None
+    s = _core.bandwidth_stats()
+    METRICS.gauge_set("hypha.bandwidth.transport.inbound_bytes", s["inbound_bytes"])
+    METRICS.gauge_set("hypha.bandwidth.transport.outbound_bytes", s["outbound_bytes"])
+    return s
+
+
 def instrument_comm(comm) -> None:
     """Wrap a parallel.Comm so collective payload bytes are counted —
     the analogue of the reference's bandwidth::Transport wrapper

@@ -150,3 +150,14 @@ def test_daemon_init_and_probe_subcommands():
         r = subprocess.run([str(bins / d), "probe", *args],
                            capture_output=True, text=True, timeout=30)
         assert r.returncode == 1 and "unreachable" in r.stderr, d
+
+
+def test_collect_transport_bytes():
+    from hypha_amd.telemetry import METRICS, collect_transport_bytes
+
+    s = collect_transport_bytes()
+    if s is None:
+        pytest.skip("native core not built")
+    assert s["inbound_bytes"] >= 0 and s["outbound_bytes"] >= 0
+    snap = METRICS.snapshot()
+    assert "hypha.bandwidth.transport.inbound_bytes" in snap["gauges"]
