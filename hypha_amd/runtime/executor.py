@@ -106,6 +106,11 @@ def main() -> int:
             worker.fp.flat.copy_(worker.fp.master)
         print("[executor] joined at current global weights", flush=True)
 
+    # round-start global weights on disk (training.py:61-63 parity:
+    # `0_global_weights`; SafeTensors as the reference's checkpoint format)
+    save_file({"flat": worker.fp.theta0.cpu()},
+              os.path.join(args.work_dir, "0_global_weights.safetensors"))
+
     done = False
     round_idx = resume_round
     while not done:
