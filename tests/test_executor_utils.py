@@ -84,3 +84,25 @@ def test_preprocessor_registry_errors():
         build_preprocessor("tokenizer", "/nonexistent/tok")
     with pytest.raises(NotImplementedError, match="offline"):
         build_preprocessor("image")
+
+
+def test_tokenizer_preprocessor_from_local_artifact(tmp_path):
+    """The wire PreprocessorType 'tokenizer' resolved against a LOCAL
+    artifact (offline parity for the reference's HF preprocessor fetch)."""
+    tokenizers = pytest.importorskip("tokenizers")
+    transformers = pytest.importorskip("transformers")
+    from tokenizers.pre_tokenizers import Whitespace
+
+    tok = tokenizers.Tokenizer(
+        tokenizers.models.WordLevel({"hello": 0, "world": 1, "[UNK]": 2},
+                                    unk_token="[UNK]"))
+    tok.pre_tokenizer = Whitespace()
+    fast = transformers.PreTrainedTokenizerFast(tokenizer_object=tok,
+                                                pad_token="[UNK]")
+    art = tmp_path / "tok"
+    fast.save_pretrained(str(art))
+
+    run = build_preprocessor("tokenizer", str(art))
+    out = run(text=["hello world", "world"])
+    assert out["input_ids"].shape[0] == 2
+    assert out["input_ids"][0].tolist()[:2] == [0, 1]
