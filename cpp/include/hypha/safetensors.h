@@ -39,6 +39,10 @@ class SafeTensors {
       fclose(f);
       throw std::runtime_error("safetensors: short header");
     }
+    if (hlen > (256u << 20)) {  // corrupt/hostile file: bound the allocation
+      fclose(f);
+      throw std::runtime_error("safetensors: absurd header length");
+    }
     std::string hdr(hlen, '\0');
     if (fread(hdr.data(), 1, hlen, f) != hlen) {
       fclose(f);
@@ -85,7 +89,7 @@ class SafeTensors {
     uint64_t hlen = h.size();
     fwrite(&hlen, 8, 1, f);
     fwrite(h.data(), 1, h.size(), f);
-    fwrite(data.data(), 1, data.size(), f);
+    if (!data.empty()) fwrite(data.data(), 1, data.size(), f);
     fclose(f);
   }
 
