@@ -320,3 +320,111 @@ def test_cluster_diloco_over_mtls(binaries, tmp_path):
                 p.wait(timeout=10)
             except Exception:
                 pass
+
+
+@pytest.mark.timeout(300)
+def test_scheduler_death_reclaims_workers(binaries, tmp_path):
+    """SURVEY §5 failure-detection parity: scheduler death means no renewals,
+    so workers' 250 ms pruner expires the 10 s leases and cancels the orphan
+    executor subprocesses (arbiter.rs:98-141 semantics — no zombie jobs)."""
+    from hypha_amd.data.synthetic import write_slice_files
+
+    data_dir = tmp_path / "slices"
+    write_slice_files(str(data_dir), "synth", num_slices=4, samples_per_slice=16,
+                      vocab_size=512, seq_len=128)
+    gw_port = free_port()
+    env = dict(os.environ, PYTHONPATH=str(REPO))
+    procs = []
+
+    def spawn(name, cmd):
+        log = open(tmp_path / f"{name}.log", "w")
+        p = subprocess.Popen(cmd, cwd=REPO, env=env, stdout=log, stderr=log)
+        procs.append(p)
+        return p
+
+    try:
+        spawn("gateway", [str(BIN / "hypha-gateway"), "--port", str(gw_port)])
+        time.sleep(0.3)
+        spawn("data", [str(BIN / "hypha-data"), "--name", "data-node",
+                       "--gateway-host", "127.0.0.1", "--gateway-port", str(gw_port),
+                       "--dataset", "synth", "--dataset-path", str(data_dir)])
+        exec_cmd = (f"{sys.executable} -m hypha_amd.runtime.executor "
+                    "--socket {SOCKET_PATH} --work-dir {WORK_DIR} --job {JOB_JSON}")
+        for i in range(3):
+            spawn(f"worker{i}", [str(BIN / "hypha-worker"), "--name", f"worker-{i}",
+                                 "--gateway-host", "127.0.0.1",
+                                 "--gateway-port", str(gw_port),
+                                 "--exec-cmd", exec_cmd,
+                                 "--work-root", str(tmp_path / f"work{i}")])
+        time.sleep(0.5)
+        cfg = tmp_path / "job.json"
+        cfg.write_text(
+            '{"model": "llama-tiny", "dataset": "synth", "num_workers": 2,'
+            ' "update_rounds": 50, "avg_samples_between_updates": 8,'
+            ' "batch_size": 2, "seq_len": 128, "inner_lr": 0.001}'
+        )
+        sched = spawn("sched", [str(BIN / "hypha-scheduler"), "--name", "scheduler",
+                                "--gateway-host", "127.0.0.1",
+                                "--gateway-port", str(gw_port),
+                                "--config", str(cfg)])
+
+        # wait until training actually started (an executor is running)
+        deadline = time.time() + 90
+        started = False
+        while time.time() < deadline and not started:
+            time.sleep(1)
+            for i in range(3):
+                log = tmp_path / f"worker{i}.log"
+                if log.exists() and "[executor] model=" in log.read_text():
+                    started = True
+        assert started, "no job was ever dispatched"
+
+        sched.send_signal(signal.SIGKILL)  # scheduler dies mid-run
+        sched.wait(timeout=10)
+
+        # train executors exit when their status posts fail (the reference's
+        # training.py would equally crash on a dead scheduler); the PS job has
+        # no scheduler dependency mid-round, so the LEASE machinery must
+        # cancel it: renewals stop -> 10 s TTL expires -> pruner kills it
+        deadline = time.time() + 40
+        cancelled = 0
+        while time.time() < deadline and cancelled < 1:
+            time.sleep(1)
+            cancelled = sum(
+                "cancelled job" in (tmp_path / f"worker{i}.log").read_text()
+                for i in range(3)
+                if (tmp_path / f"worker{i}.log").exists()
+            )
+        assert cancelled >= 1, [
+            (tmp_path / f"worker{i}.log").read_text()[-1500:] for i in range(3)
+        ]
+
+        # the definitive reclamation check: a SECOND scheduler can buy the
+        # same workers again and run a fresh job to completion
+        cfg2 = tmp_path / "job2.json"
+        cfg2.write_text(
+            '{"model": "llama-tiny", "dataset": "synth", "num_workers": 2,'
+            ' "update_rounds": 2, "avg_samples_between_updates": 8,'
+            ' "batch_size": 2, "seq_len": 128, "inner_lr": 0.001}'
+        )
+        time.sleep(12)  # let every stale lease age out
+        sched2 = subprocess.Popen(
+            [str(BIN / "hypha-scheduler"), "--name", "scheduler2",
+             "--gateway-host", "127.0.0.1", "--gateway-port", str(gw_port),
+             "--config", str(cfg2)],
+            cwd=REPO, env=env, stdout=subprocess.PIPE,
+            stderr=open(tmp_path / "sched2.log", "w"), text=True,
+        )
+        procs.append(sched2)
+        out, _ = sched2.communicate(timeout=180)
+        assert "Job is completed." in out, (
+            out, (tmp_path / "sched2.log").read_text()[-3000:])
+    finally:
+        for p in procs:
+            if p.poll() is None:
+                p.send_signal(signal.SIGKILL)
+        for p in procs:
+            try:
+                p.wait(timeout=10)
+            except Exception:
+                pass
