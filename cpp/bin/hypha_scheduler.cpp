@@ -8,6 +8,7 @@
 #include <signal.h>
 #include <unistd.h>
 
+#include <algorithm>
 #include <atomic>
 #include <chrono>
 #include <condition_variable>
@@ -139,6 +140,7 @@ int main(int argc, char** argv) {
   const int64_t update_rounds = cfg.at("update_rounds").as_int();
   const int64_t samples_between = cfg.at("avg_samples_between_updates").as_int();
   const int64_t batch_size = cfg.get_or("batch_size", Json(4)).as_int();
+  const int64_t max_batch = cfg.get_or("max_batch_size", Json((int64_t)600)).as_int();
   const int64_t seq_len = cfg.get_or("seq_len", Json(128)).as_int();
   const double worker_bid = cfg.get_or("worker_bid", Json(1.0)).as_double();
   const double worker_max = cfg.get_or("worker_max_price", Json(10.0)).as_double();
@@ -423,7 +425,12 @@ int main(int argc, char** argv) {
     Json opt;
     opt["adam"] = adam;
     tr["optimizer"] = opt;
-    tr["batch_size"] = batch_size;
+    // per-worker batch scales with OFFERED gpu capacity, capped
+    // (hypha-scheduler.rs:320-322: floor(gpu_avail/gpu_req).min(max_batch))
+    int64_t mult = worker_req.gpu > 0 ? (int64_t)(w->resources.gpu / worker_req.gpu) : 1;
+    if (mult < 1) mult = 1;
+    int64_t wbatch = std::min<int64_t>(max_batch, batch_size * mult);
+    tr["batch_size"] = wbatch;
     tr["seq_len"] = seq_len;
     if (cfg.has("checkpoint_every_rounds"))
       tr["checkpoint_every_rounds"] = cfg.at("checkpoint_every_rounds");
@@ -439,7 +446,7 @@ int main(int argc, char** argv) {
     d["lease"] = w->lease_id;
     node.request(w->peer, "dispatch_job", d, 10.0);
     std::lock_guard<std::mutex> lk(fsm_mu);
-    fsm.add_worker(w->peer, batch_size);
+    fsm.add_worker(w->peer, wbatch);
   };
 
   for (size_t i = 0; i < train_workers.size(); ++i)

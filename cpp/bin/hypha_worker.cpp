@@ -264,21 +264,27 @@ struct WorkerDaemon {
       auto picked = select_requests(ads, policy, resman.available());
       std::set<std::string> answered;
       for (auto& d : picked) {
-        if (!resman.reserve(d.request.resources)) continue;
+        // "whole" strategy: offer (and reserve) the full remaining capacity
+        Resources offer_res = policy.strategy == "whole" ? resman.available()
+                                                         : d.request.resources;
+        if (!offer_res.fits_in(resman.available()) ||
+            !d.request.resources.fits_in(offer_res))
+          continue;
+        if (!resman.reserve(offer_res)) continue;
         answered.insert(d.request.id);
         std::string lease_id =
             name + "-lease-" + std::to_string(lease_seq.fetch_add(1));
         {
           std::lock_guard<std::mutex> lk(mu);
           // temporary offer lease: 500 ms TTL (arbiter.rs:383-435)
-          leases.insert(lease_id, LeaseInfo{d.request.scheduler, d.request.resources, ""},
+          leases.insert(lease_id, LeaseInfo{d.request.scheduler, offer_res, ""},
                         0.5);
         }
         Json offer;
         offer["id"] = lease_id;
         offer["request_id"] = d.request.id;
         offer["price"] = d.offer_price;
-        offer["resources"] = d.request.resources.to_json();
+        offer["resources"] = offer_res.to_json();
         offer["timeout_s"] = 0.5;
         try {
           node.request(d.request.scheduler, "worker_offer", offer, 5.0);
@@ -745,6 +751,18 @@ int main(int argc, char** argv) {
     else if (a == "--storage") total.storage = std::stod(next());
     else if (a == "--price") policy.price = std::stod(next());
     else if (a == "--floor") policy.floor = std::stod(next());
+    else if (a == "--offer-strategy") policy.strategy = next();
+    else if (a == "--executors") {  // comma-separated supported executor list
+      policy.supported_executors.clear();
+      std::string v = next();
+      size_t pos = 0;
+      while (pos <= v.size()) {
+        size_t c = v.find(',', pos);
+        if (c == std::string::npos) c = v.size();
+        if (c > pos) policy.supported_executors.push_back(v.substr(pos, c - pos));
+        pos = c + 1;
+      }
+    }
     else if (a == "--exec-cmd") cmd = next();
     else if (a == "--work-root") work_root = next();
     else if (a == "--tls-cert") tls.cert_path = next();
@@ -762,6 +780,8 @@ int main(int argc, char** argv) {
            "# --gateway-host/--gateway-port gateway broker address\n"
            "# --gpu 1 --cpu 4 --memory 16 --storage 100   sellable resources\n"
            "# --price 1.0 --floor 0.0       auction offer policy\n"
+           "# --offer-strategy flexible|whole  offer requested vs full capacity\n"
+           "# --executors diloco-transformer,parameter-server  roles to sell\n"
            "# --exec-cmd 'python -m hypha_amd.runtime.executor --socket "
            "{SOCKET_PATH} --work-dir {WORK_DIR} --job {JOB_JSON}'\n"
            "# --work-root /tmp/hypha-work   per-job working directories\n"

@@ -93,6 +93,10 @@ struct OfferPolicy {
   double price = 1.0;   // asking price per weighted unit scale
   double floor = 0.0;   // minimum acceptable bid
   std::vector<std::string> supported_executors;
+  // OfferStrategy (worker/config.rs:54-61): "flexible" offers exactly the
+  // requested resources; "whole" offers the worker's FULL capacity
+  // (all-or-nothing co-tenancy; the scheduler scales the batch to it).
+  std::string strategy = "flexible";
 };
 
 struct ArbiterDecision {

@@ -428,3 +428,84 @@ def test_scheduler_death_reclaims_workers(binaries, tmp_path):
                 p.wait(timeout=10)
             except Exception:
                 pass
+
+
+@pytest.mark.timeout(300)
+def test_heterogeneous_batch_sizes(binaries, tmp_path):
+    """Performance-aware scheduling parity (hypha-scheduler.rs:320-322 +
+    rfc/2025-10-16): a worker offering 2 GPUs (--offer-strategy whole) gets
+    2x the per-worker batch of a 1-GPU worker; the FSM still closes rounds."""
+    from hypha_amd.data.synthetic import write_slice_files
+
+    data_dir = tmp_path / "slices"
+    write_slice_files(str(data_dir), "synth", num_slices=4, samples_per_slice=16,
+                      vocab_size=512, seq_len=128)
+    gw_port = free_port()
+    env = dict(os.environ, PYTHONPATH=str(REPO))
+    procs = []
+
+    def spawn(name, cmd):
+        log = open(tmp_path / f"{name}.log", "w")
+        p = subprocess.Popen(cmd, cwd=REPO, env=env, stdout=log, stderr=log)
+        procs.append(p)
+        return p
+
+    try:
+        spawn("gateway", [str(BIN / "hypha-gateway"), "--port", str(gw_port)])
+        time.sleep(0.3)
+        spawn("data", [str(BIN / "hypha-data"), "--name", "data-node",
+                       "--gateway-host", "127.0.0.1", "--gateway-port", str(gw_port),
+                       "--dataset", "synth", "--dataset-path", str(data_dir)])
+        exec_cmd = (f"{sys.executable} -m hypha_amd.runtime.executor "
+                    "--socket {SOCKET_PATH} --work-dir {WORK_DIR} --job {JOB_JSON}")
+        # big: 2 GPUs, sold whole; small: 1 GPU; third: PS
+        spawn("workerbig", [str(BIN / "hypha-worker"), "--name", "worker-big",
+                            "--gateway-host", "127.0.0.1",
+                            "--gateway-port", str(gw_port), "--gpu", "2",
+                            "--offer-strategy", "whole",
+                            "--executors", "diloco-transformer",
+                            "--exec-cmd", exec_cmd,
+                            "--work-root", str(tmp_path / "workbig")])
+        for nm, execs in (("small", "diloco-transformer"),
+                          ("ps", "parameter-server")):
+            spawn(f"worker{nm}", [str(BIN / "hypha-worker"), "--name", f"worker-{nm}",
+                                  "--gateway-host", "127.0.0.1",
+                                  "--gateway-port", str(gw_port), "--gpu", "1",
+                                  "--executors", execs,
+                                  "--exec-cmd", exec_cmd,
+                                  "--work-root", str(tmp_path / f"work{nm}")])
+        time.sleep(0.5)
+        cfg = tmp_path / "job.json"
+        cfg.write_text(
+            '{"model": "llama-tiny", "dataset": "synth", "num_workers": 2,'
+            ' "update_rounds": 2, "avg_samples_between_updates": 12,'
+            ' "batch_size": 2, "seq_len": 128, "inner_lr": 0.001}'
+        )
+        sched = subprocess.Popen(
+            [str(BIN / "hypha-scheduler"), "--name", "scheduler",
+             "--gateway-host", "127.0.0.1", "--gateway-port", str(gw_port),
+             "--config", str(cfg)],
+            cwd=REPO, env=env, stdout=subprocess.PIPE,
+            stderr=open(tmp_path / "sched.log", "w"), text=True,
+        )
+        procs.append(sched)
+        out, _ = sched.communicate(timeout=240)
+        assert "Job is completed." in out, (
+            out, (tmp_path / "sched.log").read_text()[-3000:])
+
+        logs = "".join((tmp_path / f"worker{n}.log").read_text()
+                       for n in ("big", "small", "ps")
+                       if (tmp_path / f"worker{n}.log").exists())
+        big_log = (tmp_path / "workerbig.log").read_text()
+        # the 2-GPU whole-offer worker trained with bs=4, some worker with bs=2
+        assert "bs=4" in big_log, big_log[-1500:]
+        assert "bs=2" in logs, logs[-1500:]
+    finally:
+        for p in procs:
+            if p.poll() is None:
+                p.send_signal(signal.SIGKILL)
+        for p in procs:
+            try:
+                p.wait(timeout=10)
+            except Exception:
+                pass
