@@ -333,6 +333,72 @@ int main(int argc, char** argv) {
 
   Resources worker_req = Resources::from_json(
       cfg.get_or("worker_resources", Json::parse("{\"gpu\":1,\"cpu\":1,\"memory\":1,\"storage\":1}")));
+
+  // --- inference mode (job_type: "generate"): allocate inference workers,
+  // dispatch generation jobs over the same bridge contract, wait for all
+  // completions. No PS, no sync FSM — slices still flow through the data
+  // scheduler. (The reference reaches inference through the same executor
+  // mechanism; here it is a first-class scheduler mode.)
+  if (cfg.get_or("job_type", Json(std::string("diloco"))).as_string() == "generate") {
+    auto ioffers = allocate(num_workers, {"inference-transformer"}, worker_req);
+    if ((int64_t)ioffers.size() < num_workers) {
+      fprintf(stderr, "[scheduler] insufficient inference workers: %zu/%lld\n",
+              ioffers.size(), (long long)num_workers);
+      return 1;
+    }
+    std::vector<std::shared_ptr<AllocatedWorker>> iworkers;
+    for (auto& o : ioffers) {
+      auto w = std::make_shared<AllocatedWorker>();
+      w->peer = o.worker;
+      w->lease_id = o.id;
+      w->resources = o.resources;
+      start_renewal(w);
+      iworkers.push_back(w);
+    }
+    for (size_t i = 0; i < iworkers.size(); ++i) {
+      Json inf;
+      inf["model"] = model;
+      Json fetch;
+      Json sref;
+      sref["peer"] = name;
+      sref["dataset"] = dataset;
+      fetch["scheduler"] = sref;
+      inf["data"] = fetch;
+      inf["batch_size"] = batch_size;
+      inf["seq_len"] = seq_len;
+      inf["max_new_tokens"] = cfg.get_or("max_new_tokens", Json((int64_t)16));
+      inf["num_batches"] = cfg.get_or("num_batches", Json((int64_t)2));
+      if (cfg.has("temperature")) inf["temperature"] = cfg.at("temperature");
+      if (cfg.has("top_k")) inf["top_k"] = cfg.at("top_k");
+      Json ex;
+      ex["infer"] = inf;
+      Json job;
+      job["id"] = std::string("job-infer-") + std::to_string(i);
+      job["executor"] = ex;
+      Json d;
+      d["job"] = job;
+      d["lease"] = iworkers[i]->lease_id;
+      node.request(iworkers[i]->peer, "dispatch_job", d, 10.0);
+    }
+    fprintf(stderr, "[scheduler] dispatched %lld inference jobs\n",
+            (long long)num_workers);
+    {
+      std::unique_lock<std::mutex> lk(status_mu);
+      status_cv.wait(lk, [&] {
+        int64_t done = 0;
+        for (auto& [id, st] : job_status)
+          if (st == "completed") ++done;
+        return done >= num_workers;
+      });
+    }
+    printf("Job is completed.\n");
+    fflush(stdout);
+    running = false;
+    for (auto& t : renewers) t.join();
+    node.stop();
+    return 0;
+  }
+
   auto train_offers = allocate(num_workers, {"diloco-transformer"}, worker_req);
   if ((int64_t)train_offers.size() < num_workers) {
     fprintf(stderr, "[scheduler] insufficient workers: %zu/%lld\n", train_offers.size(),

@@ -73,6 +73,7 @@ struct WorkerDaemon {
   StaticResourceManager resman;
   OfferPolicy policy;
   std::string exec_cmd;  // template with {SOCKET_PATH} {WORK_DIR} {JOB_JSON}
+  std::string infer_cmd;  // same template, launched for "infer" executors
   std::string work_root;
 
   std::mutex mu;
@@ -376,7 +377,12 @@ struct WorkerDaemon {
     }
     Json ex = jobj.at("executor");
     if (ex.has("train")) {
-      job->runner = std::thread([this, job, ex] { run_process_job(job, ex.at("train")); });
+      job->runner = std::thread(
+          [this, job, ex] { run_process_job(job, ex.at("train"), exec_cmd); });
+    } else if (ex.has("infer")) {
+      if (infer_cmd.empty()) throw std::runtime_error("inference not configured");
+      job->runner = std::thread(
+          [this, job, ex] { run_process_job(job, ex.at("infer"), infer_cmd); });
     } else if (ex.has("aggregate")) {
       job->is_aggregate = true;
       job->runner = std::thread(
@@ -392,7 +398,7 @@ struct WorkerDaemon {
 
   // --- process executor (executor/process.rs:78-198) ---------------------
 
-  void run_process_job(std::shared_ptr<Job> job, Json config) {
+  void run_process_job(std::shared_ptr<Job> job, Json config, const std::string& cmd_tmpl) {
     std::string sock_path = job->work_dir + "/bridge.sock";
     std::string job_json = job->work_dir + "/job.json";
     {
@@ -420,7 +426,7 @@ struct WorkerDaemon {
     bridge->start();
 
     // {SOCKET_PATH}/{WORK_DIR}/{JOB_JSON} substitution (process.rs:201-205)
-    std::string cmd = exec_cmd;
+    std::string cmd = cmd_tmpl;
     auto subst = [&](const std::string& key, const std::string& val) {
       size_t p;
       while ((p = cmd.find(key)) != std::string::npos) cmd.replace(p, key.size(), val);
@@ -731,7 +737,7 @@ struct WorkerDaemon {
 };
 
 int main(int argc, char** argv) {
-  std::string name = "worker", gw_host = "127.0.0.1", cmd, work_root = "/tmp/hypha-work";
+  std::string name = "worker", gw_host = "127.0.0.1", cmd, icmd, work_root = "/tmp/hypha-work";
   int gw_port = 0, port = 0;
   bool probe = false, init = false;
   std::vector<std::string> exclude_cidrs;
@@ -764,6 +770,7 @@ int main(int argc, char** argv) {
       }
     }
     else if (a == "--exec-cmd") cmd = next();
+    else if (a == "--infer-cmd") icmd = next();
     else if (a == "--work-root") work_root = next();
     else if (a == "--tls-cert") tls.cert_path = next();
     else if (a == "--tls-key") tls.key_path = next();
@@ -784,6 +791,8 @@ int main(int argc, char** argv) {
            "# --executors diloco-transformer,parameter-server  roles to sell\n"
            "# --exec-cmd 'python -m hypha_amd.runtime.executor --socket "
            "{SOCKET_PATH} --work-dir {WORK_DIR} --job {JOB_JSON}'\n"
+           "# --infer-cmd '... hypha_amd.runtime.infer_executor ...'  enables "
+           "the inference-transformer executor\n"
            "# --work-root /tmp/hypha-work   per-job working directories\n"
            "# --tls-cert/--tls-key/--tls-ca [--tls-crl]  mTLS identity\n");
     return 0;
@@ -798,7 +807,14 @@ int main(int argc, char** argv) {
     printf("probe: healthy\n");
     return 0;
   }
+  if (!icmd.empty()) {
+    auto& se = policy.supported_executors;
+    bool has = false;
+    for (auto& e : se) has = has || e == "inference-transformer";
+    if (!has) se.push_back("inference-transformer");
+  }
   WorkerDaemon daemon(name, gw_host, gw_port, total, policy, cmd, work_root, tls);
+  daemon.infer_cmd = icmd;
   daemon.node.set_exclude_cidrs(exclude_cidrs);
   daemon.start(port);
   printf("hypha-worker %s ready on port %d\n", name.c_str(), daemon.node.port());

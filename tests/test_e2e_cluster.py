@@ -509,3 +509,77 @@ def test_heterogeneous_batch_sizes(binaries, tmp_path):
                 p.wait(timeout=10)
             except Exception:
                 pass
+
+
+@pytest.mark.timeout(300)
+def test_cluster_inference_jobs(binaries, tmp_path):
+    """Inference through the full cluster: the scheduler's `generate` mode
+    auctions inference-capable workers, dispatches generation jobs over the
+    bridge contract, slices flow through the data scheduler, and completions
+    land as SafeTensors in the job work dirs."""
+    from hypha_amd.data.synthetic import write_slice_files
+
+    data_dir = tmp_path / "slices"
+    write_slice_files(str(data_dir), "synth", num_slices=4, samples_per_slice=8,
+                      vocab_size=512, seq_len=128)
+    gw_port = free_port()
+    env = dict(os.environ, PYTHONPATH=str(REPO))
+    procs = []
+
+    def spawn(name, cmd):
+        log = open(tmp_path / f"{name}.log", "w")
+        p = subprocess.Popen(cmd, cwd=REPO, env=env, stdout=log, stderr=log)
+        procs.append(p)
+        return p
+
+    try:
+        spawn("gateway", [str(BIN / "hypha-gateway"), "--port", str(gw_port)])
+        time.sleep(0.3)
+        spawn("data", [str(BIN / "hypha-data"), "--name", "data-node",
+                       "--gateway-host", "127.0.0.1", "--gateway-port", str(gw_port),
+                       "--dataset", "synth", "--dataset-path", str(data_dir)])
+        infer_cmd = (f"{sys.executable} -m hypha_amd.runtime.infer_executor "
+                     "--socket {SOCKET_PATH} --work-dir {WORK_DIR} --job {JOB_JSON}")
+        for i in range(2):
+            spawn(f"worker{i}", [str(BIN / "hypha-worker"), "--name", f"worker-{i}",
+                                 "--gateway-host", "127.0.0.1",
+                                 "--gateway-port", str(gw_port),
+                                 "--exec-cmd", "true", "--infer-cmd", infer_cmd,
+                                 "--work-root", str(tmp_path / f"work{i}")])
+        time.sleep(0.5)
+        cfg = tmp_path / "job.json"
+        cfg.write_text(
+            '{"job_type": "generate", "model": "llama-tiny", "dataset": "synth",'
+            ' "num_workers": 2, "update_rounds": 1,'
+            ' "avg_samples_between_updates": 1, "batch_size": 2, "seq_len": 128,'
+            ' "max_new_tokens": 8, "num_batches": 2}'
+        )
+        sched = subprocess.Popen(
+            [str(BIN / "hypha-scheduler"), "--name", "scheduler",
+             "--gateway-host", "127.0.0.1", "--gateway-port", str(gw_port),
+             "--config", str(cfg)],
+            cwd=REPO, env=env, stdout=subprocess.PIPE,
+            stderr=open(tmp_path / "sched.log", "w"), text=True,
+        )
+        procs.append(sched)
+        out, _ = sched.communicate(timeout=240)
+        assert "Job is completed." in out, (
+            out, (tmp_path / "sched.log").read_text()[-3000:],
+            *[(tmp_path / f"worker{i}.log").read_text()[-1500:] for i in range(2)],
+        )
+        # completions exist and extend the prompts by max_new_tokens
+        from safetensors.torch import load_file
+
+        comps = list(tmp_path.glob("work*/hypha-*/completion-*.safetensors"))
+        assert len(comps) >= 4, comps  # 2 workers x 2 batches
+        toks = load_file(str(comps[0]))["tokens"]
+        assert toks.shape[-1] == 128 + 8
+    finally:
+        for p in procs:
+            if p.poll() is None:
+                p.send_signal(signal.SIGKILL)
+        for p in procs:
+            try:
+                p.wait(timeout=10)
+            except Exception:
+                pass
