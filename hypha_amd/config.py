@@ -47,6 +47,10 @@ class JobConfig:
     worker_max_price: float = 10.0
     checkpoint_dir: str = ""
     checkpoint_every_rounds: int = 0  # 0 = disabled
+    max_batch_size: int = 600  # cap for capacity-proportional batches
+    job_type: str = "diloco"  # diloco | generate (inference dispatch)
+    max_new_tokens: int = 16  # generate mode
+    num_batches: int = 2  # generate mode: batches per worker
 
     def validate(self) -> "JobConfig":
         if self.num_workers < 1:
@@ -59,6 +63,8 @@ class JobConfig:
             raise ConfigError("outer_momentum", "must be in [0, 1)")
         if self.batch_size < 1 or self.seq_len < 1:
             raise ConfigError("batch_size/seq_len", "must be positive")
+        if self.job_type not in ("diloco", "generate"):
+            raise ConfigError("job_type", f"unknown job type {self.job_type!r}")
         return self
 
 
