@@ -450,6 +450,8 @@ int main(int argc, char** argv) {
     Json opt;
     opt["nesterov"] = nesterov;
     agg_cfg["optimizer"] = opt;
+    if (cfg.has("weighted_aggregation"))
+      agg_cfg["weighted_aggregation"] = cfg.at("weighted_aggregation");
     Json results;
     Json peers;
     peers["peers"] = train_peer_names;

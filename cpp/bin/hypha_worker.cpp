@@ -63,7 +63,7 @@ struct Job {
   std::mutex agg_mu;
   std::condition_variable agg_cv;
   std::set<std::string> members;                // peers whose update closes a round
-  std::map<std::string, std::string> round_files;  // peer -> pushed file
+  std::map<std::string, std::pair<std::string, double>> round_files;  // peer -> (file, samples)
   std::vector<std::string> pending_sync_peers;  // joiners awaiting the offset
 };
 
@@ -534,6 +534,7 @@ struct WorkerDaemon {
       hdr["name"] = base;
       hdr["size"] = (int64_t)size;
       hdr["job"] = ref.get_or("job", Json(job->id));
+      if (ref.has("samples")) hdr["samples"] = ref.at("samples");
       auto stream = node.open_stream(peer.as_string(), "push_resource", hdr);
       FILE* fp = fopen(path.c_str(), "rb");
       if (!fp) throw std::runtime_error("send: cannot reopen " + path);
@@ -608,7 +609,10 @@ struct WorkerDaemon {
     if (job->is_aggregate) {
       {
         std::lock_guard<std::mutex> lk(job->agg_mu);
-        job->round_files[from] = out;  // one update per peer per round
+        // one update per peer per round; "samples" = the round's sample
+        // count for optional weighted aggregation (heterogeneous batches)
+        double w = header.get_or("samples", Json(1.0)).as_double();
+        job->round_files[from] = {out, w > 0 ? w : 1.0};
       }
       job->agg_cv.notify_all();
       fprintf(stderr, "[%s] PS got file from %s\n", name.c_str(), from.c_str());
@@ -632,6 +636,7 @@ struct WorkerDaemon {
     }
     double lr = config.at("optimizer").at("nesterov").at("learning_rate").as_double();
     double mu_ = config.at("optimizer").at("nesterov").at("momentum").as_double();
+    const bool weighted = config.get_or("weighted_aggregation", Json(false)).as_bool();
     SafeTensors momentum;  // persists across rounds (parameter_server.rs:393-398)
     SafeTensors cum;       // cumulative sum of updates: joiner catch-up state
     bool have_momentum = false;
@@ -669,6 +674,7 @@ struct WorkerDaemon {
     while (!job->cancelled) {
       // wait until every current member has pushed this round's update
       std::vector<std::string> files;
+      std::vector<double> file_weights;
       std::vector<std::string> targets;
       {
         std::unique_lock<std::mutex> lk(job->agg_mu);
@@ -687,14 +693,19 @@ struct WorkerDaemon {
           serve_joiners();
           continue;
         }
-        for (auto& m : job->members) files.push_back(job->round_files.at(m));
+        for (auto& m : job->members) {
+          files.push_back(job->round_files.at(m).first);
+          file_weights.push_back(job->round_files.at(m).second);
+        }
         targets.assign(job->members.begin(), job->members.end());
         job->round_files.clear();  // stale non-member leftovers dropped too
       }
       fprintf(stderr, "[%s] PS round: %zu files\n", name.c_str(), files.size());
       // running average + outer Nesterov (shared ps_math.h; golden-tested
-      // vs torch SGD(nesterov=True) in tests/test_control_plane.py)
-      SafeTensors avg = ps_average(files);
+      // vs torch SGD(nesterov=True) in tests/test_control_plane.py);
+      // weighted_aggregation weights each delta by its sample count
+      SafeTensors avg = weighted ? ps_weighted_average(files, file_weights)
+                                 : ps_average(files);
       if (!have_momentum) {
         momentum = ps_zeros_like(avg);
         cum = momentum;  // zeros, same shapes

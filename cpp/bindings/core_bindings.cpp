@@ -268,6 +268,12 @@ PYBIND11_MODULE(_core, m) {
   m.def("select_requests", &select_requests);
 
   // parameter-server file pipeline (the production code path, for golden tests)
+  m.def("ps_weighted_average_files",
+        [](const std::vector<std::string>& delta_files, const std::vector<double>& w,
+           const std::string& out) {
+          SafeTensors avg = ps_weighted_average(delta_files, w);
+          avg.save(out);
+        });
   m.def("ps_aggregate_files",
         [](const std::vector<std::string>& delta_files, const std::string& momentum_io,
            const std::string& update_out, double lr, double mu) {

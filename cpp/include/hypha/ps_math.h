@@ -28,6 +28,35 @@ inline SafeTensors ps_average(const std::vector<std::string>& files) {
   return avg;
 }
 
+// Weighted mean: avg <- sum(w_i * delta_i) / sum(w_i). The reference notes
+// sample-count weighting as a missing TODO (parameter_server.rs:192-193);
+// with capacity-proportional batch sizes a 2x-batch worker's pseudo-gradient
+// represents 2x the samples, so weighting by contributed samples is the
+// principled mean. Opt-in (job config "weighted_aggregation"): the unweighted
+// path stays default for reference parity.
+inline SafeTensors ps_weighted_average(const std::vector<std::string>& files,
+                                       const std::vector<double>& weights) {
+  SafeTensors avg = SafeTensors::load(files[0]);
+  double wsum = weights[0];
+  for (auto& [nm, tm] : avg.tensors)
+    for (int64_t e = 0; e < tm.numel(); ++e)
+      avg.set_elem(tm, e, (float)(avg.get_elem(tm, e) * weights[0]));
+  for (size_t i = 1; i < files.size(); ++i) {
+    SafeTensors next = SafeTensors::load(files[i]);
+    for (auto& [nm, tm] : avg.tensors) {
+      auto& ntm = next.tensors.at(nm);
+      for (int64_t e = 0; e < tm.numel(); ++e)
+        avg.set_elem(tm, e,
+                     (float)(avg.get_elem(tm, e) + weights[i] * next.get_elem(ntm, e)));
+    }
+    wsum += weights[i];
+  }
+  for (auto& [nm, tm] : avg.tensors)
+    for (int64_t e = 0; e < tm.numel(); ++e)
+      avg.set_elem(tm, e, (float)(avg.get_elem(tm, e) / wsum));
+  return avg;
+}
+
 // m <- mu*m + g ; update = lr*(mu*m + g)   (parameter_server.rs:386-446)
 // momentum is updated in place; returns the update tensors.
 inline SafeTensors ps_nesterov(const SafeTensors& g, SafeTensors& momentum, double lr,

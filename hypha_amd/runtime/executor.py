@@ -116,9 +116,11 @@ def main() -> int:
     while not done:
         # ---- inner loop: train until the scheduler's counter is exhausted ----
         remaining = None
+        round_samples = 0
         while remaining is None or remaining > 0:
             ids, labels = next(batch_iter)
             loss = worker.train_step(ids, labels)
+            round_samples += batch_size
             if remaining is not None:
                 remaining -= 1
             resp = session.send_status({"kind": "status", "batch_size": batch_size})
@@ -133,7 +135,8 @@ def main() -> int:
         delta = (worker.fp.master - worker.fp.theta0).cpu()
         fname = f"{round_idx}_local_gradients.safetensors"
         save_file({"delta": delta}, os.path.join(args.work_dir, fname))
-        session.send_resource(updates_ref, fname)
+        # sample count rides with the push for optional weighted aggregation
+        session.send_resource({**updates_ref, "samples": round_samples}, fname)
 
         # ---- wait for the aggregated Nesterov update, merge ----
         ev = updates_q.get(timeout=600)

@@ -416,3 +416,24 @@ def test_slice_tracker_properties():
         assert 0 <= idx < num_slices
 
     check()
+
+
+def test_ps_weighted_average(tmp_path):
+    """Sample-count-weighted pseudo-gradient mean (the reference leaves this
+    as TODO at parameter_server.rs:192-193; here it backs the heterogeneous
+    batch path): avg = sum(w_i d_i) / sum(w_i)."""
+    import torch
+    from safetensors.torch import load_file, save_file
+
+    a = torch.tensor([1.0, 2.0, 3.0])
+    b = torch.tensor([4.0, -2.0, 0.0])
+    fa, fb = str(tmp_path / "a.safetensors"), str(tmp_path / "b.safetensors")
+    save_file({"d": a}, fa)
+    save_file({"d": b}, fb)
+    out = str(tmp_path / "avg.safetensors")
+    core.ps_weighted_average_files([fa, fb], [2.0, 1.0], out)
+    got = load_file(out)["d"]
+    torch.testing.assert_close(got, (2 * a + b) / 3)
+    # equal weights == plain mean
+    core.ps_weighted_average_files([fa, fb], [5.0, 5.0], out)
+    torch.testing.assert_close(load_file(out)["d"], (a + b) / 2)

@@ -479,7 +479,8 @@ def test_heterogeneous_batch_sizes(binaries, tmp_path):
         cfg.write_text(
             '{"model": "llama-tiny", "dataset": "synth", "num_workers": 2,'
             ' "update_rounds": 2, "avg_samples_between_updates": 12,'
-            ' "batch_size": 2, "seq_len": 128, "inner_lr": 0.001}'
+            ' "batch_size": 2, "seq_len": 128, "inner_lr": 0.001,'
+            ' "weighted_aggregation": true}'
         )
         sched = subprocess.Popen(
             [str(BIN / "hypha-scheduler"), "--name", "scheduler",
