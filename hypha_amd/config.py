@@ -48,6 +48,7 @@ class JobConfig:
     checkpoint_dir: str = ""
     checkpoint_every_rounds: int = 0  # 0 = disabled
     max_batch_size: int = 600  # cap for capacity-proportional batches
+    weighted_aggregation: bool = False  # weight deltas by sample counts
     job_type: str = "diloco"  # diloco | generate (inference dispatch)
     max_new_tokens: int = 16  # generate mode
     num_batches: int = 2  # generate mode: batches per worker
