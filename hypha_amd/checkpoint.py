@@ -19,10 +19,32 @@ MANIFEST = "manifest.json"
 
 
 def save_checkpoint(worker, out_dir: str) -> None:
-    """Save a DiLoCoWorker's (or LeanDiLoCoWorker's) full training state."""
+    """Save a DiLoCoWorker's (or LeanDiLoCoWorker's) full training state.
+
+    Crash-safe: everything is written into `<out_dir>.tmp` first (manifest
+    last), then swapped in with the previous checkpoint briefly parked at
+    `<out_dir>.bak` — `load_checkpoint` falls back to `.bak` if a crash
+    struck mid-swap, so a partial save never destroys the last good state.
+    """
+    out_dir = out_dir.rstrip("/")
+    tmp = out_dir + ".tmp"
+    import shutil
+
+    shutil.rmtree(tmp, ignore_errors=True)
+    if hasattr(worker, "fp"):
+        _save_full(worker, tmp)
+    else:  # lean engine
+        _save_lean(worker, tmp)
+    bak = out_dir + ".bak"
+    shutil.rmtree(bak, ignore_errors=True)
+    if os.path.isdir(out_dir):
+        os.rename(out_dir, bak)
+    os.rename(tmp, out_dir)
+    shutil.rmtree(bak, ignore_errors=True)
+
+
+def _save_full(worker, out_dir: str) -> None:
     os.makedirs(out_dir, exist_ok=True)
-    if not hasattr(worker, "fp"):  # lean engine
-        return _save_lean(worker, out_dir)
     fp = worker.fp
     save_file({"theta_global": fp.theta0.cpu()}, os.path.join(out_dir, "0_global_weights.safetensors"))
     state = {
@@ -50,6 +72,7 @@ def save_checkpoint(worker, out_dir: str) -> None:
 
 
 def _save_lean(worker, out_dir: str) -> None:
+    os.makedirs(out_dir, exist_ok=True)
     save_file({"theta_global": worker.theta0_host.clone()},
               os.path.join(out_dir, "0_global_weights.safetensors"))
     save_file(
@@ -79,6 +102,11 @@ def _save_lean(worker, out_dir: str) -> None:
 
 def load_checkpoint(worker, ckpt_dir: str) -> dict:
     """Restore a DiLoCoWorker's state in place; returns the manifest."""
+    ckpt_dir = ckpt_dir.rstrip("/")
+    if not os.path.exists(os.path.join(ckpt_dir, MANIFEST)) and os.path.exists(
+        os.path.join(ckpt_dir + ".bak", MANIFEST)
+    ):
+        ckpt_dir = ckpt_dir + ".bak"  # crash mid-swap: last good checkpoint
     with open(os.path.join(ckpt_dir, MANIFEST)) as f:
         manifest = json.load(f)
     if manifest.get("format", "").startswith("hypha_amd.checkpoint.lean"):

@@ -61,7 +61,9 @@ def main() -> int:
 
     # resume from a previous run's checkpoint if one exists
     resume_round = 0
-    if os.path.exists(os.path.join(ckpt_dir, "manifest.json")):
+    if os.path.exists(os.path.join(ckpt_dir, "manifest.json")) or os.path.exists(
+        os.path.join(ckpt_dir + ".bak", "manifest.json")
+    ):
         from hypha_amd import checkpoint as ckpt_mod
 
         manifest = ckpt_mod.load_checkpoint(worker, ckpt_dir)

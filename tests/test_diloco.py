@@ -82,3 +82,32 @@ def test_checkpoint_roundtrip(tmp_path):
     l1 = w.train_step(ids.clone(), labels.clone())
     l2 = w2.train_step(ids, labels)
     assert abs(l1 - l2) < 1e-6
+
+
+def test_checkpoint_crash_safety(tmp_path):
+    """Atomic checkpoint swap: a crash mid-swap leaves `.bak`, which
+    load_checkpoint falls back to; a second save cleans it up."""
+    import os
+
+    from hypha_amd.parallel import DiLoCoConfig, DiLoCoWorker, InnerOptConfig
+
+    torch.manual_seed(3)
+    m = models.build("llama-tiny")
+    cfg = DiLoCoConfig(h=4, inner=InnerOptConfig(warmup_steps=0, schedule="constant"))
+    w = DiLoCoWorker(m, cfg, device=torch.device("cpu"))
+    w.round = 7
+    ck = str(tmp_path / "ckpt")
+    checkpoint.save_checkpoint(w, ck)
+    assert not os.path.exists(ck + ".tmp") and not os.path.exists(ck + ".bak")
+
+    # simulate a crash after parking the old dir but before the new rename
+    os.rename(ck, ck + ".bak")
+    m2 = models.build("llama-tiny")
+    w2 = DiLoCoWorker(m2, cfg, device=torch.device("cpu"))
+    manifest = checkpoint.load_checkpoint(w2, ck)
+    assert manifest["round"] == 7  # restored from .bak
+
+    # a fresh save replaces everything and removes the bak
+    checkpoint.save_checkpoint(w, ck)
+    assert os.path.exists(os.path.join(ck, "manifest.json"))
+    assert not os.path.exists(ck + ".bak")
