@@ -62,6 +62,17 @@ static py::object json_to_py(const hypha::Json& j) {
   return d;
 }
 
+// shared callback holder whose LAST release may happen on a non-Python
+// thread (e.g. unsubscribe erasing the map, or a server thread dropping its
+// copy): the deleter re-acquires the GIL before touching the refcount.
+static std::shared_ptr<py::function> make_cb_holder(py::function cb) {
+  return std::shared_ptr<py::function>(new py::function(std::move(cb)),
+                                       [](py::function* f) {
+                                         py::gil_scoped_acquire gil;
+                                         delete f;
+                                       });
+}
+
 PYBIND11_MODULE(_core, m) {
   m.doc() = "hypha_amd C++ control-plane core";
 
@@ -323,7 +334,7 @@ PYBIND11_MODULE(_core, m) {
            [](Node& n, const std::string& type, py::function cb) {
              // shared_ptr so std::function copies in server threads never
              // touch Python refcounts without the GIL
-             auto cbp = std::make_shared<py::function>(std::move(cb));
+             auto cbp = make_cb_holder(std::move(cb));
              n.on(type, [cbp](const std::string& from, const Json& body) -> Json {
                py::gil_scoped_acquire gil;
                try {
@@ -357,7 +368,7 @@ PYBIND11_MODULE(_core, m) {
            })
       .def("subscribe",
            [](Node& n, const std::string& topic, py::function cb) {
-             auto cbp = std::make_shared<py::function>(std::move(cb));
+             auto cbp = make_cb_holder(std::move(cb));
              n.subscribe(topic, [cbp](const std::string& from, const Json& data) {
                py::gil_scoped_acquire gil;
                try {
@@ -368,6 +379,8 @@ PYBIND11_MODULE(_core, m) {
                }
              });
            })
+      .def("unsubscribe", &Node::unsubscribe, py::arg("topic"),
+           py::call_guard<py::gil_scoped_release>())
       .def("kv_put",
            [](Node& n, const std::string& k, py::object v) {
              Json j = py_to_json(v);
@@ -439,7 +452,7 @@ PYBIND11_MODULE(_core, m) {
            py::arg("payload") = py::none())
       .def("on_blob",
            [](Node& n, const std::string& type, py::function cb) {
-             auto cbp = std::make_shared<py::function>(std::move(cb));
+             auto cbp = make_cb_holder(std::move(cb));
              n.on_stream(type, [cbp](const std::string& from, const Json& header,
                                      MsgSocket& sock) {
                size_t size = (size_t)header.get_or("size", Json((int64_t)0)).as_int();

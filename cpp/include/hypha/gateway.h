@@ -92,6 +92,9 @@ class Gateway {
       } else if (kind == "subscribe") {
         std::lock_guard<std::mutex> lk(mu_);
         subs_[msg->at("topic").as_string()].insert(peer);
+      } else if (kind == "unsubscribe") {
+        std::lock_guard<std::mutex> lk(mu_);
+        subs_[msg->at("topic").as_string()].erase(peer);
       } else if (kind == "request") {
         Json resp;
         resp["kind"] = "response";

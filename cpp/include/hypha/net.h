@@ -109,6 +109,7 @@ class Node {
   // gateway-brokered pub/sub + KV
   void publish(const std::string& topic, const Json& data);
   void subscribe(const std::string& topic, std::function<void(const std::string&, const Json&)> cb);
+  void unsubscribe(const std::string& topic);  // gossipsub.rs:232-300 parity
   void kv_put(const std::string& key, const Json& value);
   std::optional<Json> kv_get(const std::string& key);
 

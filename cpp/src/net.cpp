@@ -505,6 +505,18 @@ void Node::subscribe(const std::string& topic,
   gw_sock_->send_json(sub);
 }
 
+void Node::unsubscribe(const std::string& topic) {
+  {
+    std::lock_guard<std::mutex> lk(mu_);
+    subs_.erase(topic);
+  }
+  Json un;
+  un["kind"] = "unsubscribe";
+  un["topic"] = topic;
+  std::lock_guard<std::mutex> lk(gw_mu_);
+  gw_sock_->send_json(un);
+}
+
 void Node::kv_put(const std::string& key, const Json& value) {
   Json b;
   b["key"] = key;

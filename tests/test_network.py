@@ -161,3 +161,24 @@ def test_bandwidth_accounting(cluster):
     # both grow because client and server share the process)
     assert after["outbound_bytes"] - before["outbound_bytes"] > 2 * len(payload)
     assert after["inbound_bytes"] - before["inbound_bytes"] > 2 * len(payload)
+
+
+def test_unsubscribe_stops_delivery(cluster):
+    """Gossipsub parity (gossipsub.rs:232-300): after unsubscribe the peer
+    receives no further topic messages; other subscribers are unaffected."""
+    a = cluster("ua")
+    b = cluster("ub")
+    c = cluster("uc")
+    got_b, got_c = [], []
+    b.subscribe("topic-x", lambda frm, d: got_b.append(d))
+    c.subscribe("topic-x", lambda frm, d: got_c.append(d))
+    time.sleep(0.05)
+    a.publish("topic-x", {"n": 1})
+    time.sleep(0.2)
+    assert got_b == [{"n": 1}] and got_c == [{"n": 1}]
+    b.unsubscribe("topic-x")
+    time.sleep(0.05)
+    a.publish("topic-x", {"n": 2})
+    time.sleep(0.2)
+    assert got_b == [{"n": 1}]  # no further delivery
+    assert got_c == [{"n": 1}, {"n": 2}]  # others unaffected
