@@ -432,6 +432,18 @@ Json Node::request(const std::string& peer, const std::string& type, const Json&
   size_t colon = addr->rfind(':');
   check_dialable(peer, addr->substr(0, colon));
   int fd = tcp_connect(addr->substr(0, colon), std::stoi(addr->substr(colon + 1)), timeout_s);
+  if (fd < 0) {
+    // stale cache: the peer may have restarted on a new port — re-resolve
+    {
+      std::lock_guard<std::mutex> lk(mu_);
+      addr_cache_.erase(peer);
+    }
+    addr = resolve(peer);
+    if (!addr) throw std::runtime_error("unknown peer " + peer);
+    colon = addr->rfind(':');
+    check_dialable(peer, addr->substr(0, colon));
+    fd = tcp_connect(addr->substr(0, colon), std::stoi(addr->substr(colon + 1)), timeout_s);
+  }
   if (fd < 0) throw std::runtime_error("peer unreachable: " + peer);
   SSL* pssl = nullptr;
   if (tls_) {
@@ -463,6 +475,17 @@ std::unique_ptr<MsgSocket> Node::open_stream(const std::string& peer, const std:
   size_t colon = addr->rfind(':');
   check_dialable(peer, addr->substr(0, colon));
   int fd = tcp_connect(addr->substr(0, colon), std::stoi(addr->substr(colon + 1)), 30.0);
+  if (fd < 0) {  // stale cache: re-resolve once (peer may have restarted)
+    {
+      std::lock_guard<std::mutex> lk(mu_);
+      addr_cache_.erase(peer);
+    }
+    addr = resolve(peer);
+    if (!addr) throw std::runtime_error("unknown peer " + peer);
+    colon = addr->rfind(':');
+    check_dialable(peer, addr->substr(0, colon));
+    fd = tcp_connect(addr->substr(0, colon), std::stoi(addr->substr(colon + 1)), 30.0);
+  }
   if (fd < 0) throw std::runtime_error("peer unreachable: " + peer);
   struct timeval tv = {600, 0};
   setsockopt(fd, SOL_SOCKET, SO_RCVTIMEO, &tv, sizeof tv);

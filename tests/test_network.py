@@ -182,3 +182,24 @@ def test_unsubscribe_stops_delivery(cluster):
     time.sleep(0.2)
     assert got_b == [{"n": 1}]  # no further delivery
     assert got_c == [{"n": 1}, {"n": 2}]  # others unaffected
+
+
+def test_reregistration_latest_wins(cluster, request):
+    """A restarted node re-registers under the same name; the registry
+    routes subsequent requests to the NEW instance (kad put overwrite)."""
+    a = cluster("rr-client")
+    b1 = cluster("rr-server")
+    b1.on("who", lambda frm, body: {"gen": 1})
+    assert a.request("rr-server", "who", {})["gen"] == 1
+    b1.stop()
+    b2 = cluster("rr-server")
+    b2.on("who", lambda frm, body: {"gen": 2})
+    assert a.request("rr-server", "who", {})["gen"] == 2
+
+
+def test_kv_overwrite_returns_latest(cluster):
+    a = cluster("kv-a")
+    a.kv_put("k", {"v": 1})
+    a.kv_put("k", {"v": 2})
+    assert a.kv_get("k") == {"v": 2}
+    assert a.kv_get("missing") is None
