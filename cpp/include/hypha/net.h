@@ -128,6 +128,7 @@ class Node {
   void accept_loop();
   void handle_conn(int fd, SSL* ssl = nullptr);
   void gateway_listen_loop();
+  bool gateway_connect();  // (re)connect + register + replay subscriptions
   Json gateway_request(const std::string& type, const Json& body);
 
   std::string name_, gw_host_;

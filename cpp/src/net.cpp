@@ -241,32 +241,52 @@ void Node::start(int port) {
   running_ = true;
   accept_thread_ = std::thread([this] { accept_loop(); });
   if (gw_port_ > 0) {
-    // persistent broker connection: register + receive pub/sub events
-    int fd = tcp_connect(gw_host_, gw_port_, 10.0);
-    if (fd < 0) throw std::runtime_error("node: cannot reach gateway");
-    // long receive timeout on the event connection
-    struct timeval tv = {86400, 0};
-    setsockopt(fd, SOL_SOCKET, SO_RCVTIMEO, &tv, sizeof tv);
-    SSL* gssl = nullptr;
-    if (tls_) {
-      gssl = tls_->wrap(fd, false);
-      if (!gssl) {
-        ::close(fd);
-        throw std::runtime_error("node: gateway TLS handshake failed");
-      }
-    }
-    gw_sock_ = std::make_unique<MsgSocket>(fd, gssl);
-    Json reg;
-    reg["kind"] = "register";
-    reg["peer"] = name_;
-    reg["addr"] = std::string("127.0.0.1:") + std::to_string(port_);
-    gw_sock_->send_json(reg);
-    // synchronous ack: the registry entry is visible before start() returns
-    auto ack = gw_sock_->recv_json();
-    if (!ack || ack->get_or("kind", Json("")).as_string() != "registered")
-      throw std::runtime_error("node: gateway registration failed");
+    if (!gateway_connect())
+      throw std::runtime_error(
+          "node: gateway registration failed (unreachable, TLS handshake "
+          "rejected, or no ack)");
     gw_thread_ = std::thread([this] { gateway_listen_loop(); });
   }
+}
+
+// Establish (or re-establish) the persistent broker connection:
+// connect + optional TLS + register + synchronous ack.
+bool Node::gateway_connect() {
+  int fd = tcp_connect(gw_host_, gw_port_, 10.0);
+  if (fd < 0) return false;
+  // long receive timeout on the event connection
+  struct timeval tv = {86400, 0};
+  setsockopt(fd, SOL_SOCKET, SO_RCVTIMEO, &tv, sizeof tv);
+  SSL* gssl = nullptr;
+  if (tls_) {
+    gssl = tls_->wrap(fd, false);
+    if (!gssl) {
+      ::close(fd);
+      return false;
+    }
+  }
+  auto sock = std::make_unique<MsgSocket>(fd, gssl);
+  Json reg;
+  reg["kind"] = "register";
+  reg["peer"] = name_;
+  reg["addr"] = std::string("127.0.0.1:") + std::to_string(port_);
+  if (!sock->send_json(reg)) return false;
+  // synchronous ack: the registry entry is visible before we proceed
+  auto ack = sock->recv_json();
+  if (!ack || ack->get_or("kind", Json("")).as_string() != "registered") return false;
+  {
+    std::lock_guard<std::mutex> lk(gw_mu_);
+    gw_sock_ = std::move(sock);
+    // replay topic subscriptions (the broker lost them with the connection)
+    std::lock_guard<std::mutex> lk2(mu_);
+    for (auto& [topic, _] : subs_) {
+      Json sub;
+      sub["kind"] = "subscribe";
+      sub["topic"] = topic;
+      gw_sock_->send_json(sub);
+    }
+  }
+  return true;
 }
 
 void Node::stop() {
@@ -276,7 +296,12 @@ void Node::stop() {
     ::close(listen_fd_);
     listen_fd_ = -1;
   }
-  if (gw_sock_) gw_sock_->close_now();
+  {
+    // gw_mu_ serializes against the reader thread swapping the socket in
+    // gateway_connect() during a reconnection
+    std::lock_guard<std::mutex> lk(gw_mu_);
+    if (gw_sock_) gw_sock_->close_now();
+  }
   if (accept_thread_.joinable()) accept_thread_.join();
   if (gw_thread_.joinable()) gw_thread_.join();
 }
@@ -363,7 +388,18 @@ void Node::handle_conn(int fd, SSL* ssl) {
 void Node::gateway_listen_loop() {
   while (running_) {
     auto msg = gw_sock_->recv_json();
-    if (!msg) break;
+    if (!msg) {
+      // broker connection lost (gateway restart?): keep re-registering until
+      // it is back — the reference's libp2p re-establishes gossipsub the
+      // same way. Registration also refreshes our addr record.
+      while (running_ && !gateway_connect()) {
+        struct timespec ts = {0, 500 * 1000 * 1000};
+        nanosleep(&ts, nullptr);
+      }
+      if (!running_) break;
+      fprintf(stderr, "[net:%s] gateway reconnected\n", name_.c_str());
+      continue;
+    }
     try {
     if (msg->get_or("kind", Json("")).as_string() == "pub") {
       std::string topic = msg->at("topic").as_string();
