@@ -622,6 +622,15 @@ int main(int argc, char** argv) {
         status_cv.wait_for(lk, std::chrono::seconds(3));
         break;
       }
+      if (!psw->alive && !fsm_done) {
+        // the aggregator is a single point per job (as in the reference):
+        // its loss mid-run is fatal — fail fast instead of hanging until
+        // executor timeouts
+        lk.unlock();
+        running = false;
+        monitor.join();
+        return shutdown_fail("parameter server lost mid-run: aborting job");
+      }
       status_cv.wait_for(lk, std::chrono::milliseconds(500));
     }
   }

@@ -584,3 +584,82 @@ def test_cluster_inference_jobs(binaries, tmp_path):
                 p.wait(timeout=10)
             except Exception:
                 pass
+
+
+@pytest.mark.timeout(300)
+def test_parameter_server_loss_fails_fast(binaries, tmp_path):
+    """The aggregate executor is a single point per job (as in the
+    reference): killing its worker mid-run must abort the job with a clear
+    error within ~a lease period, not hang until executor timeouts."""
+    from hypha_amd.data.synthetic import write_slice_files
+
+    data_dir = tmp_path / "slices"
+    write_slice_files(str(data_dir), "synth", num_slices=4, samples_per_slice=16,
+                      vocab_size=512, seq_len=128)
+    gw_port = free_port()
+    env = dict(os.environ, PYTHONPATH=str(REPO))
+    procs = []
+    worker_procs = {}
+
+    def spawn(name, cmd):
+        log = open(tmp_path / f"{name}.log", "w")
+        p = subprocess.Popen(cmd, cwd=REPO, env=env, stdout=log, stderr=log)
+        procs.append(p)
+        return p
+
+    try:
+        spawn("gateway", [str(BIN / "hypha-gateway"), "--port", str(gw_port)])
+        time.sleep(0.3)
+        spawn("data", [str(BIN / "hypha-data"), "--name", "data-node",
+                       "--gateway-host", "127.0.0.1", "--gateway-port", str(gw_port),
+                       "--dataset", "synth", "--dataset-path", str(data_dir)])
+        exec_cmd = (f"{sys.executable} -m hypha_amd.runtime.executor "
+                    "--socket {SOCKET_PATH} --work-dir {WORK_DIR} --job {JOB_JSON}")
+        for i in range(3):
+            worker_procs[i] = spawn(
+                f"worker{i}",
+                [str(BIN / "hypha-worker"), "--name", f"worker-{i}",
+                 "--gateway-host", "127.0.0.1", "--gateway-port", str(gw_port),
+                 "--exec-cmd", exec_cmd, "--work-root", str(tmp_path / f"work{i}")])
+        time.sleep(0.5)
+        cfg = tmp_path / "job.json"
+        cfg.write_text(
+            '{"model": "llama-tiny", "dataset": "synth", "num_workers": 2,'
+            ' "update_rounds": 50, "avg_samples_between_updates": 8,'
+            ' "batch_size": 2, "seq_len": 128, "inner_lr": 0.001}'
+        )
+        sched = subprocess.Popen(
+            [str(BIN / "hypha-scheduler"), "--name", "scheduler",
+             "--gateway-host", "127.0.0.1", "--gateway-port", str(gw_port),
+             "--config", str(cfg)],
+            cwd=REPO, env=env, stdout=subprocess.PIPE,
+            stderr=open(tmp_path / "sched.log", "w"), text=True,
+        )
+        procs.append(sched)
+
+        # find the PS worker (its log shows "PS start"), then kill it
+        ps_idx = None
+        deadline = time.time() + 90
+        while ps_idx is None and time.time() < deadline:
+            time.sleep(1)
+            for i in range(3):
+                log = tmp_path / f"worker{i}.log"
+                if log.exists() and "PS start" in log.read_text():
+                    ps_idx = i
+                    break
+        assert ps_idx is not None, "no PS job started"
+        worker_procs[ps_idx].send_signal(signal.SIGKILL)
+
+        out, _ = sched.communicate(timeout=120)
+        assert sched.returncode == 1, (out, sched.returncode)
+        sched_log = (tmp_path / "sched.log").read_text()
+        assert "parameter server lost" in sched_log, sched_log[-2000:]
+    finally:
+        for p in procs:
+            if p.poll() is None:
+                p.send_signal(signal.SIGKILL)
+        for p in procs:
+            try:
+                p.wait(timeout=10)
+            except Exception:
+                pass
