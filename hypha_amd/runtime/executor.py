@@ -77,11 +77,37 @@ def main() -> int:
         try:
             for ev in session.receive():
                 updates_q.put(ev)
-        except Exception as e:  # bridge socket closes at job teardown;
-            # anything else is worth a trace on stderr (thread would die silently)
+        except Exception as e:  # bridge gone = worker daemon died: abort
+            # instead of training on as an orphan (grad-release of the whole
+            # job happens via the daemon's lease machinery, not us)
             print(f"# sse listener ended: {type(e).__name__}: {e}", file=sys.stderr)
+        updates_q.put({"__bridge_closed__": True})
 
     threading.Thread(target=listen, daemon=True).start()
+
+    def watchdog():
+        # the worker daemon owns this job: if its bridge stops answering,
+        # exit rather than train on as an orphan (daemon SIGKILL leaves no
+        # one to SIGTERM us; the reference has the same subprocess model)
+        import httpx as _httpx
+
+        misses = 0
+        while True:
+            import time as _time
+
+            _time.sleep(15)
+            try:
+                _httpx.Client(transport=_httpx.HTTPTransport(uds=args.socket),
+                              timeout=5.0).get("http://bridge/openapi.json")
+                misses = 0
+            except Exception:
+                misses += 1
+                if misses >= 2:
+                    print("# bridge unreachable: exiting orphaned executor",
+                          file=sys.stderr)
+                    os._exit(3)
+
+    threading.Thread(target=watchdog, daemon=True).start()
 
     # infinite batch stream over scheduler-assigned slices (utils.py fetch_data)
     def batches():
@@ -101,6 +127,8 @@ def main() -> int:
         # global weights via the PS's cumulative offset before training
         print("[executor] joining: waiting for global offset", flush=True)
         ev = updates_q.get(timeout=600)
+        if ev.get("__bridge_closed__"):
+            raise RuntimeError("bridge connection lost (worker daemon gone)")
         offset = load_file(ev["path"])
         if "delta" in offset:
             worker.fp.theta0.add_(offset["delta"].to(worker.fp.theta0.device))
@@ -142,6 +170,8 @@ def main() -> int:
 
         # ---- wait for the aggregated Nesterov update, merge ----
         ev = updates_q.get(timeout=600)
+        if ev.get("__bridge_closed__"):
+            raise RuntimeError("bridge connection lost (worker daemon gone)")
         u = load_file(ev["path"])["delta"].to(worker.fp.theta0.device)
         worker.fp.theta0.add_(u)
         worker.fp.master.copy_(worker.fp.theta0)

@@ -51,7 +51,8 @@ def test_cluster_diloco_round_trip(binaries, tmp_path):
     def spawn(name, cmd):
         log = open(tmp_path / f"{name}.log", "w")
         logs[name] = tmp_path / f"{name}.log"
-        p = subprocess.Popen(cmd, cwd=REPO, env=env, stdout=log, stderr=log)
+        p = subprocess.Popen(cmd, cwd=REPO, env=env, stdout=log, stderr=log,
+                             start_new_session=True)
         procs.append(p)
         return p
 
@@ -93,8 +94,12 @@ def test_cluster_diloco_round_trip(binaries, tmp_path):
         assert sched.returncode == 0
     finally:
         for p in procs:
-            if p.poll() is None:
-                p.send_signal(signal.SIGKILL)
+            try:  # kill the whole group: daemons AND their executors —
+                # even if the daemon itself already died (orphan children)
+                os.killpg(p.pid, signal.SIGKILL)
+            except (ProcessLookupError, PermissionError):
+                if p.poll() is None:
+                    p.send_signal(signal.SIGKILL)
         for p in procs:
             try:
                 p.wait(timeout=10)
@@ -119,7 +124,8 @@ def test_cluster_worker_kill_and_rejoin(binaries, tmp_path):
 
     def spawn(name, cmd):
         log = open(tmp_path / f"{name}.log", "w")
-        p = subprocess.Popen(cmd, cwd=REPO, env=env, stdout=log, stderr=log)
+        p = subprocess.Popen(cmd, cwd=REPO, env=env, stdout=log, stderr=log,
+                             start_new_session=True)
         procs.append(p)
         return p
 
@@ -152,6 +158,7 @@ def test_cluster_worker_kill_and_rejoin(binaries, tmp_path):
              "--config", str(cfg)],
             cwd=REPO, env=env, stdout=subprocess.PIPE,
             stderr=open(tmp_path / "sched.log", "w"), text=True,
+            start_new_session=True,
         )
         procs.append(sched)
 
@@ -174,8 +181,12 @@ def test_cluster_worker_kill_and_rejoin(binaries, tmp_path):
         assert "lost" in sched_log and "joined" in sched_log, sched_log[-3000:]
     finally:
         for p in procs:
-            if p.poll() is None:
-                p.send_signal(signal.SIGKILL)
+            try:  # kill the whole group: daemons AND their executors —
+                # even if the daemon itself already died (orphan children)
+                os.killpg(p.pid, signal.SIGKILL)
+            except (ProcessLookupError, PermissionError):
+                if p.poll() is None:
+                    p.send_signal(signal.SIGKILL)
         for p in procs:
             try:
                 p.wait(timeout=10)
@@ -199,7 +210,8 @@ def test_cluster_checkpointing(binaries, tmp_path):
 
     def spawn(name, cmd):
         log = open(tmp_path / f"{name}.log", "w")
-        p = subprocess.Popen(cmd, cwd=REPO, env=env, stdout=log, stderr=log)
+        p = subprocess.Popen(cmd, cwd=REPO, env=env, stdout=log, stderr=log,
+                             start_new_session=True)
         procs.append(p)
         return p
 
@@ -240,8 +252,12 @@ def test_cluster_checkpointing(binaries, tmp_path):
         assert (ckpt_root / "optimizer_state.safetensors").exists()
     finally:
         for p in procs:
-            if p.poll() is None:
-                p.send_signal(signal.SIGKILL)
+            try:  # kill the whole group: daemons AND their executors —
+                # even if the daemon itself already died (orphan children)
+                os.killpg(p.pid, signal.SIGKILL)
+            except (ProcessLookupError, PermissionError):
+                if p.poll() is None:
+                    p.send_signal(signal.SIGKILL)
         for p in procs:
             try:
                 p.wait(timeout=10)
@@ -276,7 +292,8 @@ def test_cluster_diloco_over_mtls(binaries, tmp_path):
 
     def spawn(name, cmd):
         log = open(tmp_path / f"{name}.log", "w")
-        p = subprocess.Popen(cmd, cwd=REPO, env=env, stdout=log, stderr=log)
+        p = subprocess.Popen(cmd, cwd=REPO, env=env, stdout=log, stderr=log,
+                             start_new_session=True)
         procs.append(p)
         return p
 
@@ -313,8 +330,12 @@ def test_cluster_diloco_over_mtls(binaries, tmp_path):
         assert "Job is completed." in out, (tmp_path / "sched.log").read_text()[-3000:]
     finally:
         for p in procs:
-            if p.poll() is None:
-                p.send_signal(signal.SIGKILL)
+            try:  # kill the whole group: daemons AND their executors —
+                # even if the daemon itself already died (orphan children)
+                os.killpg(p.pid, signal.SIGKILL)
+            except (ProcessLookupError, PermissionError):
+                if p.poll() is None:
+                    p.send_signal(signal.SIGKILL)
         for p in procs:
             try:
                 p.wait(timeout=10)
@@ -338,7 +359,8 @@ def test_scheduler_death_reclaims_workers(binaries, tmp_path):
 
     def spawn(name, cmd):
         log = open(tmp_path / f"{name}.log", "w")
-        p = subprocess.Popen(cmd, cwd=REPO, env=env, stdout=log, stderr=log)
+        p = subprocess.Popen(cmd, cwd=REPO, env=env, stdout=log, stderr=log,
+                             start_new_session=True)
         procs.append(p)
         return p
 
@@ -421,8 +443,12 @@ def test_scheduler_death_reclaims_workers(binaries, tmp_path):
             out, (tmp_path / "sched2.log").read_text()[-3000:])
     finally:
         for p in procs:
-            if p.poll() is None:
-                p.send_signal(signal.SIGKILL)
+            try:  # kill the whole group: daemons AND their executors —
+                # even if the daemon itself already died (orphan children)
+                os.killpg(p.pid, signal.SIGKILL)
+            except (ProcessLookupError, PermissionError):
+                if p.poll() is None:
+                    p.send_signal(signal.SIGKILL)
         for p in procs:
             try:
                 p.wait(timeout=10)
@@ -446,7 +472,8 @@ def test_heterogeneous_batch_sizes(binaries, tmp_path):
 
     def spawn(name, cmd):
         log = open(tmp_path / f"{name}.log", "w")
-        p = subprocess.Popen(cmd, cwd=REPO, env=env, stdout=log, stderr=log)
+        p = subprocess.Popen(cmd, cwd=REPO, env=env, stdout=log, stderr=log,
+                             start_new_session=True)
         procs.append(p)
         return p
 
@@ -488,6 +515,7 @@ def test_heterogeneous_batch_sizes(binaries, tmp_path):
              "--config", str(cfg)],
             cwd=REPO, env=env, stdout=subprocess.PIPE,
             stderr=open(tmp_path / "sched.log", "w"), text=True,
+            start_new_session=True,
         )
         procs.append(sched)
         out, _ = sched.communicate(timeout=240)
@@ -503,8 +531,12 @@ def test_heterogeneous_batch_sizes(binaries, tmp_path):
         assert "bs=2" in logs, logs[-1500:]
     finally:
         for p in procs:
-            if p.poll() is None:
-                p.send_signal(signal.SIGKILL)
+            try:  # kill the whole group: daemons AND their executors —
+                # even if the daemon itself already died (orphan children)
+                os.killpg(p.pid, signal.SIGKILL)
+            except (ProcessLookupError, PermissionError):
+                if p.poll() is None:
+                    p.send_signal(signal.SIGKILL)
         for p in procs:
             try:
                 p.wait(timeout=10)
@@ -529,7 +561,8 @@ def test_cluster_inference_jobs(binaries, tmp_path):
 
     def spawn(name, cmd):
         log = open(tmp_path / f"{name}.log", "w")
-        p = subprocess.Popen(cmd, cwd=REPO, env=env, stdout=log, stderr=log)
+        p = subprocess.Popen(cmd, cwd=REPO, env=env, stdout=log, stderr=log,
+                             start_new_session=True)
         procs.append(p)
         return p
 
@@ -561,6 +594,7 @@ def test_cluster_inference_jobs(binaries, tmp_path):
              "--config", str(cfg)],
             cwd=REPO, env=env, stdout=subprocess.PIPE,
             stderr=open(tmp_path / "sched.log", "w"), text=True,
+            start_new_session=True,
         )
         procs.append(sched)
         out, _ = sched.communicate(timeout=240)
@@ -577,8 +611,12 @@ def test_cluster_inference_jobs(binaries, tmp_path):
         assert toks.shape[-1] == 128 + 8
     finally:
         for p in procs:
-            if p.poll() is None:
-                p.send_signal(signal.SIGKILL)
+            try:  # kill the whole group: daemons AND their executors —
+                # even if the daemon itself already died (orphan children)
+                os.killpg(p.pid, signal.SIGKILL)
+            except (ProcessLookupError, PermissionError):
+                if p.poll() is None:
+                    p.send_signal(signal.SIGKILL)
         for p in procs:
             try:
                 p.wait(timeout=10)
@@ -603,7 +641,8 @@ def test_parameter_server_loss_fails_fast(binaries, tmp_path):
 
     def spawn(name, cmd):
         log = open(tmp_path / f"{name}.log", "w")
-        p = subprocess.Popen(cmd, cwd=REPO, env=env, stdout=log, stderr=log)
+        p = subprocess.Popen(cmd, cwd=REPO, env=env, stdout=log, stderr=log,
+                             start_new_session=True)
         procs.append(p)
         return p
 
@@ -634,6 +673,7 @@ def test_parameter_server_loss_fails_fast(binaries, tmp_path):
              "--config", str(cfg)],
             cwd=REPO, env=env, stdout=subprocess.PIPE,
             stderr=open(tmp_path / "sched.log", "w"), text=True,
+            start_new_session=True,
         )
         procs.append(sched)
 
@@ -656,8 +696,12 @@ def test_parameter_server_loss_fails_fast(binaries, tmp_path):
         assert "parameter server lost" in sched_log, sched_log[-2000:]
     finally:
         for p in procs:
-            if p.poll() is None:
-                p.send_signal(signal.SIGKILL)
+            try:  # kill the whole group: daemons AND their executors —
+                # even if the daemon itself already died (orphan children)
+                os.killpg(p.pid, signal.SIGKILL)
+            except (ProcessLookupError, PermissionError):
+                if p.poll() is None:
+                    p.send_signal(signal.SIGKILL)
         for p in procs:
             try:
                 p.wait(timeout=10)
