@@ -384,12 +384,25 @@ int main(int argc, char** argv) {
             (long long)num_workers);
     {
       std::unique_lock<std::mutex> lk(status_mu);
-      status_cv.wait(lk, [&] {
-        int64_t done = 0;
-        for (auto& [id, st] : job_status)
+      while (true) {
+        int64_t done = 0, failed = 0;
+        for (auto& [id, st] : job_status) {
           if (st == "completed") ++done;
-        return done >= num_workers;
-      });
+          if (st == "failed" || st == "cancelled") ++failed;
+        }
+        if (done >= num_workers) break;
+        bool lost = false;
+        for (auto& w : iworkers) lost = lost || !w->alive;
+        if (lost || failed > 0) {
+          lk.unlock();
+          running = false;
+          for (auto& t : renewers) t.join();
+          fprintf(stderr, "[scheduler] inference worker lost/failed: aborting\n");
+          node.stop();
+          return 1;
+        }
+        status_cv.wait_for(lk, std::chrono::milliseconds(500));
+      }
     }
     printf("Job is completed.\n");
     fflush(stdout);
