@@ -88,3 +88,61 @@ def test_param_kv_push_pull(tmp_path):
         for p in procs:
             p.wait(timeout=10)
         log.close()
+
+
+@pytest.mark.timeout(120)
+def test_param_kv_concurrent_stress(tmp_path):
+    """Concurrent pushes/pulls from multiple threads: the per-connection
+    server threads and the versioned store must not race or drop data."""
+    import threading
+
+    gw_port = free_port()
+    env = dict(os.environ, PYTHONPATH=str(REPO))
+    log = open(tmp_path / "daemons.log", "w")
+    procs = [subprocess.Popen([str(BIN / "hypha-gateway"), "--port", str(gw_port)],
+                              cwd=REPO, env=env, stdout=log, stderr=log)]
+    time.sleep(0.3)
+    procs.append(subprocess.Popen(
+        [str(BIN / "hypha-worker"), "--name", "kv-stress",
+         "--gateway-host", "127.0.0.1", "--gateway-port", str(gw_port),
+         "--exec-cmd", "true", "--work-root", str(tmp_path / "work")],
+        cwd=REPO, env=env, stdout=log, stderr=log))
+    client = core.Node("kv-stress-client", "127.0.0.1", gw_port)
+    try:
+        deadline = time.time() + 10
+        while time.time() < deadline:
+            try:
+                client.start(0)
+                break
+            except RuntimeError:
+                time.sleep(0.2)
+
+        errors = []
+
+        def hammer(tid):
+            try:
+                payload = bytes([tid]) * 4096
+                for i in range(20):
+                    r, _ = client.stream_call(
+                        "kv-stress", "param_push",
+                        {"job": "stress", "key": f"k{tid}"}, payload)
+                    assert r["ok"] and r["version"] == i + 1, r
+                r, blob = client.stream_call(
+                    "kv-stress", "param_pull", {"job": "stress", "key": f"k{tid}"})
+                assert r["found"] and r["version"] == 20 and blob == payload
+            except Exception as e:  # surfaced in the main thread
+                errors.append((tid, repr(e)))
+
+        threads = [threading.Thread(target=hammer, args=(t,)) for t in range(8)]
+        for t in threads:
+            t.start()
+        for t in threads:
+            t.join(timeout=60)
+        assert not errors, errors
+    finally:
+        client.stop()
+        for p in procs:
+            p.terminate()
+        for p in procs:
+            p.wait(timeout=10)
+        log.close()
