@@ -176,6 +176,27 @@ PYBIND11_MODULE(_core, m) {
       .def_readonly("count", &RunningMean::count);
 
   // ---- simulation ----
+  // direct simulation entry for property tests: workers = {peer: (batch,
+  // mean_ms_or_None)}; returns a Projection
+  m.def("simulate_project",
+        [](const std::map<std::string, std::pair<int64_t, py::object>>& workers,
+           int64_t counter, double time_cap_ms, int64_t update_cap) {
+          std::map<std::string, WorkerEntry> entries;
+          for (auto& [peer, be] : workers) {
+            WorkerEntry e;
+            e.peer = peer;
+            e.batch_size = be.first;
+            if (!be.second.is_none()) e.stat.record(be.second.cast<double>());
+            entries[peer] = e;
+          }
+          BasicSimulation sim;
+          sim.time_cap_ms = time_cap_ms;
+          sim.update_cap = update_cap;
+          return sim.project(entries, counter);
+        },
+        py::arg("workers"), py::arg("counter"), py::arg("time_cap_ms") = 10000.0,
+        py::arg("update_cap") = 3);
+
   py::class_<Projection>(m, "Projection")
       .def_readonly("time_ms", &Projection::time_ms)
       .def_readonly("remaining", &Projection::remaining)

@@ -437,3 +437,52 @@ def test_ps_weighted_average(tmp_path):
     # equal weights == plain mean
     core.ps_weighted_average_files([fa, fb], [5.0, 5.0], out)
     torch.testing.assert_close(load_file(out)["d"], (a + b) / 2)
+
+
+def test_simulation_matches_brute_force():
+    """Property test for the discrete-event projection (simulation.rs:16-68
+    semantics): compared against an independent brute-force reimplementation
+    over random worker sets."""
+    import heapq
+
+    from hypothesis import given, settings, strategies as st
+
+    def brute(workers, counter, time_cap, update_cap):
+        # workers: {peer: (batch, mean_or_None)}
+        heap = [((m if m is not None else 1e18), peer)
+                for peer, (b, m) in workers.items()]
+        heapq.heapify(heap)
+        batches = {p: 0 for p in workers}
+        time_ms, capped = 0.0, False
+        remaining = counter
+        while remaining > 0:
+            t, peer = heapq.heappop(heap)
+            if t > time_cap:
+                capped = True
+                break
+            if batches[peer] + 1 > update_cap:
+                capped = True
+                break
+            batches[peer] += 1
+            b, m = workers[peer]
+            remaining -= b
+            time_ms = t
+            heapq.heappush(heap, (t + (m if m is not None else 1e18), peer))
+        return time_ms, remaining, batches, capped
+
+    mean = st.one_of(st.none(), st.floats(1.0, 5000.0))
+    worker = st.tuples(st.integers(1, 8), mean)
+
+    @settings(max_examples=150, deadline=None)
+    @given(st.dictionaries(st.sampled_from(["a", "b", "c", "d"]), worker,
+                           min_size=1, max_size=4),
+           st.integers(1, 64))
+    def check(workers, counter):
+        p = core.simulate_project(workers, counter)
+        t, rem, batches, capped = brute(workers, counter, 10000.0, 3)
+        assert p.remaining == rem
+        assert p.capped == capped
+        assert p.batches_per_worker == batches
+        assert p.time_ms == pytest.approx(t)
+
+    check()
