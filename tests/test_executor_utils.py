@@ -106,3 +106,49 @@ def test_tokenizer_preprocessor_from_local_artifact(tmp_path):
     out = run(text=["hello world", "world"])
     assert out["input_ids"].shape[0] == 2
     assert out["input_ids"][0].tolist()[:2] == [0, 1]
+
+
+def test_lr_schedules_vs_transformers_oracle():
+    """Shape parity against the SAME schedule implementations the reference
+    executor uses (utils.py:90-106 -> transformers get_*_schedule_with_warmup):
+    warmup ramp and decay trajectory agree within tolerance (our cosine/linear
+    floors at min_lr_frac, the oracle at 0 — compare above the floor)."""
+    import torch as t
+    from transformers import (get_cosine_schedule_with_warmup,
+                              get_linear_schedule_with_warmup)
+
+    warmup, total = 10, 100
+
+    def oracle_lrs(make):
+        opt = t.optim.SGD([t.nn.Parameter(t.zeros(1))], lr=1.0)
+        sched = make(opt)
+        out = []
+        for _ in range(total):
+            out.append(opt.param_groups[0]["lr"])
+            opt.step()
+            sched.step()
+        return out
+
+    cos = oracle_lrs(lambda o: get_cosine_schedule_with_warmup(o, warmup, total))
+    lin = oracle_lrs(lambda o: get_linear_schedule_with_warmup(o, warmup, total))
+
+    cfg = InnerOptConfig(lr=1.0, min_lr_frac=0.0)
+    apply_wire_schedule(cfg, {"type": "cosine-with-warmup", "warmup_steps": warmup,
+                              "training_steps": total})
+    ours_cos = [lr_at(cfg, i) for i in range(total)]
+    apply_wire_schedule(cfg, {"type": "linear-with-warmup", "warmup_steps": warmup,
+                              "training_steps": total})
+    ours_lin = [lr_at(cfg, i) for i in range(total)]
+
+    # warmup: both ramp 0->1 linearly over `warmup` steps (off-by-one
+    # conventions differ by <= 1/warmup)
+    for i in range(warmup):
+        assert abs(ours_cos[i] - cos[i]) <= 1.0 / warmup + 1e-9
+    # decay: same trajectory within a coarse tolerance (conventions differ
+    # in the (step - warmup)/(total - warmup) vs step/total denominators)
+    for i in range(warmup + 1, total, 7):
+        assert abs(ours_cos[i] - cos[i]) < 0.08, (i, ours_cos[i], cos[i])
+        assert abs(ours_lin[i] - lin[i]) < 0.08, (i, ours_lin[i], lin[i])
+    # endpoints agree exactly: peak 1.0 after warmup, ~0 at the end
+    assert ours_cos[-1] < 0.01 and cos[-1] < 0.01
+    assert ours_lin[-1] < 0.02 and lin[-1] < 0.02
