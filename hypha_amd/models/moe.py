@@ -63,7 +63,6 @@ class MoEMLP(nn.Module):
         self.w_down = nn.Parameter(torch.empty(e, h, f))
         for w in (self.w_gate, self.w_up, self.w_down):
             nn.init.normal_(w, std=cfg.init_std)
-        self.aux_loss = torch.zeros(())
 
     def forward(self, x):
         cfg = self.cfg
@@ -75,11 +74,15 @@ class MoEMLP(nn.Module):
         topv, topi = probs.topk(cfg.top_k, dim=-1)  # [T, K]
         topv = (topv / topv.sum(-1, keepdim=True)).to(x.dtype)
 
-        # load-balancing auxiliary loss (Switch-style)
+        # load-balancing auxiliary loss (Switch-style). Returned through the
+        # forward output (not stashed on the module) so it stays connected to
+        # autograd under activation checkpointing, where this code runs in a
+        # no-grad first pass and a grad-enabled recompute.
+        aux = xf.new_zeros(())
         if self.training:
             me = probs.mean(0)
             ce = F.one_hot(topi[:, 0], cfg.n_experts).float().mean(0)
-            self.aux_loss = cfg.router_aux_loss_coef * cfg.n_experts * (me * ce).sum()
+            aux = (cfg.router_aux_loss_coef * cfg.n_experts * (me * ce).sum()).to(xf.dtype)
 
         # group tokens by expert: one dense GEMM chain per expert
         flat_expert = topi.reshape(-1)  # [T*K]
@@ -130,7 +133,7 @@ class MoEMLP(nn.Module):
         weights = topv.reshape(-1)[order].unsqueeze(1)
         out = torch.zeros_like(xf)
         out.index_add_(0, flat_tok[order], grouped_out * weights)
-        return out.reshape(b, s, h)
+        return out.reshape(b, s, h), aux
 
 
 class MoEBlock(nn.Module):
@@ -143,8 +146,9 @@ class MoEBlock(nn.Module):
 
     def forward(self, x, cos, sin):
         x = x + self.attn(self.attn_norm(x), cos, sin)
-        x = x + self.mlp(self.mlp_norm(x))
-        return x
+        mlp_out, aux = self.mlp(self.mlp_norm(x))
+        x = x + mlp_out
+        return x, aux
 
 
 class MoEForCausalLM(nn.Module):
@@ -162,17 +166,24 @@ class MoEForCausalLM(nn.Module):
         nn.init.normal_(self.lm_head.weight, std=cfg.init_std)
 
     def forward(self, input_ids, labels=None):
+        import torch.utils.checkpoint
+
         x = self.embed(input_ids)
+        aux_total = None
         for blk in self.blocks:
-            x = blk(x, self.rope_cos, self.rope_sin)
+            if self.cfg.gradient_checkpointing and self.training:
+                x, aux = torch.utils.checkpoint.checkpoint(
+                    blk, x, self.rope_cos, self.rope_sin, use_reentrant=False)
+            else:
+                x, aux = blk(x, self.rope_cos, self.rope_sin)
+            aux_total = aux if aux_total is None else aux_total + aux
         x = self.norm(x)
         logits = self.lm_head(x)
         if labels is None:
             return logits
         loss = ops.cross_entropy_loss(logits[:, :-1], labels[:, 1:])
-        if self.training:
-            for blk in self.blocks:
-                loss = loss + blk.mlp.aux_loss
+        if self.training and aux_total is not None:
+            loss = loss + aux_total.float()
         return loss
 
 
