@@ -486,3 +486,43 @@ def test_simulation_matches_brute_force():
         assert p.time_ms == pytest.approx(t)
 
     check()
+
+
+def test_aggregator_properties():
+    """Property test for the auction buy side (allocator.rs:276-419
+    semantics): the finalized set is exactly the `count` cheapest
+    per-weighted-unit acceptable offers, deduped per worker, with the
+    deadline shrunk to the earliest offer expiry."""
+    from hypothesis import given, settings, strategies as st
+
+    offer = st.tuples(
+        st.sampled_from([f"w{i}" for i in range(6)]),   # worker
+        st.floats(0.1, 12.0),                            # price
+        st.floats(0.5, 4.0),                             # gpu
+        st.floats(1.0, 100.0),                           # expiry (abs)
+    )
+
+    @settings(max_examples=150, deadline=None)
+    @given(st.lists(offer, min_size=1, max_size=12), st.integers(1, 4))
+    def check(raw, count):
+        price_max = 5.0
+        agg = core.GreedyOfferAggregator(count, core.PriceRange(bid=1.0, max=price_max), 1000.0)
+        ev = core.WeightedResourceEvaluator()
+        accepted, seen, deadline = [], set(), 1000.0
+        for i, (w, price, gpu, exp) in enumerate(raw):
+            r = core.Resources(gpu, 1, 1, 1)
+            agg.add(core.WorkerOffer(f"l{i}", "rq", w, price, r, exp), 0.0)
+            if price <= price_max and w not in seen:
+                seen.add(w)
+                accepted.append((ev.score(price, r), i, w))
+                if exp > 0.0:
+                    deadline = min(deadline, exp)
+        best = agg.finalize()
+        model = sorted(accepted, key=lambda t: t[0])[:count]
+        assert [o.worker for o in best] == [w for _, _, w in model]
+        assert agg.deadline == pytest.approx(deadline)
+        # every returned offer is within the price ceiling and unique
+        assert len({o.worker for o in best}) == len(best)
+        assert all(o.price <= price_max for o in best)
+
+    check()
