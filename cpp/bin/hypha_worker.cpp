@@ -713,11 +713,7 @@ struct WorkerDaemon {
       }
       SafeTensors update = ps_nesterov(avg, momentum, lr, mu_);
       // cumulative offset for joiners: theta_global = theta_init + cum
-      for (auto& [nm, tm] : update.tensors) {
-        auto& ctm = cum.tensors.at(nm);
-        for (int64_t e = 0; e < tm.numel(); ++e)
-          cum.set_elem(ctm, e, cum.get_elem(ctm, e) + update.get_elem(tm, e));
-      }
+      ps_add_(cum, update);
       std::string upath = job->work_dir + "/update.safetensors";
       update.save(upath);
       fprintf(stderr, "[%s] PS broadcasting to %zu targets\n", name.c_str(), targets.size());
