@@ -145,6 +145,9 @@ def main() -> int:
     round_idx = resume_round
     while not done:
         # ---- inner loop: train until the scheduler's counter is exhausted ----
+        import time as _time
+
+        round_t0 = _time.perf_counter()
         remaining = None
         round_samples = 0
         while remaining is None or remaining > 0:
@@ -156,8 +159,12 @@ def main() -> int:
             resp = session.send_status({"kind": "status", "batch_size": batch_size})
             if resp.get("kind") == "schedule-update" and remaining is None:
                 remaining = int(resp.get("counter", 0))
+        round_s = max(1e-9, _time.perf_counter() - round_t0)
         session.send_status(
-            {"kind": "metrics", "round": round_idx, "metrics": {"loss": loss}}
+            {"kind": "metrics", "round": round_idx,
+             "metrics": {"loss": loss,
+                         "tokens_per_sec": round_samples * seq_len / round_s,
+                         "samples": round_samples}}
         )
 
         # ---- extract and push the pseudo-gradient ----
