@@ -97,3 +97,17 @@ def test_generate_sampling_reproducible():
     a = m.generate(ids, max_new_tokens=5, temperature=0.8, top_k=20, seed=7)
     b = m.generate(ids, max_new_tokens=5, temperature=0.8, top_k=20, seed=7)
     torch.testing.assert_close(a, b)
+
+
+def test_gpt2_generate():
+    """Every registry family exposes the same generate surface (inference
+    jobs may name any model); GPT-2 decodes by full recompute."""
+    torch.manual_seed(4)
+    m = models.build("gpt2-tiny")
+    ids = torch.randint(0, 512, (2, 8))
+    out = m.generate(ids, max_new_tokens=5)
+    assert out.shape == (2, 13)
+    assert torch.equal(out[:, :8], ids)
+    a = m.generate(ids, max_new_tokens=4, temperature=0.9, top_k=10, seed=3)
+    b = m.generate(ids, max_new_tokens=4, temperature=0.9, top_k=10, seed=3)
+    assert torch.equal(a, b)  # seeded sampling is reproducible
