@@ -186,6 +186,32 @@ class MoEForCausalLM(nn.Module):
             loss = loss + aux_total.float()
         return loss
 
+    @torch.no_grad()
+    def generate(self, input_ids: torch.Tensor, max_new_tokens: int = 32,
+                 temperature: float = 0.0, top_k: int = 0,
+                 seed: int | None = None) -> torch.Tensor:
+        """Full-recompute decoding (same sampling surface as the Llama
+        family so inference jobs can run any registry model)."""
+        self.eval()
+        gen = None
+        if seed is not None:
+            gen = torch.Generator(device=input_ids.device).manual_seed(seed)
+        tokens = input_ids
+        for _ in range(max_new_tokens):
+            window = tokens[:, -self.cfg.max_seq_len:]
+            logits = self(window)[:, -1]
+            if temperature <= 0:
+                nxt = logits.argmax(-1, keepdim=True)
+            else:
+                logits = logits / temperature
+                if top_k > 0:
+                    kth = logits.topk(top_k, dim=-1).values[..., -1, None]
+                    logits = logits.masked_fill(logits < kth, float("-inf"))
+                probs = torch.softmax(logits.float(), dim=-1)
+                nxt = torch.multinomial(probs, 1, generator=gen)
+            tokens = torch.cat([tokens, nxt], dim=1)
+        return tokens
+
 
 def build_model(name: str, **overrides) -> MoEForCausalLM:
     cfg = PRESETS[name]

@@ -111,3 +111,11 @@ def test_gpt2_generate():
     a = m.generate(ids, max_new_tokens=4, temperature=0.9, top_k=10, seed=3)
     b = m.generate(ids, max_new_tokens=4, temperature=0.9, top_k=10, seed=3)
     assert torch.equal(a, b)  # seeded sampling is reproducible
+
+
+def test_moe_generate():
+    torch.manual_seed(5)
+    m = models.build("moe-tiny")
+    ids = torch.randint(0, 512, (1, 8))
+    out = m.generate(ids, max_new_tokens=4)
+    assert out.shape == (1, 12) and torch.equal(out[:, :8], ids)
