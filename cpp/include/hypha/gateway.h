@@ -108,9 +108,15 @@ class Gateway {
       }
     }
     if (!peer.empty()) {
+      // identity-guarded cleanup: if the peer already re-registered over a
+      // NEW connection, this (old) handler must not erase the fresh entry
+      // or its re-subscribed topics
       std::lock_guard<std::mutex> lk(mu_);
-      peers_.erase(peer);
-      for (auto& [_, set] : subs_) set.erase(peer);
+      auto it = peers_.find(peer);
+      if (it != peers_.end() && it->second == sock) {
+        peers_.erase(it);
+        for (auto& [_, set] : subs_) set.erase(peer);
+      }
     }
   }
 
