@@ -159,6 +159,15 @@ struct WorkerDaemon {
     node.on_stream("param_push",
                    [&](const std::string&, const Json& header, MsgSocket& sock) {
                      int64_t size = header.at("size").as_int();
+                     // bound the allocation like recv_json's 256MB frame cap:
+                     // a single hostile header must not exhaust worker memory
+                     if (size < 0 || size > (int64_t)(256u << 20)) {
+                       Json r;
+                       r["ok"] = false;
+                       r["error"] = "param_push size exceeds cap";
+                       sock.send_json(r);
+                       return;
+                     }
                      std::string data((size_t)size, '\0');
                      if (size > 0 && !sock.recv_raw(data.data(), (size_t)size)) return;
                      Json r;

@@ -75,37 +75,64 @@ class Gateway {
   void handle_conn(int fd, SSL* ssl = nullptr) {
     auto sock = std::make_shared<MsgSocket>(fd, ssl);
     std::string peer;  // set once registered (persistent connection)
-    while (running_) {
-      auto msg = sock->recv_json();
-      if (!msg) break;
-      std::string kind = msg->get_or("kind", Json("")).as_string();
-      if (kind == "register") {
-        peer = msg->at("peer").as_string();
-        {
+    // With mTLS the verified cert CN is the only identity we trust: the
+    // self-claimed 'peer'/'from' fields must match it (reference model,
+    // rfc/2025-05-30_mtls.md — PeerId is derived from the cert key).
+    const std::string cn = ssl ? sock->peer_identity() : std::string();
+    // A malformed message (Json::at throws) must drop THIS connection, not
+    // std::terminate the whole broker: the per-connection thread is detached.
+    try {
+      while (running_) {
+        auto msg = sock->recv_json();
+        if (!msg) break;
+        std::string kind = msg->get_or("kind", Json("")).as_string();
+        if (kind == "register") {
+          std::string claimed = msg->at("peer").as_string();
+          if (ssl && !cn.empty() && claimed != cn) {
+            Json nak;
+            nak["kind"] = "error";
+            nak["error"] = "peer name does not match certificate CN";
+            sock->send_json(nak);
+            break;
+          }
+          peer = claimed;
+          {
+            std::lock_guard<std::mutex> lk(mu_);
+            peers_[peer] = sock;
+            kv_["addr:" + peer] = msg->at("addr");
+          }
+          Json ack;
+          ack["kind"] = "registered";
+          sock->send_json(ack);
+        } else if (kind == "subscribe") {
           std::lock_guard<std::mutex> lk(mu_);
-          peers_[peer] = sock;
-          kv_["addr:" + peer] = msg->at("addr");
+          subs_[msg->at("topic").as_string()].insert(peer);
+        } else if (kind == "unsubscribe") {
+          std::lock_guard<std::mutex> lk(mu_);
+          subs_[msg->at("topic").as_string()].erase(peer);
+        } else if (kind == "request") {
+          std::string from = msg->get_or("from", Json("")).as_string();
+          if (ssl && !cn.empty() && !from.empty() && from != cn) {
+            Json resp;
+            resp["kind"] = "response";
+            resp["ok"] = false;
+            resp["error"] = "from does not match certificate CN";
+            sock->send_json(resp);
+            continue;
+          }
+          Json resp;
+          resp["kind"] = "response";
+          resp["ok"] = true;
+          resp["body"] = handle_request(from,
+                                        msg->get_or("type", Json("")).as_string(),
+                                        msg->get_or("body", Json(JsonObject{})));
+          sock->send_json(resp);
+        } else {
+          break;
         }
-        Json ack;
-        ack["kind"] = "registered";
-        sock->send_json(ack);
-      } else if (kind == "subscribe") {
-        std::lock_guard<std::mutex> lk(mu_);
-        subs_[msg->at("topic").as_string()].insert(peer);
-      } else if (kind == "unsubscribe") {
-        std::lock_guard<std::mutex> lk(mu_);
-        subs_[msg->at("topic").as_string()].erase(peer);
-      } else if (kind == "request") {
-        Json resp;
-        resp["kind"] = "response";
-        resp["ok"] = true;
-        resp["body"] = handle_request(msg->get_or("from", Json("")).as_string(),
-                                      msg->get_or("type", Json("")).as_string(),
-                                      msg->get_or("body", Json(JsonObject{})));
-        sock->send_json(resp);
-      } else {
-        break;
       }
+    } catch (const std::exception&) {
+      // drop the offending connection; registry cleanup below still runs
     }
     if (!peer.empty()) {
       // identity-guarded cleanup: if the peer already re-registered over a

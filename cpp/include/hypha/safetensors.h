@@ -26,6 +26,16 @@ struct TensorMeta {
   }
 };
 
+inline size_t dtype_size(const std::string& dtype) {
+  if (dtype == "F32" || dtype == "I32" || dtype == "U32") return 4;
+  if (dtype == "BF16" || dtype == "F16" || dtype == "I16" || dtype == "U16") return 2;
+  if (dtype == "F64" || dtype == "I64" || dtype == "U64") return 8;
+  if (dtype == "F8_E4M3" || dtype == "F8_E5M2" || dtype == "I8" || dtype == "U8" ||
+      dtype == "BOOL")
+    return 1;
+  throw std::runtime_error("safetensors: unsupported dtype " + dtype);
+}
+
 class SafeTensors {
  public:
   std::map<std::string, TensorMeta> tensors;  // ordered by name
@@ -55,11 +65,39 @@ class SafeTensors {
       if (name == "__metadata__") continue;
       TensorMeta tm;
       tm.dtype = meta.at("dtype").as_string();
-      for (auto& d : meta.at("shape").as_array()) tm.shape.push_back(d.as_int());
-      tm.begin = (size_t)meta.at("data_offsets").as_array()[0].as_int();
-      tm.end = (size_t)meta.at("data_offsets").as_array()[1].as_int();
+      for (auto& d : meta.at("shape").as_array()) {
+        int64_t dim = d.as_int();
+        if (dim < 0) throw std::runtime_error("safetensors: negative dim in " + name);
+        tm.shape.push_back(dim);
+      }
+      int64_t b = meta.at("data_offsets").as_array()[0].as_int();
+      int64_t e = meta.at("data_offsets").as_array()[1].as_int();
+      if (b < 0 || e < b)
+        throw std::runtime_error("safetensors: bad data_offsets in " + name);
+      tm.begin = (size_t)b;
+      tm.end = (size_t)e;
+      // The byte span must match shape*itemsize exactly: the PS math in
+      // ps_math.h walks numel() elements via raw pointers at tm.begin, so a
+      // malformed file pushed by a peer must be rejected here, not there.
+      size_t isz = dtype_size(tm.dtype);
+      uint64_t want = (uint64_t)tm.numel() * isz;
+      if ((uint64_t)(tm.end - tm.begin) != want)
+        throw std::runtime_error("safetensors: data_offsets span " +
+                                 std::to_string(tm.end - tm.begin) +
+                                 " != numel*itemsize " + std::to_string(want) +
+                                 " for " + name);
       if (tm.end > data_size) data_size = tm.end;
       st.tensors[name] = tm;
+    }
+    // Bound total data by what the file can actually hold.
+    long pos = ftell(f);
+    if (pos >= 0 && fseek(f, 0, SEEK_END) == 0) {
+      long fend = ftell(f);
+      fseek(f, pos, SEEK_SET);
+      if (fend >= 0 && (uint64_t)data_size > (uint64_t)(fend - pos)) {
+        fclose(f);
+        throw std::runtime_error("safetensors: offsets exceed data section size");
+      }
     }
     st.data.resize(data_size);
     if (data_size && fread(st.data.data(), 1, data_size, f) != data_size) {

@@ -342,12 +342,29 @@ void Node::accept_loop() {
 
 void Node::handle_conn(int fd, SSL* ssl) {
   MsgSocket sock(fd, ssl);
+  // With mTLS the only identity we trust is the verified certificate CN
+  // (reference model: PeerId is derived from the cert key,
+  // rfc/2025-05-30_mtls.md). A self-declared `from` that contradicts it is a
+  // spoof attempt and the message is rejected; handlers always see the CN.
+  const std::string verified = ssl ? sock.peer_identity() : std::string();
   while (running_) {
     auto msg = sock.recv_json();
     if (!msg) break;
     std::string kind = msg->get_or("kind", Json("")).as_string();
     std::string type = msg->get_or("type", Json("")).as_string();
     std::string from = msg->get_or("from", Json("")).as_string();
+    if (!verified.empty()) {
+      if (!from.empty() && from != verified) {
+        Json resp;
+        resp["kind"] = "response";
+        resp["ok"] = false;
+        resp["error"] = "from '" + from + "' does not match certificate CN '" +
+                        verified + "'";
+        sock.send_json(resp);
+        break;
+      }
+      from = verified;
+    }
     if (kind == "request") {
       Handler h;
       {

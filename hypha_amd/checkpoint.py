@@ -18,6 +18,33 @@ from safetensors.torch import load_file, save_file
 MANIFEST = "manifest.json"
 
 
+def _fsync_dir_tree(path: str) -> None:
+    """fsync every file in `path` and the directory itself so the rename that
+    follows publishes fully-durable contents (a power loss after the swap must
+    not leave a truncated 'good' checkpoint)."""
+    for root, _dirs, files in os.walk(path):
+        for name in files:
+            fd = os.open(os.path.join(root, name), os.O_RDONLY)
+            try:
+                os.fsync(fd)
+            finally:
+                os.close(fd)
+        fd = os.open(root, os.O_RDONLY)
+        try:
+            os.fsync(fd)
+        finally:
+            os.close(fd)
+
+
+def _fsync_parent(path: str) -> None:
+    parent = os.path.dirname(os.path.abspath(path)) or "."
+    fd = os.open(parent, os.O_RDONLY)
+    try:
+        os.fsync(fd)
+    finally:
+        os.close(fd)
+
+
 def save_checkpoint(worker, out_dir: str) -> None:
     """Save a DiLoCoWorker's (or LeanDiLoCoWorker's) full training state.
 
@@ -35,11 +62,13 @@ def save_checkpoint(worker, out_dir: str) -> None:
         _save_full(worker, tmp)
     else:  # lean engine
         _save_lean(worker, tmp)
+    _fsync_dir_tree(tmp)
     bak = out_dir + ".bak"
     shutil.rmtree(bak, ignore_errors=True)
     if os.path.isdir(out_dir):
         os.rename(out_dir, bak)
     os.rename(tmp, out_dir)
+    _fsync_parent(out_dir)  # make both renames durable before dropping .bak
     shutil.rmtree(bak, ignore_errors=True)
 
 

@@ -526,3 +526,51 @@ def test_aggregator_properties():
         assert all(o.price <= price_max for o in best)
 
     check()
+
+
+def test_ps_rejects_malformed_safetensors(tmp_path):
+    """A pseudo-gradient file whose data_offsets disagree with shape*itemsize
+    (or overrun the data section) is rejected at parse time instead of being
+    walked out-of-bounds by the PS math (ADVICE high, safetensors.h)."""
+    import json as pyjson
+    import struct
+
+    import pytest as _pytest
+
+    def craft(path, dtype, shape, begin, end, payload_len):
+        hdr = pyjson.dumps(
+            {"t": {"dtype": dtype, "shape": shape, "data_offsets": [begin, end]}}
+        ).encode()
+        with open(path, "wb") as f:
+            f.write(struct.pack("<Q", len(hdr)))
+            f.write(hdr)
+            f.write(b"\x00" * payload_len)
+
+    mom = str(tmp_path / "m.safetensors")
+    upd = str(tmp_path / "u.safetensors")
+
+    # span (8 bytes) != numel*itemsize (40 bytes)
+    bad1 = str(tmp_path / "bad1.safetensors")
+    craft(bad1, "F32", [10], 0, 8, 8)
+    with _pytest.raises(RuntimeError, match="safetensors"):
+        core.ps_aggregate_files([bad1], mom, upd, 0.7, 0.9)
+
+    # offsets overrun the actual data section in the file
+    bad2 = str(tmp_path / "bad2.safetensors")
+    craft(bad2, "F32", [10], 0, 40, 4)
+    with _pytest.raises(RuntimeError, match="safetensors"):
+        core.ps_aggregate_files([bad2], mom, upd, 0.7, 0.9)
+
+    # negative begin
+    bad3 = str(tmp_path / "bad3.safetensors")
+    craft(bad3, "F32", [1], -4, 0, 4)
+    with _pytest.raises(RuntimeError, match="safetensors"):
+        core.ps_aggregate_files([bad3], mom, upd, 0.7, 0.9)
+
+    # a well-formed file still aggregates
+    import torch
+    from safetensors.torch import save_file
+
+    good = str(tmp_path / "good.safetensors")
+    save_file({"t": torch.ones(10)}, good)
+    core.ps_aggregate_files([good], mom, upd, 1.0, 0.0)
