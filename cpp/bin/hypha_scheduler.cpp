@@ -38,6 +38,7 @@ static double now_s() {
 struct AllocatedWorker {
   std::string peer;
   std::string lease_id;
+  std::string job_id;  // train job dispatched on this worker
   Resources resources;
   std::atomic<bool> alive{true};
 };
@@ -144,6 +145,11 @@ int main(int argc, char** argv) {
   const int64_t seq_len = cfg.get_or("seq_len", Json(128)).as_int();
   const double worker_bid = cfg.get_or("worker_bid", Json(1.0)).as_double();
   const double worker_max = cfg.get_or("worker_max_price", Json(10.0)).as_double();
+  // sync mode: "ps" = parameter-server star over push streams (WAN path);
+  // "rccl" = scheduler-assigned {rank, world_size, rendezvous} per worker,
+  // outer sync as a bucketed RCCL all-reduce over xGMI (on-node path).
+  const std::string sync_mode = cfg.get_or("sync", Json(std::string("ps"))).as_string();
+  const double rccl_timeout_s = cfg.get_or("rccl_timeout_s", Json(600.0)).as_double();
 
   Node node(name, gw_host, gw_port, tls);
   node.set_exclude_cidrs(exclude_cidrs);
@@ -434,17 +440,27 @@ int main(int argc, char** argv) {
     start_renewal(w);
     train_workers.push_back(w);
   }
+  // ephemeral rendezvous ports for the RCCL group (fresh one per formation)
+  auto pick_port = [&]() {
+    int fd = tcp_listen(0);
+    int p = listen_port(fd);
+    ::close(fd);
+    return p;
+  };
+
+  std::string ps_peer;
+  std::shared_ptr<AllocatedWorker> psw;
+  std::string ps_job_id = "job-ps";
+  if (sync_mode != "rccl") {
   auto ps_offers = allocate(1, {"parameter-server"}, worker_req);
   if (ps_offers.empty()) return shutdown_fail("no parameter server offer");
-  std::string ps_peer = ps_offers[0].worker;
+  ps_peer = ps_offers[0].worker;
 
   // (train workers already renewing)
-  std::vector<std::shared_ptr<AllocatedWorker>> workers;
-  auto psw = std::make_shared<AllocatedWorker>();
+  psw = std::make_shared<AllocatedWorker>();
   psw->peer = ps_peer;
   psw->lease_id = ps_offers[0].id;
   start_renewal(psw);
-  workers.push_back(psw);
   // give renewers a moment to upgrade the offer leases
   std::this_thread::sleep_for(std::chrono::milliseconds(200));
 
@@ -453,7 +469,6 @@ int main(int argc, char** argv) {
   for (auto& w : train_workers) train_peer_names.push_back(Json(w->peer));
 
   // the aggregate job first, so the PS is listening before updates flow
-  std::string ps_job_id = "job-ps";
   {
     Json agg_cfg;
     agg_cfg["num_workers"] = num_workers;
@@ -483,9 +498,13 @@ int main(int argc, char** argv) {
     d["lease"] = psw->lease_id;
     node.request(ps_peer, "dispatch_job", d, 10.0);
   }
+  }  // sync_mode != "rccl"
 
+  // dispatch a train job; rank >= 0 attaches the RCCL rendezvous assignment
+  // (rank/world_size/master addr+port) the executor forms its communicator
+  // from — the control plane bootstrapping the data plane
   auto dispatch_train = [&](std::shared_ptr<AllocatedWorker> w, const std::string& jid,
-                            bool join) {
+                            bool join, int rank = -1, int world = 0, int rdv_port = 0) {
     Json tr;
     tr["model"] = model;
     Json fetch;
@@ -494,13 +513,23 @@ int main(int argc, char** argv) {
     sref["dataset"] = dataset;
     fetch["scheduler"] = sref;
     tr["data"] = fetch;
-    Json updates;
-    Json upeers;
-    upeers["peers"] = JsonArray{Json(ps_peer)};
-    upeers["strategy"] = std::string("all");
-    updates["peers"] = upeers;
-    updates["job"] = ps_job_id;  // tag pushes for the PS job
-    tr["updates"] = updates;
+    if (rank >= 0) {
+      Json rc;
+      rc["rank"] = (int64_t)rank;
+      rc["world_size"] = (int64_t)world;
+      rc["master_addr"] = std::string("127.0.0.1");
+      rc["master_port"] = (int64_t)rdv_port;
+      rc["timeout_s"] = rccl_timeout_s;
+      tr["rccl"] = rc;
+    } else {
+      Json updates;
+      Json upeers;
+      upeers["peers"] = JsonArray{Json(ps_peer)};
+      upeers["strategy"] = std::string("all");
+      updates["peers"] = upeers;
+      updates["job"] = ps_job_id;  // tag pushes for the PS job
+      tr["updates"] = updates;
+    }
     Json adam;
     adam["learning_rate"] = cfg.get_or("inner_lr", Json(4e-4)).as_double();
     Json opt;
@@ -526,15 +555,23 @@ int main(int argc, char** argv) {
     d["job"] = job;
     d["lease"] = w->lease_id;
     node.request(w->peer, "dispatch_job", d, 10.0);
+    w->job_id = jid;
     std::lock_guard<std::mutex> lk(fsm_mu);
     fsm.add_worker(w->peer, wbatch);
   };
 
-  for (size_t i = 0; i < train_workers.size(); ++i)
-    dispatch_train(train_workers[i], std::string("job-train-") + std::to_string(i),
-                   false);
-  fprintf(stderr, "[scheduler] dispatched %lld train jobs + 1 aggregate\n",
-          (long long)num_workers);
+  int rdv_port = sync_mode == "rccl" ? pick_port() : 0;
+  for (size_t i = 0; i < train_workers.size(); ++i) {
+    if (sync_mode == "rccl")
+      dispatch_train(train_workers[i], std::string("job-train-") + std::to_string(i),
+                     false, (int)i, (int)train_workers.size(), rdv_port);
+    else
+      dispatch_train(train_workers[i], std::string("job-train-") + std::to_string(i),
+                     false);
+  }
+  fprintf(stderr, "[scheduler] dispatched %lld train jobs (%s sync)%s\n",
+          (long long)num_workers, sync_mode.c_str(),
+          sync_mode == "rccl" ? "" : " + 1 aggregate");
 
   // --- fault tolerance: detect lost workers (lease renewal failure) and
   //     allocate + join a replacement (BASELINE config 3; the reference only
@@ -567,6 +604,70 @@ int main(int argc, char** argv) {
           std::lock_guard<std::mutex> lk(fsm_mu);
           fsm.remove_worker(w->peer);
           slices.remove_worker(w->peer);
+        }
+        if (sync_mode == "rccl") {
+          // RCCL elastic path: drop the dead rank, optionally admit a
+          // replacement, and re-form the communicator on a fresh rendezvous.
+          // Survivors keep ranks 0..n-1 (rank 0 always a survivor: it seeds
+          // the post-reform state broadcast); a joiner takes the last rank.
+          std::vector<std::shared_ptr<AllocatedWorker>> survivors;
+          {
+            std::lock_guard<std::mutex> lk(tw_mu);
+            train_workers.erase(
+                std::remove_if(train_workers.begin(), train_workers.end(),
+                               [&](auto& x) { return x->peer == w->peer; }),
+                train_workers.end());
+            survivors = train_workers;
+          }
+          std::shared_ptr<AllocatedWorker> joiner;
+          auto offers2 = allocate(1, {"diloco-transformer"}, worker_req);
+          if (!offers2.empty()) {
+            joiner = std::make_shared<AllocatedWorker>();
+            joiner->peer = offers2[0].worker;
+            joiner->lease_id = offers2[0].id;
+            start_renewal(joiner);
+            std::this_thread::sleep_for(std::chrono::milliseconds(200));
+          } else {
+            fprintf(stderr, "[scheduler] no replacement worker available\n");
+          }
+          int port2 = pick_port();
+          int world = (int)survivors.size() + (joiner ? 1 : 0);
+          if (joiner) {
+            std::string jid = "job-train-r" + std::to_string(replace_seq.fetch_add(1));
+            try {
+              dispatch_train(joiner, jid, true, (int)survivors.size(), world, port2);
+              live_workers += 1;
+              {
+                std::lock_guard<std::mutex> lk(tw_mu);
+                train_workers.push_back(joiner);
+              }
+              fprintf(stderr, "[scheduler] replacement %s joins as rank %zu\n",
+                      joiner->peer.c_str(), survivors.size());
+            } catch (const std::exception& e) {
+              fprintf(stderr, "[scheduler] replacement dispatch failed: %s\n",
+                      e.what());
+              joiner.reset();
+              world = (int)survivors.size();
+            }
+          }
+          for (size_t i = 0; i < survivors.size(); ++i) {
+            Json rf;
+            rf["job"] = survivors[i]->job_id;
+            rf["rank"] = (int64_t)i;
+            rf["world_size"] = (int64_t)world;
+            rf["master_addr"] = std::string("127.0.0.1");
+            rf["master_port"] = (int64_t)port2;
+            rf["timeout_s"] = rccl_timeout_s;
+            try {
+              node.request(survivors[i]->peer, "rccl_reform", rf, 10.0);
+            } catch (const std::exception& e) {
+              fprintf(stderr, "[scheduler] reform to %s failed: %s\n",
+                      survivors[i]->peer.c_str(), e.what());
+            }
+          }
+          fprintf(stderr, "[scheduler] re-formed RCCL group: world=%d port=%d\n",
+                  world, port2);
+          continue;
         }
         Json pc;
         pc["job"] = ps_job_id;
@@ -635,7 +736,7 @@ int main(int argc, char** argv) {
         status_cv.wait_for(lk, std::chrono::seconds(3));
         break;
       }
-      if (!psw->alive && !fsm_done) {
+      if (psw && !psw->alive && !fsm_done) {
         // the aggregator is a single point per job (as in the reference):
         // its loss mid-run is fatal — fail fast instead of hanging until
         // executor timeouts

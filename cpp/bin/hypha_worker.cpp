@@ -54,6 +54,7 @@ struct Job {
   std::string lease_id;
   std::string scheduler;
   std::string work_dir;
+  int gpu_id = -1;  // device pinned for this job (-1 = none)
   pid_t pid = -1;
   std::shared_ptr<Bridge> bridge;
   std::thread runner;
@@ -92,6 +93,24 @@ struct WorkerDaemon {
   // "<job>/<key>" -> version -> blob. Monotonic versions, latest by default.
   std::mutex kv_mu;
   std::map<std::string, std::map<uint64_t, std::string>> param_store;
+  // GPU pool: device indices this daemon owns (--gpu-ids). A dispatched
+  // process job is pinned to one free device via HIP_VISIBLE_DEVICES so each
+  // worker peer IS one GPU (the 8-GPUs-on-one-node deployment: 8 daemons or
+  // one daemon with 8 ids). Guarded by `mu`.
+  std::vector<int> gpu_pool;
+
+  int acquire_gpu() {
+    std::lock_guard<std::mutex> lk(mu);
+    if (gpu_pool.empty()) return -1;
+    int id = gpu_pool.back();
+    gpu_pool.pop_back();
+    return id;
+  }
+  void release_gpu(int id) {
+    if (id < 0) return;
+    std::lock_guard<std::mutex> lk(mu);
+    gpu_pool.push_back(id);
+  }
 
   WorkerDaemon(std::string nm, const std::string& gw_host, int gw_port, Resources total,
                OfferPolicy pol, std::string cmd, std::string root, TlsConfig tls = {})
@@ -146,6 +165,31 @@ struct WorkerDaemon {
         if (body.has("remove_peer")) job->members.erase(body.at("remove_peer").as_string());
       }
       job->agg_cv.notify_all();
+      Json r;
+      r["ok"] = true;
+      return r;
+    });
+    // Scheduler-driven RCCL membership change: forward to the executor as a
+    // bridge SSE `reform` event; it re-forms the communicator at its next
+    // outer-sync boundary (elastic DiLoCo on the RCCL data plane).
+    node.on("rccl_reform", [&](const std::string& from, const Json& body) {
+      std::shared_ptr<Job> job;
+      {
+        std::lock_guard<std::mutex> lk(mu);
+        auto it = jobs.find(body.at("job").as_string());
+        if (it == jobs.end()) throw std::runtime_error("no such job");
+        job = it->second;
+      }
+      if (job->scheduler != from) throw std::runtime_error("not the job owner");
+      if (!job->bridge) throw std::runtime_error("job has no bridge");
+      Json ev;
+      ev["kind"] = std::string("reform");
+      ev["rank"] = body.at("rank");
+      ev["world_size"] = body.at("world_size");
+      ev["master_addr"] = body.get_or("master_addr", Json(std::string("127.0.0.1")));
+      ev["master_port"] = body.at("master_port");
+      if (body.has("timeout_s")) ev["timeout_s"] = body.at("timeout_s");
+      job->bridge->push_event(ev);
       Json r;
       r["ok"] = true;
       return r;
@@ -444,6 +488,17 @@ struct WorkerDaemon {
     subst("{WORK_DIR}", job->work_dir);
     subst("{JOB_JSON}", job_json);
 
+    // pin one GPU from this daemon's pool to the job for its whole lease
+    // lifetime: the executor sees exactly one device (cuda:0) regardless of
+    // which physical GPU it landed on
+    job->gpu_id = acquire_gpu();
+    if (job->gpu_id >= 0) {
+      cmd = "HIP_VISIBLE_DEVICES=" + std::to_string(job->gpu_id) +
+            " CUDA_VISIBLE_DEVICES=" + std::to_string(job->gpu_id) + " " + cmd;
+      fprintf(stderr, "[%s] job %s pinned to GPU %d\n", name.c_str(),
+              job->id.c_str(), job->gpu_id);
+    }
+
     pid_t pid = fork();
     if (pid == 0) {
       execl("/bin/sh", "sh", "-c", cmd.c_str(), (char*)nullptr);
@@ -453,6 +508,8 @@ struct WorkerDaemon {
     int status = 0;
     waitpid(pid, &status, 0);
     job->pid = -1;
+    release_gpu(job->gpu_id);
+    job->gpu_id = -1;
     bridge->stop();
     bool ok = WIFEXITED(status) && WEXITSTATUS(status) == 0;
     report_job_status(job, job->cancelled ? "cancelled" : (ok ? "completed" : "failed"));
@@ -756,6 +813,7 @@ int main(int argc, char** argv) {
   std::string name = "worker", gw_host = "127.0.0.1", cmd, icmd, work_root = "/tmp/hypha-work";
   int gw_port = 0, port = 0;
   bool probe = false, init = false;
+  std::vector<int> gpu_ids;
   std::vector<std::string> exclude_cidrs;
   TlsConfig tls;
   Resources total{1, 4, 16, 100};
@@ -768,6 +826,16 @@ int main(int argc, char** argv) {
     else if (a == "--gateway-port") gw_port = std::stoi(next());
     else if (a == "--port") port = std::stoi(next());
     else if (a == "--gpu") total.gpu = std::stod(next());
+    else if (a == "--gpu-ids") {  // physical device indices this daemon owns
+      std::string v = next();
+      size_t pos = 0;
+      while (pos <= v.size()) {
+        size_t c = v.find(',', pos);
+        if (c == std::string::npos) c = v.size();
+        if (c > pos) gpu_ids.push_back(std::stoi(v.substr(pos, c - pos)));
+        pos = c + 1;
+      }
+    }
     else if (a == "--cpu") total.cpu = std::stod(next());
     else if (a == "--memory") total.memory = std::stod(next());
     else if (a == "--storage") total.storage = std::stod(next());
@@ -830,6 +898,7 @@ int main(int argc, char** argv) {
     if (!has) se.push_back("inference-transformer");
   }
   WorkerDaemon daemon(name, gw_host, gw_port, total, policy, cmd, work_root, tls);
+  daemon.gpu_pool = gpu_ids;
   daemon.infer_cmd = icmd;
   daemon.node.set_exclude_cidrs(exclude_cidrs);
   daemon.start(port);

@@ -24,6 +24,162 @@ import threading
 import torch
 
 
+def _state_sync(worker, comm, round_idx: int) -> int:
+    """Broadcast {theta_global, outer momentum, round} from rank 0 after a
+    communicator (re)formation. For an unchanged membership this is a cheap
+    no-op semantically (all ranks identical); for a joiner it IS the catch-up
+    path — it replaces the reference PS's cumulative-offset file
+    (parameter_server.rs sync_to) with one xGMI broadcast."""
+    import torch as _torch
+
+    if not comm.is_distributed:
+        return round_idx
+    fp = worker.fp
+    comm.broadcast_flat(fp.theta0, src=0)
+    comm.broadcast_flat(fp.outer_momentum, src=0)
+    meta = _torch.tensor([round_idx, worker.inner_step_count], dtype=_torch.float64)
+    if fp.theta0.is_cuda:
+        meta = meta.to(fp.theta0.device)
+    comm.broadcast_flat(meta, src=0)
+    round_idx = int(meta[0].item())
+    # adopt: master/params <- global weights (joiner starts the round clean;
+    # survivors are at a round boundary where master==theta0 already)
+    fp.master.copy_(fp.theta0)
+    fp.flat.copy_(fp.master)
+    worker.round = round_idx
+    return round_idx
+
+
+def run_rccl(worker, comm, session, batch_iter, reform_q, rccl, batch_size,
+             seq_len, resume_round, ckpt_dir, ckpt_every, work_dir) -> int:
+    """Control-plane-orchestrated RCCL DiLoCo loop: the scheduler assigned
+    {rank, world_size, rendezvous}; the outer sync is a bucketed all-reduce of
+    the pseudo-gradient over xGMI plus a replicated fused Nesterov step on
+    every rank (no parameter-server star — SURVEY.md §2.10 C1/C2). Membership
+    changes (worker kill/rejoin, BASELINE config 3) arrive as `reform` events;
+    they are applied at the outer-sync boundary, where a collective timeout is
+    also recovered by re-forming on the scheduler's newest rendezvous and
+    retrying the sync.
+    """
+    import time as _time
+
+    from hypha_amd import ops
+    from safetensors.torch import save_file as _save_file
+
+    fp = worker.fp
+    # pseudo-gradient staging buffer for the full model: the elastic path
+    # cannot mutate theta0 until the WHOLE all-reduce succeeded (a retry after
+    # a membership change recomputes delta from the untouched master/theta0)
+    delta = torch.empty(fp.numel, dtype=worker.cfg.comm_dtype, device=fp.master.device)
+
+    def apply_reform(ev) -> None:
+        print(f"[executor] reform -> rank {ev['rank']}/{ev['world_size']} "
+              f"@ {ev.get('master_addr', '127.0.0.1')}:{ev['master_port']}", flush=True)
+        comm.reform(int(ev["rank"]), int(ev["world_size"]),
+                    ev.get("master_addr", "127.0.0.1"), int(ev["master_port"]),
+                    timeout_s=float(ev.get("timeout_s", comm.timeout_s)))
+
+    def drain_reforms(block_s: float = 0.0):
+        """Apply the newest pending reform order, if any."""
+        newest = None
+        try:
+            newest = reform_q.get(timeout=block_s) if block_s > 0 else reform_q.get_nowait()
+            while True:
+                newest = reform_q.get_nowait()
+        except queue.Empty:
+            pass
+        if newest is not None:
+            if newest.get("__bridge_closed__"):
+                raise RuntimeError("bridge connection lost (worker daemon gone)")
+            apply_reform(newest)
+            return True
+        return False
+
+    # form the scheduler-assigned group (initial members and joiners alike;
+    # a joiner blocks here until the survivors reach their sync boundary and
+    # re-form onto the same rendezvous)
+    comm.reform(int(rccl["rank"]), int(rccl["world_size"]),
+                rccl.get("master_addr", "127.0.0.1"),
+                int(rccl.get("master_port", 29531)), timeout_s=comm.timeout_s)
+    round_idx = resume_round
+    # initial state sync: no-op for a fresh group (identical seeds), the
+    # catch-up path for a joiner dispatched into a running job
+    round_idx = _state_sync(worker, comm, round_idx)
+    _save_file({"flat": fp.theta0.cpu()},
+               os.path.join(work_dir, "0_global_weights.safetensors"))
+
+    done = False
+    while not done:
+        round_t0 = _time.perf_counter()
+        remaining = None
+        round_samples = 0
+        loss = float("nan")
+        while remaining is None or remaining > 0:
+            ids, labels = next(batch_iter)
+            loss = worker.train_step(ids, labels)
+            round_samples += batch_size
+            if remaining is not None:
+                remaining -= 1
+            resp = session.send_status({"kind": "status", "batch_size": batch_size})
+            if resp.get("kind") == "schedule-update" and remaining is None:
+                remaining = int(resp.get("counter", 0))
+        round_s = max(1e-9, _time.perf_counter() - round_t0)
+        session.send_status(
+            {"kind": "metrics", "round": round_idx,
+             "metrics": {"loss": loss,
+                         "tokens_per_sec": round_samples * seq_len / round_s,
+                         "samples": round_samples}})
+        session.send_status({"kind": "update"})
+
+        # ---- outer sync boundary: apply membership changes, then sync ----
+        if drain_reforms():
+            round_idx = _state_sync(worker, comm, round_idx)
+        for attempt in range(4):
+            ops.interface.extract_delta(fp.master, fp.theta0, delta)
+            try:
+                comm.all_reduce_mean_flat(delta)
+                break
+            except Exception as e:
+                # a member died mid-round: the collective timed out. Wait for
+                # the scheduler's re-formation order (lease expiry detection
+                # is ~10 s, well inside the comm timeout), re-form, retry.
+                print(f"[executor] outer sync failed ({type(e).__name__}); "
+                      f"waiting for reform (attempt {attempt + 1})", flush=True)
+                if attempt == 3:
+                    raise
+                if not drain_reforms(block_s=comm.timeout_s + 30.0):
+                    raise RuntimeError("no reform order after sync failure")
+                round_idx = _state_sync(worker, comm, round_idx)
+        ops.fused_nesterov(fp.theta0, delta, fp.outer_momentum,
+                           lr=worker.cfg.outer.lr, mu=worker.cfg.outer.momentum)
+        fp.master.copy_(fp.theta0)
+        fp.flat.copy_(fp.master)
+        worker.outer_sync_payload_bytes += delta.numel() * delta.element_size()
+
+        # FSM ordering parity with the PS path: `updated` (advances the round)
+        # must reach the scheduler before any `update-received` — rank 0 sends
+        # it, then the barrier releases the other ranks
+        if comm.rank == 0:
+            session.send_status({"kind": "updated"})
+        comm.barrier()
+        resp = session.send_status({"kind": "update-received"})
+        done = resp.get("kind") == "done"
+        round_idx += 1
+        worker.round = round_idx
+        worker.steps_in_round = 0
+        if ckpt_every and (round_idx % ckpt_every == 0 or done):
+            from hypha_amd import checkpoint as ckpt_mod
+
+            ckpt_mod.save_checkpoint(worker, ckpt_dir)
+            print(f"[executor] checkpoint saved at round {round_idx}", flush=True)
+        print(f"[executor] rccl round {round_idx} merged, loss={loss:.4f} "
+              f"done={done}", flush=True)
+
+    comm.shutdown()
+    session.close()
+    return 0
+
+
 def main() -> int:
     p = argparse.ArgumentParser()
     p.add_argument("--socket", required=True)
@@ -49,13 +205,23 @@ def main() -> int:
         lr=cfg.get("optimizer", {}).get("adam", {}).get("learning_rate", 4e-4),
         warmup_steps=0, schedule="constant",
     )
+    # sync mode: "rccl" = scheduler-assigned rendezvous, outer sync is a
+    # bucketed RCCL all-reduce + replicated Nesterov (the on-node data plane,
+    # VERDICT r1 item 1); absent = parameter-server star via push streams
+    # (the WAN path, parameter_server.rs parity).
+    rccl = cfg.get("rccl")
+    # The group is always formed via comm.reform() inside run_rccl, AFTER the
+    # model/flat-buffer build: initial members and mid-run joiners then follow
+    # the exact same collective sequence (reform -> state broadcast), which a
+    # constructor-time init broadcast would desynchronize.
+    comm = Comm(timeout_s=float(rccl.get("timeout_s", 600.0)) if rccl else 600.0)
     worker = DiLoCoWorker(
-        model, DiLoCoConfig(h=1 << 30, inner=inner), comm=Comm(), device=device
+        model, DiLoCoConfig(h=1 << 30, inner=inner), comm=comm, device=device
     )
     batch_size = int(cfg.get("batch_size", 4))
     seq_len = int(cfg.get("seq_len", 128))
     data_ref = cfg["data"]
-    updates_ref = cfg["updates"]
+    updates_ref = cfg.get("updates")  # absent in rccl sync mode (no PS star)
     ckpt_dir = cfg.get("checkpoint_dir") or os.path.join(args.work_dir, "checkpoint")
     ckpt_every = int(cfg.get("checkpoint_every_rounds", 0))
 
@@ -70,18 +236,25 @@ def main() -> int:
         resume_round = int(manifest["round"])
         print(f"[executor] resumed from round {resume_round}", flush=True)
 
-    # background SSE listener for aggregated updates
+    # background SSE listener: aggregated-update file pointers go to
+    # updates_q; communicator re-formation orders (scheduler-driven elastic
+    # membership) go to reform_q
     updates_q: "queue.Queue[dict]" = queue.Queue()
+    reform_q: "queue.Queue[dict]" = queue.Queue()
 
     def listen():
         try:
             for ev in session.receive():
-                updates_q.put(ev)
+                if ev.get("kind") == "reform":
+                    reform_q.put(ev)
+                else:
+                    updates_q.put(ev)
         except Exception as e:  # bridge gone = worker daemon died: abort
             # instead of training on as an orphan (grad-release of the whole
             # job happens via the daemon's lease machinery, not us)
             print(f"# sse listener ended: {type(e).__name__}: {e}", file=sys.stderr)
         updates_q.put({"__bridge_closed__": True})
+        reform_q.put({"__bridge_closed__": True})
 
     threading.Thread(target=listen, daemon=True).start()
 
@@ -121,6 +294,11 @@ def main() -> int:
 
     batch_iter = batches()
     print(f"[executor] model={cfg['model']} device={device} bs={batch_size}", flush=True)
+
+    if rccl:
+        return run_rccl(worker, comm, session, batch_iter, reform_q, rccl,
+                        batch_size, seq_len, resume_round, ckpt_dir, ckpt_every,
+                        args.work_dir)
 
     if cfg.get("join"):
         # replacement worker (kill/rejoin path): catch up to the current

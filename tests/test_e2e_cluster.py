@@ -707,3 +707,180 @@ def test_parameter_server_loss_fails_fast(binaries, tmp_path):
                 p.wait(timeout=10)
             except Exception:
                 pass
+
+
+@pytest.mark.timeout(300)
+def test_cluster_rccl_sync_round_trip(binaries, tmp_path):
+    """Control-plane-orchestrated collective data plane:
+    with `sync: "rccl"` the scheduler assigns {rank, world_size, rendezvous}
+    in each dispatched job, there is NO parameter-server job, and the outer
+    sync is a bucketed all-reduce (gloo here, RCCL on GPU) + replicated
+    Nesterov. Both workers must finish with bitwise-identical global weights."""
+    from hypha_amd.data.synthetic import write_slice_files
+
+    data_dir = tmp_path / "slices"
+    write_slice_files(str(data_dir), "synth", num_slices=4, samples_per_slice=16,
+                      vocab_size=512, seq_len=128)
+    gw_port = free_port()
+    env = dict(os.environ, PYTHONPATH=str(REPO))
+    procs = []
+
+    def spawn(name, cmd):
+        log = open(tmp_path / f"{name}.log", "w")
+        p = subprocess.Popen(cmd, cwd=REPO, env=env, stdout=log, stderr=log,
+                             start_new_session=True)
+        procs.append(p)
+        return p
+
+    try:
+        spawn("gateway", [str(BIN / "hypha-gateway"), "--port", str(gw_port)])
+        time.sleep(0.3)
+        spawn("data", [str(BIN / "hypha-data"), "--name", "data-node",
+                       "--gateway-host", "127.0.0.1", "--gateway-port", str(gw_port),
+                       "--dataset", "synth", "--dataset-path", str(data_dir)])
+        exec_cmd = (f"{sys.executable} -m hypha_amd.runtime.executor "
+                    "--socket {SOCKET_PATH} --work-dir {WORK_DIR} --job {JOB_JSON}")
+        for i in range(2):
+            spawn(f"worker{i}", [str(BIN / "hypha-worker"), "--name", f"worker-{i}",
+                                 "--gateway-host", "127.0.0.1", "--gateway-port", str(gw_port),
+                                 "--exec-cmd", exec_cmd,
+                                 "--work-root", str(tmp_path / f"work{i}")])
+        time.sleep(0.5)
+
+        cfg = tmp_path / "job.json"
+        cfg.write_text(
+            '{"model": "llama-tiny", "dataset": "synth", "num_workers": 2,'
+            ' "update_rounds": 2, "avg_samples_between_updates": 8,'
+            ' "batch_size": 2, "seq_len": 128, "inner_lr": 0.001,'
+            ' "sync": "rccl", "rccl_timeout_s": 60,'
+            ' "checkpoint_every_rounds": 1}'
+        )
+        sched = subprocess.Popen(
+            [str(BIN / "hypha-scheduler"), "--name", "scheduler",
+             "--gateway-host", "127.0.0.1", "--gateway-port", str(gw_port),
+             "--config", str(cfg)],
+            cwd=REPO, env=env, stdout=subprocess.PIPE,
+            stderr=open(tmp_path / "sched.log", "w"), text=True,
+        )
+        procs.append(sched)
+        out, _ = sched.communicate(timeout=240)
+        logs = "".join((tmp_path / f"worker{i}.log").read_text() for i in range(2))
+        assert "Job is completed." in out, (
+            out, (tmp_path / "sched.log").read_text()[-3000:], logs[-3000:])
+        assert sched.returncode == 0
+        # the RCCL path actually ran (not the PS file path)
+        assert "rccl round" in logs, logs[-2000:]
+        assert "PS start" not in logs
+        # bitwise-identical global weights on both workers after the final sync
+        from safetensors.torch import load_file
+
+        ckpts = sorted(tmp_path.glob("work*/hypha-job-train-*/checkpoint/"
+                                     "0_global_weights.safetensors"))
+        assert len(ckpts) == 2, ckpts
+        t0 = load_file(str(ckpts[0]))["theta_global"]
+        t1 = load_file(str(ckpts[1]))["theta_global"]
+        assert (t0 == t1).all(), float((t0 - t1).abs().max())
+    finally:
+        for p in procs:
+            try:
+                os.killpg(p.pid, signal.SIGKILL)
+            except (ProcessLookupError, PermissionError):
+                if p.poll() is None:
+                    p.send_signal(signal.SIGKILL)
+        for p in procs:
+            try:
+                p.wait(timeout=10)
+            except Exception:
+                pass
+
+
+@pytest.mark.timeout(300)
+def test_cluster_rccl_kill_and_rejoin(binaries, tmp_path):
+    """BASELINE config 3 on the collective data plane: kill a train worker
+    mid-run under `sync: "rccl"`. The surviving worker's collective fails, the
+    scheduler auctions a replacement and re-forms the communicator on a fresh
+    rendezvous (new ranks, new port), the joiner catches up via the rank-0
+    state broadcast, and training completes."""
+    from hypha_amd.data.synthetic import write_slice_files
+
+    data_dir = tmp_path / "slices"
+    write_slice_files(str(data_dir), "synth", num_slices=4, samples_per_slice=16,
+                      vocab_size=512, seq_len=128)
+    gw_port = free_port()
+    env = dict(os.environ, PYTHONPATH=str(REPO))
+    procs = []
+    worker_procs = {}
+
+    def spawn(name, cmd):
+        log = open(tmp_path / f"{name}.log", "w")
+        p = subprocess.Popen(cmd, cwd=REPO, env=env, stdout=log, stderr=log,
+                             start_new_session=True)
+        procs.append(p)
+        return p
+
+    try:
+        spawn("gateway", [str(BIN / "hypha-gateway"), "--port", str(gw_port)])
+        time.sleep(0.3)
+        spawn("data", [str(BIN / "hypha-data"), "--name", "data-node",
+                       "--gateway-host", "127.0.0.1", "--gateway-port", str(gw_port),
+                       "--dataset", "synth", "--dataset-path", str(data_dir)])
+        exec_cmd = (f"{sys.executable} -m hypha_amd.runtime.executor "
+                    "--socket {SOCKET_PATH} --work-dir {WORK_DIR} --job {JOB_JSON}")
+        for i in range(3):  # 2 train + 1 spare for the replacement
+            worker_procs[f"worker-{i}"] = spawn(
+                f"worker{i}",
+                [str(BIN / "hypha-worker"), "--name", f"worker-{i}",
+                 "--gateway-host", "127.0.0.1", "--gateway-port", str(gw_port),
+                 "--exec-cmd", exec_cmd, "--work-root", str(tmp_path / f"work{i}")])
+        time.sleep(0.5)
+        cfg = tmp_path / "job.json"
+        cfg.write_text(
+            '{"model": "llama-tiny", "dataset": "synth", "num_workers": 2,'
+            ' "update_rounds": 12, "avg_samples_between_updates": 8,'
+            ' "batch_size": 2, "seq_len": 128, "inner_lr": 0.001,'
+            ' "sync": "rccl", "rccl_timeout_s": 30}'
+        )
+        sched = subprocess.Popen(
+            [str(BIN / "hypha-scheduler"), "--name", "scheduler",
+             "--gateway-host", "127.0.0.1", "--gateway-port", str(gw_port),
+             "--config", str(cfg)],
+            cwd=REPO, env=env, stdout=subprocess.PIPE,
+            stderr=open(tmp_path / "sched.log", "w"), text=True,
+            start_new_session=True,
+        )
+        procs.append(sched)
+
+        # wait until some train worker finishes round 1, then kill its daemon
+        victim = None
+        deadline = time.time() + 120
+        while victim is None and time.time() < deadline:
+            time.sleep(1)
+            for i in range(3):
+                log = (tmp_path / f"worker{i}.log")
+                if log.exists() and "rccl round 1 merged" in log.read_text():
+                    victim = f"worker-{i}"
+                    break
+        assert victim is not None, "no worker reached round 1"
+        os.killpg(worker_procs[victim].pid, signal.SIGKILL)
+
+        out, _ = sched.communicate(timeout=220)
+        sched_log = (tmp_path / "sched.log").read_text()
+        logs = "".join((tmp_path / f"worker{i}.log").read_text()
+                       for i in range(3) if (tmp_path / f"worker{i}.log").exists())
+        assert "Job is completed." in out, (out, sched_log[-3000:], logs[-3000:])
+        assert "re-formed RCCL group" in sched_log, sched_log[-3000:]
+        assert "joins as rank" in sched_log, sched_log[-3000:]
+        # the survivors actually re-formed (executor-side evidence)
+        assert "reform -> rank" in logs, logs[-3000:]
+    finally:
+        for p in procs:
+            try:
+                os.killpg(p.pid, signal.SIGKILL)
+            except (ProcessLookupError, PermissionError):
+                if p.poll() is None:
+                    p.send_signal(signal.SIGKILL)
+        for p in procs:
+            try:
+                p.wait(timeout=10)
+            except Exception:
+                pass
