@@ -129,6 +129,14 @@ def main() -> int:
     outer_sync_s = time.perf_counter() - t0
     rounds_before = worker.round
 
+    # Phase-shift the H cadence so at least one outer sync lands INSIDE the
+    # driver-timed window even when steps < H (it fires after ~steps/2 timed
+    # steps). The whole-run cadence is still one sync per H inner steps; a
+    # short window therefore OVER-charges the sync cost in `value` (1/K
+    # instead of 1/H of a sync per step) — the conservative side. The true
+    # H-amortized figure is reported as amortized_tokens_per_sec.
+    worker.steps_in_round = max(0, args.h - max(1, min(args.steps, args.h) // 2))
+
     # ---- timed region: exactly K steps ----
     sync()
     t_start = time.perf_counter()
@@ -141,6 +149,9 @@ def main() -> int:
     tokens_per_step_job = args.batch * args.seq_len * n_gpus
     value = tokens_per_step_job * args.steps / elapsed
     ms_per_step = elapsed / args.steps * 1000.0
+    # pure step time with the in-window syncs backed out (for H-amortization)
+    synced_now = worker.round - rounds_before
+    step_s_ex_sync = max(1e-9, elapsed - synced_now * outer_sync_s) / args.steps
 
     model_numel = worker.numel if args.memory_mode == "lean" else worker.fp.numel
     payload_bytes_per_sync = model_numel * 2  # bf16 comm dtype
@@ -172,7 +183,7 @@ def main() -> int:
                 "outer_sync_wire_bytes_per_rank": comm.wire_bytes_per_rank(payload_bytes_per_sync),
                 "outer_sync_ms": outer_sync_s * 1000.0,
                 "amortized_tokens_per_sec": tokens_per_step_job
-                / (elapsed / args.steps + outer_sync_s / args.h),
+                / (step_s_ex_sync + outer_sync_s / args.h),
                 "comm_reduction_vs_ddp": f"{args.h}x fewer syncs, bf16 payload",
                 "native_ops": ops.has_native(),
                 "last_loss": worker.last_loss,
