@@ -786,6 +786,281 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_v3_kernel(
   }
 }
 
+// Backward v4 (default for S%128==0): kv-split cooperative structure.
+// One workgroup = 4 waves; each wave OWNS one 32-row kv tile (128 kv rows
+// per workgroup, 2x v2), and ALL waves walk ALL q-tiles together:
+//   * Q^T/dO^T staging is cooperative (1/4 of v2's per-wave cost) and
+//     register-prefetched one q-tile ahead (T14 split: HBM latency hides
+//     under the previous tile's MFMA phases);
+//   * each wave's dK/dV accumulators see every q-tile, so they are COMPLETE
+//     for its 32 kv rows -> direct bf16 stores, no cross-wave combine;
+//   * dQ covers kv 0..127 in ONE atomic pass (1/4 of v2's atomic traffic
+//     per flop) with d-blocks split across waves;
+//   * P^T/dS^T images stay per-wave (no barrier); the shared dS [q][kv]
+//     image and the staged q-images need 2 barriers per q-tile.
+// Rationale (r1 profile): v2 at 1 wave/SIMD spent ~95% of cycles outside
+// MFMA issue on per-wave staging, 4x the dQ atomics, and exposed HBM
+// latency; this structure attacks all three without duplicating any MFMA
+// (v3's mistake, measured -10%).
+// `ablate` is a perf-diagnosis bitmask (HYPHA_ATTN_ABLATE, default 0=full):
+// 1 = skip the dQ atomics, 2 = skip the whole dQ phase, 4 = skip the exp.
+template <int HD>
+__global__ __launch_bounds__(256, 1) void attn_bwd_v4_kernel(
+    const short* __restrict__ qg, const short* __restrict__ kg,
+    const short* __restrict__ vg, const short* __restrict__ dog,
+    const float* __restrict__ lseg, const float* __restrict__ dig,
+    float* __restrict__ dqg, short* __restrict__ dkg, short* __restrict__ dvg,
+    int B, int Hq, int Hkv, int S, float scale, bool causal,
+    long long q_sb, long long q_sh, long long q_ss,
+    long long kv_sb, long long kv_sh, long long kv_ss, int ablate) {
+  constexpr int KVT = 32;        // kv rows per wave
+  constexpr int NT = 4;          // kv tiles (= waves) per workgroup
+  constexpr int KVW = NT * KVT;  // 128 kv rows per workgroup
+  constexpr int QT = 32;         // q rows per tile
+  constexpr int KC = HD / 16;
+  constexpr int DBLK = HD / 32;
+  constexpr int TP = 40;           // per-wave ptds/dst pitch
+  constexpr int DSP = KVW + 8;     // shared dS [q][kv] pitch (136: 16B rows)
+  constexpr int KTP = KVW + 8;     // kt image pitch
+  __shared__ __attribute__((aligned(16))) short k_img[KVW * HD];      // [kv][d] swizzled
+  __shared__ __attribute__((aligned(16))) short v_img[KVW * HD];      // [kv][d] swizzled
+  __shared__ __attribute__((aligned(16))) short kt_img[HD * KTP];     // [d][kv]
+  __shared__ __attribute__((aligned(16))) short qt_img[HD * 32];      // [d][q] swizzled, shared
+  __shared__ __attribute__((aligned(16))) short dot_img[HD * 32];     // [d][q] swizzled, shared
+  __shared__ __attribute__((aligned(16))) short ptds_img[NWAVE][KVT * TP];  // per-wave ptT
+  __shared__ __attribute__((aligned(16))) short dst_img[NWAVE][KVT * TP];   // per-wave dsT
+  // double-buffered: tile t's dQ pass is DEFERRED into tile t+1's compute
+  // phase (its MFMAs and atomics interleave with the next tile's work, and
+  // causal-idle waves get dQ work) while tile t+1 writes the other buffer
+  __shared__ __attribute__((aligned(16))) short ds_img[2][QT * DSP];  // dS [q][kv 0..127]
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int hi = lane >> 5;
+  const int ln = lane & 31;
+
+  const int bhkv = blockIdx.y;
+  const int b = bhkv / Hkv;
+  const int hkv = bhkv % Hkv;
+  const int G = Hq / Hkv;
+  const int kv_base = blockIdx.x * KVW;
+  const long long kvbase = (long long)b * kv_sb + (long long)hkv * kv_sh;
+
+  auto qimg_off = [](int d, int qe) {
+    return d * 32 + ((((qe >> 3) ^ ((d >> 2) & 3)) << 3) | (qe & 7));
+  };
+
+  // ---- stage K, V (swizzled row-major) and K^T for all 128 kv rows ----
+  {
+    constexpr int CH = KVW * HD / 8;
+    for (int c = tid; c < CH; c += 256) {
+      int row = c / (HD / 8), e0 = (c % (HD / 8)) * 8;
+      s16x8 k8 = *reinterpret_cast<const s16x8*>(kg + kvbase + (long long)(kv_base + row) * kv_ss + e0);
+      *reinterpret_cast<s16x8*>((char*)k_img + k_lds_off<HD>(row, e0)) = k8;
+      s16x8 v8 = *reinterpret_cast<const s16x8*>(vg + kvbase + (long long)(kv_base + row) * kv_ss + e0);
+      *reinterpret_cast<s16x8*>((char*)v_img + k_lds_off<HD>(row, e0)) = v8;
+    }
+#pragma unroll
+    for (int half = 0; half < KVW / 64; ++half) {
+      for (int i = 0; i < HD / 32; ++i) {
+        int kvr = (tid & 63) + 64 * half;
+        int e0 = (i * 4 + (tid >> 6)) * 8;
+        s16x8 k8 = *reinterpret_cast<const s16x8*>(kg + kvbase + (long long)(kv_base + kvr) * kv_ss + e0);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) kt_img[(e0 + j) * KTP + kvr] = k8[j];
+      }
+    }
+  }
+  __syncthreads();
+
+  f32x16 dk_acc[DBLK] = {};  // D[m=kv][n=d], this wave's 32 kv rows
+  f32x16 dv_acc[DBLK] = {};  // D[m=d][n=kv]
+
+  short* ptds = ptds_img[wid];
+  short* dst = dst_img[wid];
+  const int kv0 = kv_base + wid * KVT;  // this wave's kv tile
+
+  const int t0 = causal ? kv_base / QT : 0;
+  const int Tq = S / QT;
+  // cooperative q-image staging: wave w writes d-chunks {w, w+4} of 8
+  constexpr int NSTG = HD / 16 / NWAVE;  // chunks per wave (2 for HD=128)
+
+  // deferred-dQ pipeline state: the previous q-tile whose dS image is
+  // complete but whose dQ pass has not run yet (crosses hq boundaries:
+  // kt_img is per-hkv, shared by every head in the GQA group)
+  int prev_q0 = -1, prev_na = 0, pbuf = 0;
+  long long prev_qbase = 0;
+
+  auto run_dq = [&](int buf) {
+    if (wid >= DBLK || prev_q0 < 0) return;
+    const int db = wid;
+    const int kck_hi = 2 * prev_na;
+    f32x16 dq = {};
+    for (int kck = 0; kck < kck_hi; ++kck) {
+      s16x8 af = *reinterpret_cast<const s16x8*>(ds_img[buf] + ln * DSP + 16 * kck + 8 * hi);
+      s16x8 bf = *reinterpret_cast<const s16x8*>(kt_img + (32 * db + ln) * KTP +
+                                                 16 * kck + 8 * hi);
+      dq = __builtin_amdgcn_mfma_f32_32x32x16_bf16(af, bf, dq, 0, 0, 0);
+    }
+    if (!(ablate & 1)) {
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        int qi = prev_q0 + (r & 3) + 8 * (r >> 2) + 4 * hi;
+        atomicAdd(dqg + (prev_qbase + (long long)qi * q_ss + 32 * db + ln), dq[r]);
+      }
+    } else {
+      asm volatile("" ::"v"(dq[0]));  // keep the MFMA chain live (perf probe)
+    }
+  };
+
+  for (int hq = hkv * G; hq < (hkv + 1) * G; ++hq) {
+    const long long qbase = (long long)b * q_sb + (long long)hq * q_sh;
+    const long long lsebase = (long long)(b * Hq + hq) * S;
+
+    // prefetch the first q-tile of this head into registers
+    s16x8 qreg[NSTG], doreg[NSTG];
+    auto issue_prefetch = [&](int t) {
+#pragma unroll
+      for (int i = 0; i < NSTG; ++i) {
+        int it = wid + NWAVE * i;
+        int d0 = 8 * hi + 16 * it;
+        qreg[i] = *reinterpret_cast<const s16x8*>(qg + qbase + (long long)(t * QT + ln) * q_ss + d0);
+        doreg[i] = *reinterpret_cast<const s16x8*>(dog + qbase + (long long)(t * QT + ln) * q_ss + d0);
+      }
+    };
+    issue_prefetch(t0);
+    float lse = lseg[lsebase + t0 * QT + ln];
+    float di = dig[lsebase + t0 * QT + ln];
+
+    for (int t = t0; t < Tq; ++t) {
+      const int q0 = t * QT;
+      const int qrow = q0 + ln;
+
+      // ---- (a) write the prefetched q-tile into the shared images ----
+#pragma unroll
+      for (int i = 0; i < NSTG; ++i) {
+        int it = wid + NWAVE * i;
+        int d0 = 8 * hi + 16 * it;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          qt_img[qimg_off(d0 + j, ln)] = qreg[i][j];
+          dot_img[qimg_off(d0 + j, ln)] = doreg[i][j];
+        }
+      }
+      __syncthreads();  // (b) staged images ready
+
+      // ---- (c) issue next tile's loads; they land under the MFMA work ----
+      const float lse_cur = lse, di_cur = di;
+      if (t + 1 < Tq) {
+        issue_prefetch(t + 1);
+        lse = lseg[lsebase + (t + 1) * QT + ln];
+        di = dig[lsebase + (t + 1) * QT + ln];
+      }
+
+      // active kv tiles for this q-tile (wave tiles above the diagonal idle)
+      const int na = causal ? min(NT, (q0 + QT - 1 - kv_base) / KVT + 1) : NT;
+      const int cbuf = pbuf ^ 1;
+
+      // deferred dQ of the PREVIOUS tile: its dS buffer is complete and
+      // nobody writes it this phase — the dQ MFMAs and atomics overlap the
+      // current tile's compute instead of sitting in their own phase
+      if (!(ablate & 2)) run_dq(pbuf);
+
+      if (wid < na) {
+        // ---- S^T = K Q^T (Q frags from global: L1-hot after staging) ----
+        f32x16 st = {};
+#pragma unroll
+        for (int kc = 0; kc < KC; ++kc) {
+          s16x8 qf = *reinterpret_cast<const s16x8*>(qg + qbase + (long long)qrow * q_ss +
+                                                     16 * kc + 8 * hi);
+          s16x8 kf = *reinterpret_cast<const s16x8*>(
+              (char*)k_img + k_lds_off<HD>(wid * KVT + ln, 16 * kc + 8 * hi));
+          st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qf, st, 0, 0, 0);
+        }
+        // ---- P^T = exp(scale*S^T - lse), causal mask; per-wave ptT ----
+        f32x16 pt;
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          int kv = kv0 + (r & 3) + 8 * (r >> 2) + 4 * hi;
+          float pv = (ablate & 4) ? st[r] * scale : __expf(st[r] * scale - lse_cur);
+          if (causal && kv > qrow) pv = 0.f;
+          pt[r] = pv;
+        }
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          int kv = (r & 3) + 8 * (r >> 2) + 4 * hi;
+          ptds[kv * TP + ln] = f2bf(pt[r]);
+        }
+        // ---- dV^T += dO^T P (A from shared dO^T image, B from own ptT) ----
+#pragma unroll
+        for (int db = 0; db < DBLK; ++db) {
+#pragma unroll
+          for (int kcq = 0; kcq < QT / 16; ++kcq) {
+            s16x8 af = *reinterpret_cast<const s16x8*>(
+                dot_img + qimg_off(32 * db + ln, 16 * kcq + 8 * hi));
+            s16x8 bf = *reinterpret_cast<const s16x8*>(ptds + ln * TP + 16 * kcq + 8 * hi);
+            dv_acc[db] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(af, bf, dv_acc[db], 0, 0, 0);
+          }
+        }
+        // ---- dP^T = V dO^T ----
+        f32x16 dpt = {};
+#pragma unroll
+        for (int kc = 0; kc < KC; ++kc) {
+          s16x8 df = *reinterpret_cast<const s16x8*>(dog + qbase + (long long)qrow * q_ss +
+                                                     16 * kc + 8 * hi);
+          s16x8 vf = *reinterpret_cast<const s16x8*>(
+              (char*)v_img + k_lds_off<HD>(wid * KVT + ln, 16 * kc + 8 * hi));
+          dpt = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vf, df, dpt, 0, 0, 0);
+        }
+        // ---- dS^T = P^T (dP^T - Di) scale; own dsT + shared dS columns ----
+#pragma unroll
+        for (int r = 0; r < 16; r += 2) {
+          int kv = (r & 3) + 8 * (r >> 2) + 4 * hi;
+          float v0 = pt[r] * (dpt[r] - di_cur) * scale;
+          float v1 = pt[r + 1] * (dpt[r + 1] - di_cur) * scale;
+          dst[kv * TP + ln] = f2bf(v0);
+          dst[(kv + 1) * TP + ln] = f2bf(v1);
+          // adjacent r -> adjacent kv: one packed b32 store into [q][kv]
+          *reinterpret_cast<unsigned*>(ds_img[cbuf] + ln * DSP + wid * KVT + kv) =
+              pack_bf16x2(v0, v1);
+        }
+        // ---- dK += dS^T Q (B from shared Q^T image) ----
+#pragma unroll
+        for (int db = 0; db < DBLK; ++db) {
+#pragma unroll
+          for (int kcq = 0; kcq < QT / 16; ++kcq) {
+            s16x8 af = *reinterpret_cast<const s16x8*>(dst + ln * TP + 16 * kcq + 8 * hi);
+            s16x8 bf = *reinterpret_cast<const s16x8*>(
+                qt_img + qimg_off(32 * db + ln, 16 * kcq + 8 * hi));
+            dk_acc[db] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(af, bf, dk_acc[db], 0, 0, 0);
+          }
+        }
+      }
+      __syncthreads();  // (e) shared dS complete; qt/dot free to rewrite
+
+      prev_q0 = q0;
+      prev_na = na;
+      prev_qbase = qbase;
+      pbuf = cbuf;
+    }
+  }
+
+  // drain: the final q-tile's dQ pass (its dS buffer is complete after (e))
+  if (!(ablate & 2)) run_dq(pbuf);
+
+  // ---- direct bf16 stores: this wave owns kv rows [kv0, kv0+32) fully ----
+#pragma unroll
+  for (int db = 0; db < DBLK; ++db) {
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      int m = (r & 3) + 8 * (r >> 2) + 4 * hi;
+      dkg[kvbase + (long long)(kv0 + m) * kv_ss + 32 * db + ln] = f2bf(dk_acc[db][r]);
+      dvg[kvbase + (long long)(kv0 + ln) * kv_ss + 32 * db + m] = f2bf(dv_acc[db][r]);
+    }
+  }
+}
+
 }  // namespace
 
 std::vector<torch::Tensor> attn_bwd_ex(torch::Tensor q, torch::Tensor k, torch::Tensor v,
@@ -815,11 +1090,29 @@ std::vector<torch::Tensor> attn_bwd_ex(torch::Tensor q, torch::Tensor k, torch::
     kv_ss = HD; kv_sh = (long long)S * HD; kv_sb = (long long)Hkv * S * HD;
   }
 
-  // experimental occupancy-first backward (see attn_bwd_v3_kernel docstring)
-  static const bool use_v3 = [] {
-    const char* e = getenv("HYPHA_ATTN_BWD_V3");
-    return e && e[0] == '1';
+  // kernel selection: v4 (kv-split cooperative, S%128) is the default;
+  // HYPHA_ATTN_BWD=v2|v3|v4 overrides for A/B runs.
+  static const int bwd_variant = [] {
+    const char* e = getenv("HYPHA_ATTN_BWD");
+    if (e && e[0] == 'v' && e[1]) return e[1] - '0';
+    if (e && e[0] >= '2' && e[0] <= '4') return e[0] - '0';
+    const char* e3 = getenv("HYPHA_ATTN_BWD_V3");  // legacy toggle
+    return (e3 && e3[0] == '1') ? 3 : 4;
   }();
+  const int variant = (bwd_variant == 4 && S % 128 != 0) ? 2 : bwd_variant;
+  static const int bwd_ablate = [] {
+    const char* e = getenv("HYPHA_ATTN_ABLATE");
+    return e ? atoi(e) : 0;
+  }();
+
+#define LAUNCH_BWD(KERN, NQ, HDV)                                                       \
+  hipLaunchKernelGGL(KERN<HDV>, dim3(S / NQ, B * Hkv), dim3(256), 0, stream,            \
+                     (const short*)q.data_ptr(), (const short*)k.data_ptr(),            \
+                     (const short*)v.data_ptr(), (const short*)dout.data_ptr(),         \
+                     lse.data_ptr<float>(), di.data_ptr<float>(),                       \
+                     dq32.data_ptr<float>(), (short*)dk.data_ptr(),                     \
+                     (short*)dv.data_ptr(), B, Hq, Hkv, S, scale, causal, q_sb,         \
+                     q_sh, q_ss, kv_sb, kv_sh, kv_ss)
 
 #define DISPATCH(HDV)                                                                   \
   do {                                                                                  \
@@ -827,22 +1120,19 @@ std::vector<torch::Tensor> attn_bwd_ex(torch::Tensor q, torch::Tensor k, torch::
                        dim3(256), 0, stream, (const short*)dout.data_ptr(),             \
                        (const short*)o.data_ptr(), di.data_ptr<float>(), nrows, Hq, S,  \
                        q_sb, q_sh, q_ss);                                               \
-    if (use_v3)                                                                         \
-      hipLaunchKernelGGL(attn_bwd_v3_kernel<HDV>, dim3(S / 64, B * Hkv), dim3(256), 0,  \
-                         stream, (const short*)q.data_ptr(), (const short*)k.data_ptr(),\
-                         (const short*)v.data_ptr(), (const short*)dout.data_ptr(),     \
-                         lse.data_ptr<float>(), di.data_ptr<float>(),                   \
-                         dq32.data_ptr<float>(), (short*)dk.data_ptr(),                 \
-                         (short*)dv.data_ptr(), B, Hq, Hkv, S, scale, causal, q_sb,     \
-                         q_sh, q_ss, kv_sb, kv_sh, kv_ss);                              \
+    if (variant == 4)                                                                   \
+      hipLaunchKernelGGL(attn_bwd_v4_kernel<HDV>, dim3(S / 128, B * Hkv), dim3(256),    \
+                         0, stream, (const short*)q.data_ptr(),                         \
+                         (const short*)k.data_ptr(), (const short*)v.data_ptr(),        \
+                         (const short*)dout.data_ptr(), lse.data_ptr<float>(),          \
+                         di.data_ptr<float>(), dq32.data_ptr<float>(),                  \
+                         (short*)dk.data_ptr(), (short*)dv.data_ptr(), B, Hq, Hkv, S,   \
+                         scale, causal, q_sb, q_sh, q_ss, kv_sb, kv_sh, kv_ss,          \
+                         bwd_ablate);                                                   \
+    else if (variant == 3)                                                              \
+      LAUNCH_BWD(attn_bwd_v3_kernel, 64, HDV);                                          \
     else                                                                                \
-      hipLaunchKernelGGL(attn_bwd_kernel<HDV>, dim3(S / 64, B * Hkv), dim3(256), 0,     \
-                         stream, (const short*)q.data_ptr(), (const short*)k.data_ptr(),\
-                         (const short*)v.data_ptr(), (const short*)dout.data_ptr(),     \
-                         lse.data_ptr<float>(), di.data_ptr<float>(),                   \
-                         dq32.data_ptr<float>(), (short*)dk.data_ptr(),                 \
-                         (short*)dv.data_ptr(), B, Hq, Hkv, S, scale, causal, q_sb,     \
-                         q_sh, q_ss, kv_sb, kv_sh, kv_ss);                              \
+      LAUNCH_BWD(attn_bwd_kernel, 64, HDV);                                             \
   } while (0)
 
   if (HD == 128)

@@ -553,3 +553,21 @@ def test_lean_checkpoint_roundtrip(tmp_path):
     torch.testing.assert_close(w2.theta0_host, w.theta0_host)
     assert torch.equal(w2.m8, w.m8) and torch.equal(w2.v8, w.v8)
     assert w2.round == w.round
+
+
+def test_attention_bwd_noncausal():
+    """Non-causal backward (v4 all-tiles-active path) vs the fp32 oracle."""
+    B, S, hq, hkv, hd = 1, 256, 4, 2, 128
+    q = rand_bf16(B, hq, S, hd, seed=50).requires_grad_(True)
+    k = rand_bf16(B, hkv, S, hd, seed=51).requires_grad_(True)
+    v = rand_bf16(B, hkv, S, hd, seed=52).requires_grad_(True)
+    o = ops.flash_attention(q, k, v, causal=False)
+    do = rand_bf16(B, hq, S, hd, seed=53)
+    o.backward(do)
+    qf = q.detach().float().requires_grad_(True)
+    kf = k.detach().float().requires_grad_(True)
+    vf = v.detach().float().requires_grad_(True)
+    R.attention(qf, kf, vf, causal=False).backward(do.float())
+    torch.testing.assert_close(q.grad.float(), qf.grad, rtol=5e-2, atol=5e-2)
+    torch.testing.assert_close(k.grad.float(), kf.grad, rtol=5e-2, atol=5e-2)
+    torch.testing.assert_close(v.grad.float(), vf.grad, rtol=5e-2, atol=5e-2)
