@@ -66,6 +66,10 @@ std::vector<torch::Tensor> attn_bwd_ex(torch::Tensor q, torch::Tensor k, torch::
 // probe.hip (MFMA fragment-layout verification)
 torch::Tensor mfma_probe_32x32x16(torch::Tensor a, torch::Tensor b);
 torch::Tensor mfma_probe_16x16x32(torch::Tensor a, torch::Tensor b);
+// fp8_cast.hip
+std::vector<torch::Tensor> fp8_cast_transpose(torch::Tensor x, torch::Tensor scale,
+                                              torch::Tensor amax);
+void fp8_scale_update_(torch::Tensor amax, torch::Tensor scale, double margin);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("adamw_step_", &adamw_step_);
@@ -95,4 +99,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attn_bwd_ex", &attn_bwd_ex);
   m.def("mfma_probe_32x32x16", &mfma_probe_32x32x16);
   m.def("mfma_probe_16x16x32", &mfma_probe_16x16x32);
+  m.def("fp8_cast_transpose", &fp8_cast_transpose);
+  m.def("fp8_scale_update_", &fp8_scale_update_);
 }

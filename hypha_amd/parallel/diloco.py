@@ -175,6 +175,9 @@ class DiLoCoWorker:
 
     def train_step(self, input_ids: torch.Tensor, labels: torch.Tensor) -> float:
         """One inner step: forward, backward, fused AdamW. Returns loss."""
+        from hypha_amd.ops.fp8 import fp8_step
+
+        fp8_step()  # new fp8 epoch: weight casts refresh once per step
         self.model.train()
         input_ids = input_ids.to(self.device, non_blocking=True)
         labels = labels.to(self.device, non_blocking=True)

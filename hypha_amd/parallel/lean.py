@@ -117,6 +117,9 @@ class LeanDiLoCoWorker:
         return hook
 
     def train_step(self, input_ids: torch.Tensor, labels: torch.Tensor) -> float:
+        from hypha_amd.ops.fp8 import fp8_step
+
+        fp8_step()  # new fp8 epoch: weight casts refresh once per step
         self.model.train()
         self.inner_step_count += 1
         self.steps_in_round += 1

@@ -94,7 +94,7 @@ def test_bench_contract_world2_gloo(tmp_path):
     cfg = out["config"]
     assert cfg["parallelism"] == "diloco-dp2"
     assert cfg["global_batch"] == 4  # whole-job aggregate
-    assert cfg["outer_syncs_in_timed_window"] == 1  # h=3 < steps+warmup
+    assert cfg["outer_syncs_in_timed_window"] >= 1  # phase-shifted cadence
     assert cfg["outer_sync_wire_bytes_per_rank"] > 0  # ring wire accounting
 
 

@@ -571,3 +571,37 @@ def test_attention_bwd_noncausal():
     torch.testing.assert_close(q.grad.float(), qf.grad, rtol=5e-2, atol=5e-2)
     torch.testing.assert_close(k.grad.float(), kf.grad, rtol=5e-2, atol=5e-2)
     torch.testing.assert_close(v.grad.float(), vf.grad, rtol=5e-2, atol=5e-2)
+
+
+def test_fp8_cast_transpose():
+    """Fused cast+transpose+amax kernel vs plain PyTorch quantization."""
+    from hypha_amd import _C
+
+    torch.manual_seed(7)
+    for R, C in ((128, 256), (100, 72), (4096, 4096)):
+        x = (torch.randn(R, C, device=DEV) * 3).bfloat16()
+        scale = torch.tensor([0.05], dtype=torch.float32, device=DEV)
+        amax = torch.zeros(1, dtype=torch.float32, device=DEV)
+        x8, x8t = _C.fp8_cast_transpose(x, scale, amax)
+        assert x8.shape == (R, C) and x8t.shape == (C, R)
+        ref = (x.float() / 0.05).clamp(-448, 448).to(torch.float8_e4m3fn)
+        torch.testing.assert_close(x8.float(), ref.float(), rtol=0, atol=0)
+        torch.testing.assert_close(x8t.float(), ref.t().float(), rtol=0, atol=0)
+        torch.testing.assert_close(amax.item(), x.float().abs().max().item(),
+                                   rtol=1e-3, atol=0)
+
+
+def test_fp8_delayed_scale_reuse():
+    """Weight casts are cached within one fp8 epoch and refresh after fp8_step."""
+    from hypha_amd.ops import fp8 as f8
+
+    lin = f8.Fp8Linear(256, 128).bfloat16().to(DEV)
+    x = rand_bf16(64, 256, seed=60)
+    lin(x)
+    st = lin._state
+    c1 = st._wcache
+    lin(x)
+    assert lin._state._wcache is c1  # same epoch: cache reused
+    f8.fp8_step()
+    lin(x)
+    assert lin._state._wcache is not c1
