@@ -40,6 +40,7 @@ _FP8_STEP = 0
 def fp8_step() -> None:
     global _FP8_STEP
     _FP8_STEP += 1
+    _XCAST.clear()
 
 
 def _C():
@@ -65,8 +66,8 @@ class _DelayedScale:
             # first call: seed amax from the live tensor (device-side, async)
             self.amax.copy_(t.detach().abs().amax().float().reshape(1))
             self.inited = True
+        # scale_update publishes scale from last window's amax AND zeroes it
         C.fp8_scale_update_(self.amax, self.scale, self.margin)
-        self.amax.zero_()
         t8, t8t = C.fp8_cast_transpose(t, self.scale, self.amax)
         return t8, t8t, self.scale
 
@@ -91,11 +92,31 @@ class _Fp8State:
         return w8, w8t, sw
 
 
+# Sibling linears consume the SAME activation (wq/wk/wv share the attention
+# input; gate/up share the MLP input): cast it once and share the quantized
+# copies. Keyed by (data_ptr, shape, version); entries hold a strong ref to
+# the source so the caching allocator cannot recycle its memory into a
+# colliding key while the entry is alive. Cleared each fp8 epoch.
+_XCAST: list = []
+
+
+def _cast_x_shared(x2d: torch.Tensor, state: "_Fp8State"):
+    key = (x2d.data_ptr(), tuple(x2d.shape), x2d._version)
+    for e in _XCAST:
+        if e[0] == key:
+            return e[2], e[3], e[4]
+    x8, x8t, sx = state.x.cast(x2d)
+    sx = sx.clone()  # survives until backward; state.x.scale moves on
+    _XCAST.append((key, x2d, x8, x8t, sx))
+    if len(_XCAST) > 4:
+        _XCAST.pop(0)
+    return x8, x8t, sx
+
+
 class _Fp8LinearFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x2d, weight, state: _Fp8State):
-        x8, x8t, sx = state.x.cast(x2d)
-        sx = sx.clone()  # survives until backward; state.x.scale moves on
+        x8, x8t, sx = _cast_x_shared(x2d, state)
         w8, w8t, sw = state.weights(weight)
         out = torch._scaled_mm(x8, w8.t(), scale_a=sx, scale_b=sw,
                                out_dtype=torch.bfloat16)

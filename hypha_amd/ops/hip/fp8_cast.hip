@@ -47,6 +47,7 @@ __global__ __launch_bounds__(256) void fp8_cast_transpose_kernel(
     unsigned char* __restrict__ out8t, const float* __restrict__ scale_io,
     float* __restrict__ amax_out, int R, int C) {
   __shared__ unsigned char tile[64][72];  // [col][row], 8-byte padded rows
+  __shared__ float wmax[4];
 
   const int tid = threadIdx.x;
   const int r0 = blockIdx.y * 64;
@@ -108,17 +109,27 @@ __global__ __launch_bounds__(256) void fp8_cast_transpose_kernel(
     }
   }
 
+  // ONE atomic per block: a per-wave atomic on a single global address
+  // serializes the whole grid at the owning L2 bank (measured 0.48 ms/call
+  // vs ~15 us roofline before this reduction)
   mx = wave_max(mx);
-  if ((tid & 63) == 0 && mx > 0.f)
-    atomicMax(reinterpret_cast<unsigned int*>(amax_out), __float_as_uint(mx));
+  if ((tid & 63) == 0) wmax[tid >> 6] = mx;
+  __syncthreads();
+  if (tid == 0) {
+    float m = fmaxf(fmaxf(wmax[0], wmax[1]), fmaxf(wmax[2], wmax[3]));
+    if (m > 0.f)
+      atomicMax(reinterpret_cast<unsigned int*>(amax_out), __float_as_uint(m));
+  }
 }
 
-// scale = clamp(amax, eps) / 448 * margin; one thread. Runs BEFORE the cast
-// kernel each step so the whole delayed-scaling loop stays on-device.
-__global__ void fp8_scale_update_kernel(const float* __restrict__ amax,
+// scale = clamp(amax, eps) / 448 * margin, then RESETS amax for the next
+// accumulation window; one thread. Runs BEFORE the cast kernel each step so
+// the whole delayed-scaling loop stays on-device (no memsets, no host sync).
+__global__ void fp8_scale_update_kernel(float* __restrict__ amax,
                                         float* __restrict__ scale, float margin) {
   float a = fmaxf(amax[0], 1e-8f);
   scale[0] = a / E4M3_MAX * margin;
+  amax[0] = 0.f;
 }
 
 }  // namespace
@@ -143,6 +154,7 @@ std::vector<torch::Tensor> fp8_cast_transpose(torch::Tensor x, torch::Tensor sca
   return {out8, out8t};
 }
 
+// Publishes scale from the accumulated amax and zeroes amax in one launch.
 void fp8_scale_update_(torch::Tensor amax, torch::Tensor scale, double margin) {
   hipStream_t stream = hypha_stream();
   hipLaunchKernelGGL(fp8_scale_update_kernel, dim3(1), dim3(1), 0, stream,
