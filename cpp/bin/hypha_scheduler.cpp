@@ -136,7 +136,15 @@ int main(int argc, char** argv) {
   ss << cf.rdbuf();
   Json cfg = Json::parse(ss.str());
   const std::string model = cfg.at("model").as_string();
-  const std::string dataset = cfg.at("dataset").as_string();
+  const std::string dataset = cfg.get_or("dataset", Json(std::string())).as_string();
+  // URI data source (connector/mod.rs HttpHfFetcher): jobs may pull their
+  // training data over HTTP(S) through the worker's fetch connector instead
+  // of a cluster data node; the worker enforces its --fetch-allow list.
+  const std::string data_uri = cfg.get_or("data_uri", Json(std::string())).as_string();
+  if (dataset.empty() && data_uri.empty()) {
+    fprintf(stderr, "[scheduler] config needs `dataset` or `data_uri`\n");
+    return 1;
+  }
   const int64_t num_workers = cfg.at("num_workers").as_int();
   const int64_t update_rounds = cfg.at("update_rounds").as_int();
   const int64_t samples_between = cfg.at("avg_samples_between_updates").as_int();
@@ -252,7 +260,7 @@ int main(int argc, char** argv) {
 
   // --- discover the data provider via the registry (DHT get, kad.rs) ---
   int64_t num_slices = 0;
-  for (int i = 0; i < 100; ++i) {
+  for (int i = 0; i < 100 && data_uri.empty(); ++i) {
     auto rec = node.kv_get("dataset:" + dataset);
     if (rec) {
       num_slices = rec->at("num_slices").as_int();
@@ -261,7 +269,7 @@ int main(int argc, char** argv) {
     }
     usleep(200000);
   }
-  if (num_slices == 0) {
+  if (num_slices == 0 && data_uri.empty()) {
     fprintf(stderr, "[scheduler] dataset %s not found\n", dataset.c_str());
     return 1;
   }
@@ -365,10 +373,16 @@ int main(int argc, char** argv) {
       Json inf;
       inf["model"] = model;
       Json fetch;
-      Json sref;
-      sref["peer"] = name;
-      sref["dataset"] = dataset;
-      fetch["scheduler"] = sref;
+      if (!data_uri.empty()) {
+        Json uref;
+        uref["value"] = data_uri;
+        fetch["uri"] = uref;
+      } else {
+        Json sref;
+        sref["peer"] = name;
+        sref["dataset"] = dataset;
+        fetch["scheduler"] = sref;
+      }
       inf["data"] = fetch;
       inf["batch_size"] = batch_size;
       inf["seq_len"] = seq_len;
@@ -508,10 +522,16 @@ int main(int argc, char** argv) {
     Json tr;
     tr["model"] = model;
     Json fetch;
-    Json sref;
-    sref["peer"] = name;
-    sref["dataset"] = dataset;
-    fetch["scheduler"] = sref;
+    if (!data_uri.empty()) {
+      Json uref;
+      uref["value"] = data_uri;
+      fetch["uri"] = uref;
+    } else {
+      Json sref;
+      sref["peer"] = name;
+      sref["dataset"] = dataset;
+      fetch["scheduler"] = sref;
+    }
     tr["data"] = fetch;
     if (rank >= 0) {
       Json rc;

@@ -29,6 +29,7 @@
 #include "hypha/auction.h"
 #include "hypha/bridge.h"
 #include "hypha/gateway.h"
+#include "hypha/http_client.h"
 #include "hypha/json.h"
 #include "hypha/leases.h"
 #include "hypha/net.h"
@@ -93,6 +94,11 @@ struct WorkerDaemon {
   // "<job>/<key>" -> version -> blob. Monotonic versions, latest by default.
   std::mutex kv_mu;
   std::map<std::string, std::map<uint64_t, std::string>> param_store;
+  // URI/HF fetch connector policy (connector/mod.rs:226-302): hosts a job
+  // may fetch from (empty = deny, closing the reference's allow-list TODO)
+  // and the HuggingFace endpoint resolve URLs are built against.
+  std::vector<std::string> fetch_allow;
+  std::string hf_endpoint = "https://huggingface.co";
   // GPU pool: device indices this daemon owns (--gpu-ids). A dispatched
   // process job is pinned to one free device via HIP_VISIBLE_DEVICES so each
   // worker peer IS one GPU (the 8-GPUs-on-one-node deployment: 8 daemons or
@@ -562,6 +568,60 @@ struct WorkerDaemon {
       r["files"] = JsonArray{Json(out)};
       return r;
     }
+    // {"uri": {"value": "http(s)://..."}} -> GET into {work_dir}/artifacts/
+    // (HttpHfFetcher, connector/mod.rs:237-255; artifacts dir + 0600 perms
+    // mirror bridge.rs fetch_resource; allow-list enforced per hop)
+    if (ref.has("uri")) {
+      const std::string uri = ref.at("uri").at("value").as_string();
+      parse_url(uri);  // validate_fetch: http(s) scheme or throw
+      std::string dir = job->work_dir + "/artifacts";
+      mkdir(dir.c_str(), 0700);
+      std::string name = uri.substr(uri.rfind('/') + 1);
+      if (name.empty() || name.find("..") != std::string::npos) name = "item-0";
+      std::string out = dir + "/" + name;
+      long long size = http_get_to_file(uri, out, fetch_allow);
+      Json item;
+      item["path"] = Json(std::string("artifacts/") + name);
+      item["size"] = Json((int64_t)size);
+      Json r;
+      r["files"] = JsonArray{Json(out)};
+      r["items"] = JsonArray{item};
+      return r;
+    }
+    // {"huggingface": {"repository": ..., "revision": ..., "filenames": [...]}}
+    // -> resolve each file against the HF endpoint (connector/mod.rs:256-297;
+    // offline deployments point --hf-endpoint at a mirror)
+    if (ref.has("huggingface")) {
+      const Json& hf = ref.at("huggingface");
+      std::string repo = hf.at("repository").as_string();
+      std::string rev = hf.has("revision") ? hf.at("revision").as_string() : "main";
+      if (repo.find("..") != std::string::npos)
+        throw std::runtime_error("fetch: invalid repository");
+      std::string dir = job->work_dir + "/artifacts";
+      mkdir(dir.c_str(), 0700);
+      Json files = JsonArray{};
+      Json items = JsonArray{};
+      if (hf.has("filenames")) {
+        for (const auto& fn : hf.at("filenames").as_array()) {
+          std::string f = fn.as_string();
+          if (f.empty() || f[0] == '/' || f.find("..") != std::string::npos)
+            throw std::runtime_error("fetch: path traversal is not allowed");
+          std::string url = hf_endpoint + "/" + repo + "/resolve/" + rev + "/" + f;
+          std::string base = f.substr(f.rfind('/') + 1);
+          std::string out = dir + "/" + base;
+          long long size = http_get_to_file(url, out, fetch_allow);
+          files.as_array().push_back(Json(out));
+          Json item;
+          item["path"] = Json(std::string("artifacts/") + base);
+          item["size"] = Json((int64_t)size);
+          items.as_array().push_back(item);
+        }
+      }
+      Json r;
+      r["files"] = files;
+      r["items"] = items;
+      return r;
+    }
     throw std::runtime_error("unsupported fetch reference");
   }
 
@@ -814,7 +874,8 @@ int main(int argc, char** argv) {
   int gw_port = 0, port = 0;
   bool probe = false, init = false;
   std::vector<int> gpu_ids;
-  std::vector<std::string> exclude_cidrs;
+  std::vector<std::string> exclude_cidrs, fetch_allow;
+  std::string hf_endpoint = "https://huggingface.co";
   TlsConfig tls;
   Resources total{1, 4, 16, 100};
   OfferPolicy policy{1.0, 0.0, {"diloco-transformer", "parameter-server"}};
@@ -861,6 +922,8 @@ int main(int argc, char** argv) {
     else if (a == "--tls-ca") tls.ca_path = next();
     else if (a == "--tls-crl") tls.crl_path = next();
     else if (a == "--exclude-cidr") exclude_cidrs.push_back(next());
+    else if (a == "--fetch-allow") fetch_allow.push_back(next());
+    else if (a == "--hf-endpoint") hf_endpoint = next();
     else if (a == "probe") probe = true;
     else if (a == "init") init = true;
   }
@@ -878,7 +941,10 @@ int main(int argc, char** argv) {
            "# --infer-cmd '... hypha_amd.runtime.infer_executor ...'  enables "
            "the inference-transformer executor\n"
            "# --work-root /tmp/hypha-work   per-job working directories\n"
-           "# --tls-cert/--tls-key/--tls-ca [--tls-crl]  mTLS identity\n");
+           "# --tls-cert/--tls-key/--tls-ca [--tls-crl]  mTLS identity\n"
+           "# --fetch-allow host[:port]     allow URI/HF fetches from this host\n"
+           "#                               (repeatable; *.domain wildcards; default: deny all)\n"
+           "# --hf-endpoint https://huggingface.co  HuggingFace resolve endpoint\n");
     return 0;
   }
   if (probe) {
@@ -900,6 +966,8 @@ int main(int argc, char** argv) {
   WorkerDaemon daemon(name, gw_host, gw_port, total, policy, cmd, work_root, tls);
   daemon.gpu_pool = gpu_ids;
   daemon.infer_cmd = icmd;
+  daemon.fetch_allow = fetch_allow;
+  daemon.hf_endpoint = hf_endpoint;
   daemon.node.set_exclude_cidrs(exclude_cidrs);
   daemon.start(port);
   printf("hypha-worker %s ready on port %d\n", name.c_str(), daemon.node.port());

@@ -10,6 +10,7 @@
 #include "hypha/auction.h"
 #include "hypha/batch_scheduler.h"
 #include "hypha/gateway.h"
+#include "hypha/http_client.h"
 #include "hypha/json.h"
 #include "hypha/leases.h"
 #include "hypha/net.h"
@@ -298,6 +299,29 @@ PYBIND11_MODULE(_core, m) {
       .def_readonly("offer_price", &ArbiterDecision::offer_price);
 
   m.def("select_requests", &select_requests);
+  // URI fetch connector internals (connector/mod.rs:226-302 + the
+  // allow-list the reference left as a TODO) - unit-testable offline
+  m.def("parse_url", [](const std::string& u) {
+    HttpUrl p = parse_url(u);
+    py::dict d;
+    d["scheme"] = p.scheme;
+    d["host"] = p.host;
+    d["port"] = p.port;
+    d["path"] = p.path;
+    return d;
+  });
+  m.def("fetch_allowed", [](const std::string& host, int port,
+                            const std::vector<std::string>& allow) {
+    return fetch_allowed(host, port, allow);
+  });
+  m.def("http_get_to_file",
+        [](const std::string& url, const std::string& out,
+           const std::vector<std::string>& allow, double timeout) {
+          py::gil_scoped_release rel;
+          return http_get_to_file(url, out, allow, timeout);
+        },
+        py::arg("url"), py::arg("out"), py::arg("allow"),
+        py::arg("timeout") = 10.0);
 
   // parameter-server file pipeline (the production code path, for golden tests)
   m.def("ps_weighted_average_files",

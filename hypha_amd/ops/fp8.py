@@ -55,7 +55,8 @@ class _DelayedScale:
 
     def __init__(self, device, margin: float = 1.0):
         self.scale = torch.ones(1, dtype=torch.float32, device=device)
-        self.amax = torch.zeros(1, dtype=torch.float32, device=device)
+        # 16 striped accumulator slots (kernel fans atomics across them)
+        self.amax = torch.zeros(16, dtype=torch.float32, device=device)
         self.margin = margin
         self.inited = False
 
@@ -64,7 +65,7 @@ class _DelayedScale:
         C = _C()
         if not self.inited:
             # first call: seed amax from the live tensor (device-side, async)
-            self.amax.copy_(t.detach().abs().amax().float().reshape(1))
+            self.amax[0] = t.detach().abs().amax().float()
             self.inited = True
         # scale_update publishes scale from last window's amax AND zeroes it
         C.fp8_scale_update_(self.amax, self.scale, self.margin)

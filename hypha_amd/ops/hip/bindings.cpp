@@ -68,7 +68,7 @@ torch::Tensor mfma_probe_32x32x16(torch::Tensor a, torch::Tensor b);
 torch::Tensor mfma_probe_16x16x32(torch::Tensor a, torch::Tensor b);
 // fp8_cast.hip
 std::vector<torch::Tensor> fp8_cast_transpose(torch::Tensor x, torch::Tensor scale,
-                                              torch::Tensor amax);
+                                              torch::Tensor amax, long skip_t);
 void fp8_scale_update_(torch::Tensor amax, torch::Tensor scale, double margin);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -99,6 +99,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attn_bwd_ex", &attn_bwd_ex);
   m.def("mfma_probe_32x32x16", &mfma_probe_32x32x16);
   m.def("mfma_probe_16x16x32", &mfma_probe_16x16x32);
-  m.def("fp8_cast_transpose", &fp8_cast_transpose);
+  m.def("fp8_cast_transpose", &fp8_cast_transpose, pybind11::arg("x"),
+        pybind11::arg("scale"), pybind11::arg("amax"), pybind11::arg("skip_t") = 0);
   m.def("fp8_scale_update_", &fp8_scale_update_);
 }

@@ -22,6 +22,7 @@
 namespace {
 
 constexpr float E4M3_MAX = 448.0f;
+constexpr int AMAX_SLOTS = 16;  // striped amax accumulator (atomic fan-out)
 
 // 2 floats -> 2 packed e4m3 bytes (low half of the returned dword).
 __device__ __forceinline__ unsigned short cvt2_fp8(float a, float b) {
@@ -36,121 +37,156 @@ __device__ __forceinline__ float wave_max(float v) {
   return v;
 }
 
-// Tile 64x64, 256 threads: thread t covers rows {t%64}, col-groups
-// {t/64, t/4+4} of 8 -> two vectorized 8-elem loads, two 8-byte stores
-// straight, LDS-staged transpose, two 8-byte stores transposed.
+// Tile 64(rows)x256(cols) per 256-thread block. Design findings (PMC +
+// A/B on hardware):
+//   * the first version barriered after only TWO 16-byte loads per thread
+//     -> 79.6% of wave cycles parked in SQ_WAIT_ANY at 1.4 TB/s: far too
+//     little memory-level parallelism per wave;
+//   * a barrier-free per-wave-slab variant with 16-byte-per-line accesses
+//     dropped to 0.36 TB/s: partial cache-line coverage per instruction
+//     costs more than the barrier it saved.
+// This version issues ALL EIGHT tile loads per thread before any use
+// (128 B in flight per lane), keeps every global instruction on full
+// 64-byte lines (one instruction = 8 rows x 512 contiguous bytes), and
+// pays one barrier per 64 KB tile.
 // scale_io[0] is READ as this call's quant divisor (computed from the
 // previous call's amax by fp8_scale_update_) and passed unchanged to
-// _scaled_mm as the dequant factor; amax_out accumulates via atomicMax.
+// _scaled_mm as the dequant factor; amax_out accumulates via atomicMax,
+// striped over 16 slots (a single-address atomic serializes the grid at
+// one L2 bank: measured 0.48 ms/call vs 15 us roofline).
 __global__ __launch_bounds__(256) void fp8_cast_transpose_kernel(
     const short* __restrict__ xg, unsigned char* __restrict__ out8,
     unsigned char* __restrict__ out8t, const float* __restrict__ scale_io,
-    float* __restrict__ amax_out, int R, int C) {
-  __shared__ unsigned char tile[64][72];  // [col][row], 8-byte padded rows
-  __shared__ float wmax[4];
+    float* __restrict__ amax_out, int R, int C, int skip_t) {
+  __shared__ unsigned char tile[256][72];  // [col][row], 8B-aligned pitch
 
   const int tid = threadIdx.x;
   const int r0 = blockIdx.y * 64;
-  const int c0 = blockIdx.x * 64;
+  const int c0 = blockIdx.x * 256;
   const float rscale = 1.0f / scale_io[0];
-
-  const int row = tid & 63;       // 0..63 within tile
-  const int cg0 = tid >> 6;       // 0..3 -> col groups {cg0, cg0+4}
+  const int row0 = tid >> 5;            // 0..7 (+8 per iteration)
+  const int col = (tid & 31) * 8;       // elem column group 0..248
   float mx = 0.f;
 
+  const bool interior = (r0 + 63 < R) && (c0 + 255 < C);
+
+  // ---- phase 1: issue all 8 row-group loads, then convert/store ----
+  s16x8 v[8];
+  if (interior) {
 #pragma unroll
-  for (int g = 0; g < 2; ++g) {
-    const int col = (cg0 + 4 * g) * 8;  // 0..56 step 8
-    const int gr = r0 + row, gc = c0 + col;
-    float f[8];
-    if (gr < R && gc + 7 < C) {
-      s16x8 v = *reinterpret_cast<const s16x8*>(xg + (long long)gr * C + gc);
+    for (int g = 0; g < 8; ++g)
+      v[g] = *reinterpret_cast<const s16x8*>(
+          xg + (long long)(r0 + row0 + 8 * g) * C + c0 + col);
+  } else {
 #pragma unroll
-      for (int j = 0; j < 8; ++j) f[j] = bf2f(v[j]);
-    } else {
+    for (int g = 0; g < 8; ++g) {
+      const int gr = r0 + row0 + 8 * g, gc = c0 + col;
 #pragma unroll
       for (int j = 0; j < 8; ++j)
-        f[j] = (gr < R && gc + j < C) ? bf2f(xg[(long long)gr * C + gc + j]) : 0.f;
+        v[g][j] = (gr < R && gc + j < C) ? xg[(long long)gr * C + gc + j] : (short)0;
     }
+  }
+
+#pragma unroll
+  for (int g = 0; g < 8; ++g) {
+    const int row = row0 + 8 * g;
+    const int gr = r0 + row, gc = c0 + col;
     unsigned char q[8];
 #pragma unroll
     for (int j = 0; j < 8; j += 2) {
-      mx = fmaxf(mx, fmaxf(fabsf(f[j]), fabsf(f[j + 1])));
-      unsigned short p = cvt2_fp8(f[j] * rscale, f[j + 1] * rscale);
+      float a = bf2f(v[g][j]), b = bf2f(v[g][j + 1]);
+      mx = fmaxf(mx, fmaxf(fabsf(a), fabsf(b)));
+      unsigned short p = cvt2_fp8(a * rscale, b * rscale);
       q[j] = (unsigned char)(p & 0xff);
       q[j + 1] = (unsigned char)(p >> 8);
     }
-    if (gr < R && gc + 7 < C) {
+    if (interior) {
       *reinterpret_cast<uint2*>(out8 + (long long)gr * C + gc) =
           *reinterpret_cast<uint2*>(q);
     } else if (gr < R) {
-      for (int j = 0; j < 8 && gc + j < C; ++j) out8[(long long)gr * C + gc + j] = q[j];
+      for (int j = 0; j < 8; ++j)
+        if (gc + j < C) out8[(long long)gr * C + gc + j] = q[j];
     }
+    if (!(skip_t & 4)) {
 #pragma unroll
-    for (int j = 0; j < 8; ++j) tile[col + j][row] = q[j];
+      for (int j = 0; j < 8; ++j) tile[col + j][row] = q[j];
+    }
   }
 
+  if (skip_t) {  // perf-probe modes: bit0 skip transpose, bit1 skip amax,
+                 // bit2 skip LDS staging (diagnosis only)
+    if (!(skip_t & 2)) {
+      mx = wave_max(mx);
+      if ((tid & 63) == 0 && mx > 0.f)
+        atomicMax(reinterpret_cast<unsigned int*>(
+                      amax_out + (blockIdx.x * 4 + blockIdx.y + (tid >> 6)) % AMAX_SLOTS),
+                  __float_as_uint(mx));
+    }
+    return;
+  }
   __syncthreads();
 
-  // transposed stores: thread t covers cols {t%64} of the ORIGINAL tile
-  // (= rows of out8t), row-groups {t/64, t/64+4} of 8
+  // ---- phase 2: transposed stores ----
+  // one instruction = 8 consecutive out8t rows x 64 contiguous bytes each
+  // (8 full lines): lane -> row (tid>>3), byte offset (tid&7)*8
 #pragma unroll
-  for (int g = 0; g < 2; ++g) {
-    const int rr = (cg0 + 4 * g) * 8;          // original-row group 0..56
-    const int gc = c0 + row, gr = r0 + rr;     // out8t[gc][gr..gr+8)
+  for (int i = 0; i < 8; ++i) {
+    const int tc = 32 * i + (tid >> 3);  // tile col = out8t row
+    const int off = (tid & 7) * 8;
+    const int gc = c0 + tc, gr = r0 + off;
     if (gc >= C) continue;
     if (gr + 7 < R) {
-      uint2 v;
-      memcpy(&v, &tile[row][rr], 8);
-      *reinterpret_cast<uint2*>(out8t + (long long)gc * R + gr) = v;
+      uint2 d;
+      memcpy(&d, &tile[tc][off], 8);
+      *reinterpret_cast<uint2*>(out8t + (long long)gc * R + gr) = d;
     } else {
       for (int j = 0; j < 8 && gr + j < R; ++j)
-        out8t[(long long)gc * R + gr + j] = tile[row][rr + j];
+        out8t[(long long)gc * R + gr + j] = tile[tc][off + j];
     }
   }
 
-  // ONE atomic per block: a per-wave atomic on a single global address
-  // serializes the whole grid at the owning L2 bank (measured 0.48 ms/call
-  // vs ~15 us roofline before this reduction)
   mx = wave_max(mx);
-  if ((tid & 63) == 0) wmax[tid >> 6] = mx;
-  __syncthreads();
-  if (tid == 0) {
-    float m = fmaxf(fmaxf(wmax[0], wmax[1]), fmaxf(wmax[2], wmax[3]));
-    if (m > 0.f)
-      atomicMax(reinterpret_cast<unsigned int*>(amax_out), __float_as_uint(m));
-  }
+  if ((tid & 63) == 0 && mx > 0.f)
+    atomicMax(reinterpret_cast<unsigned int*>(
+                  amax_out + (blockIdx.x * 4 + blockIdx.y + (tid >> 6)) % AMAX_SLOTS),
+              __float_as_uint(mx));
 }
 
-// scale = clamp(amax, eps) / 448 * margin, then RESETS amax for the next
-// accumulation window; one thread. Runs BEFORE the cast kernel each step so
-// the whole delayed-scaling loop stays on-device (no memsets, no host sync).
+// scale = clamp(max over amax slots, eps) / 448 * margin, then RESETS the
+// slots for the next accumulation window; one thread. Runs BEFORE the cast
+// kernel each step so the delayed-scaling loop has no memsets or host syncs.
 __global__ void fp8_scale_update_kernel(float* __restrict__ amax,
                                         float* __restrict__ scale, float margin) {
-  float a = fmaxf(amax[0], 1e-8f);
-  scale[0] = a / E4M3_MAX * margin;
-  amax[0] = 0.f;
+  float a = 0.f;
+#pragma unroll
+  for (int i = 0; i < AMAX_SLOTS; ++i) {
+    a = fmaxf(a, amax[i]);
+    amax[i] = 0.f;
+  }
+  scale[0] = fmaxf(a, 1e-8f) / E4M3_MAX * margin;
 }
 
 }  // namespace
 
 // Dual-layout quantization: x (bf16 [R, C]) -> (x8 [R, C], x8t [C, R]) e4m3,
-// quantized by scale[0]; |x| max accumulated into amax (caller zeroes it).
+// quantized by scale[0]; |x| max accumulated into amax's 16 striped slots
+// (caller zeroes them; fp8_scale_update_ does so as part of each epoch).
 std::vector<torch::Tensor> fp8_cast_transpose(torch::Tensor x, torch::Tensor scale,
-                                              torch::Tensor amax) {
+                                              torch::Tensor amax, long skip_t) {
   TORCH_CHECK(x.is_cuda() && x.dtype() == torch::kBFloat16 && x.dim() == 2 &&
               x.is_contiguous(), "fp8_cast_transpose: bf16 2-D contiguous input");
-  TORCH_CHECK(scale.dtype() == torch::kFloat32 && amax.dtype() == torch::kFloat32);
+  TORCH_CHECK(scale.dtype() == torch::kFloat32 && amax.dtype() == torch::kFloat32 &&
+              amax.numel() >= AMAX_SLOTS, "amax must have 16 slots");
   const int R = x.size(0), C = x.size(1);
   auto opts = x.options().dtype(torch::kFloat8_e4m3fn);
   auto out8 = torch::empty({R, C}, opts);
   auto out8t = torch::empty({C, R}, opts);
-  dim3 grid((C + 63) / 64, (R + 63) / 64);
+  dim3 grid((C + 255) / 256, (R + 63) / 64);
   hipStream_t stream = hypha_stream();
   hipLaunchKernelGGL(fp8_cast_transpose_kernel, grid, dim3(256), 0, stream,
                      (const short*)x.data_ptr(), (unsigned char*)out8.data_ptr(),
                      (unsigned char*)out8t.data_ptr(), scale.data_ptr<float>(),
-                     amax.data_ptr<float>(), R, C);
+                     amax.data_ptr<float>(), R, C, (int)skip_t);
   return {out8, out8t};
 }
 
