@@ -68,8 +68,10 @@ torch::Tensor mfma_probe_32x32x16(torch::Tensor a, torch::Tensor b);
 torch::Tensor mfma_probe_16x16x32(torch::Tensor a, torch::Tensor b);
 // fp8_cast.hip
 std::vector<torch::Tensor> fp8_cast_transpose(torch::Tensor x, torch::Tensor scale,
-                                              torch::Tensor amax, long skip_t);
-void fp8_scale_update_(torch::Tensor amax, torch::Tensor scale, double margin);
+                                              torch::Tensor partials, long skip_t);
+void fp8_scale_update_(torch::Tensor partials, torch::Tensor scale, double margin,
+                       long n);
+long long fp8_cast_grid_size(long long R, long long C);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("adamw_step_", &adamw_step_);
@@ -100,6 +102,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("mfma_probe_32x32x16", &mfma_probe_32x32x16);
   m.def("mfma_probe_16x16x32", &mfma_probe_16x16x32);
   m.def("fp8_cast_transpose", &fp8_cast_transpose, pybind11::arg("x"),
-        pybind11::arg("scale"), pybind11::arg("amax"), pybind11::arg("skip_t") = 0);
-  m.def("fp8_scale_update_", &fp8_scale_update_);
+        pybind11::arg("scale"), pybind11::arg("partials"),
+        pybind11::arg("skip_t") = 0);
+  m.def("fp8_scale_update_", &fp8_scale_update_, pybind11::arg("partials"),
+        pybind11::arg("scale"), pybind11::arg("margin"), pybind11::arg("n") = 0);
+  m.def("fp8_cast_grid_size", &fp8_cast_grid_size);
 }

@@ -22,7 +22,6 @@
 namespace {
 
 constexpr float E4M3_MAX = 448.0f;
-constexpr int AMAX_SLOTS = 16;  // striped amax accumulator (atomic fan-out)
 
 // 2 floats -> 2 packed e4m3 bytes (low half of the returned dword).
 __device__ __forceinline__ unsigned short cvt2_fp8(float a, float b) {
@@ -51,13 +50,17 @@ __device__ __forceinline__ float wave_max(float v) {
 // pays one barrier per 64 KB tile.
 // scale_io[0] is READ as this call's quant divisor (computed from the
 // previous call's amax by fp8_scale_update_) and passed unchanged to
-// _scaled_mm as the dequant factor; amax_out accumulates via atomicMax,
-// striped over 16 slots (a single-address atomic serializes the grid at
-// one L2 bank: measured 0.48 ms/call vs 15 us roofline).
+// _scaled_mm as the dequant factor. amax partials are PLAIN per-block
+// stores into partials[blockIdx] — measured on hardware: atomicMax on one
+// address 0.48 ms/call, striped over 16 slots still 0.10 ms (same-address
+// RMWs serialize at the owning L2 bank); contention-free stores reduced by
+// the next epoch's scale-update kernel cost ~nothing. Each delayed-scaling
+// window is exactly one cast call, so partials are fully overwritten and
+// never need zeroing.
 __global__ __launch_bounds__(256) void fp8_cast_transpose_kernel(
     const short* __restrict__ xg, unsigned char* __restrict__ out8,
     unsigned char* __restrict__ out8t, const float* __restrict__ scale_io,
-    float* __restrict__ amax_out, int R, int C, int skip_t) {
+    float* __restrict__ partials, int R, int C, int skip_t) {
   __shared__ unsigned char tile[256][72];  // [col][row], 8B-aligned pitch
 
   const int tid = threadIdx.x;
@@ -113,14 +116,16 @@ __global__ __launch_bounds__(256) void fp8_cast_transpose_kernel(
     }
   }
 
+  __shared__ float wm[4];
   if (skip_t) {  // perf-probe modes: bit0 skip transpose, bit1 skip amax,
                  // bit2 skip LDS staging (diagnosis only)
     if (!(skip_t & 2)) {
       mx = wave_max(mx);
-      if ((tid & 63) == 0 && mx > 0.f)
-        atomicMax(reinterpret_cast<unsigned int*>(
-                      amax_out + (blockIdx.x * 4 + blockIdx.y + (tid >> 6)) % AMAX_SLOTS),
-                  __float_as_uint(mx));
+      if ((tid & 63) == 0) wm[tid >> 6] = mx;
+      __syncthreads();
+      if (tid == 0)
+        partials[blockIdx.y * gridDim.x + blockIdx.x] =
+            fmaxf(fmaxf(wm[0], wm[1]), fmaxf(wm[2], wm[3]));
     }
     return;
   }
@@ -146,38 +151,49 @@ __global__ __launch_bounds__(256) void fp8_cast_transpose_kernel(
   }
 
   mx = wave_max(mx);
-  if ((tid & 63) == 0 && mx > 0.f)
-    atomicMax(reinterpret_cast<unsigned int*>(
-                  amax_out + (blockIdx.x * 4 + blockIdx.y + (tid >> 6)) % AMAX_SLOTS),
-              __float_as_uint(mx));
+  if ((tid & 63) == 0) wm[tid >> 6] = mx;
+  __syncthreads();
+  if (tid == 0)
+    partials[blockIdx.y * gridDim.x + blockIdx.x] =
+        fmaxf(fmaxf(wm[0], wm[1]), fmaxf(wm[2], wm[3]));
 }
 
-// scale = clamp(max over amax slots, eps) / 448 * margin, then RESETS the
-// slots for the next accumulation window; one thread. Runs BEFORE the cast
-// kernel each step so the delayed-scaling loop has no memsets or host syncs.
-__global__ void fp8_scale_update_kernel(float* __restrict__ amax,
-                                        float* __restrict__ scale, float margin) {
+// scale = clamp(max over the previous cast's per-block partials, eps)
+// / 448 * margin; ONE workgroup, grid-stride. Runs BEFORE the cast kernel
+// each step so the delayed-scaling loop has no memsets or host syncs.
+__global__ __launch_bounds__(256) void fp8_scale_update_kernel(
+    const float* __restrict__ partials, float* __restrict__ scale,
+    float margin, long long n) {
+  __shared__ float wm[4];
   float a = 0.f;
-#pragma unroll
-  for (int i = 0; i < AMAX_SLOTS; ++i) {
-    a = fmaxf(a, amax[i]);
-    amax[i] = 0.f;
-  }
-  scale[0] = fmaxf(a, 1e-8f) / E4M3_MAX * margin;
+  for (long long i = threadIdx.x; i < n; i += 256) a = fmaxf(a, partials[i]);
+  a = wave_max(a);
+  if ((threadIdx.x & 63) == 0) wm[threadIdx.x >> 6] = a;
+  __syncthreads();
+  if (threadIdx.x == 0)
+    scale[0] = fmaxf(fmaxf(fmaxf(wm[0], wm[1]), fmaxf(wm[2], wm[3])), 1e-8f) /
+               E4M3_MAX * margin;
 }
 
 }  // namespace
 
+// Grid (and so the partials-buffer length) for a given input shape.
+long long fp8_cast_grid_size(long long R, long long C) {
+  return ((C + 255) / 256) * ((R + 63) / 64);
+}
+
 // Dual-layout quantization: x (bf16 [R, C]) -> (x8 [R, C], x8t [C, R]) e4m3,
-// quantized by scale[0]; |x| max accumulated into amax's 16 striped slots
-// (caller zeroes them; fp8_scale_update_ does so as part of each epoch).
+// quantized by scale[0]; per-block |x| maxes overwrite `partials`
+// (length >= fp8_cast_grid_size(R, C); reduced by fp8_scale_update_).
 std::vector<torch::Tensor> fp8_cast_transpose(torch::Tensor x, torch::Tensor scale,
-                                              torch::Tensor amax, long skip_t) {
+                                              torch::Tensor partials, long skip_t) {
   TORCH_CHECK(x.is_cuda() && x.dtype() == torch::kBFloat16 && x.dim() == 2 &&
               x.is_contiguous(), "fp8_cast_transpose: bf16 2-D contiguous input");
-  TORCH_CHECK(scale.dtype() == torch::kFloat32 && amax.dtype() == torch::kFloat32 &&
-              amax.numel() >= AMAX_SLOTS, "amax must have 16 slots");
   const int R = x.size(0), C = x.size(1);
+  TORCH_CHECK(scale.dtype() == torch::kFloat32 &&
+              partials.dtype() == torch::kFloat32 &&
+              partials.numel() >= fp8_cast_grid_size(R, C),
+              "partials must cover the cast grid");
   auto opts = x.options().dtype(torch::kFloat8_e4m3fn);
   auto out8 = torch::empty({R, C}, opts);
   auto out8t = torch::empty({C, R}, opts);
@@ -186,14 +202,16 @@ std::vector<torch::Tensor> fp8_cast_transpose(torch::Tensor x, torch::Tensor sca
   hipLaunchKernelGGL(fp8_cast_transpose_kernel, grid, dim3(256), 0, stream,
                      (const short*)x.data_ptr(), (unsigned char*)out8.data_ptr(),
                      (unsigned char*)out8t.data_ptr(), scale.data_ptr<float>(),
-                     amax.data_ptr<float>(), R, C, (int)skip_t);
+                     partials.data_ptr<float>(), R, C, (int)skip_t);
   return {out8, out8t};
 }
 
-// Publishes scale from the accumulated amax and zeroes amax in one launch.
-void fp8_scale_update_(torch::Tensor amax, torch::Tensor scale, double margin) {
+// Publishes scale from the previous cast's per-block partials (one launch,
+// one workgroup). `n` = number of valid partials (the cast's grid size).
+void fp8_scale_update_(torch::Tensor partials, torch::Tensor scale, double margin,
+                       long n) {
   hipStream_t stream = hypha_stream();
-  hipLaunchKernelGGL(fp8_scale_update_kernel, dim3(1), dim3(1), 0, stream,
-                     amax.data_ptr<float>(), scale.data_ptr<float>(),
-                     (float)margin);
+  hipLaunchKernelGGL(fp8_scale_update_kernel, dim3(1), dim3(256), 0, stream,
+                     partials.data_ptr<float>(), scale.data_ptr<float>(),
+                     (float)margin, (long long)(n > 0 ? n : partials.numel()));
 }

@@ -581,7 +581,8 @@ def test_fp8_cast_transpose():
     for R, C in ((128, 256), (100, 72), (4096, 4096)):
         x = (torch.randn(R, C, device=DEV) * 3).bfloat16()
         scale = torch.tensor([0.05], dtype=torch.float32, device=DEV)
-        amax = torch.zeros(16, dtype=torch.float32, device=DEV)
+        n = _C.fp8_cast_grid_size(R, C)
+        amax = torch.zeros(n, dtype=torch.float32, device=DEV)
         x8, x8t = _C.fp8_cast_transpose(x, scale, amax)
         assert x8.shape == (R, C) and x8t.shape == (C, R)
         ref = (x.float() / 0.05).clamp(-448, 448).to(torch.float8_e4m3fn)

@@ -17,7 +17,8 @@ def main():
     iters = int(sys.argv[3]) if len(sys.argv) > 3 else 50
     x = torch.randn(R, C, device=dev).bfloat16()
     scale = torch.tensor([0.05], dtype=torch.float32, device=dev)
-    amax = torch.zeros(16, dtype=torch.float32, device=dev)
+    amax = torch.zeros(_C.fp8_cast_grid_size(R, C), dtype=torch.float32,
+                       device=dev)
     for _ in range(5):
         _C.fp8_cast_transpose(x, scale, amax)
     torch.cuda.synchronize()
