@@ -46,6 +46,8 @@ def main() -> int:
     p.add_argument("--h", type=int, default=100, help="DiLoCo inner steps per outer sync")
     p.add_argument("--device", type=str, default=None)
     p.add_argument("--fp8", action="store_true", help="fp8 e4m3 GEMMs (config 5 path)")
+    p.add_argument("--fp8-weights", action="store_true",
+                   help="fp8 WEIGHT STORAGE + fp8 GEMMs in the lean engine (config 5)")
     p.add_argument("--state-bits", type=int, default=32, choices=(32, 8))
     p.add_argument("--grad-checkpoint", action="store_true",
                    help="per-block activation checkpointing (70B-class fits)")
@@ -97,10 +99,13 @@ def main() -> int:
         inner=InnerOptConfig(lr=4e-4, warmup_steps=10, schedule="constant",
                              state_bits=args.state_bits),
     )
+    if args.fp8_weights and on_gpu:
+        args.memory_mode = "lean"
     if args.memory_mode == "lean":
         from hypha_amd.parallel import LeanDiLoCoWorker
 
-        worker = LeanDiLoCoWorker(model, cfg, comm=comm, device=device)
+        worker = LeanDiLoCoWorker(model, cfg, comm=comm, device=device,
+                                  fp8_weights=args.fp8_weights and on_gpu)
     else:
         worker = DiLoCoWorker(model, cfg, comm=comm, device=device)
     data = SyntheticTokens(
@@ -168,7 +173,9 @@ def main() -> int:
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,
-            "dtype": ("fp8-gemm/bf16" if args.fp8 else "bf16") if on_gpu else "fp32",
+            "dtype": ("fp8-weights/fp8-gemm" if args.fp8_weights
+                      else "fp8-gemm/bf16" if args.fp8 else "bf16")
+                     if on_gpu else "fp32",
             "data": "synthetic",
             "config": {
                 "model": args.model,

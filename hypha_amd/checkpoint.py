@@ -104,20 +104,30 @@ def _save_lean(worker, out_dir: str) -> None:
     os.makedirs(out_dir, exist_ok=True)
     save_file({"theta_global": worker.theta0_host.clone()},
               os.path.join(out_dir, "0_global_weights.safetensors"))
-    save_file(
-        {
-            "outer_momentum": worker.outer_m_host.clone(),
-            "params": worker.flat.cpu(),
-            "m8": worker.m8.cpu(),
-            "v8": worker.v8.cpu(),
-            "m_scale": worker.m_scale.cpu(),
-            "v_scale": worker.v_scale.cpu(),
-        },
-        os.path.join(out_dir, "optimizer_state.safetensors"),
-    )
+    opt_state = {
+        "outer_momentum": worker.outer_m_host.clone(),
+        "params": worker.flat.cpu(),
+        "m8": worker.m8.cpu(),
+        "v8": worker.v8.cpu(),
+        "m_scale": worker.m_scale.cpu(),
+        "v_scale": worker.v_scale.cpu(),
+    }
+    if getattr(worker, "fp8_numel", 0):
+        opt_state.update({
+            "fp8_w8": worker.flat_w8.cpu(),
+            "fp8_wscale": worker.flat_wscale.cpu(),
+            "fp8_m8": worker.m8_f.cpu(),
+            "fp8_v8": worker.v8_f.cpu(),
+            "fp8_m_scale": worker.ms_f.cpu(),
+            "fp8_v_scale": worker.vs_f.cpu(),
+            "fp8_theta0": worker.theta0_fp8_host.clone(),
+            "fp8_outer_momentum": worker.outer_m_fp8_host.clone(),
+        })
+    save_file(opt_state, os.path.join(out_dir, "optimizer_state.safetensors"))
     manifest = {
         "format": "hypha_amd.checkpoint.lean.v1",
         "numel": worker.numel,
+        "fp8_numel": getattr(worker, "fp8_numel", 0),
         "inner_step_count": worker.inner_step_count,
         "round": worker.round,
         "steps_in_round": worker.steps_in_round,
@@ -177,6 +187,17 @@ def _load_lean(worker, ckpt_dir: str, manifest: dict) -> dict:
     worker.v8.copy_(opt["v8"].to(worker.device))
     worker.m_scale.copy_(opt["m_scale"].to(worker.device))
     worker.v_scale.copy_(opt["v_scale"].to(worker.device))
+    if manifest.get("fp8_numel", 0):
+        if manifest["fp8_numel"] != worker.fp8_numel:
+            raise ValueError("checkpoint fp8 numel mismatch")
+        worker.flat_w8.copy_(opt["fp8_w8"].to(worker.device))
+        worker.flat_wscale.copy_(opt["fp8_wscale"].to(worker.device))
+        worker.m8_f.copy_(opt["fp8_m8"].to(worker.device))
+        worker.v8_f.copy_(opt["fp8_v8"].to(worker.device))
+        worker.ms_f.copy_(opt["fp8_m_scale"].to(worker.device))
+        worker.vs_f.copy_(opt["fp8_v_scale"].to(worker.device))
+        worker.theta0_fp8_host.copy_(opt["fp8_theta0"])
+        worker.outer_m_fp8_host.copy_(opt["fp8_outer_momentum"])
     worker.inner_step_count = manifest["inner_step_count"]
     worker.round = manifest["round"]
     worker.steps_in_round = manifest["steps_in_round"]

@@ -97,6 +97,27 @@ class _Fp8State:
         self._wcache = (_FP8_STEP, w8, w8t, sw)
         return w8, w8t, sw
 
+    def lean_weights(self, w8s: torch.Tensor, wscale: torch.Tensor):
+        """GEMM operands from fp8 block-scaled STORAGE (config 5): one
+        fused dequant+per-tensor-requant+transpose pass, cached per step."""
+        if self._wcache is not None and self._wcache[0] == _FP8_STEP:
+            return self._wcache[1], self._wcache[2], self._wcache[3]
+        C = _C()
+        st = self.w
+        n = C.fp8_cast_grid_size(w8s.shape[0], w8s.shape[1])
+        if st.partials is None or st.partials.numel() < n:
+            st.partials = torch.zeros(n, dtype=torch.float32, device=w8s.device)
+            st.inited = False
+        if not st.inited:
+            # amax upper bound without a dequant pass: 448 * max block scale
+            st.partials[0] = wscale.max().float() * E4M3_MAX
+            st.inited = True
+        C.fp8_scale_update_(st.partials, st.scale, st.margin, n)
+        w8, w8t = C.fp8_weight_cast_transpose(w8s, wscale, st.scale, st.partials)
+        sw = st.scale.clone()
+        self._wcache = (_FP8_STEP, w8, w8t, sw)
+        return w8, w8t, sw
+
 
 # Sibling linears consume the SAME activation (wq/wk/wv share the attention
 # input; gate/up share the MLP input): cast it once and share the quantized
@@ -142,14 +163,55 @@ class _Fp8LinearFn(torch.autograd.Function):
         return dx, dw, None
 
 
+class _Fp8LeanLinearFn(torch.autograd.Function):
+    """Linear whose weight lives in fp8 block-scaled STORAGE (config 5).
+
+    The weight is not an nn.Parameter: dw is computed here in backward and
+    handed straight to the lean engine's fused fp8 AdamW (grad-release — the
+    gradient never outlives this call), mirroring lean.py's
+    post-accumulate-hook design one level deeper."""
+
+    @staticmethod
+    def forward(ctx, x2d, mod):
+        state = mod._state
+        x8, x8t, sx = _cast_x_shared(x2d, state)
+        w8, w8t, sw = state.lean_weights(mod.lean_w8, mod.lean_wscale)
+        out = torch._scaled_mm(x8, w8.t(), scale_a=sx, scale_b=sw,
+                               out_dtype=torch.bfloat16)
+        ctx.mod = mod
+        ctx.save_for_backward(x8t, sx, w8t, sw)
+        return out
+
+    @staticmethod
+    def backward(ctx, dy):
+        x8t, sx, w8t, sw = ctx.saved_tensors
+        mod = ctx.mod
+        dy = dy.contiguous()
+        dy8, dy8t, sdy = mod._state.dy.cast(dy)
+        dx = torch._scaled_mm(dy8, w8t.t(), scale_a=sdy, scale_b=sw,
+                              out_dtype=torch.bfloat16)
+        if mod.lean_opt_hook is not None:
+            dw = torch._scaled_mm(dy8t, x8t.t(), scale_a=sdy, scale_b=sx,
+                                  out_dtype=torch.bfloat16)
+            mod.lean_opt_hook(dw)  # fused fp8 AdamW; dw freed on return
+        return dx, None
+
+
 class Fp8Linear(nn.Module):
-    """Drop-in nn.Linear (no bias) running its GEMMs in fp8 on GPU."""
+    """Drop-in nn.Linear (no bias) running its GEMMs in fp8 on GPU.
+
+    Two storage modes: bf16 nn.Parameter master (default — DiLoCo flat
+    buffers unchanged), or fp8 block-scaled storage (`lean_w8`/`lean_wscale`
+    set by LeanDiLoCoWorker(fp8_weights=True); dw routes to lean_opt_hook)."""
 
     def __init__(self, in_features: int, out_features: int):
         super().__init__()
         self.weight = nn.Parameter(torch.empty(out_features, in_features))
         nn.init.normal_(self.weight, std=0.02)
         self._state = None
+        self.lean_w8 = None
+        self.lean_wscale = None
+        self.lean_opt_hook = None
 
     @classmethod
     def from_linear(cls, lin: nn.Linear) -> "Fp8Linear":
@@ -157,6 +219,9 @@ class Fp8Linear(nn.Module):
         nn.Module.__init__(m)
         m.weight = lin.weight
         m._state = None
+        m.lean_w8 = None
+        m.lean_wscale = None
+        m.lean_opt_hook = None
         return m
 
     def forward(self, x):
@@ -165,8 +230,11 @@ class Fp8Linear(nn.Module):
         if self._state is None:
             self._state = _Fp8State(x.device)
         shape = x.shape
-        out = _Fp8LinearFn.apply(x.reshape(-1, shape[-1]).contiguous(),
-                                 self.weight, self._state)
+        x2d = x.reshape(-1, shape[-1]).contiguous()
+        if self.lean_w8 is not None:
+            out = _Fp8LeanLinearFn.apply(x2d, self)
+        else:
+            out = _Fp8LinearFn.apply(x2d, self.weight, self._state)
         return out.reshape(*shape[:-1], -1)
 
 

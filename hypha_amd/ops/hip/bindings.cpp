@@ -72,6 +72,19 @@ std::vector<torch::Tensor> fp8_cast_transpose(torch::Tensor x, torch::Tensor sca
 void fp8_scale_update_(torch::Tensor partials, torch::Tensor scale, double margin,
                        long n);
 long long fp8_cast_grid_size(long long R, long long C);
+// fp8_lean.hip (fp8 weight storage, BASELINE config 5)
+void adamw8_fp8_lean_(torch::Tensor w8, torch::Tensor wscale, torch::Tensor grad,
+                      torch::Tensor m8, torch::Tensor v8, torch::Tensor m_scale,
+                      torch::Tensor v_scale, double lr, double beta1, double beta2,
+                      double eps, double wd, long step, long seed);
+void fp8_extract_delta(torch::Tensor w8, torch::Tensor wscale, torch::Tensor theta0,
+                       torch::Tensor out);
+void fp8_requant_(torch::Tensor theta, torch::Tensor w8, torch::Tensor wscale,
+                  long seed);
+std::vector<torch::Tensor> fp8_weight_cast_transpose(torch::Tensor w8s,
+                                                     torch::Tensor wscale,
+                                                     torch::Tensor scale,
+                                                     torch::Tensor partials);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("adamw_step_", &adamw_step_);
@@ -107,4 +120,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fp8_scale_update_", &fp8_scale_update_, pybind11::arg("partials"),
         pybind11::arg("scale"), pybind11::arg("margin"), pybind11::arg("n") = 0);
   m.def("fp8_cast_grid_size", &fp8_cast_grid_size);
+  m.def("adamw8_fp8_lean_", &adamw8_fp8_lean_);
+  m.def("fp8_extract_delta", &fp8_extract_delta);
+  m.def("fp8_requant_", &fp8_requant_);
+  m.def("fp8_weight_cast_transpose", &fp8_weight_cast_transpose);
 }

@@ -32,7 +32,7 @@ QBLOCK = 2048
 
 class LeanDiLoCoWorker:
     def __init__(self, model: nn.Module, cfg: DiLoCoConfig, comm: Comm | None = None,
-                 device: torch.device | None = None):
+                 device: torch.device | None = None, fp8_weights: bool = False):
         from hypha_amd import ops
 
         assert torch.cuda.is_available(), "lean mode needs a GPU"
@@ -49,8 +49,26 @@ class LeanDiLoCoWorker:
             if buf.dtype in (torch.bfloat16, torch.float16):
                 buf.data = buf.data.float()
 
+        # fp8 WEIGHT STORAGE (BASELINE config 5): every Fp8Linear's weight
+        # moves into e4m3 + per-2048-block scales (1 B/param instead of 2);
+        # their dw is applied by the fused fp8 AdamW inside backward. The
+        # remaining (bf16) params follow the original lean path below.
+        self.fp8_weights = fp8_weights
+        self._fp8_mods = []
+        if fp8_weights:
+            from hypha_amd.ops.fp8 import Fp8Linear, convert_linears_to_fp8
+
+            convert_linears_to_fp8(model)
+            for m in model.modules():
+                if isinstance(m, Fp8Linear) and m.weight is not None:
+                    assert m.weight.numel() % QBLOCK == 0, (
+                        "fp8 weight storage needs QBLOCK-aligned shapes")
+                    self._fp8_mods.append(m)
+
         # flatten bf16 params (the single weight copy)
-        self.params = [p for p in model.parameters() if p.requires_grad]
+        fp8_param_ids = {id(m.weight) for m in self._fp8_mods}
+        self.params = [p for p in model.parameters()
+                       if p.requires_grad and id(p) not in fp8_param_ids]
         self.numel = sum(p.numel() for p in self.params)
         self.flat = torch.empty(self.numel, dtype=torch.bfloat16, device=self.device)
         offset = 0
@@ -77,6 +95,50 @@ class LeanDiLoCoWorker:
         self.m_scale = torch.full((nblocks,), 1e-12, device=self.device)
         self.v_scale = torch.full((nblocks,), 1e-12, device=self.device)
 
+        # ---- fp8 weight-storage group (config 5) ----
+        self.fp8_numel = sum(m.weight.numel() for m in self._fp8_mods)
+        if self.fp8_numel:
+            n8 = self.fp8_numel
+            self.flat_w8 = torch.empty(n8, dtype=torch.uint8, device=self.device)
+            self.flat_wscale = torch.empty(n8 // QBLOCK, dtype=torch.float32,
+                                           device=self.device)
+            self.m8_f = torch.full((n8,), 127, dtype=torch.uint8, device=self.device)
+            self.v8_f = torch.zeros(n8, dtype=torch.uint8, device=self.device)
+            self.ms_f = torch.full((n8 // QBLOCK,), 1e-12, device=self.device)
+            self.vs_f = torch.full((n8 // QBLOCK,), 1e-12, device=self.device)
+            off = 0
+            self._fp8_off = []
+            for m in self._fp8_mods:
+                n = m.weight.numel()
+                shape = m.weight.shape
+                # quantize the initial bf16 weight into storage, then DROP
+                # the bf16 parameter — fp8 is the only weight copy
+                _C.fp8_requant_(m.weight.data.reshape(-1).contiguous(),
+                                self.flat_w8[off:off + n],
+                                self.flat_wscale[off // QBLOCK:(off + n) // QBLOCK],
+                                40503)
+                m.weight = None
+                m.lean_w8 = self.flat_w8[off:off + n].view(shape)
+                m.lean_wscale = self.flat_wscale[off // QBLOCK:(off + n) // QBLOCK]
+                m.lean_opt_hook = self._make_fp8_hook(off, n)
+                self._fp8_off.append(off)
+                off += n
+            if self.comm.is_distributed:
+                self.comm.broadcast_flat(self.flat_w8, src=0)
+                self.comm.broadcast_flat(self.flat_wscale, src=0)
+            # host-resident outer state for the fp8 group (bf16 dequant)
+            t0f = torch.empty(n8, dtype=torch.bfloat16, device=self.device)
+            chunk0 = min(n8, 64 * 1024 * 1024)
+            zero = torch.zeros(chunk0, dtype=torch.bfloat16, device=self.device)
+            for s in range(0, n8, chunk0):
+                e = min(s + chunk0, n8)
+                _C.fp8_extract_delta(self.flat_w8[s:e],
+                                     self.flat_wscale[s // QBLOCK:(e + QBLOCK - 1) // QBLOCK],
+                                     zero[:e - s], t0f[s:e])
+            self.theta0_fp8_host = t0f.cpu()
+            self.outer_m_fp8_host = torch.zeros_like(self.theta0_fp8_host)
+            del t0f, zero
+
         # host-resident outer state (bf16)
         self.theta0_host = self.flat.detach().cpu().clone()
         self.outer_m_host = torch.zeros_like(self.theta0_host)
@@ -91,11 +153,26 @@ class LeanDiLoCoWorker:
         for idx, p in enumerate(self.params):
             p.register_post_accumulate_grad_hook(self._make_hook(idx))
 
-        chunk = min(self.numel, 64 * 1024 * 1024)
+        chunk = min(max(self.numel, self.fp8_numel), 64 * 1024 * 1024)
         self._chunk = chunk
         self._stage_t0 = torch.empty(chunk, dtype=torch.bfloat16, device=self.device)
         self._stage_m = torch.empty(chunk, dtype=torch.bfloat16, device=self.device)
         self._stage_d = torch.empty(chunk, dtype=torch.bfloat16, device=self.device)
+
+    def _make_fp8_hook(self, off: int, n: int):
+        c = self.cfg.inner
+        b0, b1 = off // QBLOCK, (off + n) // QBLOCK
+
+        def hook(dw: torch.Tensor):
+            self._C.adamw8_fp8_lean_(
+                self.flat_w8[off:off + n], self.flat_wscale[b0:b1],
+                dw.reshape(-1), self.m8_f[off:off + n], self.v8_f[off:off + n],
+                self.ms_f[b0:b1], self.vs_f[b0:b1],
+                self._cur_lr, c.beta1, c.beta2, c.eps, c.weight_decay,
+                self.inner_step_count, self.inner_step_count * 2654435761 % (2**31),
+            )
+
+        return hook
 
     def _make_hook(self, idx: int):
         c = self.cfg.inner
@@ -157,6 +234,31 @@ class LeanDiLoCoWorker:
             self.theta0_host[start:start + m].copy_(t0)
             self.outer_m_host[start:start + m].copy_(mo)
             self.flat[start:start + m].copy_(t0)
+        # fp8 weight-storage group: dequant-delta, all-reduce, Nesterov on
+        # the bf16 global weights, SR-requant back into storage
+        if self.fp8_numel:
+            n8 = self.fp8_numel
+            rseed = (self.round + 1) * 48611 % (2**31)
+            for start in range(0, n8, self._chunk):
+                m = min(self._chunk, n8 - start)
+                t0 = self._stage_t0[:m]
+                mo = self._stage_m[:m]
+                d = self._stage_d[:m]
+                t0.copy_(self.theta0_fp8_host[start:start + m], non_blocking=False)
+                mo.copy_(self.outer_m_fp8_host[start:start + m], non_blocking=False)
+                b0 = start // QBLOCK
+                b1 = (start + m + QBLOCK - 1) // QBLOCK
+                self._C.fp8_extract_delta(self.flat_w8[start:start + m],
+                                          self.flat_wscale[b0:b1], t0, d)
+                self.comm.all_reduce_mean_flat(d)
+                self._C.nesterov_bf16_(t0, d, mo, self.cfg.outer.lr,
+                                       self.cfg.outer.momentum, rseed)
+                self.theta0_fp8_host[start:start + m].copy_(t0)
+                self.outer_m_fp8_host[start:start + m].copy_(mo)
+                self._C.fp8_requant_(t0, self.flat_w8[start:start + m],
+                                     self.flat_wscale[b0:b1], rseed ^ 0x5bd1e995)
+            self.outer_sync_payload_bytes += n8 * 2
+
         self.outer_sync_payload_bytes += n * 2
         self.round += 1
         self.steps_in_round = 0
@@ -168,6 +270,6 @@ class LeanDiLoCoWorker:
             "outer_sync_payload_bytes": payload,
             "outer_sync_wire_bytes_per_rank": self.comm.wire_bytes_per_rank(payload),
             "h": self.cfg.h,
-            "model_numel": self.numel,
-            "memory_mode": "lean",
+            "model_numel": self.numel + self.fp8_numel,
+            "memory_mode": "lean-fp8" if self.fp8_numel else "lean",
         }
