@@ -69,6 +69,8 @@ def main() -> int:
     if on_gpu:
         ops.native_available_or_raise()
 
+    if args.fp8_weights and on_gpu:
+        args.memory_mode = "lean"  # fp8 weight storage lives in the lean engine
     torch.manual_seed(1234)
     # build directly on the GPU: 8 ranks x 32 GB of fp32 CPU-side init would
     # strain host RAM and add ~a minute per rank; rank-0's broadcast makes
@@ -99,8 +101,6 @@ def main() -> int:
         inner=InnerOptConfig(lr=4e-4, warmup_steps=10, schedule="constant",
                              state_bits=args.state_bits),
     )
-    if args.fp8_weights and on_gpu:
-        args.memory_mode = "lean"
     if args.memory_mode == "lean":
         from hypha_amd.parallel import LeanDiLoCoWorker
 
@@ -158,7 +158,8 @@ def main() -> int:
     synced_now = worker.round - rounds_before
     step_s_ex_sync = max(1e-9, elapsed - synced_now * outer_sync_s) / args.steps
 
-    model_numel = worker.numel if args.memory_mode == "lean" else worker.fp.numel
+    model_numel = (worker.numel + getattr(worker, "fp8_numel", 0)
+                   if args.memory_mode == "lean" else worker.fp.numel)
     payload_bytes_per_sync = model_numel * 2  # bf16 comm dtype
     synced_in_window = worker.round - rounds_before
     if rank == 0:

@@ -102,42 +102,46 @@ class LeanDiLoCoWorker:
             self.flat_w8 = torch.empty(n8, dtype=torch.uint8, device=self.device)
             self.flat_wscale = torch.empty(n8 // QBLOCK, dtype=torch.float32,
                                            device=self.device)
-            self.m8_f = torch.full((n8,), 127, dtype=torch.uint8, device=self.device)
-            self.v8_f = torch.zeros(n8, dtype=torch.uint8, device=self.device)
-            self.ms_f = torch.full((n8 // QBLOCK,), 1e-12, device=self.device)
-            self.vs_f = torch.full((n8 // QBLOCK,), 1e-12, device=self.device)
+            # quantize + FREE the bf16 weights module by module BEFORE
+            # allocating optimizer state: peak device memory stays
+            # bf16(live) + w8, never bf16 + w8 + m8/v8 (70B would OOM)
             off = 0
             self._fp8_off = []
             for m in self._fp8_mods:
                 n = m.weight.numel()
                 shape = m.weight.shape
-                # quantize the initial bf16 weight into storage, then DROP
-                # the bf16 parameter — fp8 is the only weight copy
                 _C.fp8_requant_(m.weight.data.reshape(-1).contiguous(),
                                 self.flat_w8[off:off + n],
                                 self.flat_wscale[off // QBLOCK:(off + n) // QBLOCK],
                                 40503)
-                m.weight = None
+                m.weight = None  # fp8 is now the only weight copy
                 m.lean_w8 = self.flat_w8[off:off + n].view(shape)
                 m.lean_wscale = self.flat_wscale[off // QBLOCK:(off + n) // QBLOCK]
                 m.lean_opt_hook = self._make_fp8_hook(off, n)
                 self._fp8_off.append(off)
                 off += n
+            torch.cuda.empty_cache()
+            self.m8_f = torch.full((n8,), 127, dtype=torch.uint8, device=self.device)
+            self.v8_f = torch.zeros(n8, dtype=torch.uint8, device=self.device)
+            self.ms_f = torch.full((n8 // QBLOCK,), 1e-12, device=self.device)
+            self.vs_f = torch.full((n8 // QBLOCK,), 1e-12, device=self.device)
             if self.comm.is_distributed:
                 self.comm.broadcast_flat(self.flat_w8, src=0)
                 self.comm.broadcast_flat(self.flat_wscale, src=0)
-            # host-resident outer state for the fp8 group (bf16 dequant)
-            t0f = torch.empty(n8, dtype=torch.bfloat16, device=self.device)
+            # host-resident outer state for the fp8 group, built chunk-wise
+            # (a full-size device dequant buffer would be another 2P bytes)
+            self.theta0_fp8_host = torch.empty(n8, dtype=torch.bfloat16)
             chunk0 = min(n8, 64 * 1024 * 1024)
             zero = torch.zeros(chunk0, dtype=torch.bfloat16, device=self.device)
+            dq = torch.empty(chunk0, dtype=torch.bfloat16, device=self.device)
             for s in range(0, n8, chunk0):
                 e = min(s + chunk0, n8)
                 _C.fp8_extract_delta(self.flat_w8[s:e],
                                      self.flat_wscale[s // QBLOCK:(e + QBLOCK - 1) // QBLOCK],
-                                     zero[:e - s], t0f[s:e])
-            self.theta0_fp8_host = t0f.cpu()
+                                     zero[:e - s], dq[:e - s])
+                self.theta0_fp8_host[s:e].copy_(dq[:e - s])
             self.outer_m_fp8_host = torch.zeros_like(self.theta0_fp8_host)
-            del t0f, zero
+            del dq, zero
 
         # host-resident outer state (bf16)
         self.theta0_host = self.flat.detach().cpu().clone()
