@@ -51,7 +51,7 @@ def _C():
 class _DelayedScale:
     """Per-tensor-role delayed-scaling state: scale + double-buffered amax."""
 
-    __slots__ = ("scale", "partials", "margin", "inited")
+    __slots__ = ("scale", "partials", "margin", "inited", "epoch")
 
     def __init__(self, device, margin: float = 1.0):
         self.scale = torch.ones(1, dtype=torch.float32, device=device)
@@ -61,9 +61,21 @@ class _DelayedScale:
         self.partials = None
         self.margin = margin
         self.inited = False
+        self.epoch = -1
+
+    def _advance(self, n: int):
+        """Publish a new scale from the last epoch's partials — ONCE per
+        fp8 epoch. Re-casts within an epoch (activation-checkpoint
+        recompute) reuse the same scale so the recomputed values are
+        bit-identical to the original forward; a per-call advance made MoE
+        routing counts differ between forward and recompute (shape-mismatch
+        abort under torch checkpointing)."""
+        if self.epoch != _FP8_STEP:
+            _C().fp8_scale_update_(self.partials, self.scale, self.margin, n)
+            self.epoch = _FP8_STEP
 
     def cast(self, t: torch.Tensor):
-        """t (bf16 2-D) -> (t8, t8t, scale). Uses last call's amax."""
+        """t (bf16 2-D) -> (t8, t8t, scale). Uses last epoch's amax."""
         C = _C()
         n = C.fp8_cast_grid_size(t.shape[0], t.shape[1])
         if self.partials is None or self.partials.numel() < n:
@@ -73,7 +85,8 @@ class _DelayedScale:
             # first call: seed amax from the live tensor (device-side, async)
             self.partials[0] = t.detach().abs().amax().float()
             self.inited = True
-        C.fp8_scale_update_(self.partials, self.scale, self.margin, n)
+            self.epoch = -1
+        self._advance(n)
         t8, t8t = C.fp8_cast_transpose(t, self.scale, self.partials)
         return t8, t8t, self.scale
 
@@ -99,9 +112,12 @@ class _Fp8State:
 
     def lean_weights(self, w8s: torch.Tensor, wscale: torch.Tensor):
         """GEMM operands from fp8 block-scaled STORAGE (config 5): one
-        fused dequant+per-tensor-requant+transpose pass, cached per step."""
-        if self._wcache is not None and self._wcache[0] == _FP8_STEP:
-            return self._wcache[1], self._wcache[2], self._wcache[3]
+        fused dequant+per-tensor-requant+transpose pass.
+
+        Deliberately NOT cached across the step: a per-step cache would pin
+        w8+w8t for every layer at once — +2x weight bytes (136 GB at 70B,
+        measured OOM at b8). Operands are produced per use and die with the
+        checkpoint segment that saved them; the recompute pass re-casts."""
         C = _C()
         st = self.w
         n = C.fp8_cast_grid_size(w8s.shape[0], w8s.shape[1])
@@ -112,11 +128,10 @@ class _Fp8State:
             # amax upper bound without a dequant pass: 448 * max block scale
             st.partials[0] = wscale.max().float() * E4M3_MAX
             st.inited = True
-        C.fp8_scale_update_(st.partials, st.scale, st.margin, n)
+            st.epoch = -1
+        st._advance(n)
         w8, w8t = C.fp8_weight_cast_transpose(w8s, wscale, st.scale, st.partials)
-        sw = st.scale.clone()
-        self._wcache = (_FP8_STEP, w8, w8t, sw)
-        return w8, w8t, sw
+        return w8, w8t, st.scale.clone()
 
 
 # Sibling linears consume the SAME activation (wq/wk/wv share the attention
