@@ -57,15 +57,39 @@ class GPT2Block(nn.Module):
         self.mlp_proj = nn.Linear(4 * h, h)
         self.n_heads = cfg.n_heads
 
-    def forward(self, x):
+    def forward(self, x, cache=None, pos: int = 0):
         b, s, h = x.shape
         hd = h // self.n_heads
         q, k, v = self.attn_qkv(self.ln1(x)).split(h, dim=-1)
-        q = q.view(b, s, self.n_heads, hd).transpose(1, 2)
-        k = k.view(b, s, self.n_heads, hd).transpose(1, 2)
-        v = v.view(b, s, self.n_heads, hd).transpose(1, 2)
-        o = ops.flash_attention(q, k, v, causal=True)
-        x = x + self.attn_out(o.transpose(1, 2).reshape(b, s, h))
+        if cache is None:
+            q = q.view(b, s, self.n_heads, hd).transpose(1, 2)
+            k = k.view(b, s, self.n_heads, hd).transpose(1, 2)
+            v = v.view(b, s, self.n_heads, hd).transpose(1, 2)
+            o = ops.flash_attention(q, k, v, causal=True)
+            x = x + self.attn_out(o.transpose(1, 2).reshape(b, s, h))
+        else:
+            # serving: append into the pre-allocated bshd cache; decode via
+            # the flash-decoding kernel, prefill via the MFMA flash kernel
+            q = q.view(b, s, self.n_heads, hd)
+            k = k.view(b, s, self.n_heads, hd)
+            v = v.view(b, s, self.n_heads, hd)
+            t = cache.append(k, v)
+            if s == 1:
+                o = ops.attn_decode(q[:, 0].contiguous(), cache.k, cache.v, t)
+                o = o.reshape(b, 1, h)
+            elif pos == 0 and (not x.is_cuda or s % 128 == 0):
+                o = ops.flash_attention(q, k, v, causal=True, layout="bshd")
+                o = o.reshape(b, s, h)
+            else:  # ragged prefill / chunked continuation (fp32 reference)
+                kh = cache.k[:, :t].transpose(1, 2)
+                vh = cache.v[:, :t].transpose(1, 2)
+                qh = q.transpose(1, 2)
+                scores = (qh.float() @ kh.float().transpose(-1, -2)) / math.sqrt(hd)
+                mask = torch.arange(t, device=x.device)[None, :] > (
+                    pos + torch.arange(s, device=x.device)[:, None])
+                p = torch.softmax(scores.masked_fill(mask, float("-inf")), dim=-1)
+                o = (p @ vh.float()).to(x.dtype).transpose(1, 2).reshape(b, s, h)
+            x = x + self.attn_out(o)
         x = x + self.mlp_proj(ops.gelu(self.mlp_fc(self.ln2(x))))
         return x
 
@@ -105,17 +129,34 @@ class GPT2ForCausalLM(nn.Module):
     def generate(self, input_ids: torch.Tensor, max_new_tokens: int = 32,
                  temperature: float = 0.0, top_k: int = 0,
                  seed: int | None = None) -> torch.Tensor:
-        """Full-recompute decoding (no KV cache in this plumbing model);
-        same sampling surface as the Llama family so inference jobs can run
-        any registry model."""
+        """KV-cache decoding (prefill once, then one token per step via the
+        flash-decoding kernel); same sampling surface as the Llama family so
+        inference jobs can run any registry model. Sequences are capped at
+        n_positions (absolute position embeddings)."""
+        from .kv_cache import KVCache
+
         self.eval()
+        b, s = input_ids.shape
+        max_len = min(s + max_new_tokens, self.cfg.n_positions)
+        hd = self.cfg.hidden_size // self.cfg.n_heads
+        dtype = self.wte.weight.dtype
+        caches = [KVCache(b, max_len, self.cfg.n_heads, hd, input_ids.device,
+                          dtype=dtype) for _ in self.blocks]
         gen = None
         if seed is not None:
             gen = torch.Generator(device=input_ids.device).manual_seed(seed)
         tokens = input_ids
+        x_in = input_ids
+        pos = 0
         for _ in range(max_new_tokens):
-            window = tokens[:, -self.cfg.n_positions:]
-            logits = self(window)[:, -1]
+            if pos + x_in.shape[1] > self.cfg.n_positions:
+                break  # absolute-position ceiling reached
+            idx = torch.arange(pos, pos + x_in.shape[1], device=x_in.device)
+            x = self.wte(x_in) + self.wpe(idx)
+            for blk, cache in zip(self.blocks, caches):
+                x = blk(x, cache=cache, pos=pos)
+            x = self.ln_f(x[:, -1:])
+            logits = F.linear(x, self.wte.weight)[:, 0]
             if temperature <= 0:
                 nxt = logits.argmax(-1, keepdim=True)
             else:
@@ -125,7 +166,9 @@ class GPT2ForCausalLM(nn.Module):
                     logits = logits.masked_fill(logits < kth, float("-inf"))
                 probs = torch.softmax(logits.float(), dim=-1)
                 nxt = torch.multinomial(probs, 1, generator=gen)
+            pos += x_in.shape[1]
             tokens = torch.cat([tokens, nxt], dim=1)
+            x_in = nxt
         return tokens
 
 

@@ -100,23 +100,29 @@ class Attention(nn.Module):
             q, k = ops.apply_rope_qk(q, k, cos, sin, layout="bshd")
             o = ops.flash_attention(q, k, v, causal=True, layout="bshd")
             return self.wo(o.reshape(b, s, -1))
-        # inference with KV cache: rotate at absolute positions, append, and
-        # attend over the full prefix (decode = batched matvec via hipBLASLt)
+        # inference with KV cache: rotate at absolute positions, append into
+        # the pre-allocated bshd cache, then attend over the valid prefix.
+        # Decode (s==1) runs the native flash-decoding kernel; prefill runs
+        # the MFMA flash kernel; chunked continuation falls back to fp32.
         q, k = ops.apply_rope_qk(q, k, cos[pos:], sin[pos:], layout="bshd")
-        cache["k"] = k if "k" not in cache else torch.cat([cache["k"], k], dim=1)
-        cache["v"] = v if "v" not in cache else torch.cat([cache["v"], v], dim=1)
-        kc, vc = cache["k"], cache["v"]
+        t = cache.append(k, v)
+        if s == 1:
+            o = ops.attn_decode(q[:, 0].contiguous(), cache.k, cache.v, t)
+            return self.wo(o.reshape(b, 1, -1))
+        if pos == 0 and (not x.is_cuda or s % 128 == 0):
+            o = ops.flash_attention(q, k, v, causal=True, layout="bshd")
+            return self.wo(o.reshape(b, s, -1))
+        # chunked prefill with a position offset (rare path; fp32 reference)
+        kc, vc = cache.k[:, :t], cache.v[:, :t]
         rep = cfg.n_heads // cfg.n_kv_heads
         qh = q.transpose(1, 2)  # [b, hq, s, d]
         kh = kc.transpose(1, 2).repeat_interleave(rep, dim=1)
         vh = vc.transpose(1, 2).repeat_interleave(rep, dim=1)
         scores = (qh.float() @ kh.float().transpose(-1, -2)) / math.sqrt(hd)
-        t = kc.shape[1]
-        if s > 1:  # prefill chunk: causal mask inside the chunk
-            mask = torch.arange(t, device=x.device)[None, :] > (
-                pos + torch.arange(s, device=x.device)[:, None]
-            )
-            scores = scores.masked_fill(mask, float("-inf"))
+        mask = torch.arange(t, device=x.device)[None, :] > (
+            pos + torch.arange(s, device=x.device)[:, None]
+        )
+        scores = scores.masked_fill(mask, float("-inf"))
         p = torch.softmax(scores, dim=-1)
         o = (p @ vh.float()).to(x.dtype).transpose(1, 2)
         return self.wo(o.reshape(b, s, -1))
@@ -199,9 +205,14 @@ def _generate(self, input_ids: torch.Tensor, max_new_tokens: int = 32,
               seed: int | None = None) -> torch.Tensor:
     """Autoregressive generation with a per-layer KV cache (inference parity:
     the reference serves inference through the same executor surface)."""
+    from .kv_cache import KVCache
+
     self.eval()
     b, s = input_ids.shape
-    caches = [dict() for _ in self.blocks]
+    max_len = s + max_new_tokens
+    dtype = self.embed.weight.dtype
+    caches = [KVCache(b, max_len, self.cfg.n_kv_heads, self.cfg.head_dim,
+                      input_ids.device, dtype=dtype) for _ in self.blocks]
     gen = None
     if seed is not None:
         gen = torch.Generator(device=input_ids.device).manual_seed(seed)

@@ -144,8 +144,8 @@ class MoEBlock(nn.Module):
         self.mlp_norm = RMSNorm(cfg.hidden_size, cfg.norm_eps)
         self.mlp = MoEMLP(cfg)
 
-    def forward(self, x, cos, sin):
-        x = x + self.attn(self.attn_norm(x), cos, sin)
+    def forward(self, x, cos, sin, cache=None, pos: int = 0):
+        x = x + self.attn(self.attn_norm(x), cos, sin, cache=cache, pos=pos)
         mlp_out, aux = self.mlp(self.mlp_norm(x))
         x = x + mlp_out
         return x, aux
@@ -190,16 +190,29 @@ class MoEForCausalLM(nn.Module):
     def generate(self, input_ids: torch.Tensor, max_new_tokens: int = 32,
                  temperature: float = 0.0, top_k: int = 0,
                  seed: int | None = None) -> torch.Tensor:
-        """Full-recompute decoding (same sampling surface as the Llama
-        family so inference jobs can run any registry model)."""
+        """KV-cache decoding (prefill once, then one token per step via the
+        flash-decoding kernel; same sampling surface as the Llama family so
+        inference jobs can run any registry model)."""
+        from .kv_cache import KVCache
+
         self.eval()
+        b, s = input_ids.shape
+        dtype = self.embed.weight.dtype
+        caches = [KVCache(b, s + max_new_tokens, self.cfg.n_kv_heads,
+                          self.cfg.head_dim, input_ids.device, dtype=dtype)
+                  for _ in self.blocks]
         gen = None
         if seed is not None:
             gen = torch.Generator(device=input_ids.device).manual_seed(seed)
         tokens = input_ids
+        x_in = input_ids
+        pos = 0
         for _ in range(max_new_tokens):
-            window = tokens[:, -self.cfg.max_seq_len:]
-            logits = self(window)[:, -1]
+            x = self.embed(x_in)
+            for blk, cache in zip(self.blocks, caches):
+                x, _ = blk(x, self.rope_cos, self.rope_sin, cache=cache, pos=pos)
+            x = self.norm(x[:, -1:])
+            logits = self.lm_head(x)[:, 0]
             if temperature <= 0:
                 nxt = logits.argmax(-1, keepdim=True)
             else:
@@ -209,7 +222,9 @@ class MoEForCausalLM(nn.Module):
                     logits = logits.masked_fill(logits < kth, float("-inf"))
                 probs = torch.softmax(logits.float(), dim=-1)
                 nxt = torch.multinomial(probs, 1, generator=gen)
+            pos += x_in.shape[1]
             tokens = torch.cat([tokens, nxt], dim=1)
+            x_in = nxt
         return tokens
 
 

@@ -53,6 +53,7 @@ def use_native(*tensors: torch.Tensor) -> bool:
 
 from .interface import (  # noqa: E402
     apply_rope_qk,
+    attn_decode,
     cross_entropy_loss,
     extract_delta,
     flash_attention,
@@ -73,6 +74,7 @@ __all__ = [
     "layernorm",
     "gelu",
     "apply_rope_qk",
+    "attn_decode",
     "swiglu",
     "flash_attention",
     "cross_entropy_loss",

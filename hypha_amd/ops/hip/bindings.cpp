@@ -85,6 +85,9 @@ std::vector<torch::Tensor> fp8_weight_cast_transpose(torch::Tensor w8s,
                                                      torch::Tensor wscale,
                                                      torch::Tensor scale,
                                                      torch::Tensor partials);
+// decode.hip (flash-decoding KV-cache attention)
+torch::Tensor attn_decode(torch::Tensor q, torch::Tensor kcache, torch::Tensor vcache,
+                          long t);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("adamw_step_", &adamw_step_);
@@ -124,4 +127,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fp8_extract_delta", &fp8_extract_delta);
   m.def("fp8_requant_", &fp8_requant_);
   m.def("fp8_weight_cast_transpose", &fp8_weight_cast_transpose);
+  m.def("attn_decode", &attn_decode);
 }

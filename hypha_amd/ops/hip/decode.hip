@@ -1,0 +1,232 @@
+// KV-cache decode attention (flash-decoding) for single-token generation —
+// the MI355X-native serving path (replaces the round-1 Python fp32 matmul
+// decode noted in VERDICT.md; reference serves inference through the same
+// executor surface, /root/reference/executors' inference flow).
+//
+// Problem shape: q [B, Hq, D] (one new token per sequence), KV cache
+// [B, T_alloc, Hkv, D] (bshd — the projections' natural layout, appended
+// in place), valid length t. Decode is BANDWIDTH-bound: the whole K and V
+// prefix is read once; all arithmetic is VALU fp32 (MFMA needs matrix
+// shapes a 1-token query cannot fill).
+//
+// Parallelization: grid (nsplits, B*Hkv) — flash-decoding split-KV so small
+// batches still fill 256 CUs. One workgroup = 4 waves covers ONE (b, hkv)
+// pair and ALL G = Hq/Hkv grouped q-heads over its kv chunk: K/V tiles are
+// staged cooperatively into LDS once and shared by every q head (GQA cuts
+// cache-read bytes by G vs a per-q-head layout). Each wave owns
+// ceil(G/4) heads; per 64-row tile it computes that head's scores (one kv
+// row per lane, online softmax) and then accumulates p*V with lanes over
+// the head dim. Split partials (o, m, l) merge in a second kernel with the
+// standard log-sum-exp rescale.
+
+#include <torch/extension.h>
+
+#include "hip_common.h"
+
+namespace {
+
+constexpr int TILE = 64;  // kv rows per staged tile
+
+// ---- main split kernel -------------------------------------------------
+// partial_o: [nsplits, B, Hq, D] fp32; partial_ml: [nsplits, B, Hq, 2]
+template <int HD>
+__global__ __launch_bounds__(256) void attn_decode_kernel(
+    const short* __restrict__ qg, const short* __restrict__ kg,
+    const short* __restrict__ vg, float* __restrict__ partial_o,
+    float* __restrict__ partial_ml, int B, int Hq, int Hkv, int T_alloc, int t,
+    int nsplits, float scale) {
+  constexpr int PITCH = HD + 8;  // shorts; 16B-aligned, conflict-free b128
+  __shared__ __attribute__((aligned(16))) short k_t[TILE * PITCH];
+  __shared__ __attribute__((aligned(16))) short v_t[TILE * PITCH];
+  __shared__ float p_t[4][TILE];
+
+  const int tid = threadIdx.x;
+  const int w = tid >> 6;
+  const int l = tid & 63;
+  const int split = blockIdx.x;
+  const int bh = blockIdx.y;
+  const int b = bh / Hkv;
+  const int hkv = bh % Hkv;
+  const int G = Hq / Hkv;
+  constexpr int GWMAX = 2;  // compile-time bound: supports G <= 8
+
+  // this split's kv range (tile-aligned chunks)
+  const int chunk = ((t + nsplits - 1) / nsplits + TILE - 1) / TILE * TILE;
+  const int s0 = split * chunk;
+  const int s1 = min(t, s0 + chunk);
+
+  // q rows for this wave's heads, kept in registers (HD/8 x 8 elems).
+  // GWMAX is a compile-time constant: a runtime-sized local array is a
+  // device VLA (dynamic alloca) and memory-faults on gfx950.
+  float qreg[GWMAX][HD];
+  int heads[GWMAX];
+#pragma unroll
+  for (int gw = 0; gw < GWMAX; ++gw) {
+    int g = w + 4 * gw;
+    heads[gw] = g < G ? hkv * G + g : -1;
+    if (heads[gw] >= 0) {
+      const short* qp = qg + ((long long)b * Hq + heads[gw]) * HD;
+      for (int i = 0; i < HD / 8; ++i) {
+        s16x8 v8 = *reinterpret_cast<const s16x8*>(qp + 8 * i);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) qreg[gw][8 * i + j] = bf2f(v8[j]);
+      }
+    }
+  }
+
+  float m_run[GWMAX], l_run[GWMAX];
+  float o_acc[GWMAX][2];  // lane owns d = {2l, 2l+1}
+#pragma unroll
+  for (int gw = 0; gw < GWMAX; ++gw) {
+    m_run[gw] = -1e30f;
+    l_run[gw] = 0.f;
+    o_acc[gw][0] = o_acc[gw][1] = 0.f;
+  }
+
+  const long long row_stride = (long long)Hkv * HD;
+  const long long base = ((long long)b * T_alloc) * row_stride + (long long)hkv * HD;
+
+  for (int r0 = s0; r0 < s1; r0 += TILE) {
+    const int rows = min(TILE, s1 - r0);
+    // ---- cooperative K/V tile staging (8 rows x 512B per instruction) ----
+    {
+      constexpr int LPR = HD * 2 / 16;  // 16B loads per row
+      for (int c = tid; c < TILE * LPR; c += 256) {
+        int rr = c / LPR, off = (c % LPR) * 8;  // off in shorts
+        if (rr < rows) {
+          long long src = base + (long long)(r0 + rr) * row_stride + off;
+          *reinterpret_cast<s16x8*>(k_t + rr * PITCH + off) =
+              *reinterpret_cast<const s16x8*>(kg + src);
+          *reinterpret_cast<s16x8*>(v_t + rr * PITCH + off) =
+              *reinterpret_cast<const s16x8*>(vg + src);
+        }
+      }
+    }
+    __syncthreads();
+
+#pragma unroll
+    for (int gw = 0; gw < GWMAX; ++gw) {
+      if (heads[gw] < 0) continue;
+      // ---- scores: one kv row per lane ----
+      float s = -1e30f;
+      if (l < rows) {
+        float acc = 0.f;
+        const short* kp = k_t + l * PITCH;
+        for (int i = 0; i < HD / 8; ++i) {
+          s16x8 kv8 = *reinterpret_cast<const s16x8*>(kp + 8 * i);
+#pragma unroll
+          for (int j = 0; j < 8; ++j) acc += qreg[gw][8 * i + j] * bf2f(kv8[j]);
+        }
+        s = acc * scale;
+      }
+      // ---- online softmax (wave-wide) ----
+      float m_tile = wave_reduce_max(s);
+      float m_new = fmaxf(m_run[gw], m_tile);
+      float p = l < rows ? __expf(s - m_new) : 0.f;
+      float corr = __expf(m_run[gw] - m_new);
+      l_run[gw] = l_run[gw] * corr + wave_reduce_sum(p);
+      m_run[gw] = m_new;
+      p_t[w][l] = p;
+      // lanes covering the head dim rescale their accumulator
+      o_acc[gw][0] *= corr;
+      o_acc[gw][1] *= corr;
+      // ---- o += p * V (lane owns d = {2l, 2l+1}) ----
+      if (2 * l < HD) {
+        float a0 = 0.f, a1 = 0.f;
+        for (int r = 0; r < rows; ++r) {
+          float pr = p_t[w][r];
+          a0 += pr * bf2f(v_t[r * PITCH + 2 * l]);
+          a1 += pr * bf2f(v_t[r * PITCH + 2 * l + 1]);
+        }
+        o_acc[gw][0] += a0;
+        o_acc[gw][1] += a1;
+      }
+    }
+    __syncthreads();
+  }
+
+  // ---- write split partials ----
+#pragma unroll
+  for (int gw = 0; gw < GWMAX; ++gw) {
+    if (heads[gw] < 0) continue;
+    long long po = (((long long)split * B + b) * Hq + heads[gw]) * HD;
+    if (2 * l < HD) {
+      partial_o[po + 2 * l] = o_acc[gw][0];
+      partial_o[po + 2 * l + 1] = o_acc[gw][1];
+    }
+    if (l == 0) {
+      long long pm = (((long long)split * B + b) * Hq + heads[gw]) * 2;
+      partial_ml[pm] = m_run[gw];
+      partial_ml[pm + 1] = l_run[gw];
+    }
+  }
+}
+
+// ---- split-merge: one block per (b, hq); thread per head-dim element ----
+template <int HD>
+__global__ __launch_bounds__(128) void attn_decode_reduce_kernel(
+    const float* __restrict__ partial_o, const float* __restrict__ partial_ml,
+    short* __restrict__ out, int B, int Hq, int nsplits) {
+  const int bh = blockIdx.x;
+  const int d = threadIdx.x;
+  float m_star = -1e30f;
+  for (int s = 0; s < nsplits; ++s)
+    m_star = fmaxf(m_star, partial_ml[((long long)s * B * Hq + bh) * 2]);
+  float num = 0.f, den = 0.f;
+  for (int s = 0; s < nsplits; ++s) {
+    long long pm = ((long long)s * B * Hq + bh) * 2;
+    float f = __expf(partial_ml[pm] - m_star);
+    den += partial_ml[pm + 1] * f;
+    if (d < HD) num += partial_o[((long long)s * B * Hq + bh) * HD + d] * f;
+  }
+  if (d < HD) out[(long long)bh * HD + d] = f2bf(num / fmaxf(den, 1e-30f));
+}
+
+}  // namespace
+
+// q [B, Hq, D] bf16; k/v caches [B, T_alloc, Hkv, D] bf16 (bshd, in-place
+// appended); t = valid prefix length INCLUDING the token q was computed
+// from. Returns o [B, Hq, D] bf16.
+torch::Tensor attn_decode(torch::Tensor q, torch::Tensor kcache, torch::Tensor vcache,
+                          long t) {
+  TORCH_CHECK(q.is_cuda() && q.dtype() == torch::kBFloat16 && q.dim() == 3);
+  TORCH_CHECK(kcache.dim() == 4 && kcache.is_contiguous() && vcache.is_contiguous());
+  const int B = q.size(0), Hq = q.size(1), HD = q.size(2);
+  const int T_alloc = kcache.size(1), Hkv = kcache.size(2);
+  TORCH_CHECK(kcache.size(0) == B && kcache.size(3) == HD);
+  TORCH_CHECK(t >= 1 && t <= T_alloc, "decode: bad cache length");
+  TORCH_CHECK(Hq % Hkv == 0 && (HD == 64 || HD == 128), "decode: unsupported shape");
+  TORCH_CHECK(Hq / Hkv <= 8, "decode: GQA group > 8 not supported");
+  TORCH_CHECK(q.is_contiguous());
+
+  // enough splits to fill the chip, but >= 2 tiles of work per split
+  int nsplits = (int)std::min<long>((512 + B * Hkv - 1) / (B * Hkv),
+                                    std::max<long>(1, (t + 2 * TILE - 1) / (2 * TILE)));
+  auto fopts = q.options().dtype(torch::kFloat32);
+  auto partial_o = torch::empty({nsplits, B, Hq, HD}, fopts);
+  auto partial_ml = torch::empty({nsplits, B, Hq, 2}, fopts);
+  auto out = torch::empty_like(q);
+  float scale = 1.0f / sqrtf((float)HD);
+  hipStream_t stream = hypha_stream();
+
+#define DECODE_DISPATCH(HDV)                                                           \
+  do {                                                                                 \
+    hipLaunchKernelGGL(attn_decode_kernel<HDV>, dim3(nsplits, B* Hkv), dim3(256), 0,   \
+                       stream, (const short*)q.data_ptr(),                             \
+                       (const short*)kcache.data_ptr(),                                \
+                       (const short*)vcache.data_ptr(), partial_o.data_ptr<float>(),   \
+                       partial_ml.data_ptr<float>(), B, Hq, Hkv, T_alloc, (int)t,      \
+                       nsplits, scale);                                                \
+    hipLaunchKernelGGL(attn_decode_reduce_kernel<HDV>, dim3(B* Hq), dim3(128), 0,      \
+                       stream, partial_o.data_ptr<float>(),                            \
+                       partial_ml.data_ptr<float>(), (short*)out.data_ptr(), B, Hq,    \
+                       nsplits);                                                       \
+  } while (0)
+
+  if (HD == 128)
+    DECODE_DISPATCH(128);
+  else
+    DECODE_DISPATCH(64);
+#undef DECODE_DISPATCH
+  return out;
+}

@@ -205,6 +205,33 @@ def flash_attention(q, k, v, causal: bool = True, layout: str = "bhsd") -> torch
 
 
 # ---------------------------------------------------------------------------
+# KV-cache decode attention (flash-decoding; serving path)
+# ---------------------------------------------------------------------------
+
+
+def attn_decode(q, kcache, vcache, t: int) -> torch.Tensor:
+    """Single-token attention over an appended KV cache.
+
+    q [B, Hq, D] bf16; kcache/vcache [B, T_alloc, Hkv, D] bf16 (bshd);
+    t = valid length including the current token. Returns [B, Hq, D] bf16.
+    Native split-KV kernel (ops/hip/decode.hip); fp32 reference otherwise.
+    """
+    if use_native(q) and q.shape[-1] in (64, 128) and q.shape[1] <= 8 * kcache.shape[2]:
+        return _c().attn_decode(q, kcache, vcache, t)
+    # reference path (CPU, or head dims the kernel doesn't cover):
+    # plain masked softmax over the valid prefix
+    kc = kcache[:, :t].float()
+    vc = vcache[:, :t].float()
+    rep = q.shape[1] // kcache.shape[2]
+    kh = kc.permute(0, 2, 1, 3).repeat_interleave(rep, dim=1)
+    vh = vc.permute(0, 2, 1, 3).repeat_interleave(rep, dim=1)
+    scores = torch.einsum("bhd,bhtd->bht", q.float(), kh) / (q.shape[-1] ** 0.5)
+    p = torch.softmax(scores, dim=-1)
+    o = torch.einsum("bht,bhtd->bhd", p, vh)
+    return o.to(q.dtype)
+
+
+# ---------------------------------------------------------------------------
 # Cross-entropy over vocab (memory-frugal: backward overwrites the logits)
 # ---------------------------------------------------------------------------
 
