@@ -48,6 +48,9 @@ def main() -> int:
     p.add_argument("--fp8", action="store_true", help="fp8 e4m3 GEMMs (config 5 path)")
     p.add_argument("--fp8-weights", action="store_true",
                    help="fp8 WEIGHT STORAGE + fp8 GEMMs in the lean engine (config 5)")
+    p.add_argument("--expert-parallel", action="store_true",
+                   help="shard MoE experts across ranks (RCCL all-to-all per step; "
+                        "outer sync covers shared params only)")
     p.add_argument("--state-bits", type=int, default=32, choices=(32, 8))
     p.add_argument("--grad-checkpoint", action="store_true",
                    help="per-block activation checkpointing (70B-class fits)")
@@ -90,6 +93,13 @@ def main() -> int:
         except TypeError:  # model family without the flag
             model = models.build(args.model)
     torch.set_default_dtype(torch.float32)
+    if args.expert_parallel:
+        from hypha_amd.models.moe import shard_experts_
+
+        n_shard = shard_experts_(model, comm.rank, max(1, comm.world_size))
+        if rank == 0:
+            print(f"# expert-parallel: sharded {n_shard} MoE layers over "
+                  f"{comm.world_size} ranks", file=sys.stderr)
     if args.fp8 and on_gpu:
         from hypha_amd.ops.fp8 import convert_linears_to_fp8
 
@@ -182,7 +192,8 @@ def main() -> int:
                 "model": args.model,
                 "global_batch": args.batch * n_gpus,
                 "seq_len": args.seq_len,
-                "parallelism": f"diloco-dp{n_gpus}",
+                "parallelism": (f"ep{n_gpus}+diloco" if args.expert_parallel
+                                else f"diloco-dp{n_gpus}"),
                 "h": args.h,
                 "model_params": model_numel,
                 "memory_mode": args.memory_mode,

@@ -50,8 +50,64 @@ PRESETS: dict[str, MoEConfig] = {
 }
 
 
+def _a2a_exchange(out: torch.Tensor, x: torch.Tensor, out_splits, in_splits, group):
+    """all_to_all_single with a gloo fallback (gloo has no all-to-all:
+    emulated with batched isend/irecv pairs; self-chunk is a local copy)."""
+    import torch.distributed as dist
+
+    if dist.get_backend(group) != "gloo":
+        dist.all_to_all_single(out, x, out_splits, in_splits, group=group)
+        return
+    rank = dist.get_rank(group)
+    size = dist.get_world_size(group)
+    in_off = [0]
+    for c in in_splits:
+        in_off.append(in_off[-1] + c)
+    out_off = [0]
+    for c in out_splits:
+        out_off.append(out_off[-1] + c)
+    reqs = []
+    for r in range(size):
+        if r == rank:
+            out[out_off[r]:out_off[r + 1]] = x[in_off[r]:in_off[r + 1]]
+            continue
+        if in_splits[r]:
+            reqs.append(dist.isend(x[in_off[r]:in_off[r + 1]].contiguous(),
+                                   dist.get_global_rank(group, r), group=group))
+        if out_splits[r]:
+            reqs.append(dist.irecv(out[out_off[r]:out_off[r + 1]],
+                                   dist.get_global_rank(group, r), group=group))
+    for q in reqs:
+        q.wait()
+
+
+class _AllToAll(torch.autograd.Function):
+    """Autograd token exchange: backward is the transposed all-to-all."""
+
+    @staticmethod
+    def forward(ctx, x, out_splits, in_splits, group):
+        ctx.group = group
+        ctx.out_splits = out_splits
+        ctx.in_splits = in_splits
+        out = x.new_empty((sum(out_splits),) + tuple(x.shape[1:]))
+        _a2a_exchange(out, x.contiguous(), out_splits, in_splits, group)
+        return out
+
+    @staticmethod
+    def backward(ctx, g):
+        out = g.new_empty((sum(ctx.in_splits),) + tuple(g.shape[1:]))
+        _a2a_exchange(out, g.contiguous(), ctx.in_splits, ctx.out_splits, ctx.group)
+        return out, None, None, None
+
+
 class MoEMLP(nn.Module):
-    """Top-k routed expert MLP with token grouping (grouped-GEMM execution)."""
+    """Top-k routed expert MLP with token grouping (grouped-GEMM execution).
+
+    Optional EXPERT PARALLELISM (shard_experts_): experts are sharded across
+    an RCCL group; tokens travel to their experts by all-to-all over xGMI
+    and return after the expert GEMMs. Router and attention stay replicated
+    (DiLoCo-synced); expert weights are singletons (each update sees every
+    rank's tokens), so they are excluded from the outer all-reduce."""
 
     def __init__(self, cfg: MoEConfig):
         super().__init__()
@@ -63,6 +119,7 @@ class MoEMLP(nn.Module):
         self.w_down = nn.Parameter(torch.empty(e, h, f))
         for w in (self.w_gate, self.w_up, self.w_down):
             nn.init.normal_(w, std=cfg.init_std)
+        self.ep = None  # (rank, size, group) after shard_experts_
 
     def forward(self, x):
         cfg = self.cfg
@@ -92,6 +149,13 @@ class MoEMLP(nn.Module):
         order = torch.argsort(flat_expert, stable=True)
         counts = torch.bincount(flat_expert, minlength=cfg.n_experts).tolist()
         gathered = xf[flat_tok[order]]  # [T*K, h] grouped by expert
+
+        if self.ep is not None:
+            grouped_out = self._ep_dispatch(gathered, counts)
+            weights = topv.reshape(-1)[order].unsqueeze(1)
+            out = torch.zeros_like(xf)
+            out.index_add_(0, flat_tok[order], grouped_out * weights)
+            return out.reshape(b, s, h), aux
         # The native grouped kernel measured SLOWER than per-expert hipBLASLt
         # at every tested size (tools/moe_gemm_bench.py: 205-687 vs 250-1154
         # TF/s), so the library loop is the default; HYPHA_NATIVE_GROUPED=1
@@ -116,24 +180,111 @@ class MoEMLP(nn.Module):
             he = ops.swiglu(ge, ue)
             grouped_out = _C.grouped_gemm(he.contiguous(), self.w_down.contiguous(), off)
         else:
-            out_groups = []
-            start = 0
-            for e in range(cfg.n_experts):
-                n = counts[e]
-                if n == 0:
-                    continue
-                xe = gathered[start : start + n]
-                ge = xe @ self.w_gate[e].t()
-                ue = xe @ self.w_up[e].t()
-                he = ops.swiglu(ge, ue)
-                out_groups.append(he @ self.w_down[e].t())
-                start += n
-            grouped_out = torch.cat(out_groups, dim=0) if out_groups else gathered
+            grouped_out = self._expert_ffn(gathered, counts)
         # scatter-add back with routing weights
         weights = topv.reshape(-1)[order].unsqueeze(1)
         out = torch.zeros_like(xf)
         out.index_add_(0, flat_tok[order], grouped_out * weights)
         return out.reshape(b, s, h), aux
+
+    def _expert_ffn(self, xs: torch.Tensor, counts: list[int]) -> torch.Tensor:
+        """Per-expert SwiGLU chains over expert-grouped rows (local shard)."""
+        out_groups = []
+        start = 0
+        for e in range(len(counts)):
+            n = counts[e]
+            if n == 0:
+                continue
+            xe = xs[start:start + n]
+            ge = xe @ self.w_gate[e].t()
+            ue = xe @ self.w_up[e].t()
+            he = ops.swiglu(ge, ue)
+            out_groups.append(he @ self.w_down[e].t())
+            start += n
+        return torch.cat(out_groups, dim=0) if out_groups else xs[:0]
+
+    def _ep_dispatch(self, gathered: torch.Tensor, counts: list[int]) -> torch.Tensor:
+        """Expert-parallel token exchange (all-to-all over xGMI/RCCL):
+
+        1. `gathered` is globally expert-sorted, so each rank's experts form
+           one contiguous send chunk;
+        2. every rank learns the full [size, E] count matrix (all_gather) to
+           size the exchange and regroup received rows expert-major;
+        3. local expert GEMMs, then the transposed all-to-all returns rows
+           in the original `gathered` order for the weighted combine."""
+        import torch.distributed as dist
+
+        rank, size, group = self.ep
+        cfg = self.cfg
+        e_local = cfg.n_experts // size
+        dev = gathered.device
+
+        counts_t = torch.tensor(counts, dtype=torch.int64, device=dev)
+        all_counts = [torch.empty_like(counts_t) for _ in range(size)]
+        dist.all_gather(all_counts, counts_t, group=group)
+        L = torch.stack(all_counts).cpu()  # [size, E]
+
+        send_splits = [int(L[rank, r * e_local:(r + 1) * e_local].sum())
+                       for r in range(size)]
+        my = slice(rank * e_local, (rank + 1) * e_local)
+        recv_splits = [int(L[src, my].sum()) for src in range(size)]
+
+        recv = _AllToAll.apply(gathered, recv_splits, send_splits, group)
+
+        # regroup src-major -> expert-major for the GEMM chains
+        seg = []  # (src, e) segment order as received
+        off = 0
+        pos = {}
+        for src in range(size):
+            for e in range(e_local):
+                n = int(L[src, rank * e_local + e])
+                pos[(src, e)] = (off, n)
+                off += n
+        perm = []
+        local_counts = []
+        for e in range(e_local):
+            c = 0
+            for src in range(size):
+                o, n = pos[(src, e)]
+                if n:
+                    perm.append(torch.arange(o, o + n, device=dev))
+                c += n
+            local_counts.append(c)
+        perm = torch.cat(perm) if perm else torch.empty(0, dtype=torch.long, device=dev)
+        inv = torch.empty_like(perm)
+        inv[perm] = torch.arange(perm.numel(), device=dev)
+
+        ys = self._expert_ffn(recv[perm], local_counts)
+        back = _AllToAll.apply(ys[inv], send_splits, recv_splits, group)
+        return back
+
+
+def shard_experts_(model: nn.Module, rank: int, size: int, group=None) -> int:
+    """EXPERT PARALLELISM (opt-in): keep only this rank's n_experts/size
+    expert slices in every MoEMLP; tokens reach remote experts by RCCL
+    all-to-all each step. Sharded parameters are tagged `_ep_local` so the
+    DiLoCo engines exclude them from the outer all-reduce (each expert is a
+    singleton that already saw every rank's tokens) and from the init
+    broadcast. Returns the number of sharded MoEMLP modules."""
+    import torch.distributed as dist
+
+    if group is None:
+        group = dist.distributed_c10d._get_default_group()
+    n = 0
+    for m in model.modules():
+        if isinstance(m, MoEMLP):
+            cfg = m.cfg
+            assert cfg.n_experts % size == 0, "experts must divide EP size"
+            sl = slice(rank * (cfg.n_experts // size),
+                       (rank + 1) * (cfg.n_experts // size))
+            for name in ("w_gate", "w_up", "w_down"):
+                p = getattr(m, name)
+                new = nn.Parameter(p.data[sl].clone())
+                new._ep_local = True
+                setattr(m, name, new)
+            m.ep = (rank, size, group)
+            n += 1
+    return n
 
 
 class MoEBlock(nn.Module):

@@ -84,7 +84,15 @@ class FlatParams:
 
     def __init__(self, model: nn.Module, device: torch.device, work_dtype: torch.dtype,
                  state_bits: int = 32):
-        self.params = [p for p in model.parameters() if p.requires_grad]
+        # expert-parallel shards (`_ep_local`, models/moe.py shard_experts_)
+        # are placed LAST: the outer all-reduce and the init broadcast cover
+        # only the shared prefix [0, sync_numel) — each expert shard is a
+        # singleton owned by one rank, not a replica to average.
+        params = [p for p in model.parameters() if p.requires_grad]
+        shared = [p for p in params if not getattr(p, "_ep_local", False)]
+        ep = [p for p in params if getattr(p, "_ep_local", False)]
+        self.params = shared + ep
+        self.sync_numel = sum(p.numel() for p in shared)
         self.numel = sum(p.numel() for p in self.params)
         self.device = device
         self.work_dtype = work_dtype
@@ -155,9 +163,10 @@ class DiLoCoWorker:
             if buf.dtype in (torch.bfloat16, torch.float16):
                 buf.data = buf.data.float()
         self.fp = FlatParams(model, device, work_dtype, state_bits=cfg.inner.state_bits)
-        # make every rank start from identical weights (rank0's init wins)
+        # make every rank start from identical weights (rank0's init wins);
+        # expert-parallel shards (the tail past sync_numel) stay rank-local
         if self.comm.is_distributed:
-            self.comm.broadcast_flat(self.fp.master, src=0)
+            self.comm.broadcast_flat(self.fp.master[: self.fp.sync_numel], src=0)
             self.fp.flat.copy_(self.fp.master)
             self.fp.theta0.copy_(self.fp.master)
         self.inner_step_count = 0  # global inner step counter (for LR/bias corr)
@@ -236,24 +245,34 @@ class DiLoCoWorker:
         is O(64MB) regardless of model size."""
         fp = self.fp
         n = fp.numel
+        n_sync = fp.sync_numel  # expert-parallel tail is not averaged
         chunk = self._comm_chunk
         for start in range(0, n, chunk):
             m = min(chunk, n - start)
-            d = self._delta_buf[:m]
-            ops.interface.extract_delta(
-                fp.master[start : start + m], fp.theta0[start : start + m], d
-            )
-            self.comm.all_reduce_mean_flat(d)
-            ops.fused_nesterov(
-                fp.theta0[start : start + m],
-                d,
-                fp.outer_momentum[start : start + m],
-                lr=self.cfg.outer.lr,
-                mu=self.cfg.outer.momentum,
-            )
+            ms = max(0, min(m, n_sync - start))  # shared portion of chunk
+            if ms > 0:
+                d = self._delta_buf[:ms]
+                ops.interface.extract_delta(
+                    fp.master[start : start + ms], fp.theta0[start : start + ms], d
+                )
+                self.comm.all_reduce_mean_flat(d)
+                ops.fused_nesterov(
+                    fp.theta0[start : start + ms],
+                    d,
+                    fp.outer_momentum[start : start + ms],
+                    lr=self.cfg.outer.lr,
+                    mu=self.cfg.outer.momentum,
+                )
+            if ms < m:
+                # expert-parallel tail: singletons, purely inner-trained —
+                # the round's weights pass through (no averaging, no outer
+                # momentum re-amplifying an un-averaged delta)
+                fp.theta0[start + ms : start + m].copy_(
+                    fp.master[start + ms : start + m]
+                )
         fp.master.copy_(fp.theta0)
         fp.flat.copy_(fp.master)
-        self.outer_sync_payload_bytes += n * self._delta_buf.element_size()
+        self.outer_sync_payload_bytes += n_sync * self._delta_buf.element_size()
         self.round += 1
         self.steps_in_round = 0
 
