@@ -108,8 +108,15 @@ def run_rccl(worker, comm, session, batch_iter, reform_q, rccl, batch_size,
     _save_file({"flat": fp.theta0.cpu()},
                os.path.join(work_dir, "0_global_weights.safetensors"))
 
+    from hypha_amd.telemetry import get_tracer
+
+    tracer = get_tracer()
+    job_span = tracer.start_span("job.execute", job_id=os.environ.get("HYPHA_JOB_ID", ""),
+                                 rank=comm.rank, world_size=comm.world_size)
     done = False
     while not done:
+        round_span = tracer.start_span("diloco.round", parent=job_span,
+                                       round=round_idx, rank=comm.rank)
         round_t0 = _time.perf_counter()
         remaining = None
         round_samples = 0
@@ -124,6 +131,9 @@ def run_rccl(worker, comm, session, batch_iter, reform_q, rccl, batch_size,
             if resp.get("kind") == "schedule-update" and remaining is None:
                 remaining = int(resp.get("counter", 0))
         round_s = max(1e-9, _time.perf_counter() - round_t0)
+        round_span.set_attribute("samples", round_samples)
+        round_span.set_attribute("loss", float(loss))
+        round_span.end()
         session.send_status(
             {"kind": "metrics", "round": round_idx,
              "metrics": {"loss": loss,
@@ -132,6 +142,8 @@ def run_rccl(worker, comm, session, batch_iter, reform_q, rccl, batch_size,
         session.send_status({"kind": "update"})
 
         # ---- outer sync boundary: apply membership changes, then sync ----
+        sync_span = tracer.start_span("diloco.outer_sync", parent=job_span,
+                                      round=round_idx, rank=comm.rank)
         if drain_reforms():
             round_idx = _state_sync(worker, comm, round_idx)
         for attempt in range(4):
@@ -155,6 +167,9 @@ def run_rccl(worker, comm, session, batch_iter, reform_q, rccl, batch_size,
         fp.master.copy_(fp.theta0)
         fp.flat.copy_(fp.master)
         worker.outer_sync_payload_bytes += delta.numel() * delta.element_size()
+        sync_span.set_attribute("payload_bytes",
+                                delta.numel() * delta.element_size())
+        sync_span.end()
 
         # FSM ordering parity with the PS path: `updated` (advances the round)
         # must reach the scheduler before any `update-received` — rank 0 sends
@@ -175,6 +190,8 @@ def run_rccl(worker, comm, session, batch_iter, reform_q, rccl, batch_size,
         print(f"[executor] rccl round {round_idx} merged, loss={loss:.4f} "
               f"done={done}", flush=True)
 
+    job_span.end()
+    tracer.flush()
     comm.shutdown()
     session.close()
     return 0

@@ -161,3 +161,92 @@ def test_collect_transport_bytes():
     assert s["inbound_bytes"] >= 0 and s["outbound_bytes"] >= 0
     snap = METRICS.snapshot()
     assert "hypha.bandwidth.transport.inbound_bytes" in snap["gauges"]
+
+
+def test_tracer_span_nesting_and_otlp_shape(tmp_path):
+    """OTLP-shaped tracing (reference telemetry/src/lib.rs:15-49 parity):
+    trace/span id propagation, nesting via the context stack, and the
+    OTLP/JSON ResourceSpans schema in the file sink."""
+    import json
+
+    from hypha_amd.telemetry import Tracer
+
+    sink = tmp_path / "traces.jsonl"
+    tr = Tracer(service_name="test-svc", sink_path=str(sink), flush_every=999)
+    with tr.start_span("job.execute", job_id="j1") as job:
+        with tr.start_span("diloco.round", round=0) as rnd:
+            assert rnd.trace_id == job.trace_id
+            assert rnd.parent_id == job.span_id
+        with tr.start_span("diloco.outer_sync", round=0) as sync:
+            sync.set_attribute("payload_bytes", 123)
+    n = tr.flush()
+    assert n == 3
+    batch = json.loads(sink.read_text().strip())
+    rs = batch["resourceSpans"][0]
+    svc = rs["resource"]["attributes"][0]
+    assert svc["key"] == "service.name"
+    assert svc["value"]["stringValue"] == "test-svc"
+    spans = rs["scopeSpans"][0]["spans"]
+    assert {s["name"] for s in spans} == {"job.execute", "diloco.round",
+                                          "diloco.outer_sync"}
+    for s in spans:
+        assert len(s["traceId"]) == 32 and len(s["spanId"]) == 16
+        assert int(s["endTimeUnixNano"]) >= int(s["startTimeUnixNano"])
+        assert s["status"]["code"] == 1
+    sync_s = next(s for s in spans if s["name"] == "diloco.outer_sync")
+    assert {"key": "payload_bytes", "value": {"intValue": "123"}} in sync_s["attributes"]
+
+
+def test_tracer_posts_otlp_http(tmp_path):
+    """Spans POST to an OTLP/HTTP collector endpoint (/v1/traces)."""
+    import http.server
+    import json
+    import threading
+
+    got = {}
+
+    class H(http.server.BaseHTTPRequestHandler):
+        def do_POST(self):
+            n = int(self.headers["Content-Length"])
+            got["path"] = self.path
+            got["body"] = json.loads(self.rfile.read(n))
+            self.send_response(200)
+            self.end_headers()
+            self.wfile.write(b"{}")
+
+        def log_message(self, *a):
+            pass
+
+    srv = http.server.ThreadingHTTPServer(("127.0.0.1", 0), H)
+    threading.Thread(target=srv.serve_forever, daemon=True).start()
+    try:
+        from hypha_amd.telemetry import Tracer
+
+        tr = Tracer(endpoint=f"http://127.0.0.1:{srv.server_address[1]}")
+        with tr.start_span("dispatch", job_id="j2"):
+            pass
+        assert tr.flush() == 1
+        assert tr.export_errors == 0
+        assert got["path"] == "/v1/traces"
+        names = [s["name"] for s in
+                 got["body"]["resourceSpans"][0]["scopeSpans"][0]["spans"]]
+        assert names == ["dispatch"]
+    finally:
+        srv.shutdown()
+
+
+def test_tracer_marks_exception_spans(tmp_path):
+    from hypha_amd.telemetry import Tracer
+
+    tr = Tracer(sink_path=str(tmp_path / "t.jsonl"), flush_every=999)
+    try:
+        with tr.start_span("failing"):
+            raise ValueError("boom")
+    except ValueError:
+        pass
+    import json
+
+    tr.flush()
+    span = json.loads((tmp_path / "t.jsonl").read_text())[
+        "resourceSpans"][0]["scopeSpans"][0]["spans"][0]
+    assert span["status"]["code"] == 2
