@@ -148,9 +148,24 @@ class Block(nn.Module):
         self.mlp = MLP(cfg)
 
     def forward(self, x, cos, sin, cache=None, pos: int = 0):
-        x = x + self.attn(self.attn_norm(x), cos, sin, cache=cache, pos=pos)
-        x = x + self.mlp(self.mlp_norm(x))
+        attn_out = self.attn(self.attn_norm(x), cos, sin, cache=cache, pos=pos)
+        # fused residual-add + norm: one kernel writes the residual stream
+        # and the normalized MLP input (saves an elementwise pass per block)
+        x, n2 = ops.add_rmsnorm(x, attn_out, self.mlp_norm.weight,
+                                self.mlp_norm.eps)
+        x = x + self.mlp(n2)
         return x
+
+    def forward_pair(self, res, delta, cos, sin):
+        """Training fast path with the residual add DEFERRED: receives the
+        previous block's (residual, sub-block output) pair so EVERY residual
+        add fuses into the next norm kernel (including across blocks)."""
+        res, n1 = ops.add_rmsnorm(res, delta, self.attn_norm.weight,
+                                  self.attn_norm.eps)
+        attn_out = self.attn(n1, cos, sin)
+        res, n2 = ops.add_rmsnorm(res, attn_out, self.mlp_norm.weight,
+                                  self.mlp_norm.eps)
+        return res, self.mlp(n2)
 
 
 class LlamaForCausalLM(nn.Module):
@@ -182,12 +197,17 @@ class LlamaForCausalLM(nn.Module):
     def forward(self, input_ids: torch.Tensor, labels: torch.Tensor | None = None):
         x = self.embed(input_ids)
         cos, sin = self.rope_cos, self.rope_sin
+        # (residual, delta) carry: every residual add — including the
+        # cross-block one — fuses into the next RMSNorm kernel; the final
+        # add fuses into the output norm
+        delta = torch.zeros_like(x)
         for blk in self.blocks:
             if self.cfg.gradient_checkpointing and self.training:
-                x = torch.utils.checkpoint.checkpoint(blk, x, cos, sin, use_reentrant=False)
+                x, delta = torch.utils.checkpoint.checkpoint(
+                    blk.forward_pair, x, delta, cos, sin, use_reentrant=False)
             else:
-                x = blk(x, cos, sin)
-        x = self.norm(x)
+                x, delta = blk.forward_pair(x, delta, cos, sin)
+        _, x = ops.add_rmsnorm(x, delta, self.norm.weight, self.norm.eps)
         if self.lm_head is not None:
             logits = self.lm_head(x)
         else:

@@ -296,10 +296,23 @@ class MoEBlock(nn.Module):
         self.mlp = MoEMLP(cfg)
 
     def forward(self, x, cos, sin, cache=None, pos: int = 0):
-        x = x + self.attn(self.attn_norm(x), cos, sin, cache=cache, pos=pos)
-        mlp_out, aux = self.mlp(self.mlp_norm(x))
+        attn_out = self.attn(self.attn_norm(x), cos, sin, cache=cache, pos=pos)
+        x, n2 = ops.add_rmsnorm(x, attn_out, self.mlp_norm.weight,
+                                self.mlp_norm.eps)
+        mlp_out, aux = self.mlp(n2)
         x = x + mlp_out
         return x, aux
+
+    def forward_pair(self, res, delta, cos, sin):
+        """Training fast path: deferred residual adds fused into the norm
+        kernels (see llama.Block.forward_pair)."""
+        res, n1 = ops.add_rmsnorm(res, delta, self.attn_norm.weight,
+                                  self.attn_norm.eps)
+        attn_out = self.attn(n1, cos, sin)
+        res, n2 = ops.add_rmsnorm(res, attn_out, self.mlp_norm.weight,
+                                  self.mlp_norm.eps)
+        mlp_out, aux = self.mlp(n2)
+        return res, mlp_out, aux
 
 
 class MoEForCausalLM(nn.Module):
@@ -321,14 +334,17 @@ class MoEForCausalLM(nn.Module):
 
         x = self.embed(input_ids)
         aux_total = None
+        delta = torch.zeros_like(x)
         for blk in self.blocks:
             if self.cfg.gradient_checkpointing and self.training:
-                x, aux = torch.utils.checkpoint.checkpoint(
-                    blk, x, self.rope_cos, self.rope_sin, use_reentrant=False)
+                x, delta, aux = torch.utils.checkpoint.checkpoint(
+                    blk.forward_pair, x, delta, self.rope_cos, self.rope_sin,
+                    use_reentrant=False)
             else:
-                x, aux = blk(x, self.rope_cos, self.rope_sin)
+                x, delta, aux = blk.forward_pair(x, delta, self.rope_cos,
+                                                 self.rope_sin)
             aux_total = aux if aux_total is None else aux_total + aux
-        x = self.norm(x)
+        _, x = ops.add_rmsnorm(x, delta, self.norm.weight, self.norm.eps)
         logits = self.lm_head(x)
         if labels is None:
             return logits

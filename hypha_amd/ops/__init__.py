@@ -52,6 +52,7 @@ def use_native(*tensors: torch.Tensor) -> bool:
 
 
 from .interface import (  # noqa: E402
+    add_rmsnorm,
     apply_rope_qk,
     attn_decode,
     cross_entropy_loss,
@@ -73,6 +74,7 @@ __all__ = [
     "rmsnorm",
     "layernorm",
     "gelu",
+    "add_rmsnorm",
     "apply_rope_qk",
     "attn_decode",
     "swiglu",

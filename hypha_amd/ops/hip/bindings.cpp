@@ -39,6 +39,11 @@ std::vector<torch::Tensor> layernorm_bwd(torch::Tensor dy, torch::Tensor x,
 torch::Tensor gelu_fwd(torch::Tensor x);
 torch::Tensor gelu_bwd(torch::Tensor dy, torch::Tensor x);
 std::vector<torch::Tensor> rmsnorm_fwd(torch::Tensor x, torch::Tensor w, double eps);
+std::vector<torch::Tensor> add_rmsnorm_fwd(torch::Tensor a, torch::Tensor b,
+                                           torch::Tensor w, double eps);
+std::vector<torch::Tensor> add_rmsnorm_bwd(torch::Tensor dy, torch::Tensor dh,
+                                           torch::Tensor h, torch::Tensor w,
+                                           torch::Tensor rstd);
 std::vector<torch::Tensor> rmsnorm_bwd(torch::Tensor dy, torch::Tensor x, torch::Tensor w,
                                        torch::Tensor rstd);
 // rope.hip
@@ -102,6 +107,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("nesterov_bf16_", &nesterov_bf16_);
   m.def("swiglu_bwd", &swiglu_bwd);
   m.def("rmsnorm_fwd", &rmsnorm_fwd);
+  m.def("add_rmsnorm_fwd", &add_rmsnorm_fwd);
+  m.def("add_rmsnorm_bwd", &add_rmsnorm_bwd);
   m.def("rmsnorm_bwd", &rmsnorm_bwd);
   m.def("layernorm_fwd", &layernorm_fwd);
   m.def("layernorm_bwd", &layernorm_bwd);

@@ -59,6 +59,41 @@ def rmsnorm(x: torch.Tensor, weight: torch.Tensor, eps: float = 1e-5) -> torch.T
     return reference.rmsnorm(x, weight, eps)
 
 
+class _AddRMSNorm(torch.autograd.Function):
+    """Fused h = a + b; y = rmsnorm(h): the residual add is computed inside
+    the norm kernel (one fewer elementwise pass per transformer sub-block);
+    backward fuses the residual gradient into the dx pass."""
+
+    @staticmethod
+    def forward(ctx, a2d, b2d, weight, eps):
+        h, y, rstd = _c().add_rmsnorm_fwd(a2d, b2d, weight, eps)
+        ctx.save_for_backward(h, weight, rstd)
+        return h, y
+
+    @staticmethod
+    def backward(ctx, dh, dy):
+        h, weight, rstd = ctx.saved_tensors
+        if dh is None:
+            dx, dw = _c().rmsnorm_bwd(dy.contiguous(), h, weight, rstd)
+        else:
+            dx, dw = _c().add_rmsnorm_bwd(dy.contiguous(), dh.contiguous(), h,
+                                          weight, rstd)
+        return dx, dx, dw.to(weight.dtype), None
+
+
+def add_rmsnorm(a: torch.Tensor, b: torch.Tensor, weight: torch.Tensor,
+                eps: float = 1e-5):
+    """Returns (h, y) with h = a + b and y = rmsnorm(h), fused on GPU."""
+    if use_native(a):
+        shape = a.shape
+        h, y = _AddRMSNorm.apply(a.reshape(-1, shape[-1]).contiguous(),
+                                 b.reshape(-1, shape[-1]).contiguous(),
+                                 weight, eps)
+        return h.view(shape), y.view(shape)
+    h = a + b
+    return h, reference.rmsnorm(h, weight, eps)
+
+
 # ---------------------------------------------------------------------------
 # RoPE on q and k together (one fused kernel launch)
 # ---------------------------------------------------------------------------

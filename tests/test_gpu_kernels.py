@@ -814,3 +814,30 @@ def test_generate_kv_cache_matches_recompute():
         assert out.shape == toks.shape
         agree = (out == toks).float().mean().item()
         assert agree >= 0.95, (name, agree, out[:, -12:], toks[:, -12:])
+
+
+def test_add_rmsnorm_fused_vs_reference():
+    """Fused residual-add+RMSNorm fwd/bwd vs the fp32 composition."""
+    from hypha_amd import ops
+
+    torch.manual_seed(23)
+    N, D = 512, 4096
+    a = rand_bf16(N, D, seed=70).requires_grad_(True)
+    b = rand_bf16(N, D, seed=71).requires_grad_(True)
+    w = (torch.randn(D) * 0.1 + 1).bfloat16().to(DEV).requires_grad_(True)
+    h, y = ops.add_rmsnorm(a, b, w, 1e-5)
+    # downstream uses BOTH h (residual) and y (norm) like a transformer block
+    loss = (h.float() * 0.3).sum() + (y.float() * 0.7).sum()
+    loss.backward()
+
+    af = a.detach().float().requires_grad_(True)
+    bf = b.detach().float().requires_grad_(True)
+    wf = w.detach().float().requires_grad_(True)
+    hf = af + bf
+    yf = hf * torch.rsqrt(hf.pow(2).mean(-1, keepdim=True) + 1e-5) * wf
+    (hf * 0.3).sum().add((yf * 0.7).sum()).backward()
+    torch.testing.assert_close(h.float(), hf, rtol=2e-2, atol=2e-2)
+    torch.testing.assert_close(y.float(), yf, rtol=2e-2, atol=2e-2)
+    torch.testing.assert_close(a.grad.float(), af.grad, rtol=3e-2, atol=3e-2)
+    torch.testing.assert_close(b.grad.float(), bf.grad, rtol=3e-2, atol=3e-2)
+    torch.testing.assert_close(w.grad.float(), wf.grad, rtol=3e-2, atol=3e-1)
