@@ -93,6 +93,8 @@ std::vector<torch::Tensor> fp8_weight_cast_transpose(torch::Tensor w8s,
 // decode.hip (flash-decoding KV-cache attention)
 torch::Tensor attn_decode(torch::Tensor q, torch::Tensor kcache, torch::Tensor vcache,
                           long t);
+torch::Tensor attn_decode_graph(torch::Tensor q, torch::Tensor kcache,
+                                torch::Tensor vcache, torch::Tensor t_dev);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("adamw_step_", &adamw_step_);
@@ -135,4 +137,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fp8_requant_", &fp8_requant_);
   m.def("fp8_weight_cast_transpose", &fp8_weight_cast_transpose);
   m.def("attn_decode", &attn_decode);
+  m.def("attn_decode_graph", &attn_decode_graph);
 }

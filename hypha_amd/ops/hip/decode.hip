@@ -33,8 +33,11 @@ template <int HD>
 __global__ __launch_bounds__(256) void attn_decode_kernel(
     const short* __restrict__ qg, const short* __restrict__ kg,
     const short* __restrict__ vg, float* __restrict__ partial_o,
-    float* __restrict__ partial_ml, int B, int Hq, int Hkv, int T_alloc, int t,
-    int nsplits, float scale) {
+    float* __restrict__ partial_ml, int B, int Hq, int Hkv, int T_alloc, int t_in,
+    const int* __restrict__ t_dev, int nsplits, float scale) {
+  // hipGraph mode: the valid length lives in device memory so one captured
+  // graph serves every decode step (t grows between replays)
+  const int t = t_dev ? *t_dev : t_in;
   constexpr int PITCH = HD + 8;  // shorts; 16B-aligned, conflict-free b128
   __shared__ __attribute__((aligned(16))) short k_t[TILE * PITCH];
   __shared__ __attribute__((aligned(16))) short v_t[TILE * PITCH];
@@ -187,8 +190,9 @@ __global__ __launch_bounds__(128) void attn_decode_reduce_kernel(
 // q [B, Hq, D] bf16; k/v caches [B, T_alloc, Hkv, D] bf16 (bshd, in-place
 // appended); t = valid prefix length INCLUDING the token q was computed
 // from. Returns o [B, Hq, D] bf16.
-torch::Tensor attn_decode(torch::Tensor q, torch::Tensor kcache, torch::Tensor vcache,
-                          long t) {
+torch::Tensor attn_decode_impl(torch::Tensor q, torch::Tensor kcache,
+                               torch::Tensor vcache, long t,
+                               const torch::Tensor* t_dev) {
   TORCH_CHECK(q.is_cuda() && q.dtype() == torch::kBFloat16 && q.dim() == 3);
   TORCH_CHECK(kcache.dim() == 4 && kcache.is_contiguous() && vcache.is_contiguous());
   const int B = q.size(0), Hq = q.size(1), HD = q.size(2);
@@ -198,10 +202,14 @@ torch::Tensor attn_decode(torch::Tensor q, torch::Tensor kcache, torch::Tensor v
   TORCH_CHECK(Hq % Hkv == 0 && (HD == 64 || HD == 128), "decode: unsupported shape");
   TORCH_CHECK(Hq / Hkv <= 8, "decode: GQA group > 8 not supported");
   TORCH_CHECK(q.is_contiguous());
+  const int* t_ptr = t_dev ? t_dev->data_ptr<int>() : nullptr;
 
-  // enough splits to fill the chip, but >= 2 tiles of work per split
+  // enough splits to fill the chip, but >= 2 tiles of work per split; in
+  // graph mode (t_dev) size for the worst case T_alloc so one capture
+  // covers every step
+  long t_sz = t_dev ? T_alloc : t;
   int nsplits = (int)std::min<long>((512 + B * Hkv - 1) / (B * Hkv),
-                                    std::max<long>(1, (t + 2 * TILE - 1) / (2 * TILE)));
+                                    std::max<long>(1, (t_sz + 2 * TILE - 1) / (2 * TILE)));
   auto fopts = q.options().dtype(torch::kFloat32);
   auto partial_o = torch::empty({nsplits, B, Hq, HD}, fopts);
   auto partial_ml = torch::empty({nsplits, B, Hq, 2}, fopts);
@@ -216,7 +224,7 @@ torch::Tensor attn_decode(torch::Tensor q, torch::Tensor kcache, torch::Tensor v
                        (const short*)kcache.data_ptr(),                                \
                        (const short*)vcache.data_ptr(), partial_o.data_ptr<float>(),   \
                        partial_ml.data_ptr<float>(), B, Hq, Hkv, T_alloc, (int)t,      \
-                       nsplits, scale);                                                \
+                       t_ptr, nsplits, scale);                                         \
     hipLaunchKernelGGL(attn_decode_reduce_kernel<HDV>, dim3(B* Hq), dim3(128), 0,      \
                        stream, partial_o.data_ptr<float>(),                            \
                        partial_ml.data_ptr<float>(), (short*)out.data_ptr(), B, Hq,    \
@@ -229,4 +237,17 @@ torch::Tensor attn_decode(torch::Tensor q, torch::Tensor kcache, torch::Tensor v
     DECODE_DISPATCH(64);
 #undef DECODE_DISPATCH
   return out;
+}
+
+torch::Tensor attn_decode(torch::Tensor q, torch::Tensor kcache, torch::Tensor vcache,
+                          long t) {
+  return attn_decode_impl(q, kcache, vcache, t, nullptr);
+}
+
+// hipGraph-capturable decode: valid length read from t_dev (int32 [1]) at
+// kernel time; splits sized for T_alloc once.
+torch::Tensor attn_decode_graph(torch::Tensor q, torch::Tensor kcache,
+                                torch::Tensor vcache, torch::Tensor t_dev) {
+  TORCH_CHECK(t_dev.dtype() == torch::kInt32 && t_dev.is_cuda());
+  return attn_decode_impl(q, kcache, vcache, kcache.size(1), &t_dev);
 }

@@ -58,6 +58,17 @@ def main() -> int:
     temperature = float(cfg.get("temperature", 0.0))
     top_k = int(cfg.get("top_k", 0))
 
+    # greedy Llama-family serving on GPU: hipGraph-captured decode step
+    # (runtime/graphed_decode.py; eager decode is launch-bound)
+    graphed = None
+    if (temperature <= 0 and device.type == "cuda"
+            and type(model).__name__ == "LlamaForCausalLM"
+            and model.cfg.head_dim in (64, 128)
+            and model.cfg.n_heads // model.cfg.n_kv_heads <= 8):
+        from hypha_amd.runtime.graphed_decode import GraphedDecoder
+
+        graphed = GraphedDecoder(model, batch_size, seq_len, max_new)
+
     done_batches = 0
     out_idx = 0
     while done_batches < num_batches:
@@ -68,8 +79,12 @@ def main() -> int:
                 if done_batches >= num_batches:
                     break
                 prompts = ids[i : i + batch_size, :seq_len].to(device)
-                out = model.generate(prompts, max_new_tokens=max_new,
-                                     temperature=temperature, top_k=top_k, seed=0)
+                if graphed is not None and prompts.shape[0] == batch_size:
+                    out = graphed.generate(prompts, max_new_tokens=max_new)
+                else:
+                    out = model.generate(prompts, max_new_tokens=max_new,
+                                         temperature=temperature, top_k=top_k,
+                                         seed=0)
                 fname = f"completion-{out_idx:05d}.safetensors"
                 save_file({"tokens": out.cpu().contiguous()},
                           os.path.join(args.work_dir, fname))

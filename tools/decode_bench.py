@@ -18,6 +18,8 @@ def main():
     p.add_argument("--prefill", type=int, default=512)
     p.add_argument("--decode", type=int, default=128)
     p.add_argument("--warmup", type=int, default=16)
+    p.add_argument("--graph", action="store_true",
+                   help="hipGraph-captured decode step (greedy)")
     args = p.parse_args()
 
     from hypha_amd import models
@@ -35,13 +37,22 @@ def main():
     ids = torch.randint(0, model.cfg.vocab_size - 1, (args.batch, args.prefill),
                         device=dev)
 
+    if args.graph:
+        from hypha_amd.runtime.graphed_decode import GraphedDecoder
+
+        dec = GraphedDecoder(model, args.batch, args.prefill,
+                             args.decode + args.warmup)
+        gen = dec.generate
+    else:
+        gen = model.generate
+
     t0 = time.perf_counter()
-    model.generate(ids, max_new_tokens=args.warmup)
+    gen(ids, max_new_tokens=args.warmup)
     torch.cuda.synchronize()
     t_warm = time.perf_counter() - t0
 
     t0 = time.perf_counter()
-    out = model.generate(ids, max_new_tokens=args.decode + args.warmup)
+    out = gen(ids, max_new_tokens=args.decode + args.warmup)
     torch.cuda.synchronize()
     t_all = time.perf_counter() - t0
     # decode-only rate from the marginal cost of the extra tokens
