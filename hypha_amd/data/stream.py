@@ -58,21 +58,25 @@ def infinite(loader: Iterable) -> Iterator:
 def build_preprocessor(task: str, artifact_path: str | None = None):
     """Resolve a wire PreprocessorType to a callable.
 
-    `tokenizer` loads a local tokenizer directory/file via `transformers`
-    (offline); the media types (`feature`/`image`/`video`) and `auto` need HF
-    hub processor classes whose weights cannot be fetched here — they raise
-    with the reason rather than silently degrading.
+    Mirrors the reference's get_preprocessor mapping
+    (executors/accelerate/.../utils.py:37-54: tokenizer/feature/image/
+    video/auto -> the matching transformers Auto* class loaded from the
+    FETCHED artifact path). All types load from LOCAL artifacts
+    (`local_files_only=True`): the artifact arrives through the fetch
+    connector (scheduler slice / URI / HF), never from the hub at
+    preprocess time.
     """
     if task not in PREPROCESSOR_TYPES:
         raise ValueError(
             f"preprocessor {task!r} not supported; available: {PREPROCESSOR_TYPES}"
         )
+    if not artifact_path or not os.path.exists(artifact_path):
+        raise FileNotFoundError(
+            f"preprocessor artifact {artifact_path!r} not found (no network: "
+            "artifacts must be local files fetched through the connector)"
+        )
+
     if task == "tokenizer":
-        if not artifact_path or not os.path.exists(artifact_path):
-            raise FileNotFoundError(
-                f"tokenizer artifact {artifact_path!r} not found (no network: "
-                "artifacts must be local files)"
-            )
         from transformers import AutoTokenizer
 
         tok = AutoTokenizer.from_pretrained(artifact_path, local_files_only=True)
@@ -83,7 +87,46 @@ def build_preprocessor(task: str, artifact_path: str | None = None):
             return dict(out)
 
         return run
-    raise NotImplementedError(
-        f"preprocessor type {task!r} requires HF hub processors; this "
-        "offline deployment supports 'tokenizer' with local artifacts"
-    )
+
+    if task in ("feature", "image", "video", "auto"):
+        kwargs = {}
+        if task == "feature":
+            from transformers import AutoFeatureExtractor as Cls
+        elif task == "image":
+            try:
+                import torchvision  # noqa: F401
+
+                from transformers import AutoImageProcessor as Cls
+            except ImportError:
+                # no torchvision: resolve the PIL-backend processor class
+                # named by the artifact config directly (transformers
+                # aliases e.g. ViTImageProcessor -> ViTImageProcessorPil)
+                import json as _json
+
+                import transformers as _tf
+
+                with open(os.path.join(artifact_path,
+                                       "preprocessor_config.json")) as f:
+                    cls_name = _json.load(f)["image_processor_type"]
+                Cls = getattr(_tf, cls_name)
+        elif task == "video":
+            try:
+                from transformers import AutoVideoProcessor as Cls
+            except ImportError as e:  # transformers build without video stack
+                raise NotImplementedError(
+                    "this transformers build has no AutoVideoProcessor"
+                ) from e
+        else:
+            from transformers import AutoProcessor as Cls
+
+        proc = Cls.from_pretrained(artifact_path, local_files_only=True, **kwargs)
+
+        def run(**kw):
+            (key, values), = kw.items()
+            if hasattr(values, "numpy"):
+                values = [v.numpy() for v in values]
+            out = proc(values, return_tensors="pt")
+            return dict(out)
+
+        return run
+    raise NotImplementedError(task)

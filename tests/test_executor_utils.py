@@ -82,8 +82,8 @@ def test_preprocessor_registry_errors():
         build_preprocessor("audio")
     with pytest.raises(FileNotFoundError, match="no network"):
         build_preprocessor("tokenizer", "/nonexistent/tok")
-    with pytest.raises(NotImplementedError, match="offline"):
-        build_preprocessor("image")
+    with pytest.raises(FileNotFoundError, match="no network"):
+        build_preprocessor("image")  # media types need a local artifact too
 
 
 def test_tokenizer_preprocessor_from_local_artifact(tmp_path):
@@ -152,3 +152,54 @@ def test_lr_schedules_vs_transformers_oracle():
     # endpoints agree exactly: peak 1.0 after warmup, ~0 at the end
     assert ours_cos[-1] < 0.01 and cos[-1] < 0.01
     assert ours_lin[-1] < 0.02 and lin[-1] < 0.02
+
+
+def test_feature_preprocessor_from_local_artifact(tmp_path):
+    """PreprocessorType::Feature resolved against a fetched local artifact
+    (reference utils.py:45-46 AutoFeatureExtractor mapping) — offline."""
+    import json
+
+    import torch
+
+    from hypha_amd.data.stream import build_preprocessor
+
+    art = tmp_path / "fe"
+    art.mkdir()
+    (art / "preprocessor_config.json").write_text(json.dumps({
+        "feature_extractor_type": "Wav2Vec2FeatureExtractor",
+        "feature_size": 1, "sampling_rate": 16000, "padding_value": 0.0,
+        "return_attention_mask": False, "do_normalize": True}))
+    pre = build_preprocessor("feature", str(art))
+    out = pre(audio=torch.randn(2, 400))
+    assert "input_values" in out and out["input_values"].shape == (2, 400)
+    # normalized output
+    assert abs(float(out["input_values"][0].mean())) < 0.1
+
+
+def test_image_preprocessor_from_local_artifact(tmp_path):
+    import json
+
+    import torch
+
+    from hypha_amd.data.stream import build_preprocessor
+
+    art = tmp_path / "im"
+    art.mkdir()
+    (art / "preprocessor_config.json").write_text(json.dumps({
+        "image_processor_type": "ViTImageProcessor", "do_resize": True,
+        "size": {"height": 32, "width": 32}, "do_normalize": True,
+        "image_mean": [0.5, 0.5, 0.5], "image_std": [0.5, 0.5, 0.5],
+        "do_rescale": True}))
+    pre = build_preprocessor("image", str(art))
+    imgs = torch.randint(0, 255, (2, 3, 48, 48), dtype=torch.uint8)
+    out = pre(images=imgs)
+    assert out["pixel_values"].shape == (2, 3, 32, 32)
+
+
+def test_preprocessor_missing_artifact_raises():
+    import pytest as _pytest
+
+    from hypha_amd.data.stream import build_preprocessor
+
+    with _pytest.raises(FileNotFoundError):
+        build_preprocessor("feature", "/nonexistent/path")
