@@ -62,6 +62,7 @@ from .interface import (  # noqa: E402
     fused_nesterov,
     gelu,
     layernorm,
+    linear_skinny,
     rmsnorm,
     swiglu,
 )
@@ -82,5 +83,6 @@ __all__ = [
     "cross_entropy_loss",
     "fused_adamw",
     "fused_nesterov",
+    "linear_skinny",
     "extract_delta",
 ]

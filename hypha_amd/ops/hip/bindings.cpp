@@ -95,6 +95,8 @@ torch::Tensor attn_decode(torch::Tensor q, torch::Tensor kcache, torch::Tensor v
                           long t);
 torch::Tensor attn_decode_graph(torch::Tensor q, torch::Tensor kcache,
                                 torch::Tensor vcache, torch::Tensor t_dev);
+// skinny_gemm.hip (decode GEMV, M <= 8)
+torch::Tensor skinny_gemm(torch::Tensor x, torch::Tensor w);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("adamw_step_", &adamw_step_);
@@ -138,4 +140,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fp8_weight_cast_transpose", &fp8_weight_cast_transpose);
   m.def("attn_decode", &attn_decode);
   m.def("attn_decode_graph", &attn_decode_graph);
+  m.def("skinny_gemm", &skinny_gemm);
 }

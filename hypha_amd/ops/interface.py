@@ -266,6 +266,16 @@ def attn_decode(q, kcache, vcache, t: int) -> torch.Tensor:
     return o.to(q.dtype)
 
 
+def linear_skinny(x2d: torch.Tensor, weight: torch.Tensor) -> torch.Tensor:
+    """Inference linear for decode shapes: out = x2d @ weight.T with the
+    weight-bandwidth-bound skinny-M kernel when M <= 8 (ops/hip/
+    skinny_gemm.hip); hipBLASLt otherwise. No autograd."""
+    if (use_native(x2d) and x2d.shape[0] <= 8
+            and weight.dtype == torch.bfloat16):
+        return _c().skinny_gemm(x2d.contiguous(), weight)
+    return x2d @ weight.t()
+
+
 # ---------------------------------------------------------------------------
 # Cross-entropy over vocab (memory-frugal: backward overwrites the logits)
 # ---------------------------------------------------------------------------

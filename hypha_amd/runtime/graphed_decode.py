@@ -48,34 +48,40 @@ class GraphedDecoder:
         self.graph = None
 
     def _step(self) -> None:
-        """One decode step over static device state (graph-capturable)."""
+        """One decode step over static device state (graph-capturable).
+        GEMMs run 2-D through hipBLASLt (the hand-written skinny-M GEMV,
+        ops/hip/skinny_gemm.hip, measured 0.4 TB/s vs the library's 1.8-5.9
+        on these shapes — shuffle-reduction issue-bound; kept as an opt-in
+        negative result like grouped_gemm)."""
         m = self.model
         cfg = m.cfg
         b = self.batch
+        lin = lambda t, wt: t @ wt.t()  # noqa: E731
         x = m.embed(self.tok)
         cos = m.rope_cos.index_select(0, self.pos)
         sin = m.rope_sin.index_select(0, self.pos)
         for blk, cache in zip(m.blocks, self.caches):
-            n1 = blk.attn_norm(x)
+            n1 = blk.attn_norm(x)[:, 0]  # [b, h]
             a = blk.attn
-            q = a.wq(n1).view(b, 1, cfg.n_heads, cfg.head_dim)
-            k = a.wk(n1).view(b, 1, cfg.n_kv_heads, cfg.head_dim)
-            v = a.wv(n1).view(b, 1, cfg.n_kv_heads, cfg.head_dim)
+            q = lin(n1, a.wq.weight).view(b, 1, cfg.n_heads, cfg.head_dim)
+            k = lin(n1, a.wk.weight).view(b, 1, cfg.n_kv_heads, cfg.head_dim)
+            v = lin(n1, a.wv.weight).view(b, 1, cfg.n_kv_heads, cfg.head_dim)
             q, k = ops.apply_rope_qk(q, k, cos, sin, layout="bshd")
             cache.k.index_copy_(1, self.pos, k)
             cache.v.index_copy_(1, self.pos, v)
             o = self._C.attn_decode_graph(q[:, 0].contiguous(), cache.k,
                                           cache.v, self.t32)
-            attn_out = a.wo(o.reshape(b, 1, -1))
+            attn_out = lin(o.reshape(b, -1), a.wo.weight).view(b, 1, -1)
             x, n2 = ops.add_rmsnorm(x, attn_out, blk.mlp_norm.weight,
                                     blk.mlp_norm.eps)
-            x = x + blk.mlp(n2)
+            n2f = n2[:, 0]
+            h = ops.swiglu(lin(n2f, blk.mlp.w_gate.weight),
+                           lin(n2f, blk.mlp.w_up.weight))
+            x = x + lin(h, blk.mlp.w_down.weight).view(b, 1, -1)
         x = m.norm(x)
-        if m.lm_head is not None:
-            logits = m.lm_head(x)
-        else:
-            logits = torch.nn.functional.linear(x, m.embed.weight)
-        nxt = logits[:, 0].argmax(-1, keepdim=True)  # [b, 1]
+        head_w = m.lm_head.weight if m.lm_head is not None else m.embed.weight
+        logits = lin(x[:, 0], head_w)
+        nxt = logits.argmax(-1, keepdim=True)  # [b, 1]
         self.step.clamp_(max=self.max_new - 1)
         self.out.index_copy_(1, self.step, nxt)
         self.tok.copy_(nxt)

@@ -865,3 +865,17 @@ def test_graphed_decode_matches_eager_generate():
     # second call reuses the captured graph
     graphed2 = dec.generate(ids, max_new_tokens=24)
     assert torch.equal(graphed2, graphed)
+
+
+def test_skinny_gemm_matches_matmul():
+    """Decode GEMV kernel vs plain matmul across decode shapes."""
+    from hypha_amd import _C
+
+    torch.manual_seed(37)
+    for M, N, K in ((8, 4096, 4096), (8, 14336, 4096), (8, 4096, 14336),
+                    (1, 1024, 4096), (4, 128256, 4096), (8, 4096, 4100)):
+        x = rand_bf16(M, K, seed=80 + N % 97, scale=0.5)
+        w = rand_bf16(N, K, seed=81 + N % 97, scale=0.5)
+        o = _C.skinny_gemm(x, w)
+        ref = (x.float() @ w.float().t())
+        torch.testing.assert_close(o.float(), ref, rtol=2e-2, atol=K ** 0.5 * 2e-2)

@@ -2,11 +2,22 @@
 single-token KV-cache decode steps. Prints decode tokens/s."""
 
 import argparse
+import os
 import sys
 import time
 from pathlib import Path
 
 sys.path.insert(0, str(Path(__file__).resolve().parent.parent))
+
+# pre-tuned hipBLASLt algos (shared table with bench.py; includes the
+# skinny decode shapes)
+_TUNED_BASE = str(Path(__file__).resolve().parent.parent /
+                  "tunableop_gfx950_llama8b.csv")
+if (os.path.exists(_TUNED_BASE.replace(".csv", "0.csv"))
+        and "PYTORCH_TUNABLEOP_ENABLED" not in os.environ):
+    os.environ["PYTORCH_TUNABLEOP_ENABLED"] = "1"
+    os.environ["PYTORCH_TUNABLEOP_TUNING"] = "0"
+    os.environ["PYTORCH_TUNABLEOP_FILENAME"] = _TUNED_BASE
 
 import torch
 
@@ -46,21 +57,23 @@ def main():
     else:
         gen = model.generate
 
+    # warm (weights/algos/graph capture), then difference two WARM runs so
+    # one-time costs never land in the per-token figure
+    gen(ids, max_new_tokens=args.warmup)
+    torch.cuda.synchronize()
     t0 = time.perf_counter()
     gen(ids, max_new_tokens=args.warmup)
     torch.cuda.synchronize()
-    t_warm = time.perf_counter() - t0
-
+    t_short = time.perf_counter() - t0
     t0 = time.perf_counter()
     out = gen(ids, max_new_tokens=args.decode + args.warmup)
     torch.cuda.synchronize()
-    t_all = time.perf_counter() - t0
-    # decode-only rate from the marginal cost of the extra tokens
-    dt = t_all - t_warm
+    t_long = time.perf_counter() - t0
+    dt = t_long - t_short
     toks = args.batch * args.decode
     print(f"{args.model} b{args.batch} prefill{args.prefill}: "
           f"{toks / dt:.0f} decode tok/s  ({dt / args.decode * 1e3:.2f} ms/step)"
-          f"  [prefill+{args.warmup} warm: {t_warm:.2f}s]")
+          f"  [prefill+{args.warmup}: {t_short:.2f}s]")
     assert out.shape[1] == args.prefill + args.decode + args.warmup
 
 
