@@ -28,6 +28,8 @@
 
 namespace {
 
+typedef __attribute__((ext_vector_type(2))) __bf16 bf16x2;
+
 constexpr int KVB = 64;   // kv tile (fwd)
 constexpr int QB = 32;    // q rows per wave
 constexpr int NWAVE = 4;  // waves per workgroup
@@ -812,7 +814,8 @@ __global__ __launch_bounds__(256, 1) void attn_bwd_v4_kernel(
     float* __restrict__ dqg, short* __restrict__ dkg, short* __restrict__ dvg,
     int B, int Hq, int Hkv, int S, float scale, bool causal,
     long long q_sb, long long q_sh, long long q_ss,
-    long long kv_sb, long long kv_sh, long long kv_ss, int ablate) {
+    long long kv_sb, long long kv_sh, long long kv_ss, int ablate,
+    int dq_bf16) {
   constexpr int KVT = 32;        // kv rows per wave
   constexpr int NT = 4;          // kv tiles (= waves) per workgroup
   constexpr int KVW = NT * KVT;  // 128 kv rows per workgroup
@@ -904,10 +907,31 @@ __global__ __launch_bounds__(256, 1) void attn_bwd_v4_kernel(
       dq = __builtin_amdgcn_mfma_f32_32x32x16_bf16(af, bf, dq, 0, 0, 0);
     }
     if (!(ablate & 1)) {
+      if (dq_bf16) {
+        // HYPHA_ATTN_DQ_BF16: halve the dQ RMW traffic with packed-bf16
+        // atomics (adjacent d columns live in lane pairs: one shuffle
+        // pairs them, even lanes issue one v2bf16 add for both)
 #pragma unroll
-      for (int r = 0; r < 16; ++r) {
-        int qi = prev_q0 + (r & 3) + 8 * (r >> 2) + 4 * hi;
-        atomicAdd(dqg + (prev_qbase + (long long)qi * q_ss + 32 * db + ln), dq[r]);
+        for (int r = 0; r < 16; ++r) {
+          int qi = prev_q0 + (r & 3) + 8 * (r >> 2) + 4 * hi;
+          float other = __shfl_xor(dq[r], 1, 64);
+          if ((ln & 1) == 0) {
+            bf16x2 pkt;
+            pkt.x = (__bf16)dq[r];
+            pkt.y = (__bf16)other;
+            __builtin_amdgcn_global_atomic_fadd_v2bf16(
+                reinterpret_cast<bf16x2*>(
+                    reinterpret_cast<short*>(dqg) +
+                    (prev_qbase + (long long)qi * q_ss + 32 * db + ln)),
+                pkt);
+          }
+        }
+      } else {
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          int qi = prev_q0 + (r & 3) + 8 * (r >> 2) + 4 * hi;
+          atomicAdd(dqg + (prev_qbase + (long long)qi * q_ss + 32 * db + ln), dq[r]);
+        }
       }
     } else {
       asm volatile("" ::"v"(dq[0]));  // keep the MFMA chain live (perf probe)
@@ -1074,7 +1098,7 @@ std::vector<torch::Tensor> attn_bwd_ex(torch::Tensor q, torch::Tensor k, torch::
   const int HD = q.size(3);
   const int Hkv = bshd ? k.size(2) : k.size(1);
   TORCH_CHECK(S % 64 == 0 && (HD == 64 || HD == 128));
-  auto dq32 = torch::zeros(q.sizes(), q.options().dtype(torch::kFloat32));
+  auto dq32 = torch::empty({0}, q.options());  // allocated after variant choice
   auto dk = torch::empty_like(k);
   auto dv = torch::empty_like(v);
   auto di = torch::empty({B, Hq, S}, q.options().dtype(torch::kFloat32));
@@ -1104,6 +1128,15 @@ std::vector<torch::Tensor> attn_bwd_ex(torch::Tensor q, torch::Tensor k, torch::
     const char* e = getenv("HYPHA_ATTN_ABLATE");
     return e ? atoi(e) : 0;
   }();
+  // packed-bf16 dQ accumulation (v4 only): halves the dominant RMW traffic;
+  // costs ~3 bits of dq mantissa over 16 accumulations — opt-in
+  static const bool dq_bf16_env = [] {
+    const char* e = getenv("HYPHA_ATTN_DQ_BF16");
+    return e && e[0] == '1';
+  }();
+  const bool dq_bf16 = dq_bf16_env && variant == 4;
+  dq32 = dq_bf16 ? torch::zeros(q.sizes(), q.options())
+                 : torch::zeros(q.sizes(), q.options().dtype(torch::kFloat32));
 
 #define LAUNCH_BWD(KERN, NQ, HDV)                                                       \
   hipLaunchKernelGGL(KERN<HDV>, dim3(S / NQ, B * Hkv), dim3(256), 0, stream,            \
@@ -1125,10 +1158,10 @@ std::vector<torch::Tensor> attn_bwd_ex(torch::Tensor q, torch::Tensor k, torch::
                          0, stream, (const short*)q.data_ptr(),                         \
                          (const short*)k.data_ptr(), (const short*)v.data_ptr(),        \
                          (const short*)dout.data_ptr(), lse.data_ptr<float>(),          \
-                         di.data_ptr<float>(), dq32.data_ptr<float>(),                  \
+                         di.data_ptr<float>(), (float*)dq32.data_ptr(),                 \
                          (short*)dk.data_ptr(), (short*)dv.data_ptr(), B, Hq, Hkv, S,   \
                          scale, causal, q_sb, q_sh, q_ss, kv_sb, kv_sh, kv_ss,          \
-                         bwd_ablate);                                                   \
+                         bwd_ablate, dq_bf16 ? 1 : 0);                                  \
     else if (variant == 3)                                                              \
       LAUNCH_BWD(attn_bwd_v3_kernel, 64, HDV);                                          \
     else                                                                                \
@@ -1141,6 +1174,7 @@ std::vector<torch::Tensor> attn_bwd_ex(torch::Tensor q, torch::Tensor k, torch::
     DISPATCH(64);
 #undef DISPATCH
 
+  if (dq_bf16) return {dq32, dk, dv};  // accumulated directly in bf16
   auto dq = torch::empty_like(q);
   long long n = dq32.numel();
   hipLaunchKernelGGL(cast_f32_to_bf16_kernel, dim3(elementwise_grid((n + 3) / 4)),
