@@ -8,6 +8,8 @@
 #include <unistd.h>
 
 #include <atomic>
+#include <chrono>
+#include <condition_variable>
 #include <map>
 #include <memory>
 #include <mutex>
@@ -59,6 +61,17 @@ class Gateway {
       std::thread([this, fd] {
         SSL* ssl = nullptr;
         if (tls_) {
+          // Relay legs are PLAIN sockets carrying length-prefixed JSON
+          // hellos; the end-to-end peer TLS runs THROUGH the circuit (the
+          // relay sees only its records — reference relay-server role,
+          // gateway/src/network.rs:44). First byte discriminates: a TLS
+          // ClientHello starts 0x16, our frame length prefix starts 0x00.
+          char b0 = 0;
+          ssize_t pk = ::recv(fd, &b0, 1, MSG_PEEK);
+          if (pk == 1 && b0 == 0x00) {
+            handle_conn(fd, nullptr, /*relay_only=*/true);
+            return;
+          }
           ssl = tls_->wrap(fd, true);
           if (!ssl) {
             ::close(fd);
@@ -72,7 +85,7 @@ class Gateway {
     listen_fd_ = -1;
   }
 
-  void handle_conn(int fd, SSL* ssl = nullptr) {
+  void handle_conn(int fd, SSL* ssl = nullptr, bool relay_only = false) {
     auto sock = std::make_shared<MsgSocket>(fd, ssl);
     std::string peer;  // set once registered (persistent connection)
     // With mTLS the verified cert CN is the only identity we trust: the
@@ -86,6 +99,15 @@ class Gateway {
         auto msg = sock->recv_json();
         if (!msg) break;
         std::string kind = msg->get_or("kind", Json("")).as_string();
+        if (kind == "relay_connect") {
+          relay_connect(sock, *msg);
+          return;  // the connection IS the circuit now (or it failed)
+        }
+        if (kind == "relay_accept") {
+          relay_accept(sock, *msg);
+          return;
+        }
+        if (relay_only) break;  // plain leg may only speak relay kinds
         if (kind == "register") {
           std::string claimed = msg->at("peer").as_string();
           if (ssl && !cn.empty() && claimed != cn) {
@@ -205,7 +227,112 @@ class Gateway {
   int port_ = 0;
   std::atomic<bool> running_{false};
   std::thread accept_thread_;
+  // ---- relay circuits (opaque byte splice between two legs) ----
+  struct Circuit {
+    std::mutex m;
+    std::condition_variable cv;
+    std::shared_ptr<MsgSocket> b;  // acceptor leg, set by relay_accept
+  };
+
+  void relay_connect(std::shared_ptr<MsgSocket> a, const Json& msg) {
+    const std::string to = msg.get_or("to", Json("")).as_string();
+    std::shared_ptr<MsgSocket> target;
+    {
+      std::lock_guard<std::mutex> lk(mu_);
+      auto it = peers_.find(to);
+      if (it != peers_.end()) target = it->second;
+    }
+    Json nak;
+    nak["ok"] = false;
+    if (!target) {
+      nak["error"] = "relay: peer not registered: " + to;
+      a->send_json(nak);
+      return;
+    }
+    long long id;
+    auto circ = std::make_shared<Circuit>();
+    {
+      std::lock_guard<std::mutex> lk(mu_);
+      id = next_circuit_++;
+      circuits_[id] = circ;
+    }
+    Json offer;
+    offer["kind"] = "relay_offer";
+    offer["circuit"] = (int64_t)id;
+    offer["from"] = msg.get_or("from", Json("")).as_string();
+    bool pushed = target->send_json(offer);
+    std::shared_ptr<MsgSocket> b;
+    if (pushed) {
+      std::unique_lock<std::mutex> lk(circ->m);
+      circ->cv.wait_for(lk, std::chrono::seconds(10), [&] { return (bool)circ->b; });
+      b = circ->b;
+    }
+    {
+      std::lock_guard<std::mutex> lk(mu_);
+      circuits_.erase(id);
+    }
+    if (!b) {
+      nak["error"] = "relay: peer did not accept";
+      a->send_json(nak);
+      return;
+    }
+    Json ok;
+    ok["ok"] = true;
+    if (!b->send_json(ok) || !a->send_json(ok)) return;
+    // ---- splice: opaque bytes both ways until either side closes ----
+    int fa = a->release();
+    int fb = b->release();
+    std::thread rev([fa, fb] { pump(fb, fa); });
+    pump(fa, fb);
+    ::shutdown(fa, 2);
+    ::shutdown(fb, 2);
+    rev.join();  // never close an fd another thread may still be using
+    ::close(fa);
+    ::close(fb);
+  }
+
+  void relay_accept(std::shared_ptr<MsgSocket> b, const Json& msg) {
+    long long id = msg.get_or("circuit", Json((int64_t)0)).as_int();
+    std::shared_ptr<Circuit> circ;
+    {
+      std::lock_guard<std::mutex> lk(mu_);
+      auto it = circuits_.find(id);
+      if (it != circuits_.end()) circ = it->second;
+    }
+    if (!circ) {
+      Json nak;
+      nak["ok"] = false;
+      nak["error"] = "relay: unknown circuit";
+      b->send_json(nak);
+      return;
+    }
+    {
+      std::lock_guard<std::mutex> lk(circ->m);
+      circ->b = b;
+    }
+    circ->cv.notify_all();
+    // ownership passes to the connect-side thread via the shared_ptr held
+    // in the Circuit: this handler returns immediately and the socket
+    // outlives it until the splice finishes.
+  }
+
+  static void pump(int src, int dst) {
+    char buf[1 << 16];
+    while (true) {
+      ssize_t n = ::read(src, buf, sizeof buf);
+      if (n <= 0) break;
+      ssize_t off = 0;
+      while (off < n) {
+        ssize_t w = ::write(dst, buf + off, n - off);
+        if (w <= 0) return;
+        off += w;
+      }
+    }
+  }
+
   std::mutex mu_;
+  std::map<long long, std::shared_ptr<Circuit>> circuits_;
+  long long next_circuit_ = 1;
   std::map<std::string, std::shared_ptr<MsgSocket>> peers_;
   std::map<std::string, std::set<std::string>> subs_;
   std::map<std::string, Json> kv_;

@@ -40,6 +40,13 @@ class MsgSocket {
   bool recv_raw(char* data, size_t n);
   void close_now();
   int fd() const { return fd_; }
+  // Detach the fd (relay handshake: a few plain frames are exchanged and
+  // then the SAME socket carries the end-to-end TLS/protocol bytes).
+  int release() {
+    int f = fd_;
+    fd_ = -1;
+    return f;
+  }
   std::string peer_identity() const { return TlsContext::peer_cn(ssl_); }
 
  private:
@@ -128,6 +135,13 @@ class Node {
   void accept_loop();
   void handle_conn(int fd, SSL* ssl = nullptr);
   void gateway_listen_loop();
+  // Relay fallback (reference gateway relay-server role, network.rs:44):
+  // when a peer cannot be dialed directly, the connection is tunneled
+  // through the gateway as an opaque byte circuit; with mTLS the
+  // end-to-end handshake still runs peer<->peer (the relay sees only
+  // TLS records). relay_dial returns a connected fd or -1.
+  int relay_dial(const std::string& peer, double timeout_s);
+  void relay_accept_run(long long circuit);
   bool gateway_connect();  // (re)connect + register + replay subscriptions
   Json gateway_request(const std::string& type, const Json& body);
 

@@ -418,7 +418,17 @@ void Node::gateway_listen_loop() {
       continue;
     }
     try {
-    if (msg->get_or("kind", Json("")).as_string() == "pub") {
+    if (msg->get_or("kind", Json("")).as_string() == "relay_offer") {
+      long long circuit = msg->get_or("circuit", Json((int64_t)0)).as_int();
+      std::thread([this, circuit] {
+        try {
+          relay_accept_run(circuit);
+        } catch (const std::exception& e) {
+          fprintf(stderr, "[net:%s] relay accept error: %s\n", name_.c_str(),
+                  e.what());
+        }
+      }).detach();
+    } else if (msg->get_or("kind", Json("")).as_string() == "pub") {
       std::string topic = msg->at("topic").as_string();
       std::string from = msg->get_or("from", Json("")).as_string();
       std::function<void(const std::string&, const Json&)> cb;
@@ -433,6 +443,57 @@ void Node::gateway_listen_loop() {
       fprintf(stderr, "[net:%s] gateway event error: %s\n", name_.c_str(), e.what());
     }
   }
+}
+
+int Node::relay_dial(const std::string& peer, double timeout_s) {
+  if (gw_port_ == 0) return -1;
+  int fd = tcp_connect(gw_host_, gw_port_, timeout_s);
+  if (fd < 0) return -1;
+  struct timeval tv;
+  tv.tv_sec = (long)(timeout_s + 15.0);
+  tv.tv_usec = 0;
+  setsockopt(fd, SOL_SOCKET, SO_RCVTIMEO, &tv, sizeof tv);
+  MsgSocket hello(fd, nullptr);  // plain framing: first byte 0x00 tells the
+                                 // gateway this is a relay leg, not TLS
+  Json req;
+  req["kind"] = "relay_connect";
+  req["to"] = peer;
+  req["from"] = name_;
+  if (!hello.send_json(req)) return -1;
+  auto ack = hello.recv_json();
+  if (!ack || !ack->get_or("ok", Json(false)).as_bool()) {
+    return -1;  // MsgSocket dtor closes fd
+  }
+  fprintf(stderr, "[net:%s] relayed circuit to %s via gateway\n", name_.c_str(),
+          peer.c_str());
+  return hello.release();  // same socket now carries the peer protocol
+}
+
+void Node::relay_accept_run(long long circuit) {
+  int fd = tcp_connect(gw_host_, gw_port_, 10.0);
+  if (fd < 0) return;
+  {
+    MsgSocket hello(fd, nullptr);
+    Json acc;
+    acc["kind"] = "relay_accept";
+    acc["circuit"] = (int64_t)circuit;
+    acc["from"] = name_;
+    if (!hello.send_json(acc)) return;
+    auto ack = hello.recv_json();
+    if (!ack || !ack->get_or("ok", Json(false)).as_bool()) return;
+    fd = hello.release();
+  }
+  // from here the circuit is an ordinary inbound connection: server-side
+  // mTLS handshake + the normal typed-message loop
+  SSL* ssl = nullptr;
+  if (tls_) {
+    ssl = tls_->wrap(fd, true);
+    if (!ssl) {
+      ::close(fd);
+      return;
+    }
+  }
+  handle_conn(fd, ssl);
 }
 
 Json Node::gateway_request(const std::string& type, const Json& body) {
@@ -497,6 +558,7 @@ Json Node::request(const std::string& peer, const std::string& type, const Json&
     check_dialable(peer, addr->substr(0, colon));
     fd = tcp_connect(addr->substr(0, colon), std::stoi(addr->substr(colon + 1)), timeout_s);
   }
+  if (fd < 0) fd = relay_dial(peer, timeout_s);  // gateway circuit fallback
   if (fd < 0) throw std::runtime_error("peer unreachable: " + peer);
   SSL* pssl = nullptr;
   if (tls_) {
@@ -539,6 +601,7 @@ std::unique_ptr<MsgSocket> Node::open_stream(const std::string& peer, const std:
     check_dialable(peer, addr->substr(0, colon));
     fd = tcp_connect(addr->substr(0, colon), std::stoi(addr->substr(colon + 1)), 30.0);
   }
+  if (fd < 0) fd = relay_dial(peer, 30.0);  // gateway circuit fallback
   if (fd < 0) throw std::runtime_error("peer unreachable: " + peer);
   struct timeval tv = {600, 0};
   setsockopt(fd, SOL_SOCKET, SO_RCVTIMEO, &tv, sizeof tv);

@@ -254,3 +254,44 @@ def test_gateway_restart_reconnect():
     finally:
         a.stop()
         b.stop()
+
+
+def test_relay_circuit_when_direct_dial_fails(cluster):
+    """Gateway relay-server role (reference gateway/src/network.rs:44):
+    when the registered address is unreachable, the dial falls back to a
+    gateway-spliced byte circuit; requests and streams work through it."""
+    a = cluster("alice")
+    b = cluster("bob")
+    b.on("echo", lambda frm, body: {"from_seen": frm, "payload": body["x"] * 3})
+    # blackhole bob's registered address: direct dialing now fails
+    a.kv_put("addr:bob", "127.0.0.1:1")
+    r = a.request("bob", "echo", {"x": 14})
+    assert r == {"from_seen": "alice", "payload": 42}
+    # several sequential relayed requests (one circuit each)
+    for i in range(3):
+        assert a.request("bob", "echo", {"x": i})["payload"] == 3 * i
+
+
+def test_relay_blob_stream(cluster):
+    a = cluster("alice")
+    b = cluster("bob")
+    got = []
+
+    def on_blob(frm, header, blob):
+        got.append((frm, header["tag"], blob))
+
+    b.on_blob("blobby", on_blob)
+    a.kv_put("addr:bob", "127.0.0.1:1")
+    payload = bytes(range(256)) * 300
+    import time as _t
+
+    a.push_blob("bob", "blobby", {"tag": "t1", "size": len(payload)}, payload)
+    _t.sleep(0.3)
+    assert got and got[0][0] == "alice" and got[0][2] == payload
+
+
+def test_relay_unregistered_peer_fails(cluster):
+    a = cluster("alice")
+    a.kv_put("addr:ghost", "127.0.0.1:1")
+    with pytest.raises(RuntimeError):
+        a.request("ghost", "echo", {}, 2.0)

@@ -276,3 +276,28 @@ def test_gateway_survives_malformed_messages(pki):
         a.stop()
     finally:
         gw.stop()
+
+
+def test_mtls_relay_circuit(pki):
+    """End-to-end mTLS THROUGH a gateway relay circuit: when bob's direct
+    address is unreachable, alice's connection is spliced by the gateway as
+    opaque bytes — the peer TLS handshake still runs alice<->bob, so the
+    relay cannot read or impersonate, and bob still sees the verified CN."""
+    out, _ = pki
+    gw = core.Gateway(str(out / "gw.chain.pem"), str(out / "gw.key"),
+                      str(out / "root.crt"))
+    gw.start(0)
+    a = core.Node("alice", "127.0.0.1", gw.port, **_tls_args(out, "alice"))
+    b = core.Node("bob", "127.0.0.1", gw.port, **_tls_args(out, "bob"))
+    try:
+        a.start(0)
+        b.start(0)
+        seen = []
+        b.on("echo", lambda frm, body: (seen.append(frm), {"x": body["x"] + 1})[1])
+        a.kv_put("addr:bob", "127.0.0.1:1")  # force the relay path
+        assert a.request("bob", "echo", {"x": 41}) == {"x": 42}
+        assert seen == ["alice"]  # identity from the VERIFIED cert CN
+    finally:
+        a.stop()
+        b.stop()
+        gw.stop()
