@@ -533,6 +533,7 @@ int main(int argc, char** argv) {
       fetch["scheduler"] = sref;
     }
     tr["data"] = fetch;
+    if (cfg.has("preprocessor")) tr["preprocessor"] = cfg.at("preprocessor");
     if (rank >= 0) {
       Json rc;
       rc["rank"] = (int64_t)rank;

