@@ -41,7 +41,7 @@ def main() -> int:
     p.add_argument("--steps", type=int, default=10)
     p.add_argument("--warmup", type=int, default=3)
     p.add_argument("--model", type=str, default="llama3-8b")
-    p.add_argument("--batch", type=int, default=6, help="per-GPU batch (sequences)")
+    p.add_argument("--batch", type=int, default=8, help="per-GPU batch (sequences)")
     p.add_argument("--seq-len", type=int, default=2048)
     p.add_argument("--h", type=int, default=100, help="DiLoCo inner steps per outer sync")
     p.add_argument("--device", type=str, default=None)
