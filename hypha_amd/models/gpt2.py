@@ -75,14 +75,16 @@ class GPT2Block(nn.Module):
             v = v.view(b, s, self.n_heads, hd)
             t = cache.append(k, v)
             if s == 1:
-                o = ops.attn_decode(q[:, 0].contiguous(), cache.k, cache.v, t)
+                o = ops.attn_decode(q[:, 0].contiguous(), cache.k, cache.v, t,
+                                    cache.k_scale, cache.v_scale)
                 o = o.reshape(b, 1, h)
             elif pos == 0 and (not x.is_cuda or s % 128 == 0):
                 o = ops.flash_attention(q, k, v, causal=True, layout="bshd")
                 o = o.reshape(b, s, h)
             else:  # ragged prefill / chunked continuation (fp32 reference)
-                kh = cache.k[:, :t].transpose(1, 2)
-                vh = cache.v[:, :t].transpose(1, 2)
+                kd, vd = cache.dequant(t)
+                kh = kd.transpose(1, 2)
+                vh = vd.transpose(1, 2)
                 qh = q.transpose(1, 2)
                 scores = (qh.float() @ kh.float().transpose(-1, -2)) / math.sqrt(hd)
                 mask = torch.arange(t, device=x.device)[None, :] > (

@@ -107,13 +107,14 @@ class Attention(nn.Module):
         q, k = ops.apply_rope_qk(q, k, cos[pos:], sin[pos:], layout="bshd")
         t = cache.append(k, v)
         if s == 1:
-            o = ops.attn_decode(q[:, 0].contiguous(), cache.k, cache.v, t)
+            o = ops.attn_decode(q[:, 0].contiguous(), cache.k, cache.v, t,
+                                cache.k_scale, cache.v_scale)
             return self.wo(o.reshape(b, 1, -1))
         if pos == 0 and (not x.is_cuda or s % 128 == 0):
             o = ops.flash_attention(q, k, v, causal=True, layout="bshd")
             return self.wo(o.reshape(b, s, -1))
         # chunked prefill with a position offset (rare path; fp32 reference)
-        kc, vc = cache.k[:, :t], cache.v[:, :t]
+        kc, vc = cache.dequant(t)
         rep = cfg.n_heads // cfg.n_kv_heads
         qh = q.transpose(1, 2)  # [b, hq, s, d]
         kh = kc.transpose(1, 2).repeat_interleave(rep, dim=1)
@@ -222,7 +223,7 @@ class LlamaForCausalLM(nn.Module):
 @torch.no_grad()
 def _generate(self, input_ids: torch.Tensor, max_new_tokens: int = 32,
               temperature: float = 0.0, top_k: int = 0,
-              seed: int | None = None) -> torch.Tensor:
+              seed: int | None = None, kv_quant: str | None = None) -> torch.Tensor:
     """Autoregressive generation with a per-layer KV cache (inference parity:
     the reference serves inference through the same executor surface)."""
     from .kv_cache import KVCache
@@ -232,7 +233,8 @@ def _generate(self, input_ids: torch.Tensor, max_new_tokens: int = 32,
     max_len = s + max_new_tokens
     dtype = self.embed.weight.dtype
     caches = [KVCache(b, max_len, self.cfg.n_kv_heads, self.cfg.head_dim,
-                      input_ids.device, dtype=dtype) for _ in self.blocks]
+                      input_ids.device, dtype=dtype, quant=kv_quant)
+              for _ in self.blocks]
     gen = None
     if seed is not None:
         gen = torch.Generator(device=input_ids.device).manual_seed(seed)

@@ -95,6 +95,15 @@ torch::Tensor attn_decode(torch::Tensor q, torch::Tensor kcache, torch::Tensor v
                           long t);
 torch::Tensor attn_decode_graph(torch::Tensor q, torch::Tensor kcache,
                                 torch::Tensor vcache, torch::Tensor t_dev);
+torch::Tensor attn_decode_fp8(torch::Tensor q, torch::Tensor kcache,
+                              torch::Tensor vcache, torch::Tensor kscale,
+                              torch::Tensor vscale, long t);
+torch::Tensor attn_decode_fp8_graph(torch::Tensor q, torch::Tensor kcache,
+                                    torch::Tensor vcache, torch::Tensor kscale,
+                                    torch::Tensor vscale, torch::Tensor t_dev);
+void kv_append_fp8_(torch::Tensor k, torch::Tensor v, torch::Tensor k8,
+                    torch::Tensor v8, torch::Tensor kscale, torch::Tensor vscale,
+                    torch::Tensor pos);
 // skinny_gemm.hip (decode GEMV, M <= 8)
 torch::Tensor skinny_gemm(torch::Tensor x, torch::Tensor w);
 
@@ -140,5 +149,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fp8_weight_cast_transpose", &fp8_weight_cast_transpose);
   m.def("attn_decode", &attn_decode);
   m.def("attn_decode_graph", &attn_decode_graph);
+  m.def("attn_decode_fp8", &attn_decode_fp8);
+  m.def("attn_decode_fp8_graph", &attn_decode_fp8_graph);
+  m.def("kv_append_fp8_", &kv_append_fp8_);
   m.def("skinny_gemm", &skinny_gemm);
 }

@@ -27,12 +27,20 @@ namespace {
 
 constexpr int TILE = 64;  // kv rows per staged tile
 
+__device__ __forceinline__ float e4m3_to_f32(unsigned char b) {
+  return __builtin_amdgcn_cvt_f32_fp8((unsigned)b, 0);
+}
+
 // ---- main split kernel -------------------------------------------------
 // partial_o: [nsplits, B, Hq, D] fp32; partial_ml: [nsplits, B, Hq, 2]
-template <int HD>
+// QUANT=false: bf16 cache rows; QUANT=true: OCP e4m3 rows + one fp32
+// scale per (b, t, hkv) row (kscale/vscale) — dequantized during the LDS
+// staging pass, so phases A/B are identical and cache reads halve.
+template <int HD, bool QUANT>
 __global__ __launch_bounds__(256) void attn_decode_kernel(
     const short* __restrict__ qg, const short* __restrict__ kg,
-    const short* __restrict__ vg, float* __restrict__ partial_o,
+    const short* __restrict__ vg, const float* __restrict__ kscale,
+    const float* __restrict__ vscale, float* __restrict__ partial_o,
     float* __restrict__ partial_ml, int B, int Hq, int Hkv, int T_alloc, int t_in,
     const int* __restrict__ t_dev, int nsplits, float scale) {
   // hipGraph mode: the valid length lives in device memory so one captured
@@ -91,8 +99,9 @@ __global__ __launch_bounds__(256) void attn_decode_kernel(
 
   for (int r0 = s0; r0 < s1; r0 += TILE) {
     const int rows = min(TILE, s1 - r0);
-    // ---- cooperative K/V tile staging (8 rows x 512B per instruction) ----
-    {
+    // ---- cooperative K/V tile staging (8 rows x 512B per instruction;
+    //      fp8 mode dequantizes per-row-scaled e4m3 bytes on the way) ----
+    if (!QUANT) {
       constexpr int LPR = HD * 2 / 16;  // 16B loads per row
       for (int c = tid; c < TILE * LPR; c += 256) {
         int rr = c / LPR, off = (c % LPR) * 8;  // off in shorts
@@ -102,6 +111,31 @@ __global__ __launch_bounds__(256) void attn_decode_kernel(
               *reinterpret_cast<const s16x8*>(kg + src);
           *reinterpret_cast<s16x8*>(v_t + rr * PITCH + off) =
               *reinterpret_cast<const s16x8*>(vg + src);
+        }
+      }
+    } else {
+      constexpr int LPR = HD / 8;  // 8-byte loads per row
+      const unsigned char* k8 = reinterpret_cast<const unsigned char*>(kg);
+      const unsigned char* v8 = reinterpret_cast<const unsigned char*>(vg);
+      for (int c = tid; c < TILE * LPR; c += 256) {
+        int rr = c / LPR, off = (c % LPR) * 8;
+        if (rr < rows) {
+          long long srow = ((long long)b * T_alloc + (r0 + rr)) * Hkv + hkv;
+          long long src = srow * HD + off;
+          float ks = kscale[srow];
+          float vs = vscale[srow];
+          uint2 kb = *reinterpret_cast<const uint2*>(k8 + src);
+          uint2 vb = *reinterpret_cast<const uint2*>(v8 + src);
+          const unsigned char* kbb = reinterpret_cast<const unsigned char*>(&kb);
+          const unsigned char* vbb = reinterpret_cast<const unsigned char*>(&vb);
+          s16x8 ko, vo;
+#pragma unroll
+          for (int j = 0; j < 8; ++j) {
+            ko[j] = f2bf(e4m3_to_f32(kbb[j]) * ks);
+            vo[j] = f2bf(e4m3_to_f32(vbb[j]) * vs);
+          }
+          *reinterpret_cast<s16x8*>(k_t + rr * PITCH + off) = ko;
+          *reinterpret_cast<s16x8*>(v_t + rr * PITCH + off) = vo;
         }
       }
     }
@@ -192,9 +226,13 @@ __global__ __launch_bounds__(128) void attn_decode_reduce_kernel(
 // from. Returns o [B, Hq, D] bf16.
 torch::Tensor attn_decode_impl(torch::Tensor q, torch::Tensor kcache,
                                torch::Tensor vcache, long t,
-                               const torch::Tensor* t_dev) {
+                               const torch::Tensor* t_dev,
+                               const torch::Tensor* kscale = nullptr,
+                               const torch::Tensor* vscale = nullptr) {
   TORCH_CHECK(q.is_cuda() && q.dtype() == torch::kBFloat16 && q.dim() == 3);
   TORCH_CHECK(kcache.dim() == 4 && kcache.is_contiguous() && vcache.is_contiguous());
+  const bool quant = kcache.dtype() == torch::kFloat8_e4m3fn;
+  TORCH_CHECK(!quant || (kscale && vscale), "fp8 cache needs row scales");
   const int B = q.size(0), Hq = q.size(1), HD = q.size(2);
   const int T_alloc = kcache.size(1), Hkv = kcache.size(2);
   TORCH_CHECK(kcache.size(0) == B && kcache.size(3) == HD);
@@ -217,12 +255,16 @@ torch::Tensor attn_decode_impl(torch::Tensor q, torch::Tensor kcache,
   float scale = 1.0f / sqrtf((float)HD);
   hipStream_t stream = hypha_stream();
 
-#define DECODE_DISPATCH(HDV)                                                           \
+  const float* ksp = kscale ? kscale->data_ptr<float>() : nullptr;
+  const float* vsp = vscale ? vscale->data_ptr<float>() : nullptr;
+
+#define DECODE_DISPATCH(HDV, QV)                                                       \
   do {                                                                                 \
-    hipLaunchKernelGGL(attn_decode_kernel<HDV>, dim3(nsplits, B* Hkv), dim3(256), 0,   \
-                       stream, (const short*)q.data_ptr(),                             \
+    hipLaunchKernelGGL((attn_decode_kernel<HDV, QV>), dim3(nsplits, B* Hkv),           \
+                       dim3(256), 0, stream, (const short*)q.data_ptr(),               \
                        (const short*)kcache.data_ptr(),                                \
-                       (const short*)vcache.data_ptr(), partial_o.data_ptr<float>(),   \
+                       (const short*)vcache.data_ptr(), ksp, vsp,                      \
+                       partial_o.data_ptr<float>(),                                    \
                        partial_ml.data_ptr<float>(), B, Hq, Hkv, T_alloc, (int)t,      \
                        t_ptr, nsplits, scale);                                         \
     hipLaunchKernelGGL(attn_decode_reduce_kernel<HDV>, dim3(B* Hq), dim3(128), 0,      \
@@ -231,10 +273,14 @@ torch::Tensor attn_decode_impl(torch::Tensor q, torch::Tensor kcache,
                        nsplits);                                                       \
   } while (0)
 
-  if (HD == 128)
-    DECODE_DISPATCH(128);
+  if (HD == 128 && !quant)
+    DECODE_DISPATCH(128, false);
+  else if (HD == 128)
+    DECODE_DISPATCH(128, true);
+  else if (!quant)
+    DECODE_DISPATCH(64, false);
   else
-    DECODE_DISPATCH(64);
+    DECODE_DISPATCH(64, true);
 #undef DECODE_DISPATCH
   return out;
 }
@@ -250,4 +296,87 @@ torch::Tensor attn_decode_graph(torch::Tensor q, torch::Tensor kcache,
                                 torch::Tensor vcache, torch::Tensor t_dev) {
   TORCH_CHECK(t_dev.dtype() == torch::kInt32 && t_dev.is_cuda());
   return attn_decode_impl(q, kcache, vcache, kcache.size(1), &t_dev);
+}
+
+// fp8 (OCP e4m3) KV cache variants: per-row fp32 scales [B, T_alloc, Hkv].
+torch::Tensor attn_decode_fp8(torch::Tensor q, torch::Tensor kcache,
+                              torch::Tensor vcache, torch::Tensor kscale,
+                              torch::Tensor vscale, long t) {
+  return attn_decode_impl(q, kcache, vcache, t, nullptr, &kscale, &vscale);
+}
+
+torch::Tensor attn_decode_fp8_graph(torch::Tensor q, torch::Tensor kcache,
+                                    torch::Tensor vcache, torch::Tensor kscale,
+                                    torch::Tensor vscale, torch::Tensor t_dev) {
+  TORCH_CHECK(t_dev.dtype() == torch::kInt32 && t_dev.is_cuda());
+  return attn_decode_impl(q, kcache, vcache, kcache.size(1), &t_dev, &kscale,
+                          &vscale);
+}
+
+namespace {
+
+// Fused single-token fp8 KV append: one launch quantizes the new k AND v
+// rows (per-row amax -> scale -> e4m3) and writes bytes + scales at the
+// device position index — replaces an ~8-launch python chain per layer
+// per token inside the captured decode graph.
+template <int HD>
+__global__ __launch_bounds__(64) void kv_append_fp8_kernel(
+    const short* __restrict__ kin, const short* __restrict__ vin,
+    unsigned char* __restrict__ k8, unsigned char* __restrict__ v8,
+    float* __restrict__ kscale, float* __restrict__ vscale,
+    const long long* __restrict__ pos, int B, int Hkv, int T_alloc) {
+  const int bh = blockIdx.x;       // (b * Hkv + h)
+  const int which = blockIdx.y;    // 0 = k, 1 = v
+  const int l = threadIdx.x;
+  const short* src = (which ? vin : kin) + (long long)bh * HD;
+  unsigned char* dst8 = which ? v8 : k8;
+  float* dsts = which ? vscale : kscale;
+  const int b = bh / Hkv, h = bh % Hkv;
+  const long long p = *pos;
+  const long long row = ((long long)b * T_alloc + p) * Hkv + h;
+
+  float vals[HD / 64];
+  float mx = 0.f;
+#pragma unroll
+  for (int i = 0; i < HD / 64; ++i) {
+    vals[i] = bf2f(src[l + 64 * i]);
+    mx = fmaxf(mx, fabsf(vals[i]));
+  }
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) mx = fmaxf(mx, __shfl_xor(mx, off, 64));
+  const float scale = fmaxf(mx, 1e-8f) / 448.0f;
+  const float inv = 1.0f / scale;
+  if (l == 0) dsts[row] = scale;
+#pragma unroll
+  for (int i = 0; i < HD / 64; ++i) {
+    float a = fminf(fmaxf(vals[i] * inv, -448.f), 448.f);
+    dst8[row * HD + l + 64 * i] =
+        (unsigned char)(__builtin_amdgcn_cvt_pk_fp8_f32(a, a, 0, false) & 0xff);
+  }
+}
+
+}  // namespace
+
+// k/v [B, 1, Hkv, D] bf16; caches fp8 [B, T_alloc, Hkv, D]; scales
+// [B, T_alloc, Hkv] fp32; pos int64 [1] device tensor (graph-capturable).
+void kv_append_fp8_(torch::Tensor k, torch::Tensor v, torch::Tensor k8,
+                    torch::Tensor v8, torch::Tensor kscale, torch::Tensor vscale,
+                    torch::Tensor pos) {
+  const int B = k8.size(0), T_alloc = k8.size(1), Hkv = k8.size(2), HD = k8.size(3);
+  TORCH_CHECK(k.numel() == (long long)B * Hkv * HD && k.is_contiguous());
+  TORCH_CHECK(pos.dtype() == torch::kInt64 && pos.is_cuda());
+  TORCH_CHECK(HD == 64 || HD == 128);
+  hipStream_t stream = hypha_stream();
+  if (HD == 128)
+    hipLaunchKernelGGL(kv_append_fp8_kernel<128>, dim3(B * Hkv, 2), dim3(64), 0,
+                       stream, (const short*)k.data_ptr(), (const short*)v.data_ptr(),
+                       (unsigned char*)k8.data_ptr(), (unsigned char*)v8.data_ptr(),
+                       kscale.data_ptr<float>(), vscale.data_ptr<float>(),
+                       (const long long*)pos.data_ptr<int64_t>(), B, Hkv, T_alloc);
+  else
+    hipLaunchKernelGGL(kv_append_fp8_kernel<64>, dim3(B * Hkv, 2), dim3(64), 0,
+                       stream, (const short*)k.data_ptr(), (const short*)v.data_ptr(),
+                       (unsigned char*)k8.data_ptr(), (unsigned char*)v8.data_ptr(),
+                       kscale.data_ptr<float>(), vscale.data_ptr<float>(),
+                       (const long long*)pos.data_ptr<int64_t>(), B, Hkv, T_alloc);
 }

@@ -244,19 +244,27 @@ def flash_attention(q, k, v, causal: bool = True, layout: str = "bhsd") -> torch
 # ---------------------------------------------------------------------------
 
 
-def attn_decode(q, kcache, vcache, t: int) -> torch.Tensor:
+def attn_decode(q, kcache, vcache, t: int, kscale=None, vscale=None) -> torch.Tensor:
     """Single-token attention over an appended KV cache.
 
-    q [B, Hq, D] bf16; kcache/vcache [B, T_alloc, Hkv, D] bf16 (bshd);
-    t = valid length including the current token. Returns [B, Hq, D] bf16.
-    Native split-KV kernel (ops/hip/decode.hip); fp32 reference otherwise.
+    q [B, Hq, D] bf16; kcache/vcache [B, T_alloc, Hkv, D] bf16 OR fp8-e4m3
+    with per-row scales (bshd); t = valid length including the current
+    token. Returns [B, Hq, D] bf16. Native split-KV kernel
+    (ops/hip/decode.hip); fp32 reference otherwise.
     """
+    quant = kcache.dtype == torch.float8_e4m3fn
     if use_native(q) and q.shape[-1] in (64, 128) and q.shape[1] <= 8 * kcache.shape[2]:
+        if quant:
+            return _c().attn_decode_fp8(q, kcache, vcache, kscale, vscale, t)
         return _c().attn_decode(q, kcache, vcache, t)
     # reference path (CPU, or head dims the kernel doesn't cover):
     # plain masked softmax over the valid prefix
-    kc = kcache[:, :t].float()
-    vc = vcache[:, :t].float()
+    if quant:
+        kc = kcache[:, :t].float() * kscale[:, :t].unsqueeze(-1).float()
+        vc = vcache[:, :t].float() * vscale[:, :t].unsqueeze(-1).float()
+    else:
+        kc = kcache[:, :t].float()
+        vc = vcache[:, :t].float()
     rep = q.shape[1] // kcache.shape[2]
     kh = kc.permute(0, 2, 1, 3).repeat_interleave(rep, dim=1)
     vh = vc.permute(0, 2, 1, 3).repeat_interleave(rep, dim=1)

@@ -26,7 +26,8 @@ from hypha_amd.models.kv_cache import KVCache
 
 
 class GraphedDecoder:
-    def __init__(self, model, batch: int, prompt_len: int, max_new: int):
+    def __init__(self, model, batch: int, prompt_len: int, max_new: int,
+                 kv_quant: str | None = None):
         from hypha_amd import _C
 
         self._C = _C
@@ -36,9 +37,11 @@ class GraphedDecoder:
         self.dev = dev
         self.batch = batch
         self.max_new = max_new
+        self.kv_quant = kv_quant
         t_alloc = prompt_len + max_new
         self.caches = [KVCache(batch, t_alloc, cfg.n_kv_heads, cfg.head_dim,
-                               dev, dtype=next(model.parameters()).dtype)
+                               dev, dtype=next(model.parameters()).dtype,
+                               quant=kv_quant)
                        for _ in model.blocks]
         self.tok = torch.zeros(batch, 1, dtype=torch.long, device=dev)
         self.pos = torch.zeros(1, dtype=torch.long, device=dev)
@@ -67,10 +70,14 @@ class GraphedDecoder:
             k = lin(n1, a.wk.weight).view(b, 1, cfg.n_kv_heads, cfg.head_dim)
             v = lin(n1, a.wv.weight).view(b, 1, cfg.n_kv_heads, cfg.head_dim)
             q, k = ops.apply_rope_qk(q, k, cos, sin, layout="bshd")
-            cache.k.index_copy_(1, self.pos, k)
-            cache.v.index_copy_(1, self.pos, v)
-            o = self._C.attn_decode_graph(q[:, 0].contiguous(), cache.k,
-                                          cache.v, self.t32)
+            cache.append_at(k, v, self.pos)
+            if self.kv_quant == "fp8":
+                o = self._C.attn_decode_fp8_graph(
+                    q[:, 0].contiguous(), cache.k, cache.v, cache.k_scale,
+                    cache.v_scale, self.t32)
+            else:
+                o = self._C.attn_decode_graph(q[:, 0].contiguous(), cache.k,
+                                              cache.v, self.t32)
             attn_out = lin(o.reshape(b, -1), a.wo.weight).view(b, 1, -1)
             x, n2 = ops.add_rmsnorm(x, attn_out, blk.mlp_norm.weight,
                                     blk.mlp_norm.eps)

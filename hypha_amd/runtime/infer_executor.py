@@ -67,7 +67,8 @@ def main() -> int:
             and model.cfg.n_heads // model.cfg.n_kv_heads <= 8):
         from hypha_amd.runtime.graphed_decode import GraphedDecoder
 
-        graphed = GraphedDecoder(model, batch_size, seq_len, max_new)
+        graphed = GraphedDecoder(model, batch_size, seq_len, max_new,
+                                 kv_quant=cfg.get("kv_cache"))
 
     done_batches = 0
     out_idx = 0

@@ -879,3 +879,49 @@ def test_skinny_gemm_matches_matmul():
         o = _C.skinny_gemm(x, w)
         ref = (x.float() @ w.float().t())
         torch.testing.assert_close(o.float(), ref, rtol=2e-2, atol=K ** 0.5 * 2e-2)
+
+
+def test_fp8_kv_cache_decode_matches_bf16():
+    """fp8 (e4m3 + per-row scales) KV cache: kernel parity vs the bf16
+    cache within quantization tolerance, and greedy generation agreement."""
+    from hypha_amd import _C, models
+    from hypha_amd.models.kv_cache import KVCache
+
+    torch.manual_seed(41)
+    B, Hq, Hkv, D, T, t = 4, 32, 8, 128, 1024, 700
+    q = rand_bf16(B, Hq, D, seed=90)
+    kraw = rand_bf16(B, t, Hkv, D, seed=91)
+    vraw = rand_bf16(B, t, Hkv, D, seed=92)
+    c16 = KVCache(B, T, Hkv, D, DEV)
+    c8 = KVCache(B, T, Hkv, D, DEV, quant="fp8")
+    c16.append(kraw, vraw)
+    c8.append(kraw, vraw)
+    o16 = _C.attn_decode(q, c16.k, c16.v, t)
+    o8 = _C.attn_decode_fp8(q, c8.k, c8.v, c8.k_scale, c8.v_scale, t)
+    rel = (o8.float() - o16.float()).norm() / o16.float().norm()
+    # e4m3 rows carry ~3% RMS quantization noise; on uniform-random KV the
+    # attention output inherits it (~4%). Generation agreement below is the
+    # functional bar.
+    assert float(rel) < 0.06, float(rel)
+
+    # end-to-end greedy generation with fp8 cache tracks the bf16 cache
+    model = models.build("llama-tiny", hidden_size=2048, n_heads=16,
+                         n_kv_heads=4, ffn_hidden=4096, n_layers=2)
+    model = model.to(DEV).bfloat16().eval()
+    for buf in model.buffers():
+        if buf.dtype is torch.bfloat16:
+            buf.data = buf.data.float()
+    ids = torch.randint(0, 500, (2, 128), device=DEV)
+    out16 = model.generate(ids, max_new_tokens=16)
+    out8 = model.generate(ids, max_new_tokens=16, kv_quant="fp8")
+    agree = (out16 == out8).float().mean().item()
+    assert agree >= 0.9, agree
+
+    # graphed decode with fp8 cache
+    from hypha_amd.runtime.graphed_decode import GraphedDecoder
+
+    dec = GraphedDecoder(model, batch=2, prompt_len=128, max_new=24,
+                         kv_quant="fp8")
+    g8 = dec.generate(ids, max_new_tokens=16)
+    agree_g = (g8 == out8).float().mean().item()
+    assert agree_g >= 0.9, agree_g

@@ -31,6 +31,8 @@ def main():
     p.add_argument("--warmup", type=int, default=16)
     p.add_argument("--graph", action="store_true",
                    help="hipGraph-captured decode step (greedy)")
+    p.add_argument("--kv-fp8", action="store_true",
+                   help="fp8 (e4m3 + per-row scales) KV cache")
     args = p.parse_args()
 
     from hypha_amd import models
@@ -48,14 +50,17 @@ def main():
     ids = torch.randint(0, model.cfg.vocab_size - 1, (args.batch, args.prefill),
                         device=dev)
 
+    kvq = "fp8" if args.kv_fp8 else None
     if args.graph:
         from hypha_amd.runtime.graphed_decode import GraphedDecoder
 
         dec = GraphedDecoder(model, args.batch, args.prefill,
-                             args.decode + args.warmup)
+                             args.decode + args.warmup, kv_quant=kvq)
         gen = dec.generate
     else:
-        gen = model.generate
+        import functools
+
+        gen = functools.partial(model.generate, kv_quant=kvq)
 
     # warm (weights/algos/graph capture), then difference two WARM runs so
     # one-time costs never land in the per-token figure
