@@ -494,6 +494,17 @@ struct WorkerDaemon {
     subst("{WORK_DIR}", job->work_dir);
     subst("{JOB_JSON}", job_json);
 
+    // dispatch-span context for the executor's OTLP tracer: job id + the
+    // dispatch wall-clock (ns) so the executor can emit a job.dispatch span
+    // covering queue/spawn latency across the process boundary
+    {
+      struct timespec ts;
+      clock_gettime(CLOCK_REALTIME, &ts);
+      long long ns = (long long)ts.tv_sec * 1000000000LL + ts.tv_nsec;
+      cmd = "HYPHA_JOB_ID=" + job->id + " HYPHA_DISPATCH_TS_NS=" +
+            std::to_string(ns) + " " + cmd;
+    }
+
     // pin one GPU from this daemon's pool to the job for its whole lease
     // lifetime: the executor sees exactly one device (cuda:0) regardless of
     // which physical GPU it landed on

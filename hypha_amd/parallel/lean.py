@@ -65,10 +65,17 @@ class LeanDiLoCoWorker:
                         "fp8 weight storage needs QBLOCK-aligned shapes")
                     self._fp8_mods.append(m)
 
-        # flatten bf16 params (the single weight copy)
+        # flatten bf16 params (the single weight copy). Expert-parallel
+        # shards (`_ep_local`, models/moe.py shard_experts_) go LAST: the
+        # outer all-reduce and init broadcast cover only [0, sync_numel) —
+        # expert singletons are never averaged (see diloco.FlatParams).
         fp8_param_ids = {id(m.weight) for m in self._fp8_mods}
-        self.params = [p for p in model.parameters()
-                       if p.requires_grad and id(p) not in fp8_param_ids]
+        all_params = [p for p in model.parameters()
+                      if p.requires_grad and id(p) not in fp8_param_ids]
+        shared = [p for p in all_params if not getattr(p, "_ep_local", False)]
+        ep_tail = [p for p in all_params if getattr(p, "_ep_local", False)]
+        self.params = shared + ep_tail
+        self.sync_numel = sum(p.numel() for p in shared)
         self.numel = sum(p.numel() for p in self.params)
         self.flat = torch.empty(self.numel, dtype=torch.bfloat16, device=self.device)
         offset = 0
@@ -81,7 +88,7 @@ class LeanDiLoCoWorker:
             offset += n
 
         if self.comm.is_distributed:
-            self.comm.broadcast_flat(self.flat, src=0)
+            self.comm.broadcast_flat(self.flat[: self.sync_numel], src=0)
 
         # per-param quant-state slices, padded to the kernel's 2048 block
         self._state_off = []
@@ -223,21 +230,29 @@ class LeanDiLoCoWorker:
         """Chunked: stream host theta0/momentum through the staging window,
         delta = theta_t - theta0, all-reduce, Nesterov, write back."""
         n = self.numel
+        n_sync = self.sync_numel  # EP expert tail is rank-local
         seed = (self.round + 1) * 40503 % (2**31)
         for start in range(0, n, self._chunk):
             m = min(self._chunk, n - start)
-            t0 = self._stage_t0[:m]
-            mo = self._stage_m[:m]
-            d = self._stage_d[:m]
-            t0.copy_(self.theta0_host[start:start + m], non_blocking=False)
-            mo.copy_(self.outer_m_host[start:start + m], non_blocking=False)
-            self._C.extract_delta_bf16(self.flat[start:start + m], t0, d)
-            self.comm.all_reduce_mean_flat(d)
-            self._C.nesterov_bf16_(t0, d, mo, self.cfg.outer.lr,
-                                   self.cfg.outer.momentum, seed)
-            self.theta0_host[start:start + m].copy_(t0)
-            self.outer_m_host[start:start + m].copy_(mo)
-            self.flat[start:start + m].copy_(t0)
+            ms = max(0, min(m, n_sync - start))  # shared portion
+            if ms > 0:
+                t0 = self._stage_t0[:ms]
+                mo = self._stage_m[:ms]
+                d = self._stage_d[:ms]
+                t0.copy_(self.theta0_host[start:start + ms], non_blocking=False)
+                mo.copy_(self.outer_m_host[start:start + ms], non_blocking=False)
+                self._C.extract_delta_bf16(self.flat[start:start + ms], t0, d)
+                self.comm.all_reduce_mean_flat(d)
+                self._C.nesterov_bf16_(t0, d, mo, self.cfg.outer.lr,
+                                       self.cfg.outer.momentum, seed)
+                self.theta0_host[start:start + ms].copy_(t0)
+                self.outer_m_host[start:start + ms].copy_(mo)
+                self.flat[start:start + ms].copy_(t0)
+            if ms < m:
+                # expert-parallel tail: purely inner-trained singletons pass
+                # through the round (no averaging, no outer momentum)
+                self.theta0_host[start + ms:start + m].copy_(
+                    self.flat[start + ms:start + m])
         # fp8 weight-storage group: dequant-delta, all-reduce, Nesterov on
         # the bf16 global weights, SR-requant back into storage
         if self.fp8_numel:
@@ -263,7 +278,7 @@ class LeanDiLoCoWorker:
                                      self.flat_wscale[b0:b1], rseed ^ 0x5bd1e995)
             self.outer_sync_payload_bytes += n8 * 2
 
-        self.outer_sync_payload_bytes += n * 2
+        self.outer_sync_payload_bytes += n_sync * 2
         self.round += 1
         self.steps_in_round = 0
 

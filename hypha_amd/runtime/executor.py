@@ -111,7 +111,15 @@ def run_rccl(worker, comm, session, batch_iter, reform_q, rccl, batch_size,
     from hypha_amd.telemetry import get_tracer
 
     tracer = get_tracer()
-    job_span = tracer.start_span("job.execute", job_id=os.environ.get("HYPHA_JOB_ID", ""),
+    job_id = os.environ.get("HYPHA_JOB_ID", "")
+    # dispatch span: covers worker-side spawn -> executor ready (start
+    # timestamp injected by the worker daemon across the process boundary)
+    disp_ns = os.environ.get("HYPHA_DISPATCH_TS_NS")
+    if disp_ns:
+        dspan = tracer.start_span("job.dispatch", job_id=job_id, rank=comm.rank)
+        dspan.start_ns = int(disp_ns)
+        dspan.end()
+    job_span = tracer.start_span("job.execute", job_id=job_id,
                                  rank=comm.rank, world_size=comm.world_size)
     done = False
     while not done:
