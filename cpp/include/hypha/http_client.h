@@ -170,7 +170,8 @@ class HttpConn {
 // allow-list to the url host AND every redirect hop. Returns bytes written.
 inline long long http_get_to_file(const std::string& url, const std::string& out_path,
                                   const std::vector<std::string>& allow,
-                                  double timeout_s = 30.0, int max_redirects = 5) {
+                                  double timeout_s = 30.0, int max_redirects = 5,
+                                  long long max_bytes = 16LL << 30) {
   std::string cur = url;
   for (int hop = 0; hop <= max_redirects; ++hop) {
     HttpUrl u = parse_url(cur);
@@ -243,7 +244,16 @@ inline long long http_get_to_file(const std::string& url, const std::string& out
         throw std::runtime_error("fetch: short write to " + out_path);
       }
       written += (long long)n;
+      if (written > max_bytes) {
+        fclose(f);
+        throw std::runtime_error("fetch: response exceeds size cap for " +
+                                 out_path);
+      }
     };
+    if (content_len > max_bytes) {
+      fclose(f);
+      throw std::runtime_error("fetch: declared size exceeds cap for " + cur);
+    }
     write_chunk(body0.data(), body0.size());
     while (content_len < 0 || written < content_len) {
       ssize_t n = conn.read_some(buf, sizeof(buf));
