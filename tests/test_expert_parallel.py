@@ -124,3 +124,51 @@ def test_ep_diloco_syncs_shared_only():
     assert res[0][4] == 2 and res[1][4] == 2
     # payload accounting covers only the shared prefix
     assert res[0][5] == res[1][5] and res[0][5] > 0
+
+
+def _ep_ckpt_worker(rank, port, q):
+    """EP + activation checkpointing: the recompute pass re-runs the token
+    exchange; every rank follows the identical segment schedule so the
+    collectives stay matched."""
+    os.environ.update(RANK=str(rank), WORLD_SIZE="2", LOCAL_RANK=str(rank),
+                      MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port))
+    try:
+        from hypha_amd import models
+        from hypha_amd.data.synthetic import SyntheticTokens
+        from hypha_amd.models.moe import shard_experts_
+        from hypha_amd.parallel import Comm, DiLoCoConfig, DiLoCoWorker, InnerOptConfig
+
+        comm = Comm(backend="gloo")
+        torch.manual_seed(80 + rank)
+        model = models.build("moe-tiny", gradient_checkpointing=True)
+        shard_experts_(model, rank, 2)
+        w = DiLoCoWorker(model, DiLoCoConfig(h=2, inner=InnerOptConfig(
+            lr=1e-3, warmup_steps=0, schedule="constant")),
+            comm=comm, device=torch.device("cpu"))
+        data = SyntheticTokens(512, 32, 2, seed=90, rank=rank)
+        losses = []
+        for _ in range(4):
+            ids, labels = data.next_batch()
+            losses.append(w.train_step(ids, labels))
+            w.maybe_outer_sync()
+        q.put(("ok", rank, w.round, all(l == l for l in losses)))
+        comm.shutdown()
+    except Exception:
+        import traceback
+
+        q.put(("err", rank, traceback.format_exc(), False))
+
+
+@pytest.mark.timeout(180)
+def test_ep_with_activation_checkpointing():
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    procs = [ctx.Process(target=_ep_ckpt_worker, args=(r, 29791, q)) for r in range(2)]
+    for p in procs:
+        p.start()
+    res = [q.get() for _ in range(2)]
+    for p in procs:
+        p.join(timeout=150)
+    for r in res:
+        assert r[0] == "ok", r
+        assert r[2] == 2 and r[3]
