@@ -96,7 +96,10 @@ int main(int argc, char** argv) {
            "# job config (scheduler_config.rs analogue); JSON or TOML:\n"
            "{\n"
            "  \"model\": \"llama3-8b\",\n"
-           "  \"dataset\": \"synth\",\n"
+           "  \"dataset\": \"synth\",            # OR \"data_uri\": \"http://host/slice.safetensors\"\n"
+           "  # \"preprocessor\": {\"task\": \"tokenizer|feature|image|video|auto\",\n"
+           "  #   \"artifact\": {\"uri\": {\"value\": \"http://...\"}},\n"
+           "  #   \"input_names\": [\"text\"], \"output_key\": \"input_ids\"},\n"
            "  \"num_workers\": 2,\n"
            "  \"update_rounds\": 100,\n"
            "  \"avg_samples_between_updates\": 1200,\n"

@@ -1,5 +1,6 @@
 """Convergence evidence: the fused-kernel DiLoCo engine vs plain torch
-AdamW on the SAME model/data (gpt2-small llama-preset, synthetic corpus).
+AdamW on the SAME model/data (default: the GPT-2-small family on a
+synthetic corpus).
 Shows the whole native stack (HIP kernels + flat-param AdamW + outer
 Nesterov) follows the reference-optimizer trajectory, not just per-kernel
 oracles. Prints loss every `--log-every` steps for both runs."""
