@@ -268,6 +268,8 @@ def shard_experts_(model: nn.Module, rank: int, size: int, group=None) -> int:
     broadcast. Returns the number of sharded MoEMLP modules."""
     import torch.distributed as dist
 
+    if size <= 1:
+        return 0  # single rank: the dense path IS expert-parallel degree 1
     if group is None:
         group = dist.distributed_c10d._get_default_group()
     n = 0
