@@ -295,3 +295,24 @@ def test_relay_unregistered_peer_fails(cluster):
     a.kv_put("addr:ghost", "127.0.0.1:1")
     with pytest.raises(RuntimeError):
         a.request("ghost", "echo", {}, 2.0)
+
+
+def test_relay_concurrent_circuits(cluster):
+    """Several relayed requests in flight at once (one circuit each) and a
+    mix of relayed + direct traffic."""
+    import concurrent.futures
+
+    a = cluster("alice")
+    b = cluster("bob")
+    c = cluster("carol")
+    b.on("work", lambda frm, body: {"v": body["v"] * 2})
+    c.on("work", lambda frm, body: {"v": body["v"] + 100})
+    a.kv_put("addr:bob", "127.0.0.1:1")  # bob only reachable via relay
+
+    def call(i):
+        if i % 2 == 0:
+            return a.request("bob", "work", {"v": i})["v"] == 2 * i
+        return a.request("carol", "work", {"v": i})["v"] == i + 100
+
+    with concurrent.futures.ThreadPoolExecutor(max_workers=6) as ex:
+        assert all(ex.map(call, range(12)))
