@@ -161,3 +161,32 @@ def test_elastic_reform_after_worker_death():
     # both survivors completed 2 rounds and hold identical global weights
     assert ok[0][3] == 2 and ok[1][3] == 2
     assert ok[0][2] == pytest.approx(ok[1][2], rel=1e-6)
+
+
+@pytest.mark.timeout(300)
+def test_bench_expert_parallel_world2_gloo():
+    """bench.py --expert-parallel under torch.distributed.run (gloo, 2
+    ranks): experts shard, tokens exchange, the JSON contract holds and
+    parallelism reports ep2+diloco."""
+    import json
+    import subprocess
+    import sys
+    from pathlib import Path
+
+    repo = Path(__file__).resolve().parent.parent
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29617", str(repo / "bench.py"), "--gpus", "2",
+         "--steps", "3", "--warmup", "1", "--model", "moe-tiny",
+         "--batch", "2", "--seq-len", "128", "--h", "2",
+         "--expert-parallel"],
+        cwd=repo, capture_output=True, text=True, timeout=280,
+    )
+    assert r.returncode == 0, r.stderr[-2000:]
+    json_lines = [l for l in r.stdout.splitlines() if l.startswith("{")]
+    assert len(json_lines) == 1, r.stdout
+    out = json.loads(json_lines[0])
+    assert out["config"]["parallelism"] == "ep2+diloco"
+    assert out["value"] > 0
+    assert out["config"]["outer_syncs_in_timed_window"] >= 1
