@@ -22,7 +22,7 @@ int main(int argc, char** argv) {
   std::string name = "data", gw_host = "127.0.0.1", dataset = "dataset", dir = ".";
   int gw_port = 0, port = 0;
   bool probe = false, init = false;
-  std::vector<std::string> exclude_cidrs;
+  std::vector<std::string> exclude_cidrs, fallback_gws;
   TlsConfig tls;
   for (int i = 1; i < argc; ++i) {
     std::string a = argv[i];
@@ -38,6 +38,7 @@ int main(int argc, char** argv) {
     else if (a == "--tls-ca") tls.ca_path = next();
     else if (a == "--tls-crl") tls.crl_path = next();
     else if (a == "--exclude-cidr") exclude_cidrs.push_back(next());
+    else if (a == "--fallback-gateway") fallback_gws.push_back(next());
     else if (a == "probe") probe = true;
     else if (a == "init") init = true;
   }
@@ -48,7 +49,8 @@ int main(int argc, char** argv) {
            "# --gateway-host/--gateway-port gateway broker address\n"
            "# --dataset synth               dataset name to announce\n"
            "# --dataset-path /data/slices   dir of SafeTensors slice files\n"
-           "# --tls-cert/--tls-key/--tls-ca [--tls-crl]  mTLS identity\n");
+           "# --tls-cert/--tls-key/--tls-ca [--tls-crl]  mTLS identity\n"
+           "# --fallback-gateway host:port  additional broker(s) tried on gateway loss\n");
     return 0;
   }
   if (probe) {  // readiness = gateway reachable (hypha-data.rs probe analogue)
@@ -99,11 +101,19 @@ int main(int argc, char** argv) {
     }
     fclose(f);
   });
-  node.start(port);
-  // announce the dataset record (kad put, hypha-data.rs:176-185)
+  for (const auto& g : fallback_gws) {
+    auto c = g.rfind(':');
+    node.add_fallback_gateway(g.substr(0, c), std::stoi(g.substr(c + 1)));
+  }
+  // the dataset record does not replicate between gateways: re-announce on
+  // every (re)connection, incl. failover to a fallback broker
   Json rec;
   rec["num_slices"] = (int64_t)files.size();
   rec["provider"] = name;
+  node.on_gateway_reconnect([&node, rec, dataset] {
+    node.kv_put("dataset:" + dataset, rec);
+  });
+  node.start(port);
   node.kv_put("dataset:" + dataset, rec);
   printf("hypha-data %s serving %zu slices of %s on port %d\n", name.c_str(),
          files.size(), dataset.c_str(), node.port());

@@ -67,7 +67,7 @@ int main(int argc, char** argv) {
   std::string name = "scheduler", gw_host = "127.0.0.1", config_path, status_bridge;
   int gw_port = 0, port = 0;
   bool probe = false, init = false;
-  std::vector<std::string> exclude_cidrs;
+  std::vector<std::string> exclude_cidrs, fallback_gws;
   TlsConfig tls;
   for (int i = 1; i < argc; ++i) {
     std::string a = argv[i];
@@ -83,6 +83,7 @@ int main(int argc, char** argv) {
     else if (a == "--tls-ca") tls.ca_path = next();
     else if (a == "--tls-crl") tls.crl_path = next();
     else if (a == "--exclude-cidr") exclude_cidrs.push_back(next());
+    else if (a == "--fallback-gateway") fallback_gws.push_back(next());
     else if (a == "probe") probe = true;
     else if (a == "init") init = true;
   }
@@ -164,6 +165,10 @@ int main(int argc, char** argv) {
 
   Node node(name, gw_host, gw_port, tls);
   node.set_exclude_cidrs(exclude_cidrs);
+  for (const auto& g : fallback_gws) {
+    auto c = g.rfind(':');
+    node.add_fallback_gateway(g.substr(0, c), std::stoi(g.substr(c + 1)));
+  }
 
   // --- offer collection state (allocator.rs) ---
   std::mutex offer_mu;

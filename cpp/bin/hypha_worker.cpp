@@ -885,7 +885,7 @@ int main(int argc, char** argv) {
   int gw_port = 0, port = 0;
   bool probe = false, init = false;
   std::vector<int> gpu_ids;
-  std::vector<std::string> exclude_cidrs, fetch_allow;
+  std::vector<std::string> exclude_cidrs, fetch_allow, fallback_gws;
   std::string hf_endpoint = "https://huggingface.co";
   TlsConfig tls;
   Resources total{1, 4, 16, 100};
@@ -933,6 +933,7 @@ int main(int argc, char** argv) {
     else if (a == "--tls-ca") tls.ca_path = next();
     else if (a == "--tls-crl") tls.crl_path = next();
     else if (a == "--exclude-cidr") exclude_cidrs.push_back(next());
+    else if (a == "--fallback-gateway") fallback_gws.push_back(next());
     else if (a == "--fetch-allow") fetch_allow.push_back(next());
     else if (a == "--hf-endpoint") hf_endpoint = next();
     else if (a == "probe") probe = true;
@@ -955,7 +956,8 @@ int main(int argc, char** argv) {
            "# --tls-cert/--tls-key/--tls-ca [--tls-crl]  mTLS identity\n"
            "# --fetch-allow host[:port]     allow URI/HF fetches from this host\n"
            "#                               (repeatable; *.domain wildcards; default: deny all)\n"
-           "# --hf-endpoint https://huggingface.co  HuggingFace resolve endpoint\n");
+           "# --hf-endpoint https://huggingface.co  HuggingFace resolve endpoint\n"
+           "# --fallback-gateway host:port  additional broker(s) tried on gateway loss\n");
     return 0;
   }
   if (probe) {
@@ -978,6 +980,10 @@ int main(int argc, char** argv) {
   daemon.gpu_pool = gpu_ids;
   daemon.infer_cmd = icmd;
   daemon.fetch_allow = fetch_allow;
+  for (const auto& g : fallback_gws) {
+    auto c = g.rfind(':');
+    daemon.node.add_fallback_gateway(g.substr(0, c), std::stoi(g.substr(c + 1)));
+  }
   daemon.hf_endpoint = hf_endpoint;
   daemon.node.set_exclude_cidrs(exclude_cidrs);
   daemon.start(port);

@@ -495,6 +495,20 @@ PYBIND11_MODULE(_core, m) {
            },
            py::arg("peer"), py::arg("type"), py::arg("header"),
            py::arg("payload") = py::none())
+      .def("add_fallback_gateway", &Node::add_fallback_gateway)
+      .def("on_gateway_reconnect",
+           [](Node& n, py::function cb) {
+             auto cbp = make_cb_holder(std::move(cb));
+             n.on_gateway_reconnect([cbp] {
+               py::gil_scoped_acquire gil;
+               try {
+                 (*cbp)();
+               } catch (py::error_already_set& e) {
+                 e.restore();
+                 PyErr_Clear();
+               }
+             });
+           })
       .def("on_blob",
            [](Node& n, const std::string& type, py::function cb) {
              auto cbp = make_cb_holder(std::move(cb));

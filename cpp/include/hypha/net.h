@@ -135,6 +135,20 @@ class Node {
   void accept_loop();
   void handle_conn(int fd, SSL* ssl = nullptr);
   void gateway_listen_loop();
+
+ public:
+  // Gateway high availability: additional brokers tried in order when the
+  // active one is unreachable (each failover re-registers and replays
+  // subscriptions; daemons use on_gateway_reconnect to re-announce their
+  // KV records — gateways do not replicate state).
+  void add_fallback_gateway(const std::string& host, int port) {
+    gw_list_.emplace_back(host, port);
+  }
+  void on_gateway_reconnect(std::function<void()> cb) {
+    reconnect_cb_ = std::move(cb);
+  }
+
+ private:
   // Relay fallback (reference gateway relay-server role, network.rs:44):
   // when a peer cannot be dialed directly, the connection is tunneled
   // through the gateway as an opaque byte circuit; with mTLS the
@@ -147,6 +161,12 @@ class Node {
 
   std::string name_, gw_host_;
   int gw_port_;
+  // failover gateway candidates (primary + fallbacks); gw_host_/gw_port_
+  // is the ACTIVE one, switched by gateway_connect on unreachability
+  std::string gw_primary_host_;
+  int gw_primary_port_ = 0;
+  std::vector<std::pair<std::string, int>> gw_list_;
+  std::function<void()> reconnect_cb_;  // fired after re-register+replay
   std::unique_ptr<TlsContext> tls_;
   int listen_fd_ = -1;
   int port_ = 0;

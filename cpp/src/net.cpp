@@ -229,6 +229,8 @@ int listen_port(int listen_fd) {
 
 Node::Node(std::string name, std::string gateway_host, int gateway_port, TlsConfig tls)
     : name_(std::move(name)), gw_host_(std::move(gateway_host)), gw_port_(gateway_port) {
+  gw_primary_host_ = gw_host_;
+  gw_primary_port_ = gw_port_;
   if (tls.enabled()) tls_ = std::make_unique<TlsContext>(tls);
 }
 
@@ -252,7 +254,24 @@ void Node::start(int port) {
 // Establish (or re-establish) the persistent broker connection:
 // connect + optional TLS + register + synchronous ack.
 bool Node::gateway_connect() {
+  // try the active gateway, then every other candidate (primary first)
   int fd = tcp_connect(gw_host_, gw_port_, 10.0);
+  if (fd < 0 && !gw_list_.empty()) {
+    std::vector<std::pair<std::string, int>> cands;
+    cands.emplace_back(gw_primary_host_, gw_primary_port_);
+    cands.insert(cands.end(), gw_list_.begin(), gw_list_.end());
+    for (const auto& [h, p] : cands) {
+      if (h == gw_host_ && p == gw_port_) continue;
+      fd = tcp_connect(h, p, 5.0);
+      if (fd >= 0) {
+        fprintf(stderr, "[net:%s] gateway failover -> %s:%d\n", name_.c_str(),
+                h.c_str(), p);
+        gw_host_ = h;
+        gw_port_ = p;
+        break;
+      }
+    }
+  }
   if (fd < 0) return false;
   // long receive timeout on the event connection
   struct timeval tv = {86400, 0};
@@ -284,6 +303,14 @@ bool Node::gateway_connect() {
       sub["kind"] = "subscribe";
       sub["topic"] = topic;
       gw_sock_->send_json(sub);
+    }
+  }
+  if (reconnect_cb_) {
+    try {
+      reconnect_cb_();  // daemons re-announce their KV records
+    } catch (const std::exception& e) {
+      fprintf(stderr, "[net:%s] reconnect callback error: %s\n", name_.c_str(),
+              e.what());
     }
   }
   return true;

@@ -316,3 +316,59 @@ def test_relay_concurrent_circuits(cluster):
 
     with concurrent.futures.ThreadPoolExecutor(max_workers=6) as ex:
         assert all(ex.map(call, range(12)))
+
+
+def test_gateway_failover_and_reannounce():
+    """Gateway high availability: a node with a fallback broker re-registers
+    there when the primary dies, replays subscriptions, and fires the
+    reconnect callback so daemons re-announce their KV records (brokers do
+    not replicate state)."""
+    gw1 = core.Gateway()
+    gw1.start(0)
+    gw2 = core.Gateway()
+    gw2.start(0)
+    a = core.Node("alice", "127.0.0.1", gw1.port)
+    b = core.Node("bob", "127.0.0.1", gw1.port)
+    try:
+        for n in (a, b):
+            n.add_fallback_gateway("127.0.0.1", gw2.port)
+        announced = []
+
+        def reannounce():
+            announced.append(1)
+            b.kv_put("dataset:ha", {"num_slices": 3, "provider": "bob"})
+
+        b.on_gateway_reconnect(reannounce)
+        a.start(0)
+        b.start(0)
+        b.kv_put("dataset:ha", {"num_slices": 3, "provider": "bob"})
+        b.on("echo", lambda frm, body: {"x": body["x"] + 1})
+        got = []
+        a.subscribe("t", lambda frm, d: got.append(d))
+        time.sleep(0.1)
+
+        gw1.stop()  # primary dies
+        deadline = time.time() + 15
+        rec = None
+        while time.time() < deadline:
+            try:
+                rec = a.kv_get("dataset:ha")
+                if rec:
+                    break
+            except RuntimeError:
+                pass
+            time.sleep(0.3)
+        assert rec and rec["num_slices"] == 3, rec  # re-announced on gw2
+        assert announced, "reconnect callback never fired"
+        # direct RR still works (registry on gw2 has bob's addr)
+        assert a.request("bob", "echo", {"x": 1}) == {"x": 2}
+        # pub/sub replayed onto the fallback broker
+        b.publish("t", {"v": 5})
+        deadline = time.time() + 5
+        while not got and time.time() < deadline:
+            time.sleep(0.1)
+        assert got == [{"v": 5}]
+    finally:
+        a.stop()
+        b.stop()
+        gw2.stop()
