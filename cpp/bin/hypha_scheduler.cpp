@@ -114,24 +114,24 @@ int main(int argc, char** argv) {
            "}\n");
     return 0;
   }
-  if (probe) {  // readiness check: gateway health RR (hypha-worker.rs probe)
-    int fd = tcp_connect(gw_host, gw_port, 3.0);
-    if (fd < 0) {
-      fprintf(stderr, "probe: gateway unreachable\n");
+  if (probe) {  // readiness = mTLS registration round-trip (worker.rs probe)
+    try {
+      Node pn(name, gw_host, gw_port, tls);  // cert CN must equal `name`
+      for (const auto& g : fallback_gws) {
+        auto c = g.rfind(':');
+        pn.add_fallback_gateway(g.substr(0, c), std::stoi(g.substr(c + 1)));
+      }
+      // health RR over the (m)TLS transport WITHOUT registering (a probe
+      // must never displace a live daemon's registry entry)
+      Json r = pn.gateway_request("health", Json(JsonObject{}));
+      if (!r.get_or("healthy", Json(false)).as_bool())
+        throw std::runtime_error("gateway unhealthy");
+    } catch (const std::exception& e) {
+      fprintf(stderr, "probe: %s\n", e.what());
       return 1;
     }
-    MsgSocket sock(fd);
-    Json req;
-    req["kind"] = "request";
-    req["type"] = "health";
-    req["from"] = name;
-    req["body"] = Json(JsonObject{});
-    sock.send_json(req);
-    auto resp = sock.recv_json();
-    bool ok = resp && resp->get_or("body", Json(JsonObject{}))
-                          .get_or("healthy", Json(false)).as_bool();
-    printf("probe: %s\n", ok ? "healthy" : "unhealthy");
-    return ok ? 0 : 1;
+    printf("probe: healthy\n");
+    return 0;
   }
 
   // job config (scheduler_config.rs analogue; JSON instead of TOML)

@@ -961,12 +961,24 @@ int main(int argc, char** argv) {
     return 0;
   }
   if (probe) {
-    int fd = tcp_connect(gw_host, gw_port, 3.0);
-    if (fd < 0) {
-      fprintf(stderr, "probe: gateway unreachable\n");
+    // readiness = a full mTLS registration round-trip (register + ack),
+    // matching the reference's health RR over the authenticated transport
+    // (hypha-worker.rs:312-354) rather than a bare TCP connect
+    try {
+      Node pn(name, gw_host, gw_port, tls);  // cert CN must equal `name`
+      for (const auto& g : fallback_gws) {
+        auto c = g.rfind(':');
+        pn.add_fallback_gateway(g.substr(0, c), std::stoi(g.substr(c + 1)));
+      }
+      // health RR over the (m)TLS transport WITHOUT registering (a probe
+      // must never displace a live daemon's registry entry)
+      Json r = pn.gateway_request("health", Json(JsonObject{}));
+      if (!r.get_or("healthy", Json(false)).as_bool())
+        throw std::runtime_error("gateway unhealthy");
+    } catch (const std::exception& e) {
+      fprintf(stderr, "probe: %s\n", e.what());
       return 1;
     }
-    ::close(fd);
     printf("probe: healthy\n");
     return 0;
   }

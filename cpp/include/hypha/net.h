@@ -147,6 +147,9 @@ class Node {
   void on_gateway_reconnect(std::function<void()> cb) {
     reconnect_cb_ = std::move(cb);
   }
+  // One-shot RR to the active gateway over the (m)TLS transport — used by
+  // the daemons' `probe` subcommand (no registration side effects).
+  Json gateway_request(const std::string& type, const Json& body);
 
  private:
   // Relay fallback (reference gateway relay-server role, network.rs:44):
@@ -157,7 +160,6 @@ class Node {
   int relay_dial(const std::string& peer, double timeout_s);
   void relay_accept_run(long long circuit);
   bool gateway_connect();  // (re)connect + register + replay subscriptions
-  Json gateway_request(const std::string& type, const Json& body);
 
   std::string name_, gw_host_;
   int gw_port_;
