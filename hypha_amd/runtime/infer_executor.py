@@ -11,7 +11,10 @@ so the scheduler's FSM and lease machinery apply unchanged.
 
 Job config: {"model", "data": <fetch ref>, "results": <send ref>?,
              "max_new_tokens", "batch_size", "seq_len", "temperature",
-             "top_k", "num_batches"}
+             "top_k", "num_batches", "kv_cache": "fp8"?}
+
+Greedy Llama-family jobs on GPU decode through the hipGraph-captured step
+(runtime/graphed_decode.py); "kv_cache": "fp8" halves KV memory.
 """
 
 from __future__ import annotations
