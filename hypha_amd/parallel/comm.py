@@ -107,6 +107,35 @@ class Comm:
         dist.all_reduce(t, op=dist.ReduceOp.MAX if op == "max" else dist.ReduceOp.SUM)
         return float(t.item())
 
+    def weighted_all_reduce_flat(self, flat: torch.Tensor, weight: float,
+                                 bucket_bytes: int = DEFAULT_BUCKET_BYTES) -> float:
+        """Sample-weighted aggregate: flat <- sum_i(w_i * flat_i) / sum_i(w_i).
+
+        The reference's PS aggregation weights worker updates by their
+        sample counts (the weighting TODO closed on the PS path in round 1);
+        this is the same semantics on the RCCL data plane — a rank that
+        processed fewer batches this round (heterogeneous FSM schedules,
+        mid-round joiners with 0) contributes proportionally less. Returns
+        the total weight. With equal weights this equals the plain mean.
+        """
+        total = self.all_reduce_scalar(float(weight), op="sum")
+        if not self.is_distributed:
+            return total
+        if total <= 0:
+            total = float(self.world_size)  # degenerate: plain mean of zeros
+        self.syncs += 1
+        self.bytes_sent_payload += flat.numel() * flat.element_size()
+        flat.mul_(weight / total * self.world_size)  # undo mean's /world
+        n_per_bucket = max(1, bucket_bytes // flat.element_size())
+        handles = []
+        for start in range(0, flat.numel(), n_per_bucket):
+            chunk = flat.narrow(0, start, min(n_per_bucket, flat.numel() - start))
+            handles.append(dist.all_reduce(chunk, op=dist.ReduceOp.SUM, async_op=True))
+        for h in handles:
+            h.wait()
+        flat.div_(self.world_size)
+        return total
+
     def wire_bytes_per_rank(self, payload_bytes: int) -> int:
         """Ring all-reduce on-wire bytes per rank for a given payload."""
         w = self.world_size

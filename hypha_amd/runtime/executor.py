@@ -157,7 +157,10 @@ def run_rccl(worker, comm, session, batch_iter, reform_q, rccl, batch_size,
         for attempt in range(4):
             ops.interface.extract_delta(fp.master, fp.theta0, delta)
             try:
-                comm.all_reduce_mean_flat(delta)
+                # sample-weighted aggregation (PS-path parity on RCCL): a
+                # rank that ran fewer batches this round — heterogeneous
+                # FSM schedules, a joiner with 0 — counts proportionally
+                comm.weighted_all_reduce_flat(delta, float(round_samples))
                 break
             except Exception as e:
                 # a member died mid-round: the collective timed out. Wait for

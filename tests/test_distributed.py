@@ -190,3 +190,47 @@ def test_bench_expert_parallel_world2_gloo():
     assert out["config"]["parallelism"] == "ep2+diloco"
     assert out["value"] > 0
     assert out["config"]["outer_syncs_in_timed_window"] >= 1
+
+
+def _weighted_worker(rank, port, q):
+    os.environ.update(RANK=str(rank), WORLD_SIZE="2", LOCAL_RANK=str(rank),
+                      MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port))
+    try:
+        from hypha_amd.parallel import Comm
+
+        comm = Comm(backend="gloo")
+        # rank 0 processed 3 samples with delta=1; rank 1 processed 1 with
+        # delta=5 -> weighted mean = (3*1 + 1*5)/4 = 2.0
+        d = torch.full((1000,), 1.0 if rank == 0 else 5.0)
+        w = 3.0 if rank == 0 else 1.0
+        total = comm.weighted_all_reduce_flat(d, w)
+        # zero-weight edge: both ranks weight 0 -> plain mean, no nan
+        z = torch.full((10,), float(rank))
+        comm.weighted_all_reduce_flat(z, 0.0)
+        q.put(("ok", rank, float(d[0]), float(d[-1]), total,
+               bool(torch.isfinite(z).all())))
+        comm.shutdown()
+    except Exception:
+        import traceback
+
+        q.put(("err", rank, traceback.format_exc(), 0, 0, False))
+
+
+@pytest.mark.timeout(120)
+def test_weighted_all_reduce_flat():
+    """Sample-weighted outer aggregation (PS weighting semantics on the
+    RCCL path): sum(w_i d_i)/sum(w_i), zero-weight degenerates to mean."""
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    procs = [ctx.Process(target=_weighted_worker, args=(r, 29821, q)) for r in range(2)]
+    for p in procs:
+        p.start()
+    res = [q.get() for _ in range(2)]
+    for p in procs:
+        p.join(timeout=90)
+    for r in res:
+        assert r[0] == "ok", r
+        assert r[2] == pytest.approx(2.0, rel=1e-5)
+        assert r[3] == pytest.approx(2.0, rel=1e-5)
+        assert r[4] == pytest.approx(4.0)
+        assert r[5]
