@@ -111,15 +111,8 @@ def run_rccl(worker, comm, session, batch_iter, reform_q, rccl, batch_size,
     from hypha_amd.telemetry import get_tracer
 
     tracer = get_tracer()
-    job_id = os.environ.get("HYPHA_JOB_ID", "")
-    # dispatch span: covers worker-side spawn -> executor ready (start
-    # timestamp injected by the worker daemon across the process boundary)
-    disp_ns = os.environ.get("HYPHA_DISPATCH_TS_NS")
-    if disp_ns:
-        dspan = tracer.start_span("job.dispatch", job_id=job_id, rank=comm.rank)
-        dspan.start_ns = int(disp_ns)
-        dspan.end()
-    job_span = tracer.start_span("job.execute", job_id=job_id,
+    job_span = tracer.start_span("job.execute",
+                                 job_id=os.environ.get("HYPHA_JOB_ID", ""),
                                  rank=comm.rank, world_size=comm.world_size)
     done = False
     while not done:
@@ -226,6 +219,18 @@ def main() -> int:
         cfg = json.load(f)
 
     session = Session(args.socket)
+    # dispatch span: covers worker-side spawn -> executor ready (start
+    # timestamp injected by the worker daemon across the process boundary)
+    from hypha_amd.telemetry import get_tracer as _get_tracer
+
+    _tr = _get_tracer()
+    _disp_ns = os.environ.get("HYPHA_DISPATCH_TS_NS")
+    if _disp_ns:
+        _d = _tr.start_span("job.dispatch",
+                            job_id=os.environ.get("HYPHA_JOB_ID", ""))
+        _d.start_ns = int(_disp_ns)
+        _d.end()
+        _tr.flush()
     torch.manual_seed(0)  # all workers start from the same init
     model = models.build(cfg["model"])
     device = torch.device("cuda", 0) if torch.cuda.is_available() else torch.device("cpu")
@@ -379,10 +384,14 @@ def main() -> int:
 
     done = False
     round_idx = resume_round
+    _ps_job_span = _tr.start_span("job.execute",
+                                  job_id=os.environ.get("HYPHA_JOB_ID", ""))
     while not done:
         # ---- inner loop: train until the scheduler's counter is exhausted ----
         import time as _time
 
+        _round_span = _tr.start_span("diloco.round", parent=_ps_job_span,
+                                     round=round_idx)
         round_t0 = _time.perf_counter()
         remaining = None
         round_samples = 0
@@ -403,7 +412,13 @@ def main() -> int:
                          "samples": round_samples}}
         )
 
+        _round_span.set_attribute("samples", round_samples)
+        _round_span.set_attribute("loss", float(loss))
+        _round_span.end()
+
         # ---- extract and push the pseudo-gradient ----
+        _sync_span = _tr.start_span("diloco.outer_sync", parent=_ps_job_span,
+                                    round=round_idx)
         session.send_status({"kind": "update"})
         delta = (worker.fp.master - worker.fp.theta0).cpu()
         fname = f"{round_idx}_local_gradients.safetensors"
@@ -419,6 +434,7 @@ def main() -> int:
         worker.fp.theta0.add_(u)
         worker.fp.master.copy_(worker.fp.theta0)
         worker.fp.flat.copy_(worker.fp.master)
+        _sync_span.end()
 
         resp = session.send_status({"kind": "update-received"})
         done = resp.get("kind") == "done"
@@ -432,6 +448,8 @@ def main() -> int:
         print(f"[executor] round {round_idx} merged, loss={loss:.4f} done={done}",
               flush=True)
 
+    _ps_job_span.end()
+    _tr.flush()
     session.close()
     return 0
 

@@ -44,7 +44,9 @@ def test_cluster_diloco_round_trip(binaries, tmp_path):
                       vocab_size=512, seq_len=128)
 
     gw_port = free_port()
-    env = dict(os.environ, PYTHONPATH=str(REPO))
+    trace_file = tmp_path / "traces.jsonl"
+    env = dict(os.environ, PYTHONPATH=str(REPO),
+               HYPHA_TRACE_FILE=str(trace_file))
     procs = []
     logs = {}
 
@@ -92,6 +94,18 @@ def test_cluster_diloco_round_trip(binaries, tmp_path):
             (tmp_path / "sched.log").read_text()[-3000:],
         )
         assert sched.returncode == 0
+        # OTLP spans from the executors: dispatch + rounds + outer syncs
+        import json as _json
+
+        spans = []
+        if trace_file.exists():
+            for line in trace_file.read_text().splitlines():
+                batch = _json.loads(line)
+                for rs in batch["resourceSpans"]:
+                    for ss in rs["scopeSpans"]:
+                        spans.extend(sp["name"] for sp in ss["spans"])
+        assert "job.dispatch" in spans, spans
+        assert "diloco.round" in spans and "diloco.outer_sync" in spans, spans
     finally:
         for p in procs:
             try:  # kill the whole group: daemons AND their executors —
