@@ -44,6 +44,17 @@ def main() -> int:
         cfg = json.load(f)
 
     session = Session(args.socket)
+    from hypha_amd.telemetry import get_tracer
+
+    tracer = get_tracer()
+    disp_ns = os.environ.get("HYPHA_DISPATCH_TS_NS")
+    if disp_ns:
+        d = tracer.start_span("job.dispatch",
+                              job_id=os.environ.get("HYPHA_JOB_ID", ""))
+        d.start_ns = int(disp_ns)
+        d.end()
+    job_span = tracer.start_span("job.execute", kind_attr="generate",
+                                 job_id=os.environ.get("HYPHA_JOB_ID", ""))
     torch.manual_seed(int(cfg.get("seed", 0)))
     device = torch.device("cuda", 0) if torch.cuda.is_available() else torch.device("cpu")
     model = models.build(cfg["model"])
@@ -104,6 +115,9 @@ def main() -> int:
          "metrics": {"completions": float(out_idx)}}
     )
     print(f"[infer] wrote {out_idx} completion batches", flush=True)
+    job_span.set_attribute("completions", out_idx)
+    job_span.end()
+    tracer.flush()
     session.close()
     return 0
 
