@@ -160,6 +160,7 @@ class Node {
   int relay_dial(const std::string& peer, double timeout_s);
   void relay_accept_run(long long circuit);
   bool gateway_connect();  // (re)connect + register + replay subscriptions
+  int dial_gateway(double timeout_s);  // active gateway, then failover
 
   std::string name_, gw_host_;
   int gw_port_;

@@ -253,9 +253,9 @@ void Node::start(int port) {
 
 // Establish (or re-establish) the persistent broker connection:
 // connect + optional TLS + register + synchronous ack.
-bool Node::gateway_connect() {
+int Node::dial_gateway(double timeout_s) {
   // try the active gateway, then every other candidate (primary first)
-  int fd = tcp_connect(gw_host_, gw_port_, 10.0);
+  int fd = tcp_connect(gw_host_, gw_port_, timeout_s);
   if (fd < 0 && !gw_list_.empty()) {
     std::vector<std::pair<std::string, int>> cands;
     cands.emplace_back(gw_primary_host_, gw_primary_port_);
@@ -272,6 +272,11 @@ bool Node::gateway_connect() {
       }
     }
   }
+  return fd;
+}
+
+bool Node::gateway_connect() {
+  int fd = dial_gateway(10.0);
   if (fd < 0) return false;
   // long receive timeout on the event connection
   struct timeval tv = {86400, 0};
@@ -524,7 +529,7 @@ void Node::relay_accept_run(long long circuit) {
 }
 
 Json Node::gateway_request(const std::string& type, const Json& body) {
-  int fd = tcp_connect(gw_host_, gw_port_, 10.0);
+  int fd = dial_gateway(10.0);
   if (fd < 0) throw std::runtime_error("gateway unreachable");
   SSL* ssl = nullptr;
   if (tls_) {
