@@ -346,3 +346,61 @@ def test_cluster_trains_with_fetched_preprocessor(tmp_path):
                 p.wait(timeout=10)
             except Exception:
                 pass
+
+
+@pytest.mark.timeout(120)
+def test_data_record_survives_gateway_restart(tmp_path):
+    """The data node re-announces its dataset record whenever its broker
+    connection is (re)established — a gateway RESTART (or failover) no
+    longer loses the registry record for good."""
+    from hypha_amd import _core
+    from hypha_amd.data.synthetic import write_slice_files
+
+    data_dir = tmp_path / "slices"
+    write_slice_files(str(data_dir), "ha", num_slices=2, samples_per_slice=4,
+                      vocab_size=64, seq_len=16)
+    gw_port = free_port()
+    env = dict(os.environ, PYTHONPATH=str(REPO))
+    gw = subprocess.Popen([str(BIN / "hypha-gateway"), "--port", str(gw_port)],
+                          env=env, start_new_session=True)
+    dn = None
+    gw2 = None
+    try:
+        time.sleep(0.4)
+        dn = subprocess.Popen(
+            [str(BIN / "hypha-data"), "--name", "data-ha", "--gateway-host",
+             "127.0.0.1", "--gateway-port", str(gw_port), "--dataset", "ha",
+             "--dataset-path", str(data_dir)],
+            env=env, start_new_session=True)
+        time.sleep(0.8)
+        gw.kill()
+        gw.wait(timeout=10)
+        time.sleep(0.5)
+        gw2 = subprocess.Popen([str(BIN / "hypha-gateway"), "--port", str(gw_port)],
+                               env=env, start_new_session=True)
+        n = _core.Node("checker", "127.0.0.1", gw_port)
+        rec = None
+        deadline = time.time() + 30
+        started = False
+        while time.time() < deadline:
+            try:
+                if not started:
+                    n.start(0)
+                    started = True
+                rec = n.kv_get("dataset:ha")
+                if rec:
+                    break
+            except RuntimeError:
+                pass
+            time.sleep(0.5)
+        if started:
+            n.stop()
+        assert rec and rec["num_slices"] == 2, rec
+    finally:
+        for p in (gw, dn, gw2):
+            if p is not None:
+                try:
+                    os.killpg(p.pid, signal.SIGKILL)
+                except (ProcessLookupError, PermissionError):
+                    if p.poll() is None:
+                        p.kill()
