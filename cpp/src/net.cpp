@@ -237,16 +237,22 @@ Node::Node(std::string name, std::string gateway_host, int gateway_port, TlsConf
 Node::~Node() { stop(); }
 
 void Node::start(int port) {
+  if (running_ || accept_thread_.joinable())
+    throw std::runtime_error("node: already started");
   listen_fd_ = tcp_listen(port);
   if (listen_fd_ < 0) throw std::runtime_error("node: cannot listen");
   port_ = listen_port(listen_fd_);
   running_ = true;
   accept_thread_ = std::thread([this] { accept_loop(); });
   if (gw_port_ > 0) {
-    if (!gateway_connect())
+    if (!gateway_connect()) {
+      // clean partial start so the object can be retried/destroyed safely
+      // (assigning over a joinable std::thread is std::terminate)
+      stop();
       throw std::runtime_error(
           "node: gateway registration failed (unreachable, TLS handshake "
           "rejected, or no ack)");
+    }
     gw_thread_ = std::thread([this] { gateway_listen_loop(); });
   }
 }

@@ -378,23 +378,21 @@ def test_data_record_survives_gateway_restart(tmp_path):
         time.sleep(0.5)
         gw2 = subprocess.Popen([str(BIN / "hypha-gateway"), "--port", str(gw_port)],
                                env=env, start_new_session=True)
-        n = _core.Node("checker", "127.0.0.1", gw_port)
         rec = None
         deadline = time.time() + 30
-        started = False
-        while time.time() < deadline:
+        while time.time() < deadline and not rec:
+            n = _core.Node("checker", "127.0.0.1", gw_port)
             try:
-                if not started:
-                    n.start(0)
-                    started = True
-                rec = n.kv_get("dataset:ha")
-                if rec:
-                    break
+                n.start(0)
+                for _ in range(6):
+                    rec = n.kv_get("dataset:ha")
+                    if rec:
+                        break
+                    time.sleep(0.5)
             except RuntimeError:
-                pass
-            time.sleep(0.5)
-        if started:
-            n.stop()
+                time.sleep(0.5)
+            finally:
+                n.stop()
         assert rec and rec["num_slices"] == 2, rec
     finally:
         for p in (gw, dn, gw2):
