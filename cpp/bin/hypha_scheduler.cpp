@@ -91,10 +91,10 @@ int main(int argc, char** argv) {
   if (init) {  // reference CLI Init subcommand: emit a commented job config
     printf("# hypha-scheduler configuration\n"
            "# --name scheduler --gateway-host H --gateway-port P\n"
-           "# --config job.json|job.toml    job spec (see below)\n"
+           "# --config job.json             job spec (JSON; see below)\n"
            "# --status-bridge 127.0.0.1:53800  AIM metrics forwarding\n"
            "# --tls-cert/--tls-key/--tls-ca [--tls-crl]  mTLS identity\n#\n"
-           "# job config (scheduler_config.rs analogue); JSON or TOML:\n"
+           "# job config (scheduler_config.rs analogue), JSON:\n"
            "{\n"
            "  \"model\": \"llama3-8b\",\n"
            "  \"dataset\": \"synth\",            # OR \"data_uri\": \"http://host/slice.safetensors\"\n"
