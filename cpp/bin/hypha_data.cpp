@@ -20,6 +20,7 @@ using namespace hypha;
 
 int main(int argc, char** argv) {
   std::string name = "data", gw_host = "127.0.0.1", dataset = "dataset", dir = ".";
+  std::string advertise_host, listen_host = "127.0.0.1";
   int gw_port = 0, port = 0;
   bool probe = false, init = false;
   std::vector<std::string> exclude_cidrs, fallback_gws;
@@ -38,6 +39,8 @@ int main(int argc, char** argv) {
     else if (a == "--tls-ca") tls.ca_path = next();
     else if (a == "--tls-crl") tls.crl_path = next();
     else if (a == "--exclude-cidr") exclude_cidrs.push_back(next());
+    else if (a == "--advertise-host") advertise_host = next();
+    else if (a == "--listen-host") listen_host = next();
     else if (a == "--fallback-gateway") fallback_gws.push_back(next());
     else if (a == "probe") probe = true;
     else if (a == "init") init = true;
@@ -50,7 +53,10 @@ int main(int argc, char** argv) {
            "# --dataset synth               dataset name to announce\n"
            "# --dataset-path /data/slices   dir of SafeTensors slice files\n"
            "# --tls-cert/--tls-key/--tls-ca [--tls-crl]  mTLS identity\n"
-           "# --fallback-gateway host:port  additional broker(s) tried on gateway loss\n");
+           "# --fallback-gateway host:port  additional broker(s) tried on gateway loss\n"
+           "# --advertise-host 10.0.0.5     host peers should dial this node at\n"
+           "# --listen-host 0.0.0.0         bind address (default loopback-only)\n"
+           "#                               (default: gateway-observed address)\n");
     return 0;
   }
   if (probe) {  // readiness = mTLS registration round-trip (data.rs probe)
@@ -88,6 +94,9 @@ int main(int argc, char** argv) {
   std::sort(files.begin(), files.end());
 
   Node node(name, gw_host, gw_port, tls);
+  node.set_advertise_host(advertise_host);
+  node.set_listen_host(listen_host);
+  node.set_exclude_cidrs(exclude_cidrs);
   node.on_stream("pull_slice", [&](const std::string& from, const Json& header,
                                    MsgSocket& sock) {
     int64_t index = header.at("index").as_int();

@@ -65,6 +65,7 @@ static void http_post_json(const std::string& bridge, const std::string& path,
 
 int main(int argc, char** argv) {
   std::string name = "scheduler", gw_host = "127.0.0.1", config_path, status_bridge;
+  std::string advertise_host, listen_host = "127.0.0.1";
   int gw_port = 0, port = 0;
   bool probe = false, init = false;
   std::vector<std::string> exclude_cidrs, fallback_gws;
@@ -83,6 +84,8 @@ int main(int argc, char** argv) {
     else if (a == "--tls-ca") tls.ca_path = next();
     else if (a == "--tls-crl") tls.crl_path = next();
     else if (a == "--exclude-cidr") exclude_cidrs.push_back(next());
+    else if (a == "--advertise-host") advertise_host = next();
+    else if (a == "--listen-host") listen_host = next();
     else if (a == "--fallback-gateway") fallback_gws.push_back(next());
     else if (a == "probe") probe = true;
     else if (a == "init") init = true;
@@ -93,7 +96,9 @@ int main(int argc, char** argv) {
            "# --name scheduler --gateway-host H --gateway-port P\n"
            "# --config job.json             job spec (JSON; see below)\n"
            "# --status-bridge 127.0.0.1:53800  AIM metrics forwarding\n"
-           "# --tls-cert/--tls-key/--tls-ca [--tls-crl]  mTLS identity\n#\n"
+           "# --tls-cert/--tls-key/--tls-ca [--tls-crl]  mTLS identity\n"
+           "# --advertise-host 10.0.0.5     host peers should dial this node at\n"
+           "# --listen-host 0.0.0.0         bind address (default loopback-only)\n#\n"
            "# job config (scheduler_config.rs analogue), JSON:\n"
            "{\n"
            "  \"model\": \"llama3-8b\",\n"
@@ -165,6 +170,8 @@ int main(int argc, char** argv) {
 
   Node node(name, gw_host, gw_port, tls);
   node.set_exclude_cidrs(exclude_cidrs);
+  node.set_advertise_host(advertise_host);
+  node.set_listen_host(listen_host);
   for (const auto& g : fallback_gws) {
     auto c = g.rfind(':');
     node.add_fallback_gateway(g.substr(0, c), std::stoi(g.substr(c + 1)));

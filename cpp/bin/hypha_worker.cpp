@@ -882,6 +882,7 @@ struct WorkerDaemon {
 
 int main(int argc, char** argv) {
   std::string name = "worker", gw_host = "127.0.0.1", cmd, icmd, work_root = "/tmp/hypha-work";
+  std::string advertise_host, listen_host = "127.0.0.1";
   int gw_port = 0, port = 0;
   bool probe = false, init = false;
   std::vector<int> gpu_ids;
@@ -933,6 +934,8 @@ int main(int argc, char** argv) {
     else if (a == "--tls-ca") tls.ca_path = next();
     else if (a == "--tls-crl") tls.crl_path = next();
     else if (a == "--exclude-cidr") exclude_cidrs.push_back(next());
+    else if (a == "--advertise-host") advertise_host = next();
+    else if (a == "--listen-host") listen_host = next();
     else if (a == "--fallback-gateway") fallback_gws.push_back(next());
     else if (a == "--fetch-allow") fetch_allow.push_back(next());
     else if (a == "--hf-endpoint") hf_endpoint = next();
@@ -957,7 +960,10 @@ int main(int argc, char** argv) {
            "# --fetch-allow host[:port]     allow URI/HF fetches from this host\n"
            "#                               (repeatable; *.domain wildcards; default: deny all)\n"
            "# --hf-endpoint https://huggingface.co  HuggingFace resolve endpoint\n"
-           "# --fallback-gateway host:port  additional broker(s) tried on gateway loss\n");
+           "# --fallback-gateway host:port  additional broker(s) tried on gateway loss\n"
+           "# --advertise-host 10.0.0.5     host peers should dial this node at\n"
+           "# --listen-host 0.0.0.0         bind address (default loopback-only)\n"
+           "#                               (default: gateway-observed address)\n");
     return 0;
   }
   if (probe) {
@@ -998,6 +1004,8 @@ int main(int argc, char** argv) {
   }
   daemon.hf_endpoint = hf_endpoint;
   daemon.node.set_exclude_cidrs(exclude_cidrs);
+  daemon.node.set_advertise_host(advertise_host);
+  daemon.node.set_listen_host(listen_host);
   daemon.start(port);
   printf("hypha-worker %s ready on port %d\n", name.c_str(), daemon.node.port());
   fflush(stdout);

@@ -357,6 +357,7 @@ PYBIND11_MODULE(_core, m) {
            py::arg("tls_cert"), py::arg("tls_key"), py::arg("tls_ca"),
            py::arg("tls_crl") = "")
       .def("start", &Gateway::start, py::arg("port") = 0,
+           py::arg("listen_host") = "127.0.0.1",
            py::call_guard<py::gil_scoped_release>())
       .def("stop", &Gateway::stop, py::call_guard<py::gil_scoped_release>())
       .def_property_readonly("port", &Gateway::port);
@@ -460,6 +461,8 @@ PYBIND11_MODULE(_core, m) {
                throw std::runtime_error("stream write failed");
            })
       .def("set_exclude_cidrs", &Node::set_exclude_cidrs, py::arg("cidrs"))
+      .def("set_advertise_host", &Node::set_advertise_host, py::arg("host"))
+      .def("set_listen_host", &Node::set_listen_host, py::arg("host"))
       .def("stream_call",
            // bidirectional stream RPC: send header (+payload), read a JSON
            // reply and an optional `size`-byte body. Client side of the

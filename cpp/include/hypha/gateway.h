@@ -4,6 +4,8 @@
 // DHT/gossip semantics collapsed into a broker as SURVEY.md §7 step 2 allows.
 #pragma once
 
+#include <arpa/inet.h>
+#include <netinet/in.h>
 #include <sys/socket.h>
 #include <unistd.h>
 
@@ -23,6 +25,20 @@
 
 namespace hypha {
 
+// Source IP of a connected socket's peer (IPv4/IPv6), "" on failure.
+inline std::string observed_ip(int fd) {
+  sockaddr_storage ss{};
+  socklen_t len = sizeof(ss);
+  if (getpeername(fd, (sockaddr*)&ss, &len) != 0) return "";
+  char buf[INET6_ADDRSTRLEN] = {0};
+  if (ss.ss_family == AF_INET) {
+    inet_ntop(AF_INET, &((sockaddr_in*)&ss)->sin_addr, buf, sizeof(buf));
+  } else if (ss.ss_family == AF_INET6) {
+    inet_ntop(AF_INET6, &((sockaddr_in6*)&ss)->sin6_addr, buf, sizeof(buf));
+  }
+  return buf;
+}
+
 class Gateway {
  public:
   Gateway() = default;
@@ -31,8 +47,8 @@ class Gateway {
   }
   ~Gateway() { stop(); }
 
-  void start(int port = 0) {
-    listen_fd_ = tcp_listen(port);
+  void start(int port = 0, const std::string& listen_host = "127.0.0.1") {
+    listen_fd_ = tcp_listen(port, listen_host);
     if (listen_fd_ < 0) throw std::runtime_error("gateway: cannot listen");
     port_ = listen_port(listen_fd_);
     running_ = true;
@@ -118,10 +134,28 @@ class Gateway {
             break;
           }
           peer = claimed;
+          // Observed-address substitution (reference external_address.rs /
+          // libp2p identify semantics): a node that registered with a
+          // loopback or wildcard host but dialed in from another machine is
+          // only dialable at the address the broker OBSERVED; keep the
+          // node's listen port, swap the host. Loopback connections (the
+          // single-host case) keep the registered address, and a node that
+          // advertised a concrete non-loopback host is never overridden.
+          std::string addr = msg->at("addr").as_string();
+          auto colon = addr.rfind(':');
+          std::string reg_host =
+              colon == std::string::npos ? addr : addr.substr(0, colon);
+          if (reg_host == "127.0.0.1" || reg_host == "localhost" ||
+              reg_host == "0.0.0.0") {
+            std::string obs = observed_ip(sock->fd());
+            if (!obs.empty() && obs != "127.0.0.1" && obs != "::1" &&
+                obs != "::ffff:127.0.0.1" && colon != std::string::npos)
+              addr = obs + addr.substr(colon);
+          }
           {
             std::lock_guard<std::mutex> lk(mu_);
             peers_[peer] = sock;
-            kv_["addr:" + peer] = msg->at("addr");
+            kv_["addr:" + peer] = Json(addr);
           }
           Json ack;
           ack["kind"] = "registered";

@@ -62,7 +62,9 @@ class MsgSocket {
 };
 
 int tcp_connect(const std::string& host, int port, double timeout_s = 5.0);
-int tcp_listen(int port);  // returns listen fd; port 0 = ephemeral
+// Returns listen fd; port 0 = ephemeral. Binds loopback by default —
+// pass "0.0.0.0" (or an interface address) to accept off-host peers.
+int tcp_listen(int port, const std::string& host = "127.0.0.1");
 int listen_port(int listen_fd);
 
 // Process-global transport byte counters (reference telemetry/src/
@@ -128,6 +130,19 @@ class Node {
     exclude_cidrs_ = std::move(cidrs);
   }
 
+  // External-address advertising (reference external_address.rs:15-137):
+  // the host other peers should dial this node at. Unset = register as
+  // 127.0.0.1 and let the gateway substitute the address it OBSERVED the
+  // connection from (identify-style), which covers multi-host LANs
+  // automatically; set it explicitly when the observed address is not the
+  // dialable one (NAT with port-forward, multi-homed hosts).
+  void set_advertise_host(std::string host) { advertise_host_ = std::move(host); }
+
+  // Bind address for this node's own listener (default loopback-only);
+  // "0.0.0.0" exposes it to the LAN for multi-host clusters. Set before
+  // start().
+  void set_listen_host(std::string host) { listen_host_ = std::move(host); }
+
  private:
   void check_dialable(const std::string& peer, const std::string& host) const;
   std::vector<std::string> exclude_cidrs_;
@@ -163,6 +178,8 @@ class Node {
   int dial_gateway(double timeout_s);  // active gateway, then failover
 
   std::string name_, gw_host_;
+  std::string advertise_host_;
+  std::string listen_host_ = "127.0.0.1";
   int gw_port_;
   // failover gateway candidates (primary + fallbacks); gw_host_/gw_port_
   // is the ACTIVE one, switched by gateway_connect on unreachability

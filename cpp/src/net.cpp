@@ -202,14 +202,20 @@ int tcp_connect(const std::string& host, int port, double timeout_s) {
   return fd;
 }
 
-int tcp_listen(int port) {
+int tcp_listen(int port, const std::string& host) {
   int fd = ::socket(AF_INET, SOCK_STREAM, 0);
   if (fd < 0) return -1;
   int one = 1;
   setsockopt(fd, SOL_SOCKET, SO_REUSEADDR, &one, sizeof one);
   struct sockaddr_in addr = {};
   addr.sin_family = AF_INET;
-  addr.sin_addr.s_addr = htonl(INADDR_LOOPBACK);
+  // Default loopback-only (reference listen.rs binds explicit addresses;
+  // nothing is exposed off-host unless asked): "0.0.0.0" or a concrete
+  // interface address opens the daemon to the LAN for multi-host clusters.
+  if (inet_pton(AF_INET, host.c_str(), &addr.sin_addr) != 1) {
+    ::close(fd);
+    return -1;
+  }
   addr.sin_port = htons((uint16_t)port);
   if (::bind(fd, (struct sockaddr*)&addr, sizeof addr) != 0 || ::listen(fd, 64) != 0) {
     ::close(fd);
@@ -239,7 +245,7 @@ Node::~Node() { stop(); }
 void Node::start(int port) {
   if (running_ || accept_thread_.joinable())
     throw std::runtime_error("node: already started");
-  listen_fd_ = tcp_listen(port);
+  listen_fd_ = tcp_listen(port, listen_host_);
   if (listen_fd_ < 0) throw std::runtime_error("node: cannot listen");
   port_ = listen_port(listen_fd_);
   running_ = true;
@@ -299,7 +305,9 @@ bool Node::gateway_connect() {
   Json reg;
   reg["kind"] = "register";
   reg["peer"] = name_;
-  reg["addr"] = std::string("127.0.0.1:") + std::to_string(port_);
+  reg["addr"] = (advertise_host_.empty() ? std::string("127.0.0.1")
+                                         : advertise_host_) +
+                ":" + std::to_string(port_);
   if (!sock->send_json(reg)) return false;
   // synchronous ack: the registry entry is visible before we proceed
   auto ack = sock->recv_json();

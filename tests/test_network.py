@@ -372,3 +372,59 @@ def test_gateway_failover_and_reannounce():
         a.stop()
         b.stop()
         gw2.stop()
+
+
+def test_advertise_host_overrides_registered_address():
+    """External-address advertising (reference external_address.rs:15-137):
+    a node can declare the host other peers should dial it at; nodes that
+    register from loopback without advertising keep 127.0.0.1."""
+    gw = core.Gateway()
+    gw.start(0)
+    a = core.Node("adv-a", "127.0.0.1", gw.port)
+    a.set_advertise_host("10.1.2.3")
+    b = core.Node("adv-b", "127.0.0.1", gw.port)
+    try:
+        a.start(0)
+        b.start(0)
+        assert b.resolve("adv-a") == f"10.1.2.3:{a.port}"
+        # loopback registration observed over loopback is left untouched
+        assert a.resolve("adv-b") == f"127.0.0.1:{b.port}"
+    finally:
+        a.stop()
+        b.stop()
+        gw.stop()
+
+
+def test_observed_address_substitution_offhost():
+    """A node that registers as 127.0.0.1 but dials the broker from a
+    non-loopback address is recorded at the OBSERVED address (identify-style
+    substitution), and is dialable there when its listener binds 0.0.0.0."""
+    import socket as pysock
+
+    probe = pysock.socket(pysock.AF_INET, pysock.SOCK_DGRAM)
+    try:
+        probe.connect(("203.0.113.9", 9))  # no packet sent; picks route+src IP
+        ip = probe.getsockname()[0]
+    except OSError:
+        pytest.skip("no non-loopback route on this host")
+    finally:
+        probe.close()
+    if ip.startswith("127."):
+        pytest.skip("only loopback networking available")
+
+    gw = core.Gateway()
+    gw.start(0, "0.0.0.0")
+    n = core.Node("obs-a", ip, gw.port)  # dials the broker off-loopback
+    n.set_listen_host("0.0.0.0")
+    m = core.Node("obs-b", "127.0.0.1", gw.port)
+    try:
+        n.start(0)
+        n.on("ping", lambda frm, body: {"pong": body["x"] + 1})
+        m.start(0)
+        assert m.resolve("obs-a") == f"{ip}:{n.port}"
+        # the substituted address is genuinely dialable end-to-end
+        assert m.request("obs-a", "ping", {"x": 41}) == {"pong": 42}
+    finally:
+        n.stop()
+        m.stop()
+        gw.stop()
