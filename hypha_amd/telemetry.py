@@ -187,8 +187,17 @@ class Tracer:
         self.flush_every = flush_every
         self._lock = threading.Lock()
         self._finished: list = []
-        self._stack: list = []
+        # implicit-parent stack is PER-THREAD: a span entered on one thread
+        # must not become the parent of spans started concurrently on another
+        self._local = threading.local()
         self.export_errors = 0
+
+    @property
+    def _stack(self) -> list:
+        st = getattr(self._local, "stack", None)
+        if st is None:
+            st = self._local.stack = []
+        return st
 
     def start_span(self, name: str, parent: "Span | None" = None, **attrs) -> Span:
         if parent is None and self._stack:
