@@ -57,14 +57,27 @@ class Gateway {
 
   void stop() {
     if (!running_.exchange(false)) return;
-    if (listen_fd_ >= 0) {
-      ::shutdown(listen_fd_, 2);
+    // Wake every blocked thread with shutdown() only (fds stay open — no
+    // fd-reuse race); each conn thread closes its own socket when it exits.
+    if (listen_fd_ >= 0) ::shutdown(listen_fd_, 2);
+    {
+      std::lock_guard<std::mutex> lk(mu_);
+      for (auto& s : conns_) s->shutdown_now();
     }
     if (accept_thread_.joinable()) accept_thread_.join();
-    std::lock_guard<std::mutex> lk(mu_);
-    for (auto& [_, s] : peers_) s->close_now();
-    peers_.clear();
-    subs_.clear();
+    {
+      // bounded drain of the detached conn handlers (a conn accepted after
+      // the sweep, or a live relay circuit, must not wedge stop())
+      std::unique_lock<std::mutex> lk(mu_);
+      conn_cv_.wait_for(lk, std::chrono::seconds(5),
+                        [&] { return live_conns_ == 0; });
+      peers_.clear();
+      subs_.clear();
+    }
+    if (listen_fd_ >= 0) {
+      ::close(listen_fd_);
+      listen_fd_ = -1;
+    }
   }
 
   int port() const { return port_; }
@@ -97,17 +110,34 @@ class Gateway {
         handle_conn(fd, ssl);
       }).detach();
     }
-    ::close(listen_fd_);
-    listen_fd_ = -1;
+    // listen_fd_ is closed by stop() after this thread is joined
   }
 
   void handle_conn(int fd, SSL* ssl = nullptr, bool relay_only = false) {
     auto sock = std::make_shared<MsgSocket>(fd, ssl);
+    {
+      // conn registry: stop() shutdowns these to wake blocked readers; if
+      // stop already swept, wake ourselves so the loop below exits at once
+      std::lock_guard<std::mutex> lk(mu_);
+      conns_.insert(sock);
+      ++live_conns_;
+      if (!running_) sock->shutdown_now();
+    }
+    handle_conn_body(sock, relay_only);
+    {
+      std::lock_guard<std::mutex> lk(mu_);
+      conns_.erase(sock);
+      --live_conns_;
+    }
+    conn_cv_.notify_all();
+  }
+
+  void handle_conn_body(std::shared_ptr<MsgSocket> sock, bool relay_only) {
     std::string peer;  // set once registered (persistent connection)
     // With mTLS the verified cert CN is the only identity we trust: the
     // self-claimed 'peer'/'from' fields must match it (reference model,
     // rfc/2025-05-30_mtls.md — PeerId is derived from the cert key).
-    const std::string cn = ssl ? sock->peer_identity() : std::string();
+    const std::string cn = sock->peer_identity();  // "" on plain sockets
     // A malformed message (Json::at throws) must drop THIS connection, not
     // std::terminate the whole broker: the per-connection thread is detached.
     try {
@@ -126,7 +156,7 @@ class Gateway {
         if (relay_only) break;  // plain leg may only speak relay kinds
         if (kind == "register") {
           std::string claimed = msg->at("peer").as_string();
-          if (ssl && !cn.empty() && claimed != cn) {
+          if (!cn.empty() && claimed != cn) {
             Json nak;
             nak["kind"] = "error";
             nak["error"] = "peer name does not match certificate CN";
@@ -168,7 +198,7 @@ class Gateway {
           subs_[msg->at("topic").as_string()].erase(peer);
         } else if (kind == "request") {
           std::string from = msg->get_or("from", Json("")).as_string();
-          if (ssl && !cn.empty() && !from.empty() && from != cn) {
+          if (!cn.empty() && !from.empty() && from != cn) {
             Json resp;
             resp["kind"] = "response";
             resp["ok"] = false;
@@ -368,6 +398,9 @@ class Gateway {
   std::map<long long, std::shared_ptr<Circuit>> circuits_;
   long long next_circuit_ = 1;
   std::map<std::string, std::shared_ptr<MsgSocket>> peers_;
+  std::set<std::shared_ptr<MsgSocket>> conns_;  // every live inbound conn
+  int live_conns_ = 0;                          // guarded by mu_
+  std::condition_variable conn_cv_;
   std::map<std::string, std::set<std::string>> subs_;
   std::map<std::string, Json> kv_;
   std::unique_ptr<TlsContext> tls_;

@@ -12,11 +12,13 @@
 #pragma once
 
 #include <atomic>
+#include <condition_variable>
 #include <functional>
 #include <map>
 #include <memory>
 #include <mutex>
 #include <optional>
+#include <set>
 #include <string>
 #include <thread>
 #include <vector>
@@ -39,6 +41,10 @@ class MsgSocket {
   bool send_raw(const char* data, size_t n);
   bool recv_raw(char* data, size_t n);
   void close_now();
+  // Wake any thread blocked in recv/SSL io WITHOUT closing the fd: the fd
+  // number stays reserved, so a concurrent reader can never race a kernel
+  // fd-reuse (close_now is only safe once no other thread uses the socket).
+  void shutdown_now();
   int fd() const { return fd_; }
   // Detach the fd (relay handshake: a few plain frames are exchanged and
   // then the SAME socket carries the end-to-end TLS/protocol bytes).
@@ -148,7 +154,8 @@ class Node {
   std::vector<std::string> exclude_cidrs_;
 
   void accept_loop();
-  void handle_conn(int fd, SSL* ssl = nullptr);
+  void handle_conn(MsgSocket& sock);
+  void unregister_conn(int fd);
   void gateway_listen_loop();
 
  public:
@@ -200,7 +207,13 @@ class Node {
   std::unique_ptr<MsgSocket> gw_sock_;       // persistent broker connection
   std::mutex gw_mu_;
   std::map<int64_t, std::shared_ptr<std::pair<std::mutex, Json>>> pending_;
-  std::vector<std::thread> conn_threads_;
+  // Inbound-connection registry: stop() shutdowns every live conn fd (waking
+  // its detached handler thread) and then waits for the handlers to drain,
+  // so sockets are only ever CLOSED by the thread that owns them.
+  std::mutex conn_mu_;
+  std::condition_variable conn_cv_;
+  std::set<int> conn_fds_;   // open fds, removed BEFORE their close
+  int live_conns_ = 0;       // guarded by conn_mu_
 };
 
 }  // namespace hypha

@@ -11,6 +11,7 @@
 #include <unistd.h>
 
 #include <atomic>
+#include <chrono>
 #include <condition_variable>
 #include <stdexcept>
 
@@ -38,6 +39,10 @@ void MsgSocket::close_now() {
     ::close(fd_);
     fd_ = -1;
   }
+}
+
+void MsgSocket::shutdown_now() {
+  if (fd_ >= 0) ::shutdown(fd_, SHUT_RDWR);
 }
 
 static bool write_all_fd(int fd, const char* p, size_t n) {
@@ -337,19 +342,36 @@ bool Node::gateway_connect() {
 
 void Node::stop() {
   if (!running_.exchange(false)) return;
-  if (listen_fd_ >= 0) {
-    ::shutdown(listen_fd_, SHUT_RDWR);
-    ::close(listen_fd_);
-    listen_fd_ = -1;
-  }
+  // Phase 1: WAKE every blocked thread with shutdown() only — the fds stay
+  // open, so no thread can race a kernel fd-number reuse. Closing happens
+  // in phase 2, after the threads are done with their sockets.
+  if (listen_fd_ >= 0) ::shutdown(listen_fd_, SHUT_RDWR);
   {
     // gw_mu_ serializes against the reader thread swapping the socket in
     // gateway_connect() during a reconnection
     std::lock_guard<std::mutex> lk(gw_mu_);
-    if (gw_sock_) gw_sock_->close_now();
+    if (gw_sock_) gw_sock_->shutdown_now();
+  }
+  {
+    std::lock_guard<std::mutex> lk(conn_mu_);
+    for (int fd : conn_fds_) ::shutdown(fd, SHUT_RDWR);
   }
   if (accept_thread_.joinable()) accept_thread_.join();
   if (gw_thread_.joinable()) gw_thread_.join();
+  {
+    // conn handler threads are detached: wait (bounded — a handler stuck in
+    // user code must not wedge stop()) for them to release their sockets
+    std::unique_lock<std::mutex> lk(conn_mu_);
+    conn_cv_.wait_for(lk, std::chrono::seconds(5),
+                      [&] { return live_conns_ == 0; });
+  }
+  // Phase 2: no reader threads remain — closing is safe now.
+  if (listen_fd_ >= 0) {
+    ::close(listen_fd_);
+    listen_fd_ = -1;
+  }
+  std::lock_guard<std::mutex> lk(gw_mu_);
+  gw_sock_.reset();
 }
 
 void Node::on(const std::string& type, Handler h) {
@@ -362,37 +384,57 @@ void Node::on_stream(const std::string& type, StreamHandler h) {
   stream_handlers_[type] = std::move(h);
 }
 
+void Node::unregister_conn(int fd) {
+  std::lock_guard<std::mutex> lk(conn_mu_);
+  conn_fds_.erase(fd);  // removed while the fd is STILL open (no reuse race)
+  --live_conns_;
+  conn_cv_.notify_all();
+}
+
 void Node::accept_loop() {
   while (running_) {
     int fd = ::accept(listen_fd_, nullptr, nullptr);
     if (fd < 0) break;
     int one = 1;
     setsockopt(fd, IPPROTO_TCP, TCP_NODELAY, &one, sizeof one);
+    {
+      // register BEFORE the handler thread exists, so a concurrent stop()
+      // always sees the fd; if stop() already swept, wake it ourselves
+      std::lock_guard<std::mutex> lk(conn_mu_);
+      conn_fds_.insert(fd);
+      ++live_conns_;
+      if (!running_) ::shutdown(fd, SHUT_RDWR);
+    }
     std::thread([this, fd] {
-      try {
-        SSL* ssl = nullptr;
-        if (tls_) {
-          ssl = tls_->wrap(fd, true);
-          if (!ssl) {
-            ::close(fd);
-            return;  // unauthenticated peer rejected at handshake
-          }
+      SSL* ssl = nullptr;
+      if (tls_) {
+        ssl = tls_->wrap(fd, true);
+        if (!ssl) {
+          unregister_conn(fd);
+          ::close(fd);
+          return;  // unauthenticated peer rejected at handshake
         }
-        handle_conn(fd, ssl);
-      } catch (const std::exception& e) {
-        fprintf(stderr, "[net:%s] conn handler error: %s\n", name_.c_str(), e.what());
       }
+      {
+        MsgSocket sock(fd, ssl);
+        try {
+          handle_conn(sock);
+        } catch (const std::exception& e) {
+          fprintf(stderr, "[net:%s] conn handler error: %s\n", name_.c_str(),
+                  e.what());
+        }
+        unregister_conn(fd);
+      }  // the socket closes HERE — after deregistration
     }).detach();
   }
 }
 
-void Node::handle_conn(int fd, SSL* ssl) {
-  MsgSocket sock(fd, ssl);
+void Node::handle_conn(MsgSocket& sock) {
   // With mTLS the only identity we trust is the verified certificate CN
   // (reference model: PeerId is derived from the cert key,
   // rfc/2025-05-30_mtls.md). A self-declared `from` that contradicts it is a
   // spoof attempt and the message is rejected; handlers always see the CN.
-  const std::string verified = ssl ? sock.peer_identity() : std::string();
+  const std::string verified = sock.peer_identity();
   while (running_) {
     auto msg = sock.recv_json();
     if (!msg) break;
@@ -530,16 +572,33 @@ void Node::relay_accept_run(long long circuit) {
     fd = hello.release();
   }
   // from here the circuit is an ordinary inbound connection: server-side
-  // mTLS handshake + the normal typed-message loop
+  // mTLS handshake + the normal typed-message loop (registered like any
+  // accepted conn so stop() can wake this thread)
+  {
+    std::lock_guard<std::mutex> lk(conn_mu_);
+    conn_fds_.insert(fd);
+    ++live_conns_;
+    if (!running_) ::shutdown(fd, SHUT_RDWR);
+  }
   SSL* ssl = nullptr;
   if (tls_) {
     ssl = tls_->wrap(fd, true);
     if (!ssl) {
+      unregister_conn(fd);
       ::close(fd);
       return;
     }
   }
-  handle_conn(fd, ssl);
+  {
+    MsgSocket sock(fd, ssl);
+    try {
+      handle_conn(sock);
+    } catch (const std::exception& e) {
+      fprintf(stderr, "[net:%s] relay conn handler error: %s\n", name_.c_str(),
+              e.what());
+    }
+    unregister_conn(fd);
+  }
 }
 
 Json Node::gateway_request(const std::string& type, const Json& body) {
