@@ -62,15 +62,19 @@ class Gateway {
     if (listen_fd_ >= 0) ::shutdown(listen_fd_, 2);
     {
       std::lock_guard<std::mutex> lk(mu_);
-      for (auto& s : conns_) s->shutdown_now();
+      for (int fd : conn_fds_) ::shutdown(fd, SHUT_RDWR);
+      for (int fd : circuit_fds_) ::shutdown(fd, SHUT_RDWR);
     }
+    circ_cv_.notify_all();  // abort any circuit handshake still waiting
     if (accept_thread_.joinable()) accept_thread_.join();
     {
-      // bounded drain of the detached conn handlers (a conn accepted after
-      // the sweep, or a live relay circuit, must not wedge stop())
+      // UNBOUNDED drain of the detached conn handlers: every conn and
+      // circuit fd was just shutdown, so each handler unblocks and exits;
+      // returning before they drain would let them touch a destroyed
+      // Gateway (TSAN caught exactly that with a bounded wait under its
+      // ~20x slowdown).
       std::unique_lock<std::mutex> lk(mu_);
-      conn_cv_.wait_for(lk, std::chrono::seconds(5),
-                        [&] { return live_conns_ == 0; });
+      conn_cv_.wait(lk, [&] { return live_conns_ == 0; });
       peers_.clear();
       subs_.clear();
     }
@@ -87,8 +91,18 @@ class Gateway {
     while (running_) {
       int fd = ::accept(listen_fd_, nullptr, nullptr);
       if (fd < 0) break;
+      {
+        // register BEFORE the handler thread exists (and before any TLS
+        // handshake can block), so a concurrent stop() always sees the fd;
+        // if stop() already swept, wake ourselves
+        std::lock_guard<std::mutex> lk(mu_);
+        conn_fds_.insert(fd);
+        ++live_conns_;
+        if (!running_) ::shutdown(fd, SHUT_RDWR);
+      }
       std::thread([this, fd] {
         SSL* ssl = nullptr;
+        bool done = false;
         if (tls_) {
           // Relay legs are PLAIN sockets carrying length-prefixed JSON
           // hellos; the end-to-end peer TLS runs THROUGH the circuit (the
@@ -99,37 +113,38 @@ class Gateway {
           ssize_t pk = ::recv(fd, &b0, 1, MSG_PEEK);
           if (pk == 1 && b0 == 0x00) {
             handle_conn(fd, nullptr, /*relay_only=*/true);
-            return;
-          }
-          ssl = tls_->wrap(fd, true);
-          if (!ssl) {
-            ::close(fd);
-            return;  // unauthenticated peer rejected at handshake
+            done = true;
+          } else {
+            ssl = tls_->wrap(fd, true);
+            if (!ssl) {
+              // unauthenticated peer rejected at handshake: deregister
+              // while the fd is still open, then close
+              unregister_conn(fd);
+              ::close(fd);
+              return;
+            }
           }
         }
-        handle_conn(fd, ssl);
+        if (!done) handle_conn(fd, ssl);
+        unregister_conn(fd);
       }).detach();
     }
     // listen_fd_ is closed by stop() after this thread is joined
   }
 
+  void unregister_conn(int fd) {
+    std::lock_guard<std::mutex> lk(mu_);
+    conn_fds_.erase(fd);  // removed while the fd is STILL open (no reuse race)
+    --live_conns_;
+    conn_cv_.notify_all();
+  }
+
+  // NOTE: the caller (accept_loop lambda) holds the conn-registry entry for
+  // `fd` and deregisters after this returns; the socket itself closes when
+  // its last shared_ptr drops, which is always after deregistration.
   void handle_conn(int fd, SSL* ssl = nullptr, bool relay_only = false) {
     auto sock = std::make_shared<MsgSocket>(fd, ssl);
-    {
-      // conn registry: stop() shutdowns these to wake blocked readers; if
-      // stop already swept, wake ourselves so the loop below exits at once
-      std::lock_guard<std::mutex> lk(mu_);
-      conns_.insert(sock);
-      ++live_conns_;
-      if (!running_) sock->shutdown_now();
-    }
     handle_conn_body(sock, relay_only);
-    {
-      std::lock_guard<std::mutex> lk(mu_);
-      conns_.erase(sock);
-      --live_conns_;
-    }
-    conn_cv_.notify_all();
   }
 
   void handle_conn_body(std::shared_ptr<MsgSocket> sock, bool relay_only) {
@@ -292,11 +307,16 @@ class Gateway {
   std::atomic<bool> running_{false};
   std::thread accept_thread_;
   // ---- relay circuits (opaque byte splice between two legs) ----
+  // Circuit state is tiny and guarded by ONE gateway-level mutex/cv pair
+  // (circ_mu_/circ_cv_) rather than per-circuit synchronization: circuit
+  // setup is rare (one handshake per undialable peer pair), and heap-reused
+  // per-circuit mutexes trip gcc-11 libtsan's stale-sync-object false
+  // positives while adding no real concurrency.
   struct Circuit {
-    std::mutex m;
-    std::condition_variable cv;
     std::shared_ptr<MsgSocket> b;  // acceptor leg, set by relay_accept
   };
+  std::mutex circ_mu_;
+  std::condition_variable circ_cv_;
 
   void relay_connect(std::shared_ptr<MsgSocket> a, const Json& msg) {
     const std::string to = msg.get_or("to", Json("")).as_string();
@@ -327,8 +347,9 @@ class Gateway {
     bool pushed = target->send_json(offer);
     std::shared_ptr<MsgSocket> b;
     if (pushed) {
-      std::unique_lock<std::mutex> lk(circ->m);
-      circ->cv.wait_for(lk, std::chrono::seconds(10), [&] { return (bool)circ->b; });
+      std::unique_lock<std::mutex> lk(circ_mu_);
+      circ_cv_.wait_for(lk, std::chrono::seconds(10),
+                        [&] { return (bool)circ->b || !running_; });
       b = circ->b;
     }
     {
@@ -346,11 +367,30 @@ class Gateway {
     // ---- splice: opaque bytes both ways until either side closes ----
     int fa = a->release();
     int fb = b->release();
+    {
+      // register the raw circuit fds so stop() can wake the pumps (their
+      // MsgSockets released ownership and no longer cover them)
+      std::lock_guard<std::mutex> lk(mu_);
+      circuit_fds_.insert(fa);
+      circuit_fds_.insert(fb);
+      if (!running_) {
+        ::shutdown(fa, 2);
+        ::shutdown(fb, 2);
+      }
+    }
     std::thread rev([fa, fb] { pump(fb, fa); });
     pump(fa, fb);
     ::shutdown(fa, 2);
     ::shutdown(fb, 2);
     rev.join();  // never close an fd another thread may still be using
+    {
+      // deregister BEFORE closing (the fds are still open here)
+      std::lock_guard<std::mutex> lk(mu_);
+      circuit_fds_.erase(fa);
+      circuit_fds_.erase(fb);
+      conn_fds_.erase(fa);
+      conn_fds_.erase(fb);
+    }
     ::close(fa);
     ::close(fb);
   }
@@ -371,10 +411,10 @@ class Gateway {
       return;
     }
     {
-      std::lock_guard<std::mutex> lk(circ->m);
+      std::lock_guard<std::mutex> lk(circ_mu_);
       circ->b = b;
     }
-    circ->cv.notify_all();
+    circ_cv_.notify_all();
     // ownership passes to the connect-side thread via the shared_ptr held
     // in the Circuit: this handler returns immediately and the socket
     // outlives it until the splice finishes.
@@ -398,8 +438,12 @@ class Gateway {
   std::map<long long, std::shared_ptr<Circuit>> circuits_;
   long long next_circuit_ = 1;
   std::map<std::string, std::shared_ptr<MsgSocket>> peers_;
-  std::set<std::shared_ptr<MsgSocket>> conns_;  // every live inbound conn
-  int live_conns_ = 0;                          // guarded by mu_
+  // Shutdown registries (guarded by mu_): every live inbound conn fd and
+  // every raw fd inside an active relay splice. Entries are always removed
+  // BEFORE their fd is closed, so stop() can never shutdown a reused fd.
+  std::set<int> conn_fds_;
+  std::set<int> circuit_fds_;
+  int live_conns_ = 0;  // guarded by mu_
   std::condition_variable conn_cv_;
   std::map<std::string, std::set<std::string>> subs_;
   std::map<std::string, Json> kv_;

@@ -188,6 +188,18 @@ class Node {
   std::string advertise_host_;
   std::string listen_host_ = "127.0.0.1";
   int gw_port_;
+  // gw_host_/gw_port_ are rewritten by failover while relay threads read
+  // them; all post-start access goes through gw_addr()/set_gw_addr()
+  mutable std::mutex gw_addr_mu_;
+  std::pair<std::string, int> gw_addr() const {
+    std::lock_guard<std::mutex> lk(gw_addr_mu_);
+    return {gw_host_, gw_port_};
+  }
+  void set_gw_addr(const std::string& h, int p) {
+    std::lock_guard<std::mutex> lk(gw_addr_mu_);
+    gw_host_ = h;
+    gw_port_ = p;
+  }
   // failover gateway candidates (primary + fallbacks); gw_host_/gw_port_
   // is the ACTIVE one, switched by gateway_connect on unreachability
   std::string gw_primary_host_;
@@ -206,7 +218,6 @@ class Node {
   std::map<std::string, std::string> addr_cache_;
   std::unique_ptr<MsgSocket> gw_sock_;       // persistent broker connection
   std::mutex gw_mu_;
-  std::map<int64_t, std::shared_ptr<std::pair<std::mutex, Json>>> pending_;
   // Inbound-connection registry: stop() shutdowns every live conn fd (waking
   // its detached handler thread) and then waits for the handlers to drain,
   // so sockets are only ever CLOSED by the thread that owns them.

@@ -272,19 +272,19 @@ void Node::start(int port) {
 // connect + optional TLS + register + synchronous ack.
 int Node::dial_gateway(double timeout_s) {
   // try the active gateway, then every other candidate (primary first)
-  int fd = tcp_connect(gw_host_, gw_port_, timeout_s);
+  auto [ah, ap] = gw_addr();
+  int fd = tcp_connect(ah, ap, timeout_s);
   if (fd < 0 && !gw_list_.empty()) {
     std::vector<std::pair<std::string, int>> cands;
     cands.emplace_back(gw_primary_host_, gw_primary_port_);
     cands.insert(cands.end(), gw_list_.begin(), gw_list_.end());
     for (const auto& [h, p] : cands) {
-      if (h == gw_host_ && p == gw_port_) continue;
+      if (h == ah && p == ap) continue;
       fd = tcp_connect(h, p, 5.0);
       if (fd >= 0) {
         fprintf(stderr, "[net:%s] gateway failover -> %s:%d\n", name_.c_str(),
                 h.c_str(), p);
-        gw_host_ = h;
-        gw_port_ = p;
+        set_gw_addr(h, p);
         break;
       }
     }
@@ -359,11 +359,11 @@ void Node::stop() {
   if (accept_thread_.joinable()) accept_thread_.join();
   if (gw_thread_.joinable()) gw_thread_.join();
   {
-    // conn handler threads are detached: wait (bounded — a handler stuck in
-    // user code must not wedge stop()) for them to release their sockets
+    // UNBOUNDED drain of the detached conn handlers: every registered fd
+    // was shutdown above, so each handler unblocks and exits. Returning
+    // before they drain would let a straggler touch a destroyed Node.
     std::unique_lock<std::mutex> lk(conn_mu_);
-    conn_cv_.wait_for(lk, std::chrono::seconds(5),
-                      [&] { return live_conns_ == 0; });
+    conn_cv_.wait(lk, [&] { return live_conns_ == 0; });
   }
   // Phase 2: no reader threads remain — closing is safe now.
   if (listen_fd_ >= 0) {
@@ -508,6 +508,13 @@ void Node::gateway_listen_loop() {
     try {
     if (msg->get_or("kind", Json("")).as_string() == "relay_offer") {
       long long circuit = msg->get_or("circuit", Json((int64_t)0)).as_int();
+      {
+        // count the relay thread BEFORE detaching: stop() joins THIS (gw)
+        // thread before draining live_conns_, so the count is always
+        // visible to it and the relay thread can never outlive the Node
+        std::lock_guard<std::mutex> lk(conn_mu_);
+        ++live_conns_;
+      }
       std::thread([this, circuit] {
         try {
           relay_accept_run(circuit);
@@ -515,6 +522,9 @@ void Node::gateway_listen_loop() {
           fprintf(stderr, "[net:%s] relay accept error: %s\n", name_.c_str(),
                   e.what());
         }
+        std::lock_guard<std::mutex> lk(conn_mu_);
+        --live_conns_;
+        conn_cv_.notify_all();
       }).detach();
     } else if (msg->get_or("kind", Json("")).as_string() == "pub") {
       std::string topic = msg->at("topic").as_string();
@@ -534,8 +544,9 @@ void Node::gateway_listen_loop() {
 }
 
 int Node::relay_dial(const std::string& peer, double timeout_s) {
-  if (gw_port_ == 0) return -1;
-  int fd = tcp_connect(gw_host_, gw_port_, timeout_s);
+  auto [gh, gp] = gw_addr();
+  if (gp == 0) return -1;
+  int fd = tcp_connect(gh, gp, timeout_s);
   if (fd < 0) return -1;
   struct timeval tv;
   tv.tv_sec = (long)(timeout_s + 15.0);
@@ -558,33 +569,46 @@ int Node::relay_dial(const std::string& peer, double timeout_s) {
 }
 
 void Node::relay_accept_run(long long circuit) {
-  int fd = tcp_connect(gw_host_, gw_port_, 10.0);
+  // The surrounding thread is already counted in live_conns_ by its
+  // spawner; this function only tracks the FD (insert/erase, no counting)
+  // so every blocking phase — hello recv, TLS handshake, message loop —
+  // is woken by stop()'s shutdown sweep. Erase always precedes close.
+  auto [gh, gp] = gw_addr();
+  int fd = tcp_connect(gh, gp, 10.0);
   if (fd < 0) return;
+  {
+    std::lock_guard<std::mutex> lk(conn_mu_);
+    conn_fds_.insert(fd);
+    if (!running_) ::shutdown(fd, SHUT_RDWR);
+  }
+  auto drop_fd = [this](int f) {
+    std::lock_guard<std::mutex> lk(conn_mu_);
+    conn_fds_.erase(f);
+  };
   {
     MsgSocket hello(fd, nullptr);
     Json acc;
     acc["kind"] = "relay_accept";
     acc["circuit"] = (int64_t)circuit;
     acc["from"] = name_;
-    if (!hello.send_json(acc)) return;
+    if (!hello.send_json(acc)) {
+      drop_fd(fd);
+      return;  // hello socket closes after deregistration
+    }
     auto ack = hello.recv_json();
-    if (!ack || !ack->get_or("ok", Json(false)).as_bool()) return;
+    if (!ack || !ack->get_or("ok", Json(false)).as_bool()) {
+      drop_fd(fd);
+      return;
+    }
     fd = hello.release();
   }
   // from here the circuit is an ordinary inbound connection: server-side
-  // mTLS handshake + the normal typed-message loop (registered like any
-  // accepted conn so stop() can wake this thread)
-  {
-    std::lock_guard<std::mutex> lk(conn_mu_);
-    conn_fds_.insert(fd);
-    ++live_conns_;
-    if (!running_) ::shutdown(fd, SHUT_RDWR);
-  }
+  // mTLS handshake + the normal typed-message loop
   SSL* ssl = nullptr;
   if (tls_) {
     ssl = tls_->wrap(fd, true);
     if (!ssl) {
-      unregister_conn(fd);
+      drop_fd(fd);
       ::close(fd);
       return;
     }
@@ -597,7 +621,7 @@ void Node::relay_accept_run(long long circuit) {
       fprintf(stderr, "[net:%s] relay conn handler error: %s\n", name_.c_str(),
               e.what());
     }
-    unregister_conn(fd);
+    drop_fd(fd);
   }
 }
 
@@ -746,7 +770,7 @@ void Node::subscribe(const std::string& topic,
   sub["kind"] = "subscribe";
   sub["topic"] = topic;
   std::lock_guard<std::mutex> lk(gw_mu_);
-  gw_sock_->send_json(sub);
+  if (gw_sock_) gw_sock_->send_json(sub);  // no broker: recorded locally only
 }
 
 void Node::unsubscribe(const std::string& topic) {
@@ -758,7 +782,7 @@ void Node::unsubscribe(const std::string& topic) {
   un["kind"] = "unsubscribe";
   un["topic"] = topic;
   std::lock_guard<std::mutex> lk(gw_mu_);
-  gw_sock_->send_json(un);
+  if (gw_sock_) gw_sock_->send_json(un);
 }
 
 void Node::kv_put(const std::string& key, const Json& value) {
