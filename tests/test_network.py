@@ -428,3 +428,52 @@ def test_observed_address_substitution_offhost():
         n.stop()
         m.stop()
         gw.stop()
+
+
+def test_protocol_robustness_malformed_frames():
+    """Hostile/garbage input must drop only the offending connection: the
+    broker and nodes keep serving (reference framing guards: 1 MiB header
+    cap stream_pull.rs:28; here a 256 MiB frame cap + per-conn try/catch)."""
+    import socket as pysock
+
+    gw = core.Gateway()
+    gw.start(0)
+    n = core.Node("rob", "127.0.0.1", gw.port)
+    try:
+        n.start(0)
+        n.on("ping", lambda frm, body: {"ok": True})
+
+        def abuse(port):
+            # oversized length prefix (4 GiB) -> must be rejected, not allocated
+            s = pysock.create_connection(("127.0.0.1", port), timeout=2)
+            s.sendall(b"\xff\xff\xff\xff" + b"junk")
+            s.close()
+            # truncated frame: claims 100 bytes, sends 10, disconnects
+            s = pysock.create_connection(("127.0.0.1", port), timeout=2)
+            s.sendall(b"\x00\x00\x00\x64" + b"0123456789")
+            s.close()
+            # valid length, non-JSON payload
+            s = pysock.create_connection(("127.0.0.1", port), timeout=2)
+            s.sendall(b"\x00\x00\x00\x04" + b"~~~~")
+            s.close()
+            # valid JSON, unknown kind
+            blob = b'{"kind": "nonsense"}'
+            s = pysock.create_connection(("127.0.0.1", port), timeout=2)
+            s.sendall(len(blob).to_bytes(4, "big") + blob)
+            s.close()
+
+        for _ in range(3):
+            abuse(gw.port)
+            abuse(n.port)
+
+        # both survive: a fresh peer registers and round-trips a request
+        m = core.Node("rob2", "127.0.0.1", gw.port)
+        try:
+            m.start(0)
+            assert m.request("rob", "ping", {})["ok"] is True
+            assert m.kv_get("addr:rob") is not None
+        finally:
+            m.stop()
+    finally:
+        n.stop()
+        gw.stop()
