@@ -423,11 +423,13 @@ class Gateway {
   static void pump(int src, int dst) {
     char buf[1 << 16];
     while (true) {
-      ssize_t n = ::read(src, buf, sizeof buf);
+      ssize_t n = ::recv(src, buf, sizeof buf, 0);
       if (n <= 0) break;
       ssize_t off = 0;
       while (off < n) {
-        ssize_t w = ::write(dst, buf + off, n - off);
+        // MSG_NOSIGNAL: a torn-down far leg must surface as EPIPE, not a
+        // process-killing SIGPIPE in embedders that don't ignore it
+        ssize_t w = ::send(dst, buf + off, n - off, MSG_NOSIGNAL);
         if (w <= 0) return;
         off += w;
       }
