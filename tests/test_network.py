@@ -477,3 +477,32 @@ def test_protocol_robustness_malformed_frames():
     finally:
         n.stop()
         gw.stop()
+
+
+def test_node_and_gateway_restart_same_objects():
+    """stop() leaves both objects restartable: a second start() re-binds,
+    re-registers, and serves (validates teardown leaves no stale state)."""
+    gw = core.Gateway()
+    gw.start(0)
+    a = core.Node("ra", "127.0.0.1", gw.port)
+    b = core.Node("rb", "127.0.0.1", gw.port)
+    try:
+        a.start(0)
+        a.on("echo", lambda frm, body: {"x": body["x"] * 2})
+        b.start(0)
+        assert b.request("ra", "echo", {"x": 3})["x"] == 6
+        a.stop()
+        a.start(0)  # same object, fresh ephemeral port + re-registration
+        deadline = time.time() + 5
+        out = None
+        while time.time() < deadline:
+            try:
+                out = b.request("ra", "echo", {"x": 4})
+                break
+            except RuntimeError:
+                time.sleep(0.2)  # b may hold a stale cached address briefly
+        assert out == {"x": 8}
+    finally:
+        a.stop()
+        b.stop()
+        gw.stop()
