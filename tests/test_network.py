@@ -506,3 +506,40 @@ def test_node_and_gateway_restart_same_objects():
         a.stop()
         b.stop()
         gw.stop()
+
+
+def test_gateway_stop_during_active_relay_circuit():
+    """stop() must wake an ACTIVE relay splice (raw circuit fds are in the
+    shutdown sweep) and return promptly — with the unbounded handler drain,
+    a missed circuit fd would hang stop() forever."""
+    gw = core.Gateway()
+    gw.start(0)
+    a = core.Node("rca", "127.0.0.1", gw.port)
+    b = core.Node("rcb", "127.0.0.1", gw.port)
+    errs = []
+    try:
+        a.start(0)
+        b.start(0)
+        b.on("slow", lambda frm, body: (time.sleep(3), {"ok": True})[1])
+        a.kv_put("addr:rcb", "127.0.0.1:1")  # force the relay path
+
+        def call():
+            try:
+                a.request("rcb", "slow", {}, 20.0)
+            except RuntimeError as e:
+                errs.append(e)
+
+        t = threading.Thread(target=call)
+        t.start()
+        time.sleep(0.8)  # circuit established, response pending
+        t0 = time.time()
+        gw.stop()
+        elapsed = time.time() - t0
+        assert elapsed < 4.0, f"gateway stop took {elapsed:.1f}s with live circuit"
+        t.join(timeout=25)
+        assert not t.is_alive(), "relayed request never unblocked"
+        assert errs, "request should have failed when the circuit died"
+    finally:
+        a.stop()
+        b.stop()
+        gw.stop()
